@@ -1,0 +1,503 @@
+"""Learner: batching, loss computation, optimization and orchestration.
+
+Semantics parity with reference train.py (forward_prediction :127-186,
+compose_losses :189-215, compute_loss :218-267, Trainer :321-400,
+Learner :403-645) with the MI355X re-architecture:
+
+* device learner with bf16 autocast compute, fp32 loss math and fp32
+  master weights (Adam);
+* data-parallel replicas are separate PROCESSES, one per GPU, synchronized
+  by a single fused RCCL all-reduce of gradients + the data-count scalar
+  over xGMI (handyrl_amd/dist.py) — replacing nn.DataParallel
+  (reference train.py:339-340);
+* the off-policy target scans run as one fused HIP kernel on GPU
+  (handyrl_amd/losses.py -> handyrl_amd/ops);
+* the episode store is an explicitly locked buffer (handyrl_amd/batch.py)
+  instead of a shared bare deque.
+"""
+
+import copy
+import os
+import queue
+import random
+import threading
+import time
+import warnings
+
+import numpy as np
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+import torch.distributions as torch_dist
+import torch.optim as optim
+
+try:
+    import psutil
+except ImportError:       # pragma: no cover - psutil is in the image
+    psutil = None
+
+from .environment import prepare_env, make_env
+from .util import map_r, bimap_r, trimap_r
+from .model import to_torch, to_gpu, ModelWrapper
+from .losses import compute_target
+from .batch import EpisodeBuffer, Batcher, make_batch
+from .worker import WorkerCluster, WorkerServer
+from . import dist as hdist
+
+
+def forward_prediction(model, hidden, batch, args):
+    """Run the network over a (B, T, P) batch; FF nets flatten to one batch
+    axis, RNNs loop T with observation-masked hidden carry and burn-in under
+    no_grad."""
+    observations = batch['observation']
+    batch_shape = batch['action'].size()[:3]    # (B, T, P or 1)
+
+    if hidden is None:
+        obs = map_r(observations, lambda o: o.flatten(0, 2))
+        outputs = model(obs, None)
+        outputs = map_r(outputs, lambda o: o.unflatten(0, batch_shape))
+    else:
+        outputs = {}
+        for t in range(batch_shape[1]):
+            obs = map_r(observations, lambda o: o[:, t].flatten(0, 1))
+            omask_t = batch['observation_mask'][:, t]
+            omask = map_r(hidden, lambda h: omask_t.view(*h.size()[:2], *([1] * (h.dim() - 2))))
+            hidden_masked = bimap_r(hidden, omask, lambda h, m: h * m)
+            if args['turn_based_training'] and not args['observation']:
+                hidden_in = map_r(hidden_masked, lambda h: h.sum(1))
+            else:
+                hidden_in = map_r(hidden_masked, lambda h: h.flatten(0, 1))
+            if t < args['burn_in_steps']:
+                model.eval()
+                with torch.no_grad():
+                    outputs_t = model(obs, hidden_in)
+            else:
+                if not model.training:
+                    model.train()
+                outputs_t = model(obs, hidden_in)
+            outputs_t = map_r(outputs_t, lambda o: o.unflatten(0, (batch_shape[0], batch_shape[2])))
+            next_hidden = None
+            for k, o in outputs_t.items():
+                if k == 'hidden':
+                    next_hidden = o
+                else:
+                    outputs.setdefault(k, []).append(o)
+            hidden = trimap_r(hidden, next_hidden, omask, lambda h, nh, m: h * (1 - m) + nh * m)
+        outputs = {k: torch.stack(o, dim=1) for k, o in outputs.items() if o[0] is not None}
+
+    for k, o in outputs.items():
+        if k == 'policy':
+            o = o.mul(batch['turn_mask'])
+            if o.size(2) > 1 and batch_shape[2] == 1:
+                o = o.sum(2, keepdim=True)        # gather the turn seat
+            outputs[k] = o - batch['action_mask']
+        else:
+            outputs[k] = o.mul(batch['observation_mask'])
+    return outputs
+
+
+def compose_losses(outputs, log_selected_policies, total_advantages, targets, batch, args):
+    """Summed (not averaged) loss terms + the data count that scales lr."""
+    tmasks = batch['turn_mask']
+    omasks = batch['observation_mask']
+
+    losses = {}
+    dcnt = tmasks.sum().item()
+
+    losses['p'] = (-log_selected_policies * total_advantages).mul(tmasks).sum()
+    if 'value' in outputs:
+        losses['v'] = ((outputs['value'] - targets['value']) ** 2).mul(omasks).sum() / 2
+    if 'return' in outputs:
+        losses['r'] = F.smooth_l1_loss(outputs['return'], targets['return'],
+                                       reduction='none').mul(omasks).sum()
+
+    entropy = torch_dist.Categorical(logits=outputs['policy']).entropy().mul(tmasks.sum(-1))
+    losses['ent'] = entropy.sum()
+
+    base_loss = losses['p'] + losses.get('v', 0) + losses.get('r', 0)
+    decay = 1 - batch['progress'] * (1 - args['entropy_regularization_decay'])
+    entropy_loss = entropy.mul(decay).sum() * -args['entropy_regularization']
+    losses['total'] = base_loss + entropy_loss
+    return losses, dcnt
+
+
+def compute_loss(batch, model, hidden, args):
+    outputs = forward_prediction(model, hidden, batch, args)
+    if args['burn_in_steps'] > 0:
+        batch = map_r(batch, lambda v: v[:, args['burn_in_steps']:] if v.size(1) > 1 else v)
+        outputs = map_r(outputs, lambda v: v[:, args['burn_in_steps']:])
+
+    # loss math in fp32 regardless of autocast compute dtype
+    outputs = {k: o.float() for k, o in outputs.items()}
+
+    actions = batch['action']
+    emasks = batch['episode_mask']
+    omasks = batch['observation_mask']
+    value_target_masks, return_target_masks = omasks, omasks
+
+    clip_rho_threshold, clip_c_threshold = 1.0, 1.0
+
+    log_selected_b_policies = torch.log(torch.clamp(batch['selected_prob'], 1e-16, 1)) * emasks
+    log_selected_t_policies = F.log_softmax(outputs['policy'], dim=-1) \
+        .gather(-1, actions) * emasks
+
+    # importance weights (behavior vs current policy), clipped
+    log_rhos = log_selected_t_policies.detach() - log_selected_b_policies
+    rhos = torch.exp(log_rhos)
+    clipped_rhos = torch.clamp(rhos, 0, clip_rho_threshold)
+    cs = torch.clamp(rhos, 0, clip_c_threshold)
+    outputs_nograd = {k: o.detach() for k, o in outputs.items()}
+
+    if 'value' in outputs_nograd:
+        values_nograd = outputs_nograd['value']
+        if args['turn_based_training'] and values_nograd.size(2) == 2:
+            # two-player zero-sum: symmetrize with the opponent's negated view
+            values_opp = -torch.flip(values_nograd, dims=[2])
+            omasks_opp = torch.flip(omasks, dims=[2])
+            values_nograd = (values_nograd * omasks + values_opp * omasks_opp) \
+                / (omasks + omasks_opp + 1e-8)
+            value_target_masks = torch.clamp(omasks + omasks_opp, 0, 1)
+        # splice the terminal outcome in past the episode end
+        outputs_nograd['value'] = values_nograd * emasks + batch['outcome'] * (1 - emasks)
+
+    targets, advantages = {}, {}
+    value_args = (outputs_nograd.get('value', None), batch['outcome'], None,
+                  args['lambda'], 1, clipped_rhos, cs, value_target_masks)
+    return_args = (outputs_nograd.get('return', None), batch['return'], batch['reward'],
+                   args['lambda'], args['gamma'], clipped_rhos, cs, return_target_masks)
+
+    targets['value'], advantages['value'] = compute_target(args['value_target'], *value_args)
+    targets['return'], advantages['return'] = compute_target(args['value_target'], *return_args)
+    if args['policy_target'] != args['value_target']:
+        _, advantages['value'] = compute_target(args['policy_target'], *value_args)
+        _, advantages['return'] = compute_target(args['policy_target'], *return_args)
+
+    total_advantages = clipped_rhos * sum(advantages.values())
+    return compose_losses(outputs, log_selected_t_policies, total_advantages,
+                          targets, batch, args)
+
+
+class Trainer:
+    """SGD loop over batches from the episode buffer.
+
+    Multi-GPU: each DP rank owns one Trainer on its own GPU; gradients and
+    the per-epoch data count are SUM-all-reduced so the dynamic lr
+    (default_lr * data_cnt_ema, decayed by steps) sees the global batch.
+    """
+
+    def __init__(self, args, model, device=None):
+        self.args = args
+        self.device = device if device is not None else (
+            torch.device('cuda', hdist.env_local_rank())
+            if torch.cuda.is_available() else torch.device('cpu'))
+        self.use_amp = bool(args.get('bf16', True)) and self.device.type == 'cuda'
+
+        self.episodes = EpisodeBuffer(args)
+        self.template_model = copy.deepcopy(model).cpu()
+        self.model = model.to(self.device)
+        self.params = list(self.model.parameters())
+        self.reducer = hdist.GradReducer(self.params)
+
+        self.default_lr = 3e-8
+        self.data_cnt_ema = args['batch_size'] * args['forward_steps']
+        lr = self.default_lr * self.data_cnt_ema
+        self.optimizer = optim.Adam(self.params, lr=lr, weight_decay=1e-5) \
+            if len(self.params) > 0 else None
+        self.steps = 0
+        self.batcher = Batcher(args, self.episodes)
+        self.update_flag = threading.Event()
+        self.update_queue = queue.Queue(maxsize=1)
+        self.wrapped_model = ModelWrapper(self.model)
+
+    def update(self):
+        self.update_flag.set()
+        model, steps = self.update_queue.get()
+        return model, steps
+
+    def snapshot(self):
+        """CPU eval-mode copy of the current weights (no device churn)."""
+        snap = copy.deepcopy(self.template_model)
+        snap.load_state_dict(
+            {k: v.detach().cpu() for k, v in self.model.state_dict().items()})
+        snap.eval()
+        return snap
+
+    def train_step(self, batch):
+        """One optimizer step on an already-built CPU batch dict."""
+        batch_size = batch['value'].size(0)
+        player_count = batch['value'].size(2)
+        hidden = self.wrapped_model.init_hidden([batch_size, player_count])
+        if self.device.type == 'cuda':
+            batch = to_gpu(batch, non_blocking=True)
+            if hidden is not None:
+                hidden = map_r(hidden, lambda h: h.cuda(non_blocking=True))
+
+        if self.use_amp:
+            with torch.autocast('cuda', dtype=torch.bfloat16):
+                losses, dcnt = compute_loss(batch, self.wrapped_model, hidden, self.args)
+        else:
+            losses, dcnt = compute_loss(batch, self.wrapped_model, hidden, self.args)
+
+        self.optimizer.zero_grad(set_to_none=False)
+        losses['total'].backward()
+        self.reducer.allreduce_()                 # fused RCCL all-reduce (DP)
+        nn.utils.clip_grad_norm_(self.params, 4.0)
+        self.optimizer.step()
+        self.steps += 1
+        return losses, dcnt
+
+    def train(self):
+        if self.optimizer is None:
+            time.sleep(0.1)
+            return self.snapshot()
+
+        batch_cnt, data_cnt, loss_sum = 0, 0, {}
+        self.model.train()
+
+        while data_cnt == 0 or not self.update_flag.is_set():
+            batch = self.batcher.batch()
+            losses, dcnt = self.train_step(batch)
+            batch_cnt += 1
+            data_cnt += dcnt
+            for k, l in losses.items():
+                loss_sum[k] = loss_sum.get(k, 0.0) + l.item()
+
+        print('loss = %s' % ' '.join(
+            [k + ':' + '%.3f' % (l / data_cnt) for k, l in loss_sum.items()]))
+
+        global_data_cnt = hdist.allreduce_scalar(data_cnt / (1e-2 + batch_cnt))
+        self.data_cnt_ema = self.data_cnt_ema * 0.8 + global_data_cnt * 0.2
+        for param_group in self.optimizer.param_groups:
+            param_group['lr'] = self.default_lr * self.data_cnt_ema / (1 + self.steps * 1e-5)
+        return self.snapshot()
+
+    def run(self):
+        print('waiting training')
+        while len(self.episodes) < self.args['minimum_episodes']:
+            time.sleep(1)
+        if self.optimizer is not None:
+            self.batcher.run()
+            print('started training')
+        while True:
+            model = self.train()
+            self.update_flag.clear()
+            self.update_queue.put((model, self.steps))
+
+
+class Learner:
+    """Central conductor: owns the model epoch, serves worker requests
+    (job args / episodes / results / model pulls) and rolls training epochs."""
+
+    def __init__(self, args, net=None, remote=False):
+        train_args = args['train_args']
+        env_args = args['env_args']
+        train_args['env'] = env_args
+        args = train_args
+
+        self.args = args
+        random.seed(args['seed'])
+
+        self.env = make_env(env_args)
+        eval_modify_rate = (args['update_episodes'] ** 0.85) / args['update_episodes']
+        self.eval_rate = max(args['eval_rate'], eval_modify_rate)
+        self.shutdown_flag = False
+        self.flags = set()
+
+        self.model_epoch = self.args['restart_epoch']
+        self.model = net if net is not None else self.env.net()
+        if self.model_epoch > 0:
+            self.model.load_state_dict(
+                torch.load(self.model_path(self.model_epoch)), strict=False)
+
+        # generation / evaluation statistics: {model_id: (n, sum_r, sum_r2)}
+        self.generation_results = {}
+        self.num_episodes = 0
+        self.num_returned_episodes = 0
+        self.results = {}
+        self.results_per_opponent = {}
+        self.num_results = 0
+
+        self.worker = WorkerServer(args) if remote else WorkerCluster(args)
+        self.trainer = Trainer(args, copy.deepcopy(self.model))
+
+    def model_path(self, model_id):
+        return os.path.join('models', str(model_id) + '.pth')
+
+    def latest_model_path(self):
+        return os.path.join('models', 'latest.pth')
+
+    def update_model(self, model, steps):
+        print('updated model(%d)' % steps)
+        self.model_epoch += 1
+        self.model = model
+        os.makedirs('models', exist_ok=True)
+        torch.save(model.state_dict(), self.model_path(self.model_epoch))
+        torch.save(model.state_dict(), self.latest_model_path())
+
+    def feed_episodes(self, episodes):
+        for episode in episodes:
+            if episode is None:
+                continue
+            for p in episode['args']['player']:
+                model_id = self.model_epoch
+                outcome = episode['outcome'][p]
+                n, r, r2 = self.generation_results.get(model_id, (0, 0, 0))
+                self.generation_results[model_id] = n + 1, r + outcome, r2 + outcome ** 2
+            self.num_returned_episodes += 1
+            if self.num_returned_episodes % 100 == 0:
+                print(self.num_returned_episodes, end=' ', flush=True)
+
+        self.trainer.episodes.extend([e for e in episodes if e is not None])
+
+        mem_percent = psutil.virtual_memory().percent if psutil is not None else 0
+        mem_ok = mem_percent <= 95
+        maximum_episodes = self.args['maximum_episodes'] if mem_ok else \
+            int(len(self.trainer.episodes) * 95 / mem_percent)
+
+        if not mem_ok and 'memory_over' not in self.flags:
+            warnings.warn('memory usage %.1f%% with buffer size %d' %
+                          (mem_percent, len(self.trainer.episodes)))
+            self.flags.add('memory_over')
+
+        self.trainer.episodes.trim(maximum_episodes)
+
+    def feed_results(self, results):
+        for result in results:
+            if result is None:
+                continue
+            for p in result['args']['player']:
+                model_id = self.model_epoch
+                res = result['result'][p]
+                n, r, r2 = self.results.get(model_id, (0, 0, 0))
+                self.results[model_id] = n + 1, r + res, r2 + res ** 2
+                opp_map = self.results_per_opponent.setdefault(model_id, {})
+                opponent = result['opponent']
+                n, r, r2 = opp_map.get(opponent, (0, 0, 0))
+                opp_map[opponent] = n + 1, r + res, r2 + res ** 2
+
+    def update(self):
+        print()
+        print('epoch %d' % self.model_epoch)
+
+        if self.model_epoch not in self.results:
+            print('win rate = Nan (0)')
+        else:
+            def output_wp(name, results):
+                n, r, r2 = results
+                mean = r / (n + 1e-6)
+                name_tag = ' (%s)' % name if name != '' else ''
+                print('win rate%s = %.3f (%.1f / %d)' % (name_tag, (mean + 1) / 2, (r + n) / 2, n))
+
+            keys = self.results_per_opponent[self.model_epoch]
+            if len(self.args.get('eval', {}).get('opponent', [])) <= 1 and len(keys) <= 1:
+                output_wp('', self.results[self.model_epoch])
+            else:
+                output_wp('total', self.results[self.model_epoch])
+                for key in sorted(list(self.results_per_opponent[self.model_epoch])):
+                    output_wp(key, self.results_per_opponent[self.model_epoch][key])
+
+        if self.model_epoch not in self.generation_results:
+            print('generation stats = Nan (0)')
+        else:
+            n, r, r2 = self.generation_results[self.model_epoch]
+            mean = r / (n + 1e-6)
+            std = (r2 / (n + 1e-6) - mean ** 2) ** 0.5
+            print('generation stats = %.3f +- %.3f' % (mean, std))
+
+        model, steps = self.trainer.update()
+        if model is None:
+            model = self.model
+        self.update_model(model, steps)
+        self.flags = set()
+
+    def server(self):
+        print('started server')
+        prev_update_episodes = self.args['minimum_episodes']
+        next_update_episodes = prev_update_episodes + self.args['update_episodes']
+
+        while self.worker.connection_count() > 0 or not self.shutdown_flag:
+            try:
+                conn, (req, data) = self.worker.recv(timeout=0.3)
+            except queue.Empty:
+                continue
+
+            multi_req = isinstance(data, list)
+            if not multi_req:
+                data = [data]
+            send_data = []
+
+            if req == 'args':
+                if self.shutdown_flag:
+                    send_data = [None] * len(data)
+                else:
+                    for _ in data:
+                        job = {'model_id': {}}
+                        # evaluation share of jobs per eval_rate
+                        if self.num_results < self.eval_rate * self.num_episodes:
+                            job['role'] = 'e'
+                        else:
+                            job['role'] = 'g'
+
+                        if job['role'] == 'g':
+                            job['player'] = self.env.players()
+                            for p in self.env.players():
+                                job['model_id'][p] = self.model_epoch
+                            self.num_episodes += 1
+                        else:
+                            # rotate the evaluated seat
+                            job['player'] = [self.env.players()[
+                                self.num_results % len(self.env.players())]]
+                            for p in self.env.players():
+                                job['model_id'][p] = self.model_epoch \
+                                    if p in job['player'] else -1
+                            self.num_results += 1
+                        send_data.append(job)
+
+            elif req == 'episode':
+                self.feed_episodes(data)
+                send_data = [None] * len(data)
+
+            elif req == 'result':
+                self.feed_results(data)
+                send_data = [None] * len(data)
+
+            elif req == 'model':
+                import pickle
+                for model_id in data:
+                    model = self.model
+                    if model_id != self.model_epoch and model_id > 0:
+                        try:
+                            model = copy.deepcopy(self.model)
+                            model.load_state_dict(
+                                torch.load(self.model_path(model_id)), strict=False)
+                        except Exception:
+                            pass   # fall back to the latest model
+                    send_data.append(pickle.dumps(model))
+
+            if not multi_req and len(send_data) == 1:
+                send_data = send_data[0]
+            self.worker.send(conn, send_data)
+
+            if self.num_returned_episodes >= next_update_episodes:
+                prev_update_episodes = next_update_episodes
+                next_update_episodes = prev_update_episodes + self.args['update_episodes']
+                self.update()
+                if self.args['epochs'] >= 0 and self.model_epoch >= self.args['epochs']:
+                    self.shutdown_flag = True
+        print('finished server')
+
+    def run(self):
+        threading.Thread(target=self.trainer.run, daemon=True).start()
+        self.worker.run()
+        self.server()
+
+
+def train_main(args):
+    prepare_env(args['env_args'])
+    learner = Learner(args=args)
+    learner.run()
+
+
+def train_server_main(args):
+    learner = Learner(args=args, remote=True)
+    learner.run()
