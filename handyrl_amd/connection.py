@@ -120,7 +120,7 @@ def open_multiprocessing_connections(num_process, target, args_func):
     parent_conns = []
     for i in range(num_process):
         conn0, conn1 = mp.Pipe(duplex=True)
-        mp.Process(target=target, args=args_func(i, conn1), daemon=True).start()
+        mp.Process(target=target, args=args_func(i, conn1)).start()
         conn1.close()
         parent_conns.append(conn0)
     return parent_conns

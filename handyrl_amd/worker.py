@@ -174,7 +174,7 @@ class WorkerCluster(QueueCommunicator):
                 1 + max(0, self.args['worker']['num_parallel'] - 1) // 16
         for i in range(self.args['worker']['num_gathers']):
             conn0, conn1 = mp.Pipe(duplex=True)
-            mp.Process(target=gather_loop, args=(self.args, conn1, i), daemon=True).start()
+            mp.Process(target=gather_loop, args=(self.args, conn1, i)).start()
             conn1.close()
             self.add_connection(conn0)
 
