@@ -1,0 +1,197 @@
+"""Flagship benchmark: Hungry Geese self-play training throughput.
+
+Measures the BASELINE.json headline metric (learner steps/sec + env
+frames/sec, Hungry Geese self-play) on N GPUs of one node: each rank runs
+the full pipeline — vectorized self-play with batched bf16 GPU inference
+(handyrl_amd/actor.py), parallel batch builders, and V-Trace learner steps
+with fused HIP target scans — with gradients all-reduced over RCCL/xGMI
+(one process per GPU; weak scaling: per-GPU work is fixed).
+
+Each timed step is a fixed work quantum:
+  ACTOR_VEC_STEPS vectorized env transitions over N_ENVS games (self-play
+  generation feeding the replay buffer) + exactly one optimizer step on a
+  (batch_size x forward_steps) V-Trace batch.
+
+Launch (driver contract):
+  python bench.py --gpus 1 --steps K --warmup W
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+      --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+"""
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+from handyrl_amd import dist as hdist
+from handyrl_amd.actor import GeeseActorPool
+from handyrl_amd.batch import Batcher
+from handyrl_amd.models.geese_net import GeeseNet
+from handyrl_amd.train import Trainer
+
+N_ENVS = 256            # self-play games per GPU
+ACTOR_VEC_STEPS = 16    # env transitions (per game) per learner step
+
+
+def bench_args(batch_size=128, forward_steps=16):
+    return {
+        'turn_based_training': False,     # 4-player simultaneous: solo seats
+        'observation': False,
+        'gamma': 0.8,
+        'forward_steps': forward_steps,
+        'burn_in_steps': 0,
+        'compress_steps': 4,
+        'entropy_regularization': 0.1,
+        'entropy_regularization_decay': 0.1,
+        'batch_size': batch_size,
+        'minimum_episodes': 160,
+        'maximum_episodes': 4000,
+        'num_batchers': 3,
+        'lambda': 0.7,
+        'policy_target': 'VTRACE',
+        'value_target': 'VTRACE',
+        'seed': 0,
+        'bf16': True,
+        'compress_episodes': False,       # local buffer: skip bz2
+    }
+
+
+def main():
+    parser = argparse.ArgumentParser()
+    parser.add_argument('--gpus', type=int, default=1)
+    parser.add_argument('--steps', type=int, default=30)
+    parser.add_argument('--warmup', type=int, default=5)
+    parser.add_argument('--envs', type=int, default=N_ENVS)
+    parser.add_argument('--batch-size', type=int, default=128)
+    parser.add_argument('--forward-steps', type=int, default=16)
+    cli = parser.parse_args()
+
+    # single-threaded CPU torch: the GPU does the math, and forked batch
+    # builders deadlock if the parent has a live OpenMP pool at fork time
+    torch.set_num_threads(1)
+    rank = hdist.env_rank()
+    local_rank = hdist.env_local_rank()
+    world = hdist.env_world_size()
+
+    args = bench_args(cli.batch_size, cli.forward_steps)
+
+    # fork the batch-builder processes FIRST, before any HIP context or
+    # torch thread-pool exists in this process
+    from handyrl_amd.batch import EpisodeBuffer
+    buffer = EpisodeBuffer(args)
+    batcher = Batcher(args, buffer)
+
+    use_cuda = torch.cuda.is_available()
+    device = torch.device('cuda', local_rank) if use_cuda else torch.device('cpu')
+    if use_cuda:
+        torch.cuda.set_device(device)
+    hdist.init_from_env(device=local_rank if use_cuda else None)
+
+    torch.manual_seed(1234)                      # identical init on all ranks
+    model = GeeseNet()
+
+    trainer = Trainer(args, model, device=device, episodes=buffer)
+    if world > 1:
+        hdist.broadcast_params(trainer.model)
+
+    actor_model = trainer.model                  # shared weights, zero staleness
+    actor_model_eval = actor_model
+    pool = GeeseActorPool(actor_model_eval, args, n_games=cli.envs,
+                          device=device, seed=1000 + rank)
+
+    def pump_actor(n_vec_steps):
+        frames = 0
+        was_training = trainer.model.training
+        trainer.model.eval()
+        for _ in range(n_vec_steps):
+            frames += pool.step_once()
+        if was_training:
+            trainer.model.train()
+        eps = pool.harvest()
+        if eps:
+            trainer.episodes.extend(eps)
+        return frames
+
+    # ---- prefill: generate the minimum episode set (untimed) ----
+    t0 = time.time()
+    while pool.episodes_done < args['minimum_episodes']:
+        pump_actor(8)
+    trainer.episodes.trim(args['maximum_episodes'])
+    batcher.run()
+    if rank == 0:
+        print('# prefill: %d episodes in %.1fs' %
+              (len(trainer.episodes), time.time() - t0), flush=True)
+
+    def one_step():
+        frames = pump_actor(ACTOR_VEC_STEPS)
+        batch = batcher.batch()
+        losses, dcnt = trainer.train_step(batch)
+        return frames, losses
+
+    # ---- warmup (untimed) ----
+    for _ in range(cli.warmup):
+        one_step()
+
+    # ---- timed region ----
+    hdist.barrier()
+    if use_cuda:
+        torch.cuda.synchronize()
+    start = time.time()
+    total_frames = 0
+    for _ in range(cli.steps):
+        frames, _ = one_step()
+        total_frames += frames
+    if use_cuda:
+        torch.cuda.synchronize()
+    hdist.barrier()
+    elapsed = time.time() - start
+
+    # max over ranks (slowest rank defines job time)
+    if world > 1:
+        import torch.distributed as tdist
+        t = torch.tensor([elapsed], device=device if use_cuda else 'cpu')
+        tdist.all_reduce(t, op=tdist.ReduceOp.MAX)
+        elapsed = t.item()
+        f = torch.tensor([float(total_frames)], device=device if use_cuda else 'cpu')
+        tdist.all_reduce(f, op=tdist.ReduceOp.SUM)
+        total_frames = int(f.item())
+
+    n_gpus = world if world > 1 else cli.gpus
+    steps_per_sec = cli.steps / elapsed
+    frames_per_sec = total_frames / elapsed
+    samples_per_sec = steps_per_sec * cli.batch_size * cli.forward_steps * n_gpus
+
+    if rank == 0:
+        result = {
+            'metric': 'hungry_geese_selfplay_env_frames_per_sec',
+            'value': round(frames_per_sec, 1),
+            'unit': 'frames/s',
+            'n_gpus': n_gpus,
+            'steps': cli.steps,
+            'warmup': cli.warmup,
+            'ms_per_step': round(1000.0 * elapsed / cli.steps, 2),
+            'higher_is_better': True,
+            'scaling': 'weak',
+            'vs_baseline': None,
+            'dtype': 'bf16' if (args['bf16'] and use_cuda) else 'fp32',
+            'data': 'synthetic',
+            'config': {
+                'model': 'GeeseNet(12x32 torus-conv residual)',
+                'global_batch': cli.batch_size * n_gpus,
+                'seq_len': cli.forward_steps,
+                'parallelism': 'dp%d' % n_gpus,
+                'envs_per_gpu': cli.envs,
+                'actor_vec_steps_per_learner_step': ACTOR_VEC_STEPS,
+                'learner_steps_per_sec': round(steps_per_sec, 2),
+                'learner_samples_per_sec': round(samples_per_sec, 1),
+                'loss': 'VTRACE (policy+value), entropy reg',
+                'note': 'self-play on random-init weights; no external data',
+            },
+        }
+        print(json.dumps(result), flush=True)
+
+
+if __name__ == '__main__':
+    main()

@@ -1,0 +1,149 @@
+"""GPU actor pool: batched self-play over vectorized environments.
+
+The MI355X replacement for the reference's one-CPU-process-per-environment
+workers (reference worker.py:26-89, generation.py:20-93): hundreds of
+Hungry Geese games advance in lock-step; every step does ONE batched
+network forward for all alive seats (bf16, inference mode) and one fused
+masked-softmax-sample kernel (handyrl_amd/ops), instead of per-env
+single-sample CPU inference.
+
+Finished games are packaged into reference-format episodes (uncompressed
+moment blocks by default — same interchange as the CPU worker path, minus
+the bz2 that a local learner does not need).
+"""
+
+import numpy as np
+import torch
+
+from . import ops
+from .batch import pack_moments
+from .envs.vec_geese import GeeseVecEnv, N_PLAYERS
+
+MOMENT_KEYS = ('observation', 'selected_prob', 'action_mask', 'action',
+               'value', 'reward', 'return')
+
+
+class GeeseActorPool:
+    """Self-play actor pool for Hungry Geese on one GPU."""
+
+    def __init__(self, model, args, n_games=256, device=None, seed=0,
+                 store_uint8_obs=True):
+        self.args = args
+        self.device = device if device is not None else (
+            torch.device('cuda') if torch.cuda.is_available() else torch.device('cpu'))
+        self.model = model
+        self.vec = GeeseVecEnv(n_games, seed=seed)
+        self.n_games = n_games
+        self.store_uint8_obs = store_uint8_obs
+        self.gamma = args.get('gamma', 0.8)
+        self.compress = args.get('compress_episodes', False)
+        self.compress_steps = args.get('compress_steps', 4)
+        # per-game trajectory: list of (alive_mask, obs[4,...], act[4], prob[4], val[4])
+        self.traj = [[] for _ in range(n_games)]
+        self.completed = []
+        self.frames = 0          # env transitions executed (sum over games)
+        self.episodes_done = 0
+        self._zero_mask = None
+
+    @torch.inference_mode()
+    def _policy_forward(self, obs_f):
+        if self.device.type == 'cuda':
+            with torch.autocast('cuda', dtype=torch.bfloat16):
+                out = self.model(obs_f, None)
+            return out['policy'].float(), out['value'].float()
+        out = self.model(obs_f, None)
+        return out['policy'].float(), out['value'].float()
+
+    def step_once(self):
+        """Advance every live game by one transition; returns #frames."""
+        vec = self.vec
+        obs_u8 = vec.observations()                     # (G, 4, 17, 7, 11)
+        live = vec.alive & ~vec.over[:, None]
+        gi, pi = np.nonzero(live)
+        if len(gi) == 0:
+            return 0
+
+        obs_sel = obs_u8[gi, pi]                        # (M, 17, 7, 11)
+        obs_t = torch.from_numpy(obs_sel)
+        if self.device.type == 'cuda':
+            obs_t = obs_t.to(self.device, non_blocking=True)
+        obs_f = obs_t.float()
+
+        policy, value = self._policy_forward(obs_f)
+        M, A = policy.shape
+        if self.device.type == 'cuda':
+            if self._zero_mask is None or self._zero_mask.shape[0] < M:
+                self._zero_mask = torch.zeros(max(M, 1), A, device=self.device)
+            uniform = torch.rand(M, device=self.device)
+            actions_t, probs_t = ops.masked_sample(policy, self._zero_mask[:M], uniform)
+            actions = actions_t.cpu().numpy()
+            probs = probs_t.cpu().numpy()
+            values = value.squeeze(-1).cpu().numpy()
+        else:
+            probs_full = torch.softmax(policy, dim=-1)
+            actions_t = torch.multinomial(probs_full, 1).squeeze(-1)
+            actions = actions_t.numpy()
+            probs = probs_full.gather(-1, actions_t.unsqueeze(-1)).squeeze(-1).numpy()
+            values = value.squeeze(-1).numpy()
+
+        act_grid = np.zeros((self.n_games, N_PLAYERS), dtype=np.int32)
+        act_grid[gi, pi] = actions
+
+        # record the step per game (only games with at least one live seat)
+        act_row = np.full((self.n_games, N_PLAYERS), -1, dtype=np.int32)
+        prob_row = np.zeros((self.n_games, N_PLAYERS), dtype=np.float32)
+        val_row = np.zeros((self.n_games, N_PLAYERS), dtype=np.float32)
+        act_row[gi, pi] = actions
+        prob_row[gi, pi] = probs
+        val_row[gi, pi] = values
+
+        game_has_live = live.any(axis=1)
+        for g in np.nonzero(game_has_live)[0]:
+            self.traj[g].append((live[g].copy(), obs_u8[g], act_row[g].copy(),
+                                 prob_row[g].copy(), val_row[g].copy()))
+
+        done = vec.step(act_grid)
+        self.frames += int(game_has_live.sum())
+
+        finished = np.nonzero(done)[0]
+        if len(finished):
+            outcomes = vec.outcomes(finished)
+            for k, g in enumerate(finished):
+                self.completed.append(self._package(g, outcomes[k]))
+                self.traj[g] = []
+            self.episodes_done += len(finished)
+            vec.reset_games(finished)
+        return int(game_has_live.sum())
+
+    def _package(self, g, outcome_row):
+        """Build a reference-format episode dict from a finished game."""
+        moments = []
+        for alive_mask, obs_all, act, prob, val in self.traj[g]:
+            moment = {key: {p: None for p in range(N_PLAYERS)} for key in MOMENT_KEYS}
+            turn = [int(p) for p in np.nonzero(alive_mask)[0]]
+            for p in turn:
+                obs = obs_all[p]
+                moment['observation'][p] = obs if self.store_uint8_obs \
+                    else obs.astype(np.float32)
+                moment['selected_prob'][p] = float(prob[p])
+                moment['action_mask'][p] = np.zeros(4, dtype=np.float32)
+                moment['action'][p] = int(act[p])
+                moment['value'][p] = np.array([val[p]], dtype=np.float32)
+                # geese have no immediate reward; returns stay at 0
+            moment['turn'] = turn
+            moments.append(moment)
+
+        job_args = {'player': list(range(N_PLAYERS)),
+                    'model_id': {p: -1 for p in range(N_PLAYERS)}}
+        return {
+            'args': job_args,
+            'steps': len(moments),
+            'outcome': {p: float(outcome_row[p]) for p in range(N_PLAYERS)},
+            'moment': pack_moments(moments, self.compress_steps, compress=self.compress),
+        }
+
+    def harvest(self):
+        """Return and clear the finished-episode list."""
+        out = self.completed
+        self.completed = []
+        return out

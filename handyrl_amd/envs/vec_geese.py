@@ -1,0 +1,230 @@
+"""Vectorized Hungry Geese engine: G games stepped at once in numpy.
+
+Same rules as handyrl_amd.envs.hungry_geese.GeeseState (the single-game
+oracle; parity is tested in tests/test_vec_geese.py), restructured as
+struct-of-arrays so the GPU actor pool (handyrl_amd/actor.py) can run
+hundreds of self-play games against ONE batched network forward per step —
+the MI355X replacement for the reference's one-process-per-environment
+workers (reference worker.py / generation.py).
+
+Goose bodies are fixed-capacity ring buffers (head at ``start``, elements
+at (start + i) % CAP), so head-push/tail-pop are O(1) index updates and
+board grids are maintained incrementally.
+"""
+
+import numpy as np
+
+from .hungry_geese import (ROWS, COLS, N_CELLS, N_PLAYERS, HUNGER_RATE,
+                           MIN_FOOD, MAX_LEN, MAX_STEPS, OPPOSITE)
+
+CAP = N_CELLS  # a goose can never exceed the board
+
+# SHIFT[cell, action] -> next cell on the torus
+_r, _c = np.divmod(np.arange(N_CELLS), COLS)
+SHIFT = np.stack([
+    ((_r - 1) % ROWS) * COLS + _c,          # NORTH
+    ((_r + 1) % ROWS) * COLS + _c,          # SOUTH
+    _r * COLS + (_c - 1) % COLS,            # WEST
+    _r * COLS + (_c + 1) % COLS,            # EAST
+], axis=1).astype(np.int32)
+
+_OPP = np.array(OPPOSITE, dtype=np.int32)
+
+# seat-relative channel gather: obs channel c of player k shows goose (c+k)%4
+_REL = (np.arange(N_PLAYERS)[None, :] + np.arange(N_PLAYERS)[:, None]) % N_PLAYERS
+
+
+class GeeseVecEnv:
+    """G simultaneous Hungry Geese games."""
+
+    def __init__(self, n_games, seed=0):
+        self.G = n_games
+        self.rng = np.random.default_rng(seed)
+        G = self.G
+        self.body = np.full((G, N_PLAYERS, CAP), -1, dtype=np.int32)
+        self.start = np.zeros((G, N_PLAYERS), dtype=np.int32)
+        self.length = np.zeros((G, N_PLAYERS), dtype=np.int32)
+        self.alive = np.zeros((G, N_PLAYERS), dtype=bool)
+        self.scores = np.zeros((G, N_PLAYERS), dtype=np.float64)
+        self.last_action = np.full((G, N_PLAYERS), -1, dtype=np.int32)
+        self.prev_head = np.full((G, N_PLAYERS), -1, dtype=np.int32)
+        self.food = np.full((G, MIN_FOOD), -1, dtype=np.int32)
+        self.step_count = np.zeros(G, dtype=np.int32)
+        self.over = np.zeros(G, dtype=bool)
+        # incremental grids
+        self.body_grid = np.zeros((G, N_PLAYERS, N_CELLS), dtype=np.uint8)
+        self.reset_games(np.arange(G))
+
+    # -- helpers -----------------------------------------------------------
+    def _head(self, gmask=None):
+        idx = self.start % CAP
+        return np.take_along_axis(self.body, idx[..., None], axis=2)[..., 0]
+
+    def _tail_cell(self):
+        idx = (self.start + self.length - 1) % CAP
+        return np.take_along_axis(self.body, idx[..., None], axis=2)[..., 0]
+
+    def reset_games(self, games):
+        """Reset the given game indices to fresh initial states."""
+        if len(games) == 0:
+            return
+        for g in games:
+            cells = self.rng.choice(N_CELLS, size=N_PLAYERS + MIN_FOOD, replace=False)
+            self.body[g] = -1
+            self.body[g, :, 0] = cells[:N_PLAYERS]
+            self.food[g] = cells[N_PLAYERS:]
+        self.start[games] = 0
+        self.length[games] = 1
+        self.alive[games] = True
+        self.scores[games] = 0.0
+        self.last_action[games] = -1
+        self.prev_head[games] = -1
+        self.step_count[games] = 0
+        self.over[games] = False
+        self.body_grid[games] = 0
+        gg = np.repeat(games, N_PLAYERS)
+        pp = np.tile(np.arange(N_PLAYERS), len(games))
+        self.body_grid[gg, pp, self.body[gg, pp, 0]] = 1
+
+    def _kill(self, g_idx, p_idx):
+        self.alive[g_idx, p_idx] = False
+        self.length[g_idx, p_idx] = 0
+        self.body_grid[g_idx, p_idx] = 0
+
+    def step(self, actions):
+        """actions: (G, 4) int32; entries for dead seats/finished games ignored."""
+        G = self.G
+        act = np.asarray(actions, dtype=np.int32)
+        live = self.alive & ~self.over[:, None]
+
+        self.prev_head = np.where(self.alive, self._head(), -1)
+
+        # 1) reverse-move deaths
+        rev = live & (self.last_action >= 0) & (act == _OPP[np.clip(self.last_action, 0, 3)])
+        gi, pi = np.nonzero(rev)
+        self._kill(gi, pi)
+        live = self.alive & ~self.over[:, None]
+        self.last_action = np.where(live, act, self.last_action)
+
+        # 2) move: new head cell
+        heads = self._head()
+        new_head = np.where(live, SHIFT[np.clip(heads, 0, N_CELLS - 1), np.clip(act, 0, 3)], heads)
+
+        # food consumption (order-free: contested food implies a head
+        # collision, and the losers die this step anyway)
+        ate = np.zeros((G, N_PLAYERS), dtype=bool)
+        for f in range(MIN_FOOD):
+            hit = live & (new_head == self.food[:, f][:, None]) & (self.food[:, f][:, None] >= 0)
+            ate |= hit
+            self.food[hit.any(axis=1), f] = -1
+
+        # pop tail unless the goose ate (bodies never self-overlap, so the
+        # grid bit of the vacated cell can be cleared directly)
+        popping = live & ~ate
+        gi, pi = np.nonzero(popping)
+        if len(gi):
+            tail = self._tail_cell()[gi, pi]
+            self.length[gi, pi] -= 1
+            self.body_grid[gi, pi, tail] = 0
+
+        # push new head; a head landing on this goose's own remaining body is
+        # a self-collision the 0/1 grid can't count — flag it explicitly
+        self_crash = np.zeros((G, N_PLAYERS), dtype=bool)
+        gi, pi = np.nonzero(live)
+        if len(gi):
+            nh = new_head[gi, pi]
+            self_crash[gi, pi] = self.body_grid[gi, pi, nh] == 1
+            new_start = (self.start[gi, pi] - 1) % CAP
+            self.start[gi, pi] = new_start
+            self.body[gi, pi, new_start] = nh
+            self.length[gi, pi] += 1
+            self.body_grid[gi, pi, nh] = 1
+
+        # 3) hunger shrink every HUNGER_RATE transitions
+        hungry_games = (self.step_count + 1) % HUNGER_RATE == 0
+        shrink = live & hungry_games[:, None]
+        gi, pi = np.nonzero(shrink)
+        if len(gi):
+            tail = self._tail_cell()[gi, pi]
+            self.length[gi, pi] -= 1
+            keep = self_crash[gi, pi] & (tail == new_head[gi, pi])  # looped head
+            self.body_grid[gi[~keep], pi[~keep], tail[~keep]] = 0
+            starved = self.length[gi, pi] <= 0
+            self._kill(gi[starved], pi[starved])
+        live = self.alive & ~self.over[:, None]
+
+        # 4) collision deaths: live head on a cell with >1 segment
+        counts = self.body_grid.sum(axis=1, dtype=np.int16)     # (G, cells)
+        heads = self._head()
+        gi, pi = np.nonzero(live)
+        if len(gi):
+            crash = (counts[gi, heads[gi, pi]] > 1) | self_crash[gi, pi]
+            self._kill(gi[crash], pi[crash])
+        live = self.alive & ~self.over[:, None]
+
+        # 5) food replenishment onto random free cells
+        need_mask = (self.food < 0) & ~self.over[:, None]
+        for g in np.nonzero(need_mask.any(axis=1))[0]:
+            occ = self.body_grid[g].any(axis=0)
+            for f in range(MIN_FOOD):
+                if self.food[g, f] >= 0:
+                    occ[self.food[g, f]] = True
+            free = np.nonzero(~occ)[0]
+            self.rng.shuffle(free)
+            k = 0
+            for f in range(MIN_FOOD):
+                if self.food[g, f] < 0 and k < len(free):
+                    self.food[g, f] = free[k]
+                    k += 1
+
+        stepped = ~self.over
+        self.step_count[stepped] += 1
+
+        # 6) scores for survivors
+        surv = self.alive & stepped[:, None]
+        self.scores[surv] = (self.step_count[:, None] * (MAX_LEN + 1) + self.length)[surv]
+
+        # 7) termination
+        done = stepped & ((self.alive.sum(axis=1) <= 1) | (self.step_count >= MAX_STEPS))
+        self.over |= done
+        return done
+
+    # -- observation / outcome ----------------------------------------------
+    def observations(self):
+        """uint8 (G, 4, 17, 7, 11): the 17-plane encoding for every seat."""
+        G = self.G
+        head_grid = np.zeros((G, N_PLAYERS, N_CELLS), dtype=np.uint8)
+        tail_grid = np.zeros_like(head_grid)
+        prev_grid = np.zeros_like(head_grid)
+        heads = self._head()
+        tails = self._tail_cell()
+        gi, pi = np.nonzero(self.alive)
+        head_grid[gi, pi, heads[gi, pi]] = 1
+        tail_grid[gi, pi, tails[gi, pi]] = 1
+        gi, pi = np.nonzero(self.prev_head >= 0)
+        prev_grid[gi, pi, self.prev_head[gi, pi]] = 1
+
+        food_grid = np.zeros((G, N_CELLS), dtype=np.uint8)
+        for f in range(MIN_FOOD):
+            ok = self.food[:, f] >= 0
+            food_grid[np.nonzero(ok)[0], self.food[ok, f]] = 1
+
+        # groups stacked on a new axis: (G, group, goose, cell)
+        groups = np.stack([head_grid, tail_grid, self.body_grid, prev_grid], axis=1)
+        # seat-relative gather over the goose axis -> (G, seat, group, goose, cell)
+        rel = groups[:, :, _REL, :]                   # (G, group, seat, goose, cell)
+        rel = rel.transpose(0, 2, 1, 3, 4)            # (G, seat, group, goose, cell)
+        obs = np.empty((G, N_PLAYERS, 17, N_CELLS), dtype=np.uint8)
+        obs[:, :, 0:4] = rel[:, :, 0]
+        obs[:, :, 4:8] = rel[:, :, 1]
+        obs[:, :, 8:12] = rel[:, :, 2]
+        obs[:, :, 12:16] = rel[:, :, 3]
+        obs[:, :, 16] = food_grid[:, None, :]
+        return obs.reshape(G, N_PLAYERS, 17, ROWS, COLS)
+
+    def outcomes(self, games):
+        """Pairwise rank outcome per seat for the given finished games."""
+        sc = self.scores[games]                        # (n, 4)
+        gt = (sc[:, :, None] > sc[:, None, :]).sum(axis=2)
+        lt = (sc[:, :, None] < sc[:, None, :]).sum(axis=2)
+        return (gt - lt) / (N_PLAYERS - 1)

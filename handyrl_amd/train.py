@@ -50,6 +50,11 @@ def forward_prediction(model, hidden, batch, args):
     axis, RNNs loop T with observation-masked hidden carry and burn-in under
     no_grad."""
     observations = batch['observation']
+    # GPU actors store 0/1 observations as uint8 (4x H2D traffic saved);
+    # cast to compute dtype at the consumer
+    observations = map_r(
+        observations,
+        lambda o: o.float() if not torch.is_floating_point(o) else o)
     batch_shape = batch['action'].size()[:3]    # (B, T, P or 1)
 
     if hidden is None:
@@ -185,14 +190,14 @@ class Trainer:
     (default_lr * data_cnt_ema, decayed by steps) sees the global batch.
     """
 
-    def __init__(self, args, model, device=None):
+    def __init__(self, args, model, device=None, episodes=None):
         self.args = args
         self.device = device if device is not None else (
             torch.device('cuda', hdist.env_local_rank())
             if torch.cuda.is_available() else torch.device('cpu'))
         self.use_amp = bool(args.get('bf16', True)) and self.device.type == 'cuda'
 
-        self.episodes = EpisodeBuffer(args)
+        self.episodes = episodes if episodes is not None else EpisodeBuffer(args)
         self.template_model = copy.deepcopy(model).cpu()
         self.model = model.to(self.device)
         self.params = list(self.model.parameters())
@@ -204,7 +209,9 @@ class Trainer:
         self.optimizer = optim.Adam(self.params, lr=lr, weight_decay=1e-5) \
             if len(self.params) > 0 else None
         self.steps = 0
-        self.batcher = Batcher(args, self.episodes)
+        # created lazily in run(): forking batch-builder processes must
+        # happen before heavy torch thread-pool / HIP context activity
+        self.batcher = None
         self.update_flag = threading.Event()
         self.update_queue = queue.Queue(maxsize=1)
         self.wrapped_model = ModelWrapper(self.model)
@@ -273,6 +280,8 @@ class Trainer:
 
     def run(self):
         print('waiting training')
+        if self.batcher is None:
+            self.batcher = Batcher(self.args, self.episodes)
         while len(self.episodes) < self.args['minimum_episodes']:
             time.sleep(1)
         if self.optimizer is not None:
