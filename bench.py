@@ -92,7 +92,7 @@ def main():
     torch.manual_seed(1234)                      # identical init on all ranks
     model = GeeseNet()
 
-    trainer = Trainer(args, model, device=device, episodes=buffer)
+    trainer = Trainer(args, model, device=device, episodes=buffer, batcher=batcher)
     if world > 1:
         hdist.broadcast_params(trainer.model)
 

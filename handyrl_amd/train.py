@@ -190,7 +190,7 @@ class Trainer:
     (default_lr * data_cnt_ema, decayed by steps) sees the global batch.
     """
 
-    def __init__(self, args, model, device=None, episodes=None):
+    def __init__(self, args, model, device=None, episodes=None, batcher=None):
         self.args = args
         self.device = device if device is not None else (
             torch.device('cuda', hdist.env_local_rank())
@@ -209,9 +209,12 @@ class Trainer:
         self.optimizer = optim.Adam(self.params, lr=lr, weight_decay=1e-5) \
             if len(self.params) > 0 else None
         self.steps = 0
-        # created lazily in run(): forking batch-builder processes must
-        # happen before heavy torch thread-pool / HIP context activity
-        self.batcher = None
+        # batch-builder processes fork HERE (construction time, main thread,
+        # before worker/server threads start): forking later from a threaded
+        # process risks a child deadlock on inherited locks.  Callers that
+        # need the fork even earlier (before any HIP context, e.g. bench.py)
+        # pass a pre-built Batcher in.
+        self.batcher = batcher if batcher is not None else Batcher(args, self.episodes)
         self.update_flag = threading.Event()
         self.update_queue = queue.Queue(maxsize=1)
         self.wrapped_model = ModelWrapper(self.model)
@@ -280,8 +283,6 @@ class Trainer:
 
     def run(self):
         print('waiting training')
-        if self.batcher is None:
-            self.batcher = Batcher(self.args, self.episodes)
         while len(self.episodes) < self.args['minimum_episodes']:
             time.sleep(1)
         if self.optimizer is not None:
