@@ -75,7 +75,9 @@ def target_scan(algorithm, values, returns, rewards, lambda_, gamma, rhos, cs):
 
     kind = _SCAN_KIND[algorithm]
     v = values.contiguous().float()
-    ret = returns.contiguous().float()
+    # only the bootstrap slice returns[:, -1] enters the scan; 'returns' may
+    # be a (B, 1, P, 1) outcome tensor broadcast along T (train.py value path)
+    ret = returns[:, -1].contiguous().float()
     rew = rewards.contiguous().float() if rewards is not None else None
     lam = lambda_.contiguous().float()
     if kind == 2:

@@ -22,7 +22,7 @@ enum ScanKind { KIND_TD = 0, KIND_UPGO = 1, KIND_VTRACE = 2 };
 // tiny (B*P lanes x T steps); the win over eager is launch-count, not FLOPs.
 __global__ void target_scan_kernel(
     const float* __restrict__ values,    // (B,T,P)
-    const float* __restrict__ returns_,  // (B,T,P)
+    const float* __restrict__ ret_boot,  // (B,P): bootstrap value (returns at T-1)
     const float* __restrict__ rewards,   // (B,T,P) or nullptr
     const float* __restrict__ lambda_,   // (B,T,P)
     const float* __restrict__ rhos,      // (B,T,P), VTRACE only
@@ -39,7 +39,7 @@ __global__ void target_scan_kernel(
 
 #define AT(a, t) a[base + (long)(t) * stride]
 
-    const float ret_last = AT(returns_, T - 1);
+    const float ret_last = ret_boot[(long)b * P + p];
 
     if (kind == KIND_VTRACE) {
         // vs(t) = V(t) + sum-scan of deltas; advantage uses vs(t+1)
@@ -123,7 +123,7 @@ __global__ void masked_sample_kernel(
 }  // namespace
 
 static std::vector<torch::Tensor> target_scan(
-    torch::Tensor values, torch::Tensor returns,
+    torch::Tensor values, torch::Tensor ret_boot,
     c10::optional<torch::Tensor> rewards, torch::Tensor lambda_,
     c10::optional<torch::Tensor> rhos, c10::optional<torch::Tensor> cs,
     double gamma, int64_t kind)
@@ -134,6 +134,8 @@ static std::vector<torch::Tensor> target_scan(
     const int B = values.size(0), T = values.size(1);
     int P = 1;
     for (int d = 2; d < values.dim(); ++d) P *= values.size(d);
+    TORCH_CHECK(ret_boot.numel() == (long)B * P,
+                "target_scan: bootstrap tensor must have B*P elements");
 
     auto targets = torch::empty_like(values);
     auto adv = torch::empty_like(values);
@@ -147,7 +149,7 @@ static std::vector<torch::Tensor> target_scan(
     const float* cs_ptr = cs.has_value() ? cs->data_ptr<float>() : nullptr;
 
     hipLaunchKernelGGL(target_scan_kernel, dim3(grid), dim3(block), 0, stream,
-        values.data_ptr<float>(), returns.data_ptr<float>(), rew_ptr,
+        values.data_ptr<float>(), ret_boot.data_ptr<float>(), rew_ptr,
         lambda_.data_ptr<float>(), rho_ptr, cs_ptr,
         targets.data_ptr<float>(), adv.data_ptr<float>(),
         n_bp, T, P, (float)gamma, (int)kind);
