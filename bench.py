@@ -124,10 +124,22 @@ def main():
         print('# prefill: %d episodes in %.1fs' %
               (len(trainer.episodes), time.time() - t0), flush=True)
 
+    phase_t = {'actor': 0.0, 'batch_wait': 0.0, 'train': 0.0, 'n': 0}
+
     def one_step():
+        t0 = time.time()
         frames = pump_actor(ACTOR_VEC_STEPS)
+        t1 = time.time()
         batch = batcher.batch()
+        t2 = time.time()
         losses, dcnt = trainer.train_step(batch)
+        if use_cuda:
+            torch.cuda.synchronize()
+        t3 = time.time()
+        phase_t['actor'] += t1 - t0
+        phase_t['batch_wait'] += t2 - t1
+        phase_t['train'] += t3 - t2
+        phase_t['n'] += 1
         return frames, losses
 
     # ---- warmup (untimed) ----
@@ -164,6 +176,10 @@ def main():
     samples_per_sec = steps_per_sec * cli.batch_size * cli.forward_steps * n_gpus
 
     if rank == 0:
+        n = max(1, phase_t['n'])
+        print('# phase ms/step: actor=%.1f batch_wait=%.1f train=%.1f' %
+              (1000 * phase_t['actor'] / n, 1000 * phase_t['batch_wait'] / n,
+               1000 * phase_t['train'] / n), file=__import__('sys').stderr, flush=True)
         result = {
             'metric': 'hungry_geese_selfplay_env_frames_per_sec',
             'value': round(frames_per_sec, 1),
