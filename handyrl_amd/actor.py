@@ -12,8 +12,13 @@ moment blocks by default — same interchange as the CPU worker path, minus
 the bz2 that a local learner does not need).
 """
 
+import os
+
 import numpy as np
 import torch
+
+# avoid per-shape MIOpen exhaustive search; find-db lookups only
+os.environ.setdefault('MIOPEN_FIND_MODE', 'FAST')
 
 from . import ops
 from .batch import pack_moments
@@ -64,13 +69,23 @@ class GeeseActorPool:
             return 0
 
         obs_sel = obs_u8[gi, pi]                        # (M, 17, 7, 11)
-        obs_t = torch.from_numpy(obs_sel)
+        M = len(gi)
+        # pad the inference batch to a fixed bucket: constant shapes keep
+        # one MIOpen solution per bucket and stay hipGraph-capturable
+        bucket = 256 * ((M + 255) // 256)
+        if bucket > M:
+            pad = np.zeros((bucket - M,) + obs_sel.shape[1:], dtype=obs_sel.dtype)
+            obs_in = np.concatenate([obs_sel, pad], axis=0)
+        else:
+            obs_in = obs_sel
+        obs_t = torch.from_numpy(obs_in)
         if self.device.type == 'cuda':
             obs_t = obs_t.to(self.device, non_blocking=True)
         obs_f = obs_t.float()
 
         policy, value = self._policy_forward(obs_f)
-        M, A = policy.shape
+        policy, value = policy[:M], value[:M]
+        A = policy.shape[1]
         if self.device.type == 'cuda':
             if self._zero_mask is None or self._zero_mask.shape[0] < M:
                 self._zero_mask = torch.zeros(max(M, 1), A, device=self.device)
