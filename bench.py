@@ -176,10 +176,17 @@ def main():
     samples_per_sec = steps_per_sec * cli.batch_size * cli.forward_steps * n_gpus
 
     if rank == 0:
+        import sys
         n = max(1, phase_t['n'])
         print('# phase ms/step: actor=%.1f batch_wait=%.1f train=%.1f' %
               (1000 * phase_t['actor'] / n, 1000 * phase_t['batch_wait'] / n,
-               1000 * phase_t['train'] / n), file=__import__('sys').stderr, flush=True)
+               1000 * phase_t['train'] / n), file=sys.stderr, flush=True)
+        tm = pool.timing
+        an = max(1, tm['n'])
+        print('# actor ms/vecstep: ' + ' '.join(
+            '%s=%.2f' % (k, 1000 * tm[k] / an) for k in
+            ('obs', 'fwd', 'sample', 'record', 'env', 'package')),
+            file=sys.stderr, flush=True)
         result = {
             'metric': 'hungry_geese_selfplay_env_frames_per_sec',
             'value': round(frames_per_sec, 1),

@@ -49,6 +49,8 @@ class GeeseActorPool:
         self.frames = 0          # env transitions executed (sum over games)
         self.episodes_done = 0
         self._zero_mask = None
+        self.timing = {'obs': 0.0, 'fwd': 0.0, 'sample': 0.0,
+                       'record': 0.0, 'env': 0.0, 'package': 0.0, 'n': 0}
 
     @torch.inference_mode()
     def _policy_forward(self, obs_f):
@@ -61,12 +63,17 @@ class GeeseActorPool:
 
     def step_once(self):
         """Advance every live game by one transition; returns #frames."""
+        import time
+        tm = self.timing
         vec = self.vec
+        t0 = time.time()
         obs_u8 = vec.observations()                     # (G, 4, 17, 7, 11)
+        tm['obs'] += time.time() - t0
         live = vec.alive & ~vec.over[:, None]
         gi, pi = np.nonzero(live)
         if len(gi) == 0:
             return 0
+        t0 = time.time()
 
         obs_sel = obs_u8[gi, pi]                        # (M, 17, 7, 11)
         M = len(gi)
@@ -86,6 +93,8 @@ class GeeseActorPool:
         policy, value = self._policy_forward(obs_f)
         policy, value = policy[:M], value[:M]
         A = policy.shape[1]
+        tm['fwd'] += time.time() - t0
+        t0 = time.time()
         if self.device.type == 'cuda':
             if self._zero_mask is None or self._zero_mask.shape[0] < M:
                 self._zero_mask = torch.zeros(max(M, 1), A, device=self.device)
@@ -101,6 +110,8 @@ class GeeseActorPool:
             probs = probs_full.gather(-1, actions_t.unsqueeze(-1)).squeeze(-1).numpy()
             values = value.squeeze(-1).numpy()
 
+        tm['sample'] += time.time() - t0
+        t0 = time.time()
         act_grid = np.zeros((self.n_games, N_PLAYERS), dtype=np.int32)
         act_grid[gi, pi] = actions
 
@@ -117,7 +128,11 @@ class GeeseActorPool:
             self.traj[g].append((live[g].copy(), obs_u8[g], act_row[g].copy(),
                                  prob_row[g].copy(), val_row[g].copy()))
 
+        tm['record'] += time.time() - t0
+        t0 = time.time()
         done = vec.step(act_grid)
+        tm['env'] += time.time() - t0
+        t0 = time.time()
         self.frames += int(game_has_live.sum())
 
         finished = np.nonzero(done)[0]
@@ -128,6 +143,8 @@ class GeeseActorPool:
                 self.traj[g] = []
             self.episodes_done += len(finished)
             vec.reset_games(finished)
+        tm['package'] += time.time() - t0
+        tm['n'] += 1
         return int(game_has_live.sum())
 
     def _package(self, g, outcome_row):
