@@ -100,9 +100,12 @@ class GeeseActorPool:
                 self._zero_mask = torch.zeros(max(M, 1), A, device=self.device)
             uniform = torch.rand(M, device=self.device)
             actions_t, probs_t = ops.masked_sample(policy, self._zero_mask[:M], uniform)
-            actions = actions_t.cpu().numpy()
-            probs = probs_t.cpu().numpy()
-            values = value.squeeze(-1).cpu().numpy()
+            # ONE device-to-host transfer for the step's three result vectors
+            packed = torch.cat([actions_t.float().unsqueeze(1),
+                                probs_t.unsqueeze(1), value], dim=1).cpu().numpy()
+            actions = packed[:, 0].astype(np.int64)
+            probs = packed[:, 1]
+            values = packed[:, 2]
         else:
             probs_full = torch.softmax(policy, dim=-1)
             actions_t = torch.multinomial(probs_full, 1).squeeze(-1)
