@@ -13,6 +13,7 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 from ..environment import BaseEnvironment
+from ..models.common import apply_bn
 
 # the eight winning index-triples of a 3x3 board (cells are x*3+y)
 _LINES = [
@@ -31,8 +32,7 @@ class ConvBN(nn.Module):
         self.bn = nn.BatchNorm2d(ch_out) if bn else None
 
     def forward(self, x):
-        h = self.conv(x)
-        return h if self.bn is None else self.bn(h)
+        return apply_bn(self.bn, self.conv(x))
 
 
 class ConvHead(nn.Module):

@@ -200,6 +200,8 @@ class Trainer:
         self.episodes = episodes if episodes is not None else EpisodeBuffer(args)
         self.template_model = copy.deepcopy(model).cpu()
         self.model = model.to(self.device)
+        if self.device.type == 'cuda' and getattr(model, 'prefers_channels_last', False):
+            self.model = self.model.to(memory_format=torch.channels_last)
         self.params = list(self.model.parameters())
         self.reducer = hdist.GradReducer(self.params)
 
