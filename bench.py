@@ -142,6 +142,14 @@ def main():
         phase_t['n'] += 1
         return frames, losses
 
+    # ---- capture the train step as a hipGraph (fixed shapes) ----
+    if use_cuda and os.environ.get('HANDYRL_NO_GRAPHS') != '1':
+        example = batcher.batch()
+        ok = trainer.enable_cuda_graph(example)
+        if rank == 0:
+            print('# train-step hipGraph: %s' % ('captured' if ok else 'EAGER'),
+                  flush=True)
+
     # ---- warmup (untimed) ----
     for _ in range(cli.warmup):
         one_step()

@@ -32,7 +32,7 @@ class GeeseActorPool:
     """Self-play actor pool for Hungry Geese on one GPU."""
 
     def __init__(self, model, args, n_games=256, device=None, seed=0,
-                 store_uint8_obs=True):
+                 store_uint8_obs=True, use_graphs=True):
         self.args = args
         self.device = device if device is not None else (
             torch.device('cuda') if torch.cuda.is_available() else torch.device('cpu'))
@@ -49,6 +49,10 @@ class GeeseActorPool:
         self.frames = 0          # env transitions executed (sum over games)
         self.episodes_done = 0
         self._zero_mask = None
+        self.graphed = None
+        if use_graphs and self.device.type == 'cuda':
+            from .hipgraph import GraphedActorForward
+            self.graphed = GraphedActorForward(model, self.device)
         self.timing = {'obs': 0.0, 'fwd': 0.0, 'sample': 0.0,
                        'record': 0.0, 'env': 0.0, 'package': 0.0, 'n': 0}
 
@@ -77,25 +81,30 @@ class GeeseActorPool:
 
         obs_sel = obs_u8[gi, pi]                        # (M, 17, 7, 11)
         M = len(gi)
-        # pad the inference batch to a fixed bucket: constant shapes keep
-        # one MIOpen solution per bucket and stay hipGraph-capturable
-        bucket = 256 * ((M + 255) // 256)
-        if bucket > M:
-            pad = np.zeros((bucket - M,) + obs_sel.shape[1:], dtype=obs_sel.dtype)
-            obs_in = np.concatenate([obs_sel, pad], axis=0)
-        else:
-            obs_in = obs_sel
-        obs_t = torch.from_numpy(obs_in)
-        if self.device.type == 'cuda':
-            obs_t = obs_t.to(self.device, non_blocking=True)
-        obs_f = obs_t.float()
-
-        policy, value = self._policy_forward(obs_f)
-        policy, value = policy[:M], value[:M]
-        A = policy.shape[1]
-        tm['fwd'] += time.time() - t0
-        t0 = time.time()
-        if self.device.type == 'cuda':
+        if self.graphed is not None:
+            # hipGraph path: one static-buffer H2D copy + one graph replay
+            # (forward + sample fused) + one packed D2H readback
+            packed = self.graphed.run(torch.from_numpy(obs_sel)).cpu().numpy()
+            tm['fwd'] += time.time() - t0
+            t0 = time.time()
+            actions = packed[:, 0].astype(np.int64)
+            probs = packed[:, 1]
+            values = packed[:, 2]
+        elif self.device.type == 'cuda':
+            # eager GPU path: pad to a fixed bucket so MIOpen keeps one
+            # solution per shape
+            bucket = 256 * ((M + 255) // 256)
+            if bucket > M:
+                pad = np.zeros((bucket - M,) + obs_sel.shape[1:], dtype=obs_sel.dtype)
+                obs_in = np.concatenate([obs_sel, pad], axis=0)
+            else:
+                obs_in = obs_sel
+            obs_t = torch.from_numpy(obs_in).to(self.device, non_blocking=True)
+            policy, value = self._policy_forward(obs_t.float())
+            policy, value = policy[:M], value[:M]
+            A = policy.shape[1]
+            tm['fwd'] += time.time() - t0
+            t0 = time.time()
             if self._zero_mask is None or self._zero_mask.shape[0] < M:
                 self._zero_mask = torch.zeros(max(M, 1), A, device=self.device)
             uniform = torch.rand(M, device=self.device)
@@ -107,6 +116,10 @@ class GeeseActorPool:
             probs = packed[:, 1]
             values = packed[:, 2]
         else:
+            # CPU path (tests / debugging)
+            policy, value = self._policy_forward(torch.from_numpy(obs_sel).float())
+            tm['fwd'] += time.time() - t0
+            t0 = time.time()
             probs_full = torch.softmax(policy, dim=-1)
             actions_t = torch.multinomial(probs_full, 1).squeeze(-1)
             actions = actions_t.numpy()

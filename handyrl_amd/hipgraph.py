@@ -1,0 +1,140 @@
+"""hipGraph capture for the launch-bound hot loops.
+
+The learner step (a ~300-kernel chain of small convs, elementwise loss
+math, fused scans and the Adam update at these model sizes) and the actor
+forward are dominated by per-launch gaps on MI355X, not kernel time
+(profiles/bench_kernel_stats_r01.txt) — so both are captured once into a
+hipGraph and replayed, with inputs staged through static device buffers.
+
+Constraints honored here:
+* no host syncs inside capture (compose_losses keeps dcnt as a tensor);
+* Adam runs in capturable mode (device-side step counters);
+* static shapes (the bench batch is fixed; actor batches are padded to
+  fixed buckets by GeeseActorPool);
+* the RCCL gradient all-reduce is captured too (graph-capturable on ROCm);
+  if capture fails anywhere the caller falls back to the eager path.
+"""
+
+import torch
+import torch.nn as nn
+
+from .util import map_r, bimap_r
+
+
+def _copy_into(static, src, non_blocking=True):
+    bimap_r(static, src, lambda dst, s: dst.copy_(s, non_blocking=non_blocking))
+
+
+class GraphedTrainStep:
+    """Whole-train-step capture: forward + loss + backward + all-reduce +
+    grad-clip + Adam, replayed per step."""
+
+    def __init__(self, trainer, example_batch_cpu, warmup_iters=3):
+        from .train import compute_loss
+        self._compute_loss = compute_loss
+        self.trainer = trainer
+        assert trainer.device.type == 'cuda'
+        assert trainer.wrapped_model.init_hidden([1, 1]) is None, \
+            'graphed train step supports feed-forward models only'
+
+        self.static = map_r(example_batch_cpu,
+                            lambda t: t.to(trainer.device).clone())
+
+        for group in trainer.optimizer.param_groups:
+            group['capturable'] = True
+
+        trainer.model.train()
+        stream = torch.cuda.Stream()
+        stream.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(stream):
+            for _ in range(warmup_iters):
+                self._run()
+        torch.cuda.current_stream().wait_stream(stream)
+
+        self.graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.graph):
+            self.losses, self.dcnt = self._run()
+
+    def _run(self):
+        tr = self.trainer
+        if tr.use_amp:
+            with torch.autocast('cuda', dtype=torch.bfloat16):
+                losses, dcnt = self._compute_loss(self.static, tr.wrapped_model,
+                                                  None, tr.args)
+        else:
+            losses, dcnt = self._compute_loss(self.static, tr.wrapped_model,
+                                              None, tr.args)
+        tr.optimizer.zero_grad(set_to_none=False)
+        losses['total'].backward()
+        tr.reducer.allreduce_()
+        nn.utils.clip_grad_norm_(tr.params, 4.0)
+        tr.optimizer.step()
+        return losses, dcnt
+
+    def step(self, batch_cpu):
+        """Copy the CPU batch into the static buffers and replay.
+
+        Returns (losses, dcnt) as DEVICE tensors that alias the graph's
+        static outputs — read them (.item()) before the next replay."""
+        _copy_into(self.static, batch_cpu)
+        self.graph.replay()
+        self.trainer.steps += 1
+        return self.losses, self.dcnt
+
+
+class GraphedActorForward:
+    """Per-bucket capture of {uint8 obs -> bf16 forward -> masked softmax
+    sample -> packed (action, prob, value)} for the GPU actor pool."""
+
+    def __init__(self, model, device, warmup_iters=2):
+        from . import ops
+        self._ops = ops
+        self.model = model
+        self.device = device
+        self.graphs = {}
+
+    def _fwd_sample(self, obs_u8, zero_mask):
+        with torch.no_grad():
+            obs_f = obs_u8.float()
+            with torch.autocast('cuda', dtype=torch.bfloat16):
+                out = self.model(obs_f, None)
+            policy = out['policy'].float()
+            value = out['value'].float()
+            uniform = torch.rand(policy.shape[0], device=self.device)
+            actions, probs = self._ops.masked_sample(policy, zero_mask, uniform)
+            return torch.cat([actions.float().unsqueeze(1),
+                              probs.unsqueeze(1), value], dim=1)
+
+    def _capture(self, bucket, n_actions):
+        was_training = self.model.training
+        self.model.eval()
+        static_obs = torch.zeros(bucket, 17, 7, 11, dtype=torch.uint8,
+                                 device=self.device)
+        zero_mask = torch.zeros(bucket, n_actions, device=self.device)
+        stream = torch.cuda.Stream()
+        stream.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(stream):
+            for _ in range(2):
+                self._fwd_sample(static_obs, zero_mask)
+        torch.cuda.current_stream().wait_stream(stream)
+        graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(graph):
+            packed = self._fwd_sample(static_obs, zero_mask)
+        if was_training:
+            self.model.train()
+        self.graphs[bucket] = (graph, static_obs, packed)
+
+    def run(self, obs_u8_cpu, n_actions=4):
+        """obs_u8_cpu: torch uint8 tensor (M, 17, 7, 11) on CPU (M <= bucket
+        after caller padding decides the bucket). Returns packed (M, 3) on
+        device."""
+        M = obs_u8_cpu.shape[0]
+        bucket = 256 * ((M + 255) // 256)
+        if bucket not in self.graphs:
+            self._capture(bucket, n_actions)
+        graph, static_obs, packed = self.graphs[bucket]
+        static_obs[:M].copy_(obs_u8_cpu, non_blocking=True)
+        # rows [M:bucket) keep stale data; every op is row-independent and
+        # the outputs are sliced to [:M]
+        graph.replay()
+        return packed[:M]

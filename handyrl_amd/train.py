@@ -107,7 +107,7 @@ def compose_losses(outputs, log_selected_policies, total_advantages, targets, ba
     omasks = batch['observation_mask']
 
     losses = {}
-    dcnt = tmasks.sum().item()
+    dcnt = tmasks.sum()          # tensor: stays device-side (hipGraph-safe)
 
     losses['p'] = (-log_selected_policies * total_advantages).mul(tmasks).sum()
     if 'value' in outputs:
@@ -234,8 +234,27 @@ class Trainer:
         snap.eval()
         return snap
 
+    def enable_cuda_graph(self, example_batch_cpu):
+        """Capture the whole train step into a hipGraph (FF models, fixed
+        batch shapes).  Returns True on success; on failure the eager path
+        stays active."""
+        if self.device.type != 'cuda':
+            return False
+        try:
+            from .hipgraph import GraphedTrainStep
+            self.graphed_step = GraphedTrainStep(self, example_batch_cpu)
+            return True
+        except Exception as e:        # noqa: BLE001 - deliberate fallback
+            print('cuda-graph capture failed, using eager train step: %r' % (e,))
+            self.graphed_step = None
+            return False
+
     def train_step(self, batch):
         """One optimizer step on an already-built CPU batch dict."""
+        graphed = getattr(self, 'graphed_step', None)
+        if graphed is not None:
+            losses_t, dcnt_t = graphed.step(batch)
+            return losses_t, float(dcnt_t)
         batch_size = batch['value'].size(0)
         player_count = batch['value'].size(2)
         hidden = self.wrapped_model.init_hidden([batch_size, player_count])
@@ -256,7 +275,7 @@ class Trainer:
         nn.utils.clip_grad_norm_(self.params, 4.0)
         self.optimizer.step()
         self.steps += 1
-        return losses, dcnt
+        return losses, float(dcnt)
 
     def train(self):
         if self.optimizer is None:
