@@ -1,0 +1,62 @@
+"""Model-block tests: the primitive BatchNorm (GPU path) must match
+nn.BatchNorm2d bit-for-tolerance in both modes, including running-stat
+updates and gradients."""
+
+import copy
+
+import pytest
+import torch
+import torch.nn as nn
+
+from handyrl_amd.models.common import primitive_bn
+
+
+@pytest.mark.parametrize('training', [True, False])
+def test_primitive_bn_matches_stock(training):
+    torch.manual_seed(0)
+    bn_ref = nn.BatchNorm2d(32)
+    bn_ref.weight.data.uniform_(0.5, 1.5)
+    bn_ref.bias.data.uniform_(-0.5, 0.5)
+    bn_ref.running_mean.uniform_(-1, 1)
+    bn_ref.running_var.uniform_(0.5, 2.0)
+    bn_mine = copy.deepcopy(bn_ref)
+    bn_ref.train(training)
+    bn_mine.train(training)
+
+    x = torch.randn(16, 32, 7, 11)
+    x_ref = x.clone().requires_grad_(True)
+    x_mine = x.clone().requires_grad_(True)
+
+    y_ref = bn_ref(x_ref)
+    y_mine = primitive_bn(bn_mine, x_mine)
+    torch.testing.assert_close(y_mine, y_ref, rtol=1e-5, atol=1e-5)
+
+    torch.testing.assert_close(bn_mine.running_mean, bn_ref.running_mean,
+                               rtol=1e-6, atol=1e-6)
+    torch.testing.assert_close(bn_mine.running_var, bn_ref.running_var,
+                               rtol=1e-6, atol=1e-6)
+    assert bn_mine.num_batches_tracked == bn_ref.num_batches_tracked
+
+    g = torch.randn_like(y_ref)
+    y_ref.backward(g)
+    y_mine.backward(g)
+    torch.testing.assert_close(x_mine.grad, x_ref.grad, rtol=1e-4, atol=1e-5)
+    torch.testing.assert_close(bn_mine.weight.grad, bn_ref.weight.grad,
+                               rtol=1e-4, atol=1e-5)
+    torch.testing.assert_close(bn_mine.bias.grad, bn_ref.bias.grad,
+                               rtol=1e-4, atol=1e-5)
+
+
+def test_primitive_bn_momentum_sequence():
+    """Running stats track across several training steps."""
+    torch.manual_seed(1)
+    bn_ref = nn.BatchNorm2d(8)
+    bn_mine = copy.deepcopy(bn_ref)
+    for _ in range(5):
+        x = torch.randn(4, 8, 5, 5)
+        bn_ref(x)
+        primitive_bn(bn_mine, x)
+    torch.testing.assert_close(bn_mine.running_mean, bn_ref.running_mean,
+                               rtol=1e-6, atol=1e-6)
+    torch.testing.assert_close(bn_mine.running_var, bn_ref.running_var,
+                               rtol=1e-6, atol=1e-6)
