@@ -116,7 +116,11 @@ def compose_losses(outputs, log_selected_policies, total_advantages, targets, ba
         losses['r'] = F.smooth_l1_loss(outputs['return'], targets['return'],
                                        reduction='none').mul(omasks).sum()
 
-    entropy = torch_dist.Categorical(logits=outputs['policy']).entropy().mul(tmasks.sum(-1))
+    # entropy = -(p * log p) over the (finitely) masked logits; written in
+    # primitives because Categorical's arg validation host-syncs, which
+    # breaks hipGraph capture
+    logp = F.log_softmax(outputs['policy'], dim=-1)
+    entropy = -(logp.exp() * logp).sum(-1).mul(tmasks.sum(-1))
     losses['ent'] = entropy.sum()
 
     base_loss = losses['p'] + losses.get('v', 0) + losses.get('r', 0)
