@@ -45,6 +45,9 @@ class GeeseActorPool:
         self.compress_steps = args.get('compress_steps', 4)
         # per-game trajectory: list of (alive_mask, obs[4,...], act[4], prob[4], val[4])
         self.traj = [[] for _ in range(n_games)]
+        # all geese actions are always legal: one shared (read-only) zero mask
+        self._amask_shared = np.zeros(4, dtype=np.float32)
+        self._amask_shared.setflags(write=False)
         self.completed = []
         self.frames = 0          # env transitions executed (sum over games)
         self.episodes_done = 0
@@ -170,13 +173,15 @@ class GeeseActorPool:
             moment = {key: {p: None for p in range(N_PLAYERS)} for key in MOMENT_KEYS}
             turn = [int(p) for p in np.nonzero(alive_mask)[0]]
             for p in turn:
-                obs = obs_all[p]
+                # copy the seat's obs out of the step-wide array so the
+                # episode buffer doesn't pin every step's (G,4,...) array
+                obs = obs_all[p].copy()
                 moment['observation'][p] = obs if self.store_uint8_obs \
                     else obs.astype(np.float32)
                 moment['selected_prob'][p] = float(prob[p])
-                moment['action_mask'][p] = np.zeros(4, dtype=np.float32)
+                moment['action_mask'][p] = self._amask_shared
                 moment['action'][p] = int(act[p])
-                moment['value'][p] = np.array([val[p]], dtype=np.float32)
+                moment['value'][p] = [float(val[p])]
                 # geese have no immediate reward; returns stay at 0
             moment['turn'] = turn
             moments.append(moment)
