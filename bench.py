@@ -31,7 +31,7 @@ from handyrl_amd.batch import Batcher
 from handyrl_amd.models.geese_net import GeeseNet
 from handyrl_amd.train import Trainer
 
-N_ENVS = 256            # self-play games per GPU
+N_ENVS = 512            # self-play games per GPU (measured best 1-GPU throughput)
 ACTOR_VEC_STEPS = 16    # env transitions (per game) per learner step
 
 
@@ -121,8 +121,9 @@ def main():
     trainer.episodes.trim(args['maximum_episodes'])
     batcher.run()
     if rank == 0:
+        import sys
         print('# prefill: %d episodes in %.1fs' %
-              (len(trainer.episodes), time.time() - t0), flush=True)
+              (len(trainer.episodes), time.time() - t0), file=sys.stderr, flush=True)
 
     phase_t = {'actor': 0.0, 'batch_wait': 0.0, 'train': 0.0, 'n': 0}
 
@@ -147,8 +148,9 @@ def main():
         example = batcher.batch()
         ok = trainer.enable_cuda_graph(example)
         if rank == 0:
+            import sys
             print('# train-step hipGraph: %s' % ('captured' if ok else 'EAGER'),
-                  flush=True)
+                  file=sys.stderr, flush=True)
 
     # ---- warmup (untimed) ----
     for _ in range(cli.warmup):

@@ -21,8 +21,8 @@ import torch
 os.environ.setdefault('MIOPEN_FIND_MODE', 'FAST')
 
 from . import ops
-from .batch import pack_moments
 from .envs.vec_geese import GeeseVecEnv, N_PLAYERS
+from .envs.hungry_geese import MAX_STEPS
 
 MOMENT_KEYS = ('observation', 'selected_prob', 'action_mask', 'action',
                'value', 'reward', 'return')
@@ -43,11 +43,15 @@ class GeeseActorPool:
         self.gamma = args.get('gamma', 0.8)
         self.compress = args.get('compress_episodes', False)
         self.compress_steps = args.get('compress_steps', 4)
-        # per-game trajectory: list of (alive_mask, obs[4,...], act[4], prob[4], val[4])
-        self.traj = [[] for _ in range(n_games)]
-        # all geese actions are always legal: one shared (read-only) zero mask
-        self._amask_shared = np.zeros(4, dtype=np.float32)
-        self._amask_shared.setflags(write=False)
+        # columnar trajectory recording: struct-of-arrays ring per game
+        # (episodes stay columnar through the replay buffer and batch maker)
+        G, CAPT = n_games, MAX_STEPS
+        self.rec_obs = np.zeros((G, CAPT, N_PLAYERS, 17, 7, 11), dtype=np.uint8)
+        self.rec_alive = np.zeros((G, CAPT, N_PLAYERS), dtype=bool)
+        self.rec_act = np.zeros((G, CAPT, N_PLAYERS), dtype=np.int32)
+        self.rec_prob = np.zeros((G, CAPT, N_PLAYERS), dtype=np.float32)
+        self.rec_val = np.zeros((G, CAPT, N_PLAYERS), dtype=np.float32)
+        self.rec_len = np.zeros(G, dtype=np.int32)
         self.completed = []
         self.frames = 0          # env transitions executed (sum over games)
         self.episodes_done = 0
@@ -134,18 +138,21 @@ class GeeseActorPool:
         act_grid = np.zeros((self.n_games, N_PLAYERS), dtype=np.int32)
         act_grid[gi, pi] = actions
 
-        # record the step per game (only games with at least one live seat)
-        act_row = np.full((self.n_games, N_PLAYERS), -1, dtype=np.int32)
         prob_row = np.zeros((self.n_games, N_PLAYERS), dtype=np.float32)
         val_row = np.zeros((self.n_games, N_PLAYERS), dtype=np.float32)
-        act_row[gi, pi] = actions
         prob_row[gi, pi] = probs
         val_row[gi, pi] = values
 
+        # columnar recording: one fancy-indexed scatter per field
         game_has_live = live.any(axis=1)
-        for g in np.nonzero(game_has_live)[0]:
-            self.traj[g].append((live[g].copy(), obs_u8[g], act_row[g].copy(),
-                                 prob_row[g].copy(), val_row[g].copy()))
+        lg = np.nonzero(game_has_live)[0]
+        t_idx = self.rec_len[lg]
+        self.rec_obs[lg, t_idx] = obs_u8[lg]
+        self.rec_alive[lg, t_idx] = live[lg]
+        self.rec_act[lg, t_idx] = act_grid[lg]
+        self.rec_prob[lg, t_idx] = prob_row[lg]
+        self.rec_val[lg, t_idx] = val_row[lg]
+        self.rec_len[lg] += 1
 
         tm['record'] += time.time() - t0
         t0 = time.time()
@@ -159,7 +166,7 @@ class GeeseActorPool:
             outcomes = vec.outcomes(finished)
             for k, g in enumerate(finished):
                 self.completed.append(self._package(g, outcomes[k]))
-                self.traj[g] = []
+                self.rec_len[g] = 0
             self.episodes_done += len(finished)
             vec.reset_games(finished)
         tm['package'] += time.time() - t0
@@ -167,32 +174,21 @@ class GeeseActorPool:
         return int(game_has_live.sum())
 
     def _package(self, g, outcome_row):
-        """Build a reference-format episode dict from a finished game."""
-        moments = []
-        for alive_mask, obs_all, act, prob, val in self.traj[g]:
-            moment = {key: {p: None for p in range(N_PLAYERS)} for key in MOMENT_KEYS}
-            turn = [int(p) for p in np.nonzero(alive_mask)[0]]
-            for p in turn:
-                # copy the seat's obs out of the step-wide array so the
-                # episode buffer doesn't pin every step's (G,4,...) array
-                obs = obs_all[p].copy()
-                moment['observation'][p] = obs if self.store_uint8_obs \
-                    else obs.astype(np.float32)
-                moment['selected_prob'][p] = float(prob[p])
-                moment['action_mask'][p] = self._amask_shared
-                moment['action'][p] = int(act[p])
-                moment['value'][p] = [float(val[p])]
-                # geese have no immediate reward; returns stay at 0
-            moment['turn'] = turn
-            moments.append(moment)
-
+        """Slice the game's columnar recording into an episode dict."""
+        S = int(self.rec_len[g])
         job_args = {'player': list(range(N_PLAYERS)),
                     'model_id': {p: -1 for p in range(N_PLAYERS)}}
         return {
             'args': job_args,
-            'steps': len(moments),
+            'steps': S,
             'outcome': {p: float(outcome_row[p]) for p in range(N_PLAYERS)},
-            'moment': pack_moments(moments, self.compress_steps, compress=self.compress),
+            'columnar': True,
+            'n_actions': 4,
+            'obs': self.rec_obs[g, :S].copy(),
+            'alive': self.rec_alive[g, :S].copy(),
+            'action': self.rec_act[g, :S].copy(),
+            'prob': self.rec_prob[g, :S].copy(),
+            'value': self.rec_val[g, :S].copy(),
         }
 
     def harvest(self):

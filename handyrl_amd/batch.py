@@ -48,6 +48,50 @@ def pack_moments(moments, compress_steps, compress=True):
     return blocks
 
 
+def _episode_fields_columnar(ep, args):
+    """Per-episode batch fields from a COLUMNAR episode window.
+
+    Columnar episodes (produced by the GPU actor pool) carry plain arrays
+    (T,4,...) instead of per-step moment dicts: obs/alive/action/prob/value.
+    Semantics mirror the dict path: a dead seat contributes zero obs/value,
+    behavior prob 1, action 0 and a fully-illegal action mask.
+    """
+    alive = ep['alive']                          # (T, seats) bool
+    n_seats = alive.shape[1]
+    all_players = list(range(n_seats))
+    if not args['turn_based_training']:
+        players = [random.choice(all_players)]
+    elif args['observation']:
+        players = all_players
+    else:
+        raise NotImplementedError(
+            'columnar episodes support solo or full-observation training')
+    pl = np.array(players)
+    T, NP = alive.shape[0], len(players)
+    A = int(ep['n_actions'])
+
+    aliveP = alive[:, pl]                                        # (T, P)
+    obs = ep['obs'][:, pl] * aliveP[..., None, None, None].astype(ep['obs'].dtype)
+    prob = np.where(aliveP, ep['prob'][:, pl], 1.0)[..., None].astype(np.float64)
+    act = (ep['action'][:, pl] * aliveP).astype(np.int64)[..., None]
+    amask = np.full((T, NP, A), 1e32, dtype=np.float32)
+    amask[aliveP] = 0.0
+
+    v = (ep['value'][:, pl] * aliveP)[..., None].astype(np.float32)
+    rew = np.zeros((T, NP, 1), dtype=np.float32)
+    ret = np.zeros((T, NP, 1), dtype=np.float32)
+    oc = np.array([ep['outcome'][p] for p in players],
+                  dtype=np.float32).reshape(1, NP, -1)
+    emask = np.ones((T, 1, 1), dtype=np.float32)
+    tmask = aliveP[..., None].astype(np.float32)
+    omask = tmask.copy()
+    progress = np.arange(ep['start'], ep['end'],
+                         dtype=np.float32)[..., np.newaxis] / ep['total']
+    obs_zeros = np.zeros_like(ep['obs'][0, 0])
+    return (obs, obs_zeros, prob, v, act, oc, rew, ret, emask, tmask, omask,
+            amask, progress)
+
+
 def make_batch(episodes, args):
     """Assemble a (B, T, P, ...) training batch from sampled episode windows."""
     obss, datum = [], []
@@ -56,6 +100,36 @@ def make_batch(episodes, args):
         return a if a is not None else b
 
     for ep in episodes:
+        if ep.get('columnar'):
+            (obs, obs_zeros, prob, v, act, oc, rew, ret, emask, tmask, omask,
+             amask, progress) = _episode_fields_columnar(ep, args)
+            T = emask.shape[0]
+            batch_steps = args['burn_in_steps'] + args['forward_steps']
+            if T < batch_steps:
+                pad_b = args['burn_in_steps'] - (ep['train_start'] - ep['start'])
+                pad_a = batch_steps - T - pad_b
+                pad3 = lambda a, val=0: np.pad(
+                    a, [(pad_b, pad_a)] + [(0, 0)] * (a.ndim - 1),
+                    'constant', constant_values=val)
+                obs = pad3(obs)
+                prob = pad3(prob, val=1)
+                v = np.concatenate(
+                    [np.pad(v, [(pad_b, 0), (0, 0), (0, 0)], 'constant'),
+                     np.tile(oc, [pad_a, 1, 1])])
+                act = pad3(act)
+                rew = pad3(rew)
+                ret = pad3(ret)
+                emask = pad3(emask)
+                tmask = pad3(tmask)
+                omask = pad3(omask)
+                amask = pad3(amask, val=1e32)
+                progress = np.pad(progress, [(pad_b, pad_a), (0, 0)],
+                                  'constant', constant_values=1)
+            obss.append(obs)
+            datum.append((prob, v, act, oc, rew, ret, emask, tmask, omask,
+                          amask, progress))
+            continue
+
         moments = unpack_moments(ep, ep['start'], ep['end'])
         players = list(moments[0]['observation'].keys())
         if not args['turn_based_training']:     # solo training on one seat
@@ -191,6 +265,17 @@ class EpisodeBuffer:
         train_st = random.randrange(turn_candidates)
         st = max(0, train_st - args['burn_in_steps'])
         ed = min(train_st + args['forward_steps'], ep['steps'])
+        if ep.get('columnar'):
+            # columnar episodes: the window is an array view, no block math
+            return {
+                'args': ep['args'], 'outcome': ep['outcome'], 'columnar': True,
+                'n_actions': ep['n_actions'],
+                'obs': ep['obs'][st:ed], 'alive': ep['alive'][st:ed],
+                'action': ep['action'][st:ed], 'prob': ep['prob'][st:ed],
+                'value': ep['value'][st:ed],
+                'start': st, 'end': ed, 'train_start': train_st,
+                'total': ep['steps'],
+            }
         st_block = st // args['compress_steps']
         ed_block = (ed - 1) // args['compress_steps'] + 1
         return {
@@ -215,7 +300,8 @@ class Batcher:
             yield [self.buffer.select_episode() for _ in range(self.args['batch_size'])]
 
     def _worker(self, conn, bid):
-        print('started batcher %d' % bid)
+        import sys
+        print('started batcher %d' % bid, file=sys.stderr)
         while True:
             episodes = conn.recv()
             conn.send(make_batch(episodes, self.args))

@@ -249,7 +249,9 @@ class Trainer:
             self.graphed_step = GraphedTrainStep(self, example_batch_cpu)
             return True
         except Exception as e:        # noqa: BLE001 - deliberate fallback
-            print('cuda-graph capture failed, using eager train step: %r' % (e,))
+            import sys
+            print('cuda-graph capture failed, using eager train step: %r' % (e,),
+                  file=sys.stderr)
             self.graphed_step = None
             return False
 

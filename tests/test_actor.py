@@ -1,7 +1,10 @@
 """GPU actor pool tests (CPU execution here; the same code runs batched
-bf16 inference on an MI355X)."""
+bf16 inference + hipGraphs on an MI355X)."""
+
+import random
 
 import numpy as np
+import pytest
 import torch
 
 from handyrl_amd.actor import GeeseActorPool
@@ -29,25 +32,86 @@ def test_actor_pool_generates_valid_episodes():
     model = GeeseNet(layers=2)
     model.eval()
     pool = GeeseActorPool(model, args, n_games=8, device=torch.device('cpu'), seed=1)
-    for _ in range(220):
+    for _ in range(250):
         pool.step_once()
         if pool.episodes_done >= 4:
             break
     episodes = pool.harvest()
     assert len(episodes) >= 4
     for ep in episodes:
-        assert ep['steps'] >= 1
+        S = ep['steps']
+        assert S >= 1
+        assert ep['columnar']
         assert set(ep['outcome'].keys()) == {0, 1, 2, 3}
         assert abs(sum(ep['outcome'].values())) < 1e-6    # pairwise zero sum
-        moments = [m for block in ep['moment'] for m in block]
-        assert len(moments) == ep['steps']
-        m0 = moments[0]
-        assert sorted(m0['turn']) == m0['turn']
-        for p in m0['turn']:
-            assert m0['observation'][p].shape == (17, 7, 11)
-            assert m0['observation'][p].dtype == np.uint8
-            assert 0 <= m0['action'][p] < 4
-            assert 0 < m0['selected_prob'][p] <= 1
+        assert ep['obs'].shape == (S, 4, 17, 7, 11) and ep['obs'].dtype == np.uint8
+        assert ep['alive'].shape == (S, 4)
+        assert ep['alive'][0].all()                        # all alive at start
+        live = ep['alive']
+        assert ((ep['action'] >= 0) & (ep['action'] < 4))[live].all()
+        assert (ep['prob'][live] > 0).all() and (ep['prob'][live] <= 1).all()
+        # dead seats never revive
+        for p in range(4):
+            col = live[:, p].astype(int)
+            assert (np.diff(col) <= 0).all()
+
+
+def _dict_episode_from_columnar(ep):
+    """Reference-format (moment dict) episode with identical content."""
+    moments = []
+    S = ep['steps']
+    for t in range(S):
+        keys = ('observation', 'selected_prob', 'action_mask', 'action',
+                'value', 'reward', 'return')
+        moment = {k: {p: None for p in range(4)} for k in keys}
+        turn = [p for p in range(4) if ep['alive'][t, p]]
+        for p in turn:
+            moment['observation'][p] = ep['obs'][t, p]
+            moment['selected_prob'][p] = float(ep['prob'][t, p])
+            moment['action_mask'][p] = np.zeros(4, dtype=np.float32)
+            moment['action'][p] = int(ep['action'][t, p])
+            moment['value'][p] = [float(ep['value'][t, p])]
+        moment['turn'] = turn
+        moments.append(moment)
+    return {
+        'args': ep['args'], 'steps': S, 'outcome': ep['outcome'],
+        'moment': [moments[i:i + 4] for i in range(0, S, 4)],
+    }
+
+
+def test_columnar_batch_matches_dict_batch():
+    """make_batch must produce identical tensors from the columnar fast
+    path and the reference moment-dict path."""
+    args = _args(forward_steps=6)
+    model = GeeseNet(layers=1)
+    model.eval()
+    pool = GeeseActorPool(model, args, n_games=4, device=torch.device('cpu'), seed=3)
+    while pool.episodes_done < 3:
+        pool.step_once()
+    cols = pool.harvest()[:3]
+
+    buf_c = EpisodeBuffer(args)
+    buf_c.extend(cols)
+    buf_d = EpisodeBuffer(args)
+    buf_d.extend([_dict_episode_from_columnar(ep) for ep in cols])
+
+    random.seed(42)
+    sel_c = [buf_c.select_episode() for _ in range(4)]
+    random.seed(42)
+    sel_d = [buf_d.select_episode() for _ in range(4)]
+    for c, d in zip(sel_c, sel_d):
+        assert (c['start'], c['end'], c['train_start']) == \
+            (d['start'], d['end'], d['train_start'])
+
+    random.seed(7)
+    batch_c = make_batch(sel_c, args)
+    random.seed(7)
+    batch_d = make_batch(sel_d, args)
+    for key in batch_d:
+        tc, td = batch_c[key], batch_d[key]
+        assert tc.shape == td.shape, key
+        torch.testing.assert_close(tc.double(), td.double(), rtol=1e-6,
+                                   atol=1e-6, msg=lambda m: '%s: %s' % (key, m))
 
 
 def test_actor_episodes_train():
