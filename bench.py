@@ -134,6 +134,7 @@ def main():
         batch = batcher.batch()
         t2 = time.time()
         losses, dcnt = trainer.train_step(batch)
+        pool.refresh_weights()          # re-fold BN into the MFMA actor path
         if use_cuda:
             torch.cuda.synchronize()
         t3 = time.time()

@@ -57,9 +57,14 @@ class GeeseActorPool:
         self.episodes_done = 0
         self._zero_mask = None
         self.graphed = None
+        self.fused = None
+        if self.device.type == 'cuda' and os.environ.get('HANDYRL_NO_FUSED') != '1' \
+                and hasattr(model, 'stem') and ops.available():
+            from .models.geese_net import GeeseFusedEval
+            self.fused = GeeseFusedEval(model, self.device)
         if use_graphs and self.device.type == 'cuda':
             from .hipgraph import GraphedActorForward
-            self.graphed = GraphedActorForward(model, self.device)
+            self.graphed = GraphedActorForward(model, self.device, fused=self.fused)
         self.timing = {'obs': 0.0, 'fwd': 0.0, 'sample': 0.0,
                        'record': 0.0, 'env': 0.0, 'package': 0.0, 'n': 0}
 
@@ -190,6 +195,12 @@ class GeeseActorPool:
             'prob': self.rec_prob[g, :S].copy(),
             'value': self.rec_val[g, :S].copy(),
         }
+
+    def refresh_weights(self):
+        """Re-fold BN into the packed MFMA weights after an optimizer step
+        (the hand-written inference path then runs <=1 step stale)."""
+        if self.fused is not None:
+            self.fused.refresh()
 
     def harvest(self):
         """Return and clear the finished-episode list."""

@@ -86,18 +86,22 @@ class GraphedActorForward:
     """Per-bucket capture of {uint8 obs -> bf16 forward -> masked softmax
     sample -> packed (action, prob, value)} for the GPU actor pool."""
 
-    def __init__(self, model, device, warmup_iters=2):
+    def __init__(self, model, device, warmup_iters=2, fused=None):
         from . import ops
         self._ops = ops
         self.model = model
         self.device = device
+        self.fused = fused        # GeeseFusedEval: hand-written MFMA path
         self.graphs = {}
 
     def _fwd_sample(self, obs_u8, zero_mask):
         with torch.no_grad():
-            obs_f = obs_u8.float()
-            with torch.autocast('cuda', dtype=torch.bfloat16):
-                out = self.model(obs_f, None)
+            if self.fused is not None:
+                out = self.fused.forward(obs_u8)
+            else:
+                obs_f = obs_u8.float()
+                with torch.autocast('cuda', dtype=torch.bfloat16):
+                    out = self.model(obs_f, None)
             policy = out['policy'].float()
             value = out['value'].float()
             uniform = torch.rand(policy.shape[0], device=self.device)

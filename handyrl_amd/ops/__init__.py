@@ -89,6 +89,57 @@ def target_scan(algorithm, values, returns, rewards, lambda_, gamma, rhos, cs):
     return targets, adv
 
 
+def mfma_probe(a_bf16, b_bf16):
+    """Test-only: D(16,16) = A(16,32) @ B(32,16) via one MFMA."""
+    return require().mfma_probe(a_bf16.contiguous(), b_bf16.contiguous())
+
+
+def obs_to_nhwc(obs_u8):
+    """(N,17,7,11) uint8 -> (N,77,32) bf16 NHWC, channels 17..31 zero."""
+    return require().obs_to_nhwc(obs_u8.reshape(obs_u8.shape[0], 17, 77).contiguous())
+
+
+def torus_conv_fused(x, wfrag, shift, nbr, residual, relu):
+    """y = act(torus_conv3x3(x) * scale + shift [+ x]) on NHWC bf16.
+
+    wfrag: (9,2,4,16,8) bf16 from pack_torus_weights; shift fp32 (32,);
+    nbr: (77,9) int32 wrap-around neighbor table (see torus_neighbor_table).
+    """
+    return require().torus_conv_fused(x, wfrag, shift, nbr, residual, relu)
+
+
+def torus_neighbor_table(device=None):
+    """(77, 9) int32: nbr[cell][ky*3+kx] on the 7x11 torus (conv2d
+    cross-correlation tap order, padding=1 circular)."""
+    rows, cols = 7, 11
+    tbl = torch.empty(rows * cols, 9, dtype=torch.int32)
+    for r in range(rows):
+        for c in range(cols):
+            for ky in range(3):
+                for kx in range(3):
+                    rr = (r + ky - 1) % rows
+                    cc = (c + kx - 1) % cols
+                    tbl[r * cols + c, ky * 3 + kx] = rr * cols + cc
+    return tbl.to(device) if device is not None else tbl
+
+
+def pack_torus_weights(conv_weight, scale):
+    """Pack (32, Cin<=32, 3, 3) conv weights (pre-scaled per out-channel)
+    into the MFMA B-fragment layout the fused kernel reads:
+    frag[tap][cotile][khi][lane_lo][e] = W[k = khi*8+e (ci), co = cotile*16+lane_lo],
+    with k tap-major (k = tap*32 + ci), ci zero-padded to 32."""
+    co, ci = conv_weight.shape[0], conv_weight.shape[1]
+    assert co == 32 and ci <= 32
+    w = conv_weight.float() * scale.float().view(-1, 1, 1, 1)
+    wt = w.permute(2, 3, 1, 0).reshape(9, ci, co)          # (tap, ci, co)
+    if ci < 32:
+        pad = torch.zeros(9, 32 - ci, co, device=w.device, dtype=w.dtype)
+        wt = torch.cat([wt, pad], dim=1)
+    wt = wt.view(9, 4, 8, 2, 16)                           # (tap, khi, e, cotile, lo)
+    frag = wt.permute(0, 3, 1, 4, 2).contiguous()          # (tap, cotile, khi, lo, e)
+    return frag.to(torch.bfloat16)
+
+
 def masked_sample(logits, action_mask, uniform):
     """Sample one action per row from softmax(logits - action_mask).
 

@@ -16,6 +16,20 @@
 
 namespace {
 
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+DEV_INLINE float bf2f(short s) {
+    unsigned u = ((unsigned)(unsigned short)s) << 16;
+    return __builtin_bit_cast(float, u);
+}
+
+DEV_INLINE short f2bf(float f) {   // round-to-nearest-even
+    unsigned u = __builtin_bit_cast(unsigned, f);
+    u += 0x7fffu + ((u >> 16) & 1u);
+    return (short)(u >> 16);
+}
+
 enum ScanKind { KIND_TD = 0, KIND_UPGO = 1, KIND_VTRACE = 2 };
 
 // One lane scans one (b, p) trajectory serially over T.  Work per call is
@@ -120,7 +134,148 @@ __global__ void masked_sample_kernel(
     prob[i] = pr;
 }
 
+// --- MFMA fragment-layout probe (test harness for the conv kernel) ------
+// Computes D(16x16) = A(16x32) @ B(32x16) with one v_mfma_f32_16x16x32_bf16
+// using the assumed lane->fragment mapping:
+//   A: lane l holds A[m = l&15][k = (l>>4)*8 + i], i in [0,8)
+//   B: lane l holds B[k = (l>>4)*8 + i][n = l&15]
+//   D: lane l, reg r -> D[row = (l>>4)*4 + r][col = l&15]
+// Verified against torch.matmul on gfx950 by tests/test_gpu.py.
+__global__ void mfma_probe_kernel(const short* __restrict__ A,
+                                  const short* __restrict__ B,
+                                  float* __restrict__ D) {
+    const int l = threadIdx.x;
+    const int khi = l >> 4, lo = l & 15;
+    bf16x8 a, b;
+    for (int i = 0; i < 8; ++i) {
+        a[i] = __builtin_bit_cast(__bf16, A[lo * 32 + khi * 8 + i]);
+        b[i] = __builtin_bit_cast(__bf16, B[(khi * 8 + i) * 16 + lo]);
+    }
+    f32x4 c = {0.f, 0.f, 0.f, 0.f};
+    c = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
+    for (int r = 0; r < 4; ++r)
+        D[(khi * 4 + r) * 16 + lo] = c[r];
+}
+
+// --- obs (N,17,7,11) uint8 NCHW -> (N,77,32) bf16 NHWC, channels padded --
+__global__ void obs_to_nhwc_kernel(const unsigned char* __restrict__ obs,
+                                   short* __restrict__ out, long total) {
+    const long idx = (long)blockIdx.x * blockDim.x + threadIdx.x;  // N*77*32
+    if (idx >= total) return;
+    const int c = idx & 31;
+    const long pc = idx >> 5;
+    const long n = pc / 77;
+    const int cell = pc % 77;
+    float v = (c < 17) ? (float)obs[(n * 17 + c) * 77 + cell] : 0.f;
+    out[idx] = f2bf(v);
+}
+
+// --- fused torus conv block: y = act(conv3x3_wrap(x)*scale + shift [+ x]) -
+// Implicit GEMM on MFMA, NHWC bf16 activations, BN folded into the packed
+// weights (wfrag) and the per-channel shift.  The 3x3 wrap-around gather is
+// the A-operand load itself: k is tap-major (k = tap*32 + ci), so one
+// A-fragment = 16 contiguous bytes of the neighbor cell's channel vector.
+// Replaces the reference eval chain circular-pad + conv + BN + (+x) + relu
+// (reference envs/kaggle/hungry_geese.py:30-35, 48-51) with ONE kernel.
+//
+// Geometry: 256 threads = 4 waves; each wave computes a 16-position x 32-
+// channel output tile (2 MFMA accumulators), block covers 64 positions;
+// 9 taps -> 18 MFMA per wave.  W (18 KB) streams from L2 (shared by all
+// blocks); per-lane fragment rows are packed host-side so every read is a
+// contiguous 16-byte load.
+__global__ __launch_bounds__(256) void torus_conv_fused_kernel(
+    const short* __restrict__ x,        // (N,77,32) bf16
+    const short* __restrict__ wfrag,    // (9,2,4,16,8) bf16, BN-folded
+    const float* __restrict__ shift,    // (32,)
+    const int* __restrict__ nbr,        // (77,9) wrap-around neighbor cells
+    short* __restrict__ y,              // (N,77,32) bf16
+    long total_pos,                     // N*77
+    int add_residual, int apply_relu) {
+    const int wid = threadIdx.x >> 6;
+    const int lane = threadIdx.x & 63;
+    const int khi = lane >> 4, lo = lane & 15;
+    const long pos_base = (long)blockIdx.x * 64 + wid * 16;
+
+    const long p = pos_base + lo;                  // this lane's A row
+    const long pp = (p < total_pos) ? p : 0;
+    const long n = pp / 77;
+    const int cell = (int)(pp % 77);
+
+    const bf16x8* wf = (const bf16x8*)wfrag;
+    f32x4 acc0 = {0.f, 0.f, 0.f, 0.f};
+    f32x4 acc1 = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+    for (int tap = 0; tap < 9; ++tap) {
+        const int nb = nbr[cell * 9 + tap];
+        const bf16x8 a = *(const bf16x8*)(x + (n * 77 + nb) * 32 + khi * 8);
+        const bf16x8 b0 = wf[((tap * 2 + 0) * 4 + khi) * 16 + lo];
+        const bf16x8 b1 = wf[((tap * 2 + 1) * 4 + khi) * 16 + lo];
+        acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b0, acc0, 0, 0, 0);
+        acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b1, acc1, 0, 0, 0);
+    }
+
+    const float sh0 = shift[lo], sh1 = shift[16 + lo];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+        const long prow = pos_base + khi * 4 + r;
+        if (prow >= total_pos) continue;
+        float v0 = acc0[r] + sh0;
+        float v1 = acc1[r] + sh1;
+        const long off = prow * 32;
+        if (add_residual) {
+            v0 += bf2f(x[off + lo]);
+            v1 += bf2f(x[off + 16 + lo]);
+        }
+        if (apply_relu) {
+            v0 = fmaxf(v0, 0.f);
+            v1 = fmaxf(v1, 0.f);
+        }
+        y[off + lo] = f2bf(v0);
+        y[off + 16 + lo] = f2bf(v1);
+    }
+}
+
 }  // namespace
+
+static torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B) {
+    TORCH_CHECK(A.is_cuda() && A.scalar_type() == torch::kBFloat16);
+    auto D = torch::empty({16, 16}, A.options().dtype(torch::kFloat32));
+    auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+    hipLaunchKernelGGL(mfma_probe_kernel, dim3(1), dim3(64), 0, stream,
+        (const short*)A.data_ptr(), (const short*)B.data_ptr(),
+        D.data_ptr<float>());
+    return D;
+}
+
+static torch::Tensor obs_to_nhwc(torch::Tensor obs) {
+    TORCH_CHECK(obs.is_cuda() && obs.scalar_type() == torch::kUInt8);
+    const long N = obs.size(0);
+    auto out = torch::empty({N, 77, 32}, obs.options().dtype(torch::kBFloat16));
+    const long total = N * 77 * 32;
+    const int block = 256;
+    const long grid = (total + block - 1) / block;
+    auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+    hipLaunchKernelGGL(obs_to_nhwc_kernel, dim3(grid), dim3(block), 0, stream,
+        obs.data_ptr<unsigned char>(), (short*)out.data_ptr(), total);
+    return out;
+}
+
+static torch::Tensor torus_conv_fused(
+    torch::Tensor x, torch::Tensor wfrag, torch::Tensor shift,
+    torch::Tensor nbr, bool add_residual, bool apply_relu) {
+    TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kBFloat16);
+    TORCH_CHECK(x.dim() == 3 && x.size(1) == 77 && x.size(2) == 32,
+                "torus_conv_fused: x must be (N,77,32) bf16");
+    const long total_pos = x.size(0) * 77;
+    auto y = torch::empty_like(x);
+    const long grid = (total_pos + 63) / 64;
+    auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+    hipLaunchKernelGGL(torus_conv_fused_kernel, dim3(grid), dim3(256), 0, stream,
+        (const short*)x.data_ptr(), (const short*)wfrag.data_ptr(),
+        shift.data_ptr<float>(), nbr.data_ptr<int>(), (short*)y.data_ptr(),
+        total_pos, (int)add_residual, (int)apply_relu);
+    return y;
+}
 
 static std::vector<torch::Tensor> target_scan(
     torch::Tensor values, torch::Tensor ret_boot,
@@ -177,4 +332,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "fused TD/UPGO/V-Trace backward scan (targets, advantages)");
     m.def("masked_sample", &masked_sample,
           "batched masked-softmax action sampling (actions, probs)");
+    m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 MFMA layout probe");
+    m.def("obs_to_nhwc", &obs_to_nhwc, "uint8 NCHW obs -> padded NHWC bf16");
+    m.def("torus_conv_fused", &torus_conv_fused,
+          "fused wrap-around conv3x3 + BN-fold + residual + relu (MFMA)");
 }

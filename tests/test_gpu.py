@@ -49,6 +49,87 @@ def test_masked_sample_matches_inverse_cdf():
 
 
 @requires_gpu
+def test_mfma_fragment_layout():
+    """The probe kernel's lane->fragment mapping must reproduce matmul.
+    Asymmetric operands so a transposed mapping cannot pass (guide G9)."""
+    from handyrl_amd import ops
+    torch.manual_seed(0)
+    A = (torch.randn(16, 32) * 0.5).bfloat16().cuda()
+    B = (torch.arange(32 * 16).reshape(32, 16).float() * 0.01 - 2.0).bfloat16().cuda()
+    D = ops.mfma_probe(A, B)
+    torch.cuda.synchronize()
+    ref = A.float() @ B.float()
+    torch.testing.assert_close(D.cpu(), ref.cpu(), rtol=2e-2, atol=2e-2)
+
+
+@requires_gpu
+@pytest.mark.parametrize('residual', [False, True])
+def test_torus_conv_fused_matches_eager(residual):
+    """Fused MFMA torus conv vs the fp32 eager eval composition."""
+    from handyrl_amd import ops
+    from handyrl_amd.models.geese_net import TorusConv2d
+    torch.manual_seed(1)
+    N = 173                                    # deliberately not /64
+    layer = TorusConv2d(32, 32).cuda().eval()
+    layer.bn.running_mean.uniform_(-0.3, 0.3)
+    layer.bn.running_var.uniform_(0.5, 1.5)
+    layer.bn.weight.data.uniform_(0.5, 1.5)
+    layer.bn.bias.data.uniform_(-0.3, 0.3)
+
+    x_nchw = torch.randn(N, 32, 7, 11, device='cuda')
+    with torch.no_grad():
+        ref = layer.bn(layer.conv(x_nchw)) if False else None
+        # eager eval reference (fp32): conv + BN affine (+x) + relu
+        rstd = torch.rsqrt(layer.bn.running_var + layer.bn.eps)
+        scale = layer.bn.weight * rstd
+        shift = layer.bn.bias - layer.bn.running_mean * scale
+        conv = torch.nn.functional.conv2d(
+            torch.nn.functional.pad(x_nchw, (1, 1, 1, 1), mode='circular'),
+            layer.conv.weight)
+        ref = conv * scale.view(1, -1, 1, 1) + shift.view(1, -1, 1, 1)
+        if residual:
+            ref = ref + x_nchw
+        ref = torch.relu(ref)
+
+    # NHWC bf16 input for the fused kernel
+    x_nhwc = x_nchw.permute(0, 2, 3, 1).reshape(N, 77, 32).contiguous().bfloat16()
+    wfrag = ops.pack_torus_weights(layer.conv.weight, scale)
+    nbr = ops.torus_neighbor_table('cuda')
+    y = ops.torus_conv_fused(x_nhwc, wfrag, shift.contiguous(), nbr,
+                             residual, True)
+    torch.cuda.synchronize()
+    y_nchw = y.float().reshape(N, 7, 11, 32).permute(0, 3, 1, 2)
+    # bf16 inputs/weights: tolerance at bf16 resolution of the accumulations
+    torch.testing.assert_close(y_nchw, ref, rtol=5e-2, atol=5e-2)
+
+
+@requires_gpu
+def test_geese_fused_eval_matches_eager():
+    """13-layer fused MFMA tower + heads vs the eager eval GeeseNet."""
+    from handyrl_amd.models.geese_net import GeeseNet, GeeseFusedEval
+    torch.manual_seed(2)
+    net = GeeseNet().cuda()
+    # non-trivial BN stats
+    for layer in [net.stem] + list(net.blocks):
+        layer.bn.running_mean.uniform_(-0.2, 0.2)
+        layer.bn.running_var.uniform_(0.7, 1.4)
+    net.eval()
+
+    obs = (torch.rand(257, 17, 7, 11, device='cuda') < 0.15).to(torch.uint8)
+    obs[:, 0] = 0
+    heads = torch.randint(0, 77, (257,), device='cuda')
+    obs.view(257, 17, 77)[torch.arange(257), 0, heads] = 1   # one head cell
+
+    fused = GeeseFusedEval(net, torch.device('cuda'))
+    with torch.no_grad():
+        ref = net(obs.float(), None)
+        out = fused.forward(obs)
+    torch.cuda.synchronize()
+    torch.testing.assert_close(out['policy'], ref['policy'], rtol=0.08, atol=0.08)
+    torch.testing.assert_close(out['value'], ref['value'], rtol=0.08, atol=0.08)
+
+
+@requires_gpu
 def test_gpu_train_step_bf16():
     from handyrl_amd.models.geese_net import GeeseNet
     from handyrl_amd.train import Trainer
