@@ -357,6 +357,12 @@ class Learner:
         self.worker = WorkerServer(args) if remote else WorkerCluster(args)
         self.trainer = Trainer(args, copy.deepcopy(self.model))
 
+        # GPU actor pool (worker: {type: 'gpu'}): self-play generation runs
+        # as batched inference on the learner's GPU instead of CPU worker
+        # processes; any CPU workers then serve evaluation jobs only.
+        self.gpu_actor = bool(args['worker'].get('type') == 'gpu')
+        self.feed_lock = threading.Lock()
+
     def model_path(self, model_id):
         return os.path.join('models', str(model_id) + '.pth')
 
@@ -372,31 +378,61 @@ class Learner:
         torch.save(model.state_dict(), self.latest_model_path())
 
     def feed_episodes(self, episodes):
-        for episode in episodes:
-            if episode is None:
-                continue
-            for p in episode['args']['player']:
-                model_id = self.model_epoch
-                outcome = episode['outcome'][p]
-                n, r, r2 = self.generation_results.get(model_id, (0, 0, 0))
-                self.generation_results[model_id] = n + 1, r + outcome, r2 + outcome ** 2
-            self.num_returned_episodes += 1
-            if self.num_returned_episodes % 100 == 0:
-                print(self.num_returned_episodes, end=' ', flush=True)
+        with self.feed_lock:
+            for episode in episodes:
+                if episode is None:
+                    continue
+                for p in episode['args']['player']:
+                    model_id = self.model_epoch
+                    outcome = episode['outcome'][p]
+                    n, r, r2 = self.generation_results.get(model_id, (0, 0, 0))
+                    self.generation_results[model_id] = \
+                        n + 1, r + outcome, r2 + outcome ** 2
+                self.num_returned_episodes += 1
+                self.num_episodes += 1 if self.gpu_actor else 0
+                if self.num_returned_episodes % 100 == 0:
+                    print(self.num_returned_episodes, end=' ', flush=True)
 
-        self.trainer.episodes.extend([e for e in episodes if e is not None])
+            self.trainer.episodes.extend([e for e in episodes if e is not None])
 
-        mem_percent = psutil.virtual_memory().percent if psutil is not None else 0
-        mem_ok = mem_percent <= 95
-        maximum_episodes = self.args['maximum_episodes'] if mem_ok else \
-            int(len(self.trainer.episodes) * 95 / mem_percent)
+            mem_percent = psutil.virtual_memory().percent if psutil is not None else 0
+            mem_ok = mem_percent <= 95
+            maximum_episodes = self.args['maximum_episodes'] if mem_ok else \
+                int(len(self.trainer.episodes) * 95 / mem_percent)
 
-        if not mem_ok and 'memory_over' not in self.flags:
-            warnings.warn('memory usage %.1f%% with buffer size %d' %
-                          (mem_percent, len(self.trainer.episodes)))
-            self.flags.add('memory_over')
+            if not mem_ok and 'memory_over' not in self.flags:
+                warnings.warn('memory usage %.1f%% with buffer size %d' %
+                              (mem_percent, len(self.trainer.episodes)))
+                self.flags.add('memory_over')
 
-        self.trainer.episodes.trim(maximum_episodes)
+            self.trainer.episodes.trim(maximum_episodes)
+
+    def _gpu_actor_loop(self):
+        """Self-play generation on the learner GPU (worker type 'gpu'):
+        a vectorized env pool stepped by batched (graphed) inference on the
+        trainer's live model, feeding columnar episodes straight into the
+        replay buffer — no per-env worker processes, no pickled models."""
+        from .actor import GeeseActorPool
+        n_envs = self.args['worker'].get('num_envs', 256)
+        pool = GeeseActorPool(self.trainer.model, self.args, n_games=n_envs,
+                              device=self.trainer.device,
+                              seed=self.args['seed'] + 1)
+        last_epoch = -1
+        print('started gpu actor pool (%d envs)' % n_envs)
+        while not self.shutdown_flag:
+            if self.model_epoch != last_epoch:
+                pool.refresh_weights()
+                last_epoch = self.model_epoch
+            was_training = self.trainer.model.training
+            self.trainer.model.eval()
+            for _ in range(8):
+                pool.step_once()
+            if was_training:
+                self.trainer.model.train()
+            episodes = pool.harvest()
+            if episodes:
+                self.feed_episodes(episodes)
+        print('finished gpu actor pool')
 
     def feed_results(self, results):
         for result in results:
@@ -453,6 +489,16 @@ class Learner:
         next_update_episodes = prev_update_episodes + self.args['update_episodes']
 
         while self.worker.connection_count() > 0 or not self.shutdown_flag:
+            # epoch rollover: checked every iteration (with GPU actors the
+            # buffer fills without any worker request traffic)
+            if self.num_returned_episodes >= next_update_episodes:
+                prev_update_episodes = next_update_episodes
+                next_update_episodes = prev_update_episodes + self.args['update_episodes']
+                self.update()
+                if self.args['epochs'] >= 0 and self.model_epoch >= self.args['epochs']:
+                    self.shutdown_flag = True
+                    if self.gpu_actor and self.worker.connection_count() == 0:
+                        break
             try:
                 conn, (req, data) = self.worker.recv(timeout=0.3)
             except queue.Empty:
@@ -469,8 +515,10 @@ class Learner:
                 else:
                     for _ in data:
                         job = {'model_id': {}}
-                        # evaluation share of jobs per eval_rate
-                        if self.num_results < self.eval_rate * self.num_episodes:
+                        # evaluation share of jobs per eval_rate; with GPU
+                        # actors generating, CPU workers only evaluate
+                        if self.gpu_actor or \
+                                self.num_results < self.eval_rate * self.num_episodes:
                             job['role'] = 'e'
                         else:
                             job['role'] = 'g'
@@ -514,18 +562,14 @@ class Learner:
             if not multi_req and len(send_data) == 1:
                 send_data = send_data[0]
             self.worker.send(conn, send_data)
-
-            if self.num_returned_episodes >= next_update_episodes:
-                prev_update_episodes = next_update_episodes
-                next_update_episodes = prev_update_episodes + self.args['update_episodes']
-                self.update()
-                if self.args['epochs'] >= 0 and self.model_epoch >= self.args['epochs']:
-                    self.shutdown_flag = True
         print('finished server')
 
     def run(self):
         threading.Thread(target=self.trainer.run, daemon=True).start()
-        self.worker.run()
+        if self.gpu_actor:
+            threading.Thread(target=self._gpu_actor_loop, daemon=True).start()
+        if self.args['worker'].get('num_parallel', 0) > 0:
+            self.worker.run()
         self.server()
 
 
