@@ -41,3 +41,14 @@ timeit('refresh_weights (eager)', lambda: pool.fused.refresh())
 timeit('full graphed train_step', lambda: trainer.train_step(batch))
 # actor step components under load
 timeit('actor step_once', lambda: pool.step_once(), n=50)
+
+from torch.profiler import profile, ProfilerActivity
+with profile(activities=[ProfilerActivity.CUDA]) as prof:
+    for _ in range(10):
+        g.graph.replay()
+    torch.cuda.synchronize()
+ka = prof.key_averages()
+tot = sum(e.device_time_total for e in ka)
+print('TRAIN REPLAY: total device time per replay: %.2f ms' % (tot/1e3/10))
+for e in sorted(ka, key=lambda e: -e.device_time_total)[:14]:
+    print('%9.1f us  x%-5d %s' % (e.device_time_total/10, e.count//10, e.key[:90]))
