@@ -130,6 +130,50 @@ def test_geese_fused_eval_matches_eager():
 
 
 @requires_gpu
+@pytest.mark.parametrize('dtype', [torch.float32, torch.bfloat16])
+def test_fused_bn_train_matches_stock(dtype):
+    """Custom BN train fwd/bwd HIP kernels vs nn.BatchNorm2d: outputs,
+    running stats and all three gradients."""
+    import copy
+    import torch.nn as nn
+    from handyrl_amd.models.common import apply_bn
+    torch.manual_seed(0)
+    bn_ref = nn.BatchNorm2d(32).cuda()
+    bn_ref.weight.data.uniform_(0.5, 1.5)
+    bn_ref.bias.data.uniform_(-0.5, 0.5)
+    bn_ref.running_mean.uniform_(-1, 1)
+    bn_ref.running_var.uniform_(0.5, 2.0)
+    bn_mine = copy.deepcopy(bn_ref)
+    bn_ref.train(); bn_mine.train()
+
+    tol = dict(rtol=1e-4, atol=1e-4) if dtype == torch.float32 \
+        else dict(rtol=5e-2, atol=5e-2)
+    x = torch.randn(64, 32, 7, 11, device='cuda', dtype=dtype)
+    x_ref = x.float().clone().requires_grad_(True)
+    x_mine = x.clone().requires_grad_(True)
+
+    y_ref = bn_ref(x_ref)
+    y_mine = apply_bn(bn_mine, x_mine)
+    torch.cuda.synchronize()
+    torch.testing.assert_close(y_mine.float(), y_ref, **tol)
+    torch.testing.assert_close(bn_mine.running_mean, bn_ref.running_mean,
+                               rtol=1e-3, atol=1e-3)
+    torch.testing.assert_close(bn_mine.running_var, bn_ref.running_var,
+                               rtol=1e-3, atol=1e-3)
+    assert int(bn_mine.num_batches_tracked) == int(bn_ref.num_batches_tracked)
+
+    g = torch.randn_like(y_ref)
+    y_ref.backward(g)
+    y_mine.backward(g.to(dtype))
+    torch.cuda.synchronize()
+    torch.testing.assert_close(x_mine.grad.float(), x_ref.grad, **tol)
+    torch.testing.assert_close(bn_mine.weight.grad, bn_ref.weight.grad,
+                               rtol=2e-2, atol=2e-2)
+    torch.testing.assert_close(bn_mine.bias.grad, bn_ref.bias.grad,
+                               rtol=2e-2, atol=2e-2)
+
+
+@requires_gpu
 def test_gpu_train_step_bf16():
     from handyrl_amd.models.geese_net import GeeseNet
     from handyrl_amd.train import Trainer
