@@ -134,12 +134,15 @@ __global__ void masked_sample_kernel(
     prob[i] = pr;
 }
 
-// --- BatchNorm2d training fwd/bwd, one kernel each ----------------------
+// --- BatchNorm2d training fwd/bwd ---------------------------------------
 // nn.BatchNorm2d semantics (biased var for normalization, unbiased for the
-// running-var update).  One block per channel, two sweeps over the
-// channel's data; stats in fp32, activations bf16 or fp32.  Replaces the
-// ~22-kernel torch-primitive composition per layer that dominated the
-// captured train step (Welford/mean reduces at 40-85us each).
+// running-var update).  Layout (N,C,HW): a "row" = one (n,c) slice of HW
+// contiguous elements; rows are contiguous in memory (flat row id =
+// n*C + c).  Stats kernels: grid C*S with one whole row per thread (lines
+// stay L1-resident across the scalar sweep) + global fp32 atomics; apply
+// kernels: one row per thread.  3 graph nodes per direction, replacing the
+// ~22-kernel torch-primitive composition per layer whose strided reduces
+// (40-85us each) dominated the captured train step.
 
 template <typename T>
 DEV_INLINE float load_as_f32(const T* p);
@@ -151,99 +154,139 @@ template <> DEV_INLINE void store_f32<short>(short* p, float v) { *p = f2bf(v); 
 template <> DEV_INLINE void store_f32<float>(float* p, float v) { *p = v; }
 
 template <typename T>
-__global__ __launch_bounds__(256) void bn_train_fwd_kernel(
-    const T* __restrict__ x,      // (N, C, HW)
-    T* __restrict__ y,
+__global__ __launch_bounds__(256) void bn_stats_kernel(
+    const T* __restrict__ x, float* __restrict__ accum,  // (2C): sum, sumsq
+    int N, int C, int HW, int S)
+{
+    const int c = blockIdx.x % C;
+    const int slice = blockIdx.x / C;
+    const int rows_per = (N + S - 1) / S;
+    const int n0 = slice * rows_per;
+    const int n1 = min(N, n0 + rows_per);
+    float acc = 0.f, acc2 = 0.f;
+    for (int n = n0 + (int)threadIdx.x; n < n1; n += 256) {
+        const T* row = x + ((long)n * C + c) * HW;
+        for (int i = 0; i < HW; ++i) {
+            const float v = load_as_f32(row + i);
+            acc += v;
+            acc2 += v * v;
+        }
+    }
+    __shared__ float s1[256], s2[256];
+    s1[threadIdx.x] = acc; s2[threadIdx.x] = acc2;
+    __syncthreads();
+    for (int s = 128; s > 0; s >>= 1) {
+        if ((int)threadIdx.x < s) {
+            s1[threadIdx.x] += s1[threadIdx.x + s];
+            s2[threadIdx.x] += s2[threadIdx.x + s];
+        }
+        __syncthreads();
+    }
+    if (threadIdx.x == 0) {
+        atomicAdd(&accum[c], s1[0]);
+        atomicAdd(&accum[C + c], s2[0]);
+    }
+}
+
+template <typename T>
+__global__ __launch_bounds__(256) void bn_apply_kernel(
+    const T* __restrict__ x, T* __restrict__ y,
+    const float* __restrict__ accum,
     const float* __restrict__ weight, const float* __restrict__ bias,
     float* __restrict__ running_mean, float* __restrict__ running_var,
     float* __restrict__ save_mean, float* __restrict__ save_rstd,
     int N, int C, int HW, float momentum, float eps)
 {
-    const int c = blockIdx.x;
-    const int tid = threadIdx.x;
-    const long M = (long)N * HW;
-    __shared__ float s_sum[256], s_sum2[256];
-    __shared__ float s_mean, s_rstd;
-
-    float acc = 0.f, acc2 = 0.f;
-    for (long i = tid; i < M; i += 256) {
-        const long n = i / HW, hw = i % HW;
-        const float v = load_as_f32(x + (n * C + c) * HW + hw);
-        acc += v;
-        acc2 += v * v;
-    }
-    s_sum[tid] = acc; s_sum2[tid] = acc2;
-    __syncthreads();
-    for (int s = 128; s > 0; s >>= 1) {
-        if (tid < s) { s_sum[tid] += s_sum[tid + s]; s_sum2[tid] += s_sum2[tid + s]; }
-        __syncthreads();
-    }
-    if (tid == 0) {
-        const float mean = s_sum[0] / M;
-        const float var = fmaxf(s_sum2[0] / M - mean * mean, 0.f);
-        s_mean = mean;
-        s_rstd = rsqrtf(var + eps);
+    const long row = (long)blockIdx.x * 256 + threadIdx.x;   // over N*C
+    const float M = (float)N * HW;
+    if (row < C) {   // one thread per channel also publishes the stats
+        const int c = (int)row;
+        const float mean = accum[c] / M;
+        const float var = fmaxf(accum[C + c] / M - mean * mean, 0.f);
         save_mean[c] = mean;
-        save_rstd[c] = s_rstd;
-        const float unbiased = var * ((float)M / fmaxf((float)(M - 1), 1.f));
+        save_rstd[c] = rsqrtf(var + eps);
+        const float unbiased = var * (M / fmaxf(M - 1.f, 1.f));
         running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * mean;
         running_var[c] = (1.f - momentum) * running_var[c] + momentum * unbiased;
     }
+    if (row >= (long)N * C) return;
+    const int c = (int)(row % C);
+    const float mean = accum[c] / M;
+    const float var = fmaxf(accum[C + c] / M - mean * mean, 0.f);
+    const float rstd = rsqrtf(var + eps);
+    const float scale = weight[c] * rstd;
+    const float shift = bias[c] - mean * scale;
+    const T* xr = x + row * HW;
+    T* yr = y + row * HW;
+    for (int i = 0; i < HW; ++i)
+        store_f32(yr + i, load_as_f32(xr + i) * scale + shift);
+}
+
+template <typename T>
+__global__ __launch_bounds__(256) void bn_bwd_stats_kernel(
+    const T* __restrict__ x, const T* __restrict__ dy,
+    float* __restrict__ accum,   // (2C): sum dy, sum dy*xhat
+    const float* __restrict__ save_mean, const float* __restrict__ save_rstd,
+    int N, int C, int HW, int S)
+{
+    const int c = blockIdx.x % C;
+    const int slice = blockIdx.x / C;
+    const int rows_per = (N + S - 1) / S;
+    const int n0 = slice * rows_per;
+    const int n1 = min(N, n0 + rows_per);
+    const float mean = save_mean[c], rstd = save_rstd[c];
+    float acc = 0.f, acc2 = 0.f;
+    for (int n = n0 + (int)threadIdx.x; n < n1; n += 256) {
+        const long off = ((long)n * C + c) * HW;
+        for (int i = 0; i < HW; ++i) {
+            const float g = load_as_f32(dy + off + i);
+            const float xh = (load_as_f32(x + off + i) - mean) * rstd;
+            acc += g;
+            acc2 += g * xh;
+        }
+    }
+    __shared__ float s1[256], s2[256];
+    s1[threadIdx.x] = acc; s2[threadIdx.x] = acc2;
     __syncthreads();
-    const float scale = weight[c] * s_rstd;
-    const float shift = bias[c] - s_mean * scale;
-    for (long i = tid; i < M; i += 256) {
-        const long n = i / HW, hw = i % HW;
-        const long off = (n * C + c) * HW + hw;
-        store_f32(y + off, load_as_f32(x + off) * scale + shift);
+    for (int s = 128; s > 0; s >>= 1) {
+        if ((int)threadIdx.x < s) {
+            s1[threadIdx.x] += s1[threadIdx.x + s];
+            s2[threadIdx.x] += s2[threadIdx.x + s];
+        }
+        __syncthreads();
+    }
+    if (threadIdx.x == 0) {
+        atomicAdd(&accum[c], s1[0]);
+        atomicAdd(&accum[C + c], s2[0]);
     }
 }
 
 template <typename T>
-__global__ __launch_bounds__(256) void bn_train_bwd_kernel(
-    const T* __restrict__ x, const T* __restrict__ dy,
-    T* __restrict__ dx,
+__global__ __launch_bounds__(256) void bn_bwd_apply_kernel(
+    const T* __restrict__ x, const T* __restrict__ dy, T* __restrict__ dx,
+    const float* __restrict__ accum,
     const float* __restrict__ weight,
     const float* __restrict__ save_mean, const float* __restrict__ save_rstd,
     float* __restrict__ dweight, float* __restrict__ dbias,
     int N, int C, int HW)
 {
-    const int c = blockIdx.x;
-    const int tid = threadIdx.x;
-    const long M = (long)N * HW;
+    const long row = (long)blockIdx.x * 256 + threadIdx.x;
+    const float M = (float)N * HW;
+    if (row < C) {
+        dbias[row] = accum[row];
+        dweight[row] = accum[C + row];
+    }
+    if (row >= (long)N * C) return;
+    const int c = (int)(row % C);
     const float mean = save_mean[c], rstd = save_rstd[c];
-    __shared__ float s_dy[256], s_dyx[256];
-    __shared__ float s_sdy, s_sdyx;
-
-    float acc_dy = 0.f, acc_dyx = 0.f;
-    for (long i = tid; i < M; i += 256) {
-        const long n = i / HW, hw = i % HW;
-        const long off = (n * C + c) * HW + hw;
-        const float g = load_as_f32(dy + off);
-        const float xh = (load_as_f32(x + off) - mean) * rstd;
-        acc_dy += g;
-        acc_dyx += g * xh;
-    }
-    s_dy[tid] = acc_dy; s_dyx[tid] = acc_dyx;
-    __syncthreads();
-    for (int s = 128; s > 0; s >>= 1) {
-        if (tid < s) { s_dy[tid] += s_dy[tid + s]; s_dyx[tid] += s_dyx[tid + s]; }
-        __syncthreads();
-    }
-    if (tid == 0) {
-        s_sdy = s_dy[0]; s_sdyx = s_dyx[0];
-        dbias[c] = s_dy[0];
-        dweight[c] = s_dyx[0];
-    }
-    __syncthreads();
-    const float k_dy = s_sdy / M, k_dyx = s_sdyx / M;
+    const float k_dy = accum[c] / M;
+    const float k_dyx = accum[C + c] / M;
     const float wr = weight[c] * rstd;
-    for (long i = tid; i < M; i += 256) {
-        const long n = i / HW, hw = i % HW;
-        const long off = (n * C + c) * HW + hw;
-        const float g = load_as_f32(dy + off);
-        const float xh = (load_as_f32(x + off) - mean) * rstd;
-        store_f32(dx + off, wr * (g - k_dy - xh * k_dyx));
+    const long off = row * HW;
+    for (int i = 0; i < HW; ++i) {
+        const float g = load_as_f32(dy + off + i);
+        const float xh = (load_as_f32(x + off + i) - mean) * rstd;
+        store_f32(dx + off + i, wr * (g - k_dy - xh * k_dyx));
     }
 }
 
@@ -350,55 +393,82 @@ __global__ __launch_bounds__(256) void torus_conv_fused_kernel(
 
 }  // namespace
 
+template <typename T>
+static void bn_fwd_launch(torch::Tensor& x, torch::Tensor& y, torch::Tensor& accum,
+                          torch::Tensor& weight, torch::Tensor& bias,
+                          torch::Tensor& rmean, torch::Tensor& rvar,
+                          torch::Tensor& smean, torch::Tensor& srstd,
+                          int N, int C, int HW, int S,
+                          double momentum, double eps, hipStream_t stream) {
+    hipLaunchKernelGGL(bn_stats_kernel<T>, dim3(C * S), dim3(256), 0, stream,
+        (const T*)x.data_ptr(), accum.data_ptr<float>(), N, C, HW, S);
+    const long rows = (long)N * C;
+    hipLaunchKernelGGL(bn_apply_kernel<T>, dim3((rows + 255) / 256), dim3(256), 0, stream,
+        (const T*)x.data_ptr(), (T*)y.data_ptr(), accum.data_ptr<float>(),
+        weight.data_ptr<float>(), bias.data_ptr<float>(),
+        rmean.data_ptr<float>(), rvar.data_ptr<float>(),
+        smean.data_ptr<float>(), srstd.data_ptr<float>(),
+        N, C, HW, (float)momentum, (float)eps);
+}
+
 static std::vector<torch::Tensor> bn_train_fwd(
     torch::Tensor x, torch::Tensor weight, torch::Tensor bias,
     torch::Tensor running_mean, torch::Tensor running_var,
     double momentum, double eps) {
     TORCH_CHECK(x.is_cuda() && x.dim() == 4, "bn_train_fwd: (N,C,H,W) expected");
     const int N = x.size(0), C = x.size(1), HW = x.size(2) * x.size(3);
+    const int S = std::min(32, std::max(1, N / 256));
     auto y = torch::empty_like(x);
+    auto accum = torch::empty({2 * C}, weight.options());
     auto save_mean = torch::empty({C}, weight.options());
     auto save_rstd = torch::empty({C}, weight.options());
     auto stream = at::cuda::getCurrentCUDAStream();
+    hipMemsetAsync(accum.data_ptr(), 0, 2 * C * sizeof(float), stream);
     if (x.scalar_type() == torch::kBFloat16) {
-        hipLaunchKernelGGL(bn_train_fwd_kernel<short>, dim3(C), dim3(256), 0, stream,
-            (const short*)x.data_ptr(), (short*)y.data_ptr(),
-            weight.data_ptr<float>(), bias.data_ptr<float>(),
-            running_mean.data_ptr<float>(), running_var.data_ptr<float>(),
-            save_mean.data_ptr<float>(), save_rstd.data_ptr<float>(),
-            N, C, HW, (float)momentum, (float)eps);
+        bn_fwd_launch<short>(x, y, accum, weight, bias, running_mean, running_var,
+                             save_mean, save_rstd, N, C, HW, S, momentum, eps, stream);
     } else {
         TORCH_CHECK(x.scalar_type() == torch::kFloat32, "bn: bf16/f32 only");
-        hipLaunchKernelGGL(bn_train_fwd_kernel<float>, dim3(C), dim3(256), 0, stream,
-            x.data_ptr<float>(), y.data_ptr<float>(),
-            weight.data_ptr<float>(), bias.data_ptr<float>(),
-            running_mean.data_ptr<float>(), running_var.data_ptr<float>(),
-            save_mean.data_ptr<float>(), save_rstd.data_ptr<float>(),
-            N, C, HW, (float)momentum, (float)eps);
+        bn_fwd_launch<float>(x, y, accum, weight, bias, running_mean, running_var,
+                             save_mean, save_rstd, N, C, HW, S, momentum, eps, stream);
     }
     return {y, save_mean, save_rstd};
+}
+
+template <typename T>
+static void bn_bwd_launch(torch::Tensor& x, torch::Tensor& dy, torch::Tensor& dx,
+                          torch::Tensor& accum, torch::Tensor& weight,
+                          torch::Tensor& smean, torch::Tensor& srstd,
+                          torch::Tensor& dweight, torch::Tensor& dbias,
+                          int N, int C, int HW, int S, hipStream_t stream) {
+    hipLaunchKernelGGL(bn_bwd_stats_kernel<T>, dim3(C * S), dim3(256), 0, stream,
+        (const T*)x.data_ptr(), (const T*)dy.data_ptr(), accum.data_ptr<float>(),
+        smean.data_ptr<float>(), srstd.data_ptr<float>(), N, C, HW, S);
+    const long rows = (long)N * C;
+    hipLaunchKernelGGL(bn_bwd_apply_kernel<T>, dim3((rows + 255) / 256), dim3(256), 0, stream,
+        (const T*)x.data_ptr(), (const T*)dy.data_ptr(), (T*)dx.data_ptr(),
+        accum.data_ptr<float>(), weight.data_ptr<float>(),
+        smean.data_ptr<float>(), srstd.data_ptr<float>(),
+        dweight.data_ptr<float>(), dbias.data_ptr<float>(), N, C, HW);
 }
 
 static std::vector<torch::Tensor> bn_train_bwd(
     torch::Tensor x, torch::Tensor dy, torch::Tensor weight,
     torch::Tensor save_mean, torch::Tensor save_rstd) {
     const int N = x.size(0), C = x.size(1), HW = x.size(2) * x.size(3);
+    const int S = std::min(32, std::max(1, N / 256));
     auto dx = torch::empty_like(x);
+    auto accum = torch::empty({2 * C}, weight.options());
     auto dweight = torch::empty({C}, weight.options());
     auto dbias = torch::empty({C}, weight.options());
     auto stream = at::cuda::getCurrentCUDAStream();
+    hipMemsetAsync(accum.data_ptr(), 0, 2 * C * sizeof(float), stream);
     if (x.scalar_type() == torch::kBFloat16) {
-        hipLaunchKernelGGL(bn_train_bwd_kernel<short>, dim3(C), dim3(256), 0, stream,
-            (const short*)x.data_ptr(), (const short*)dy.data_ptr(),
-            (short*)dx.data_ptr(), weight.data_ptr<float>(),
-            save_mean.data_ptr<float>(), save_rstd.data_ptr<float>(),
-            dweight.data_ptr<float>(), dbias.data_ptr<float>(), N, C, HW);
+        bn_bwd_launch<short>(x, dy, dx, accum, weight, save_mean, save_rstd,
+                             dweight, dbias, N, C, HW, S, stream);
     } else {
-        hipLaunchKernelGGL(bn_train_bwd_kernel<float>, dim3(C), dim3(256), 0, stream,
-            x.data_ptr<float>(), dy.data_ptr<float>(),
-            dx.data_ptr<float>(), weight.data_ptr<float>(),
-            save_mean.data_ptr<float>(), save_rstd.data_ptr<float>(),
-            dweight.data_ptr<float>(), dbias.data_ptr<float>(), N, C, HW);
+        bn_bwd_launch<float>(x, dy, dx, accum, weight, save_mean, save_rstd,
+                             dweight, dbias, N, C, HW, S, stream);
     }
     return {dx, dweight, dbias};
 }

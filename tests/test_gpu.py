@@ -162,15 +162,17 @@ def test_fused_bn_train_matches_stock(dtype):
                                rtol=1e-3, atol=1e-3)
     assert int(bn_mine.num_batches_tracked) == int(bn_ref.num_batches_tracked)
 
-    g = torch.randn_like(y_ref)
-    y_ref.backward(g)
-    y_mine.backward(g.to(dtype))
+    # both sides must see the SAME (dtype-rounded) upstream gradient
+    g = torch.randn(y_ref.shape, device='cuda', dtype=dtype)
+    y_ref.backward(g.float())
+    y_mine.backward(g)
     torch.cuda.synchronize()
     torch.testing.assert_close(x_mine.grad.float(), x_ref.grad, **tol)
-    torch.testing.assert_close(bn_mine.weight.grad, bn_ref.weight.grad,
-                               rtol=2e-2, atol=2e-2)
-    torch.testing.assert_close(bn_mine.bias.grad, bn_ref.bias.grad,
-                               rtol=2e-2, atol=2e-2)
+    # dweight/dbias are ~170k-term sums: bf16 input rounding accumulates
+    wtol = dict(rtol=1e-4, atol=1e-3) if dtype == torch.float32 \
+        else dict(rtol=5e-2, atol=1.0)
+    torch.testing.assert_close(bn_mine.weight.grad, bn_ref.weight.grad, **wtol)
+    torch.testing.assert_close(bn_mine.bias.grad, bn_ref.bias.grad, **wtol)
 
 
 @requires_gpu
