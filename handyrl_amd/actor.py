@@ -32,7 +32,7 @@ class GeeseActorPool:
     """Self-play actor pool for Hungry Geese on one GPU."""
 
     def __init__(self, model, args, n_games=256, device=None, seed=0,
-                 store_uint8_obs=True, use_graphs=True):
+                 store_uint8_obs=True, use_graphs=True, engine=None):
         self.args = args
         self.device = device if device is not None else (
             torch.device('cuda') if torch.cuda.is_available() else torch.device('cpu'))
@@ -58,13 +58,27 @@ class GeeseActorPool:
         self._zero_mask = None
         self.graphed = None
         self.fused = None
-        if self.device.type == 'cuda' and os.environ.get('HANDYRL_NO_FUSED') != '1' \
-                and hasattr(model, 'stem') and ops.available():
-            from .models.geese_net import GeeseFusedEval
-            self.fused = GeeseFusedEval(model, self.device)
-        if use_graphs and self.device.type == 'cuda':
-            from .hipgraph import GraphedActorForward
-            self.graphed = GraphedActorForward(model, self.device, fused=self.fused)
+        if engine is not None:
+            self.fused, self.graphed = engine
+        else:
+            if self.device.type == 'cuda' and os.environ.get('HANDYRL_NO_FUSED') != '1' \
+                    and hasattr(model, 'stem') and ops.available():
+                from .models.geese_net import GeeseFusedEval
+                self.fused = GeeseFusedEval(model, self.device)
+            if use_graphs and self.device.type == 'cuda':
+                from .hipgraph import GraphedActorForward
+                self.graphed = GraphedActorForward(model, self.device, fused=self.fused)
+        # pipelined-transfer staging (graphed path)
+        if self.graphed is not None:
+            cap = N_PLAYERS * n_games
+            self._obs_pin = torch.empty(cap, 17, 7, 11, dtype=torch.uint8,
+                                        pin_memory=True)
+            self._obs_pin_np = self._obs_pin.numpy()
+            self._out_pin = torch.empty(cap, 3, dtype=torch.float32,
+                                        pin_memory=True)
+            self._out_pin_np = self._out_pin.numpy()
+            self._event = torch.cuda.Event()
+        self._pending = None
         self.timing = {'obs': 0.0, 'fwd': 0.0, 'sample': 0.0,
                        'record': 0.0, 'env': 0.0, 'package': 0.0, 'n': 0}
 
@@ -79,6 +93,12 @@ class GeeseActorPool:
 
     def step_once(self):
         """Advance every live game by one transition; returns #frames."""
+        self._phase1()
+        return self._phase2()
+
+    def _phase1(self):
+        """Build observations and ISSUE the GPU forward (async on the
+        graphed path: pinned H2D + graph replay + pinned D2H + event)."""
         import time
         tm = self.timing
         vec = self.vec
@@ -88,21 +108,29 @@ class GeeseActorPool:
         live = vec.alive & ~vec.over[:, None]
         gi, pi = np.nonzero(live)
         if len(gi) == 0:
-            return 0
+            self._pending = None
+            return
         t0 = time.time()
-
         obs_sel = obs_u8[gi, pi]                        # (M, 17, 7, 11)
         M = len(gi)
+
+        # record the observation/alive columns now (obs is in hand and this
+        # CPU work overlaps the GPU forward on the pipelined path)
+        game_has_live = live.any(axis=1)
+        lg = np.nonzero(game_has_live)[0]
+        t_idx = self.rec_len[lg]
+        self.rec_obs[lg, t_idx] = obs_u8[lg]
+        self.rec_alive[lg, t_idx] = live[lg]
+        self._rec_slot = (lg, t_idx)
+
         if self.graphed is not None:
-            # hipGraph path: one static-buffer H2D copy + one graph replay
-            # (forward + sample fused) + one packed D2H readback
-            packed = self.graphed.run(torch.from_numpy(obs_sel)).cpu().numpy()
+            np.copyto(self._obs_pin_np[:M], obs_sel)
+            self.graphed.run_async(self._obs_pin, M, self._out_pin, self._event)
+            self._pending = ('async', gi, pi, live, M)
             tm['fwd'] += time.time() - t0
-            t0 = time.time()
-            actions = packed[:, 0].astype(np.int64)
-            probs = packed[:, 1]
-            values = packed[:, 2]
-        elif self.device.type == 'cuda':
+            return
+
+        if self.device.type == 'cuda':
             # eager GPU path: pad to a fixed bucket so MIOpen keeps one
             # solution per shape
             bucket = 256 * ((M + 255) // 256)
@@ -115,13 +143,10 @@ class GeeseActorPool:
             policy, value = self._policy_forward(obs_t.float())
             policy, value = policy[:M], value[:M]
             A = policy.shape[1]
-            tm['fwd'] += time.time() - t0
-            t0 = time.time()
             if self._zero_mask is None or self._zero_mask.shape[0] < M:
                 self._zero_mask = torch.zeros(max(M, 1), A, device=self.device)
             uniform = torch.rand(M, device=self.device)
             actions_t, probs_t = ops.masked_sample(policy, self._zero_mask[:M], uniform)
-            # ONE device-to-host transfer for the step's three result vectors
             packed = torch.cat([actions_t.float().unsqueeze(1),
                                 probs_t.unsqueeze(1), value], dim=1).cpu().numpy()
             actions = packed[:, 0].astype(np.int64)
@@ -130,14 +155,33 @@ class GeeseActorPool:
         else:
             # CPU path (tests / debugging)
             policy, value = self._policy_forward(torch.from_numpy(obs_sel).float())
-            tm['fwd'] += time.time() - t0
-            t0 = time.time()
             probs_full = torch.softmax(policy, dim=-1)
             actions_t = torch.multinomial(probs_full, 1).squeeze(-1)
             actions = actions_t.numpy()
             probs = probs_full.gather(-1, actions_t.unsqueeze(-1)).squeeze(-1).numpy()
             values = value.squeeze(-1).numpy()
+        tm['fwd'] += time.time() - t0
+        self._pending = ('done', gi, pi, live, M, actions, probs, values)
 
+    def _phase2(self):
+        """Wait for the issued forward, then apply actions: record, step
+        the vectorized env, package finished games.  Returns #frames."""
+        if self._pending is None:
+            return 0
+        import time
+        tm = self.timing
+        vec = self.vec
+        pend, self._pending = self._pending, None
+        t0 = time.time()
+        if pend[0] == 'async':
+            _, gi, pi, live, M = pend
+            self._event.synchronize()
+            packed = self._out_pin_np[:M]
+            actions = packed[:, 0].astype(np.int64)
+            probs = packed[:, 1]
+            values = packed[:, 2]
+        else:
+            _, gi, pi, live, M, actions, probs, values = pend
         tm['sample'] += time.time() - t0
         t0 = time.time()
         act_grid = np.zeros((self.n_games, N_PLAYERS), dtype=np.int32)
@@ -148,12 +192,8 @@ class GeeseActorPool:
         prob_row[gi, pi] = probs
         val_row[gi, pi] = values
 
-        # columnar recording: one fancy-indexed scatter per field
-        game_has_live = live.any(axis=1)
-        lg = np.nonzero(game_has_live)[0]
-        t_idx = self.rec_len[lg]
-        self.rec_obs[lg, t_idx] = obs_u8[lg]
-        self.rec_alive[lg, t_idx] = live[lg]
+        # finish the columnar step record started in _phase1
+        lg, t_idx = self._rec_slot
         self.rec_act[lg, t_idx] = act_grid[lg]
         self.rec_prob[lg, t_idx] = prob_row[lg]
         self.rec_val[lg, t_idx] = val_row[lg]
@@ -164,7 +204,7 @@ class GeeseActorPool:
         done = vec.step(act_grid)
         tm['env'] += time.time() - t0
         t0 = time.time()
-        self.frames += int(game_has_live.sum())
+        self.frames += len(lg)
 
         finished = np.nonzero(done)[0]
         if len(finished):
@@ -176,7 +216,7 @@ class GeeseActorPool:
             vec.reset_games(finished)
         tm['package'] += time.time() - t0
         tm['n'] += 1
-        return int(game_has_live.sum())
+        return len(lg)
 
     def _package(self, g, outcome_row):
         """Slice the game's columnar recording into an episode dict."""
@@ -207,3 +247,62 @@ class GeeseActorPool:
         out = self.completed
         self.completed = []
         return out
+
+
+class PipelinedGeesePool:
+    """Two actor shards in software pipeline: while shard A's forward runs
+    on the GPU (issued async through pinned staging), shard B's CPU side
+    (action unpack, columnar record, vectorized env step, episode
+    packaging) executes — hiding one side under the other.  One shared
+    fused-MFMA/hipGraph engine serves both shards."""
+
+    def __init__(self, model, args, n_games=512, device=None, seed=0):
+        half = max(1, n_games // 2)
+        first = GeeseActorPool(model, args, n_games=half, device=device, seed=seed)
+        engine = (first.fused, first.graphed)
+        second = GeeseActorPool(model, args, n_games=n_games - half,
+                                device=device, seed=seed + 7777, engine=engine)
+        self.pools = [first, second]
+        self.cur = 0
+        self.calls_per_vec_step = 2     # one call completes half the games
+
+    def step_once(self):
+        """Issue shard A's forward; complete shard B's step. Returns the
+        frames finished this call (~n_games/2)."""
+        self.pools[self.cur]._phase1()
+        frames = self.pools[1 - self.cur]._phase2()
+        self.cur ^= 1
+        return frames
+
+    def drain(self):
+        """Complete any in-flight shard step (call before harvesting all)."""
+        total = 0
+        for pool in self.pools:
+            total += pool._phase2()
+        return total
+
+    def refresh_weights(self):
+        self.pools[0].refresh_weights()   # the engine is shared
+
+    def harvest(self):
+        out = []
+        for pool in self.pools:
+            out.extend(pool.harvest())
+        return out
+
+    @property
+    def episodes_done(self):
+        return sum(p.episodes_done for p in self.pools)
+
+    @property
+    def frames(self):
+        return sum(p.frames for p in self.pools)
+
+    @property
+    def timing(self):
+        merged = {k: 0.0 for k in self.pools[0].timing}
+        for p in self.pools:
+            for k, v in p.timing.items():
+                merged[k] += v
+        merged['n'] = max(1, merged['n'] // 2)
+        return merged

@@ -142,3 +142,17 @@ class GraphedActorForward:
         # the outputs are sliced to [:M]
         graph.replay()
         return packed[:M]
+
+    def run_async(self, obs_pinned, M, out_pinned, event, n_actions=4):
+        """Pipelined variant: H2D from a pinned staging tensor, replay, and
+        an async D2H of the packed result into ``out_pinned``; ``event``
+        records completion.  No host sync — the caller overlaps CPU work
+        and waits on the event."""
+        bucket = 256 * ((M + 255) // 256)
+        if bucket not in self.graphs:
+            self._capture(bucket, n_actions)
+        graph, static_obs, packed = self.graphs[bucket]
+        static_obs[:M].copy_(obs_pinned[:M], non_blocking=True)
+        graph.replay()
+        out_pinned[:M].copy_(packed[:M], non_blocking=True)
+        event.record()
