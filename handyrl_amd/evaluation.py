@@ -10,7 +10,8 @@ from .environment import prepare_env, make_env
 from .connection import send_recv, accept_socket_connections, connect_socket_connection
 from .agent import RandomAgent, RuleBasedAgent, Agent, EnsembleAgent, SoftAgent
 
-NETWORK_MATCH_PORT = 9876
+import os as _os
+NETWORK_MATCH_PORT = int(_os.environ.get('HANDYRL_MATCH_PORT', 9876))
 
 
 def view(env, player=None):

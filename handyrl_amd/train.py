@@ -568,7 +568,10 @@ class Learner:
         threading.Thread(target=self.trainer.run, daemon=True).start()
         if self.gpu_actor:
             threading.Thread(target=self._gpu_actor_loop, daemon=True).start()
-        if self.args['worker'].get('num_parallel', 0) > 0:
+        # a LOCAL cluster with zero workers has nothing to spawn; the remote
+        # WorkerServer must always run (it accepts workers that join later)
+        if isinstance(self.worker, WorkerServer) or \
+                self.args['worker'].get('num_parallel', 0) > 0:
             self.worker.run()
         self.server()
 

@@ -26,8 +26,9 @@ from .evaluation import Evaluator
 from .generation import Generator
 from .model import ModelWrapper, RandomModel
 
-ENTRY_PORT = 9999
-WORKER_PORT = 9998
+import os as _os
+ENTRY_PORT = int(_os.environ.get('HANDYRL_ENTRY_PORT', 9999))
+WORKER_PORT = int(_os.environ.get('HANDYRL_WORKER_PORT', 9998))
 
 
 class Worker:
