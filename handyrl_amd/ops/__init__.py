@@ -99,13 +99,35 @@ def obs_to_nhwc(obs_u8):
     return require().obs_to_nhwc(obs_u8.reshape(obs_u8.shape[0], 17, 77).contiguous())
 
 
-def torus_conv_fused(x, wfrag, shift, nbr, residual, relu):
-    """y = act(torus_conv3x3(x) * scale + shift [+ x]) on NHWC bf16.
+def torus_conv_fused(x, wfrag, shift, nbr, residual=None, relu=False):
+    """y = act(torus_conv3x3(x) * scale + shift [+ residual]) on NHWC bf16.
 
     wfrag: (9,2,4,16,8) bf16 from pack_torus_weights; shift fp32 (32,);
-    nbr: (77,9) int32 wrap-around neighbor table (see torus_neighbor_table).
+    nbr: (77,9) int32 wrap-around neighbor table (see torus_neighbor_table);
+    residual: optional (N,77,32) bf16 tensor added before the activation.
     """
     return require().torus_conv_fused(x, wfrag, shift, nbr, residual, relu)
+
+
+def pack_weights_hip(w, scale=None, dgrad=False):
+    """GPU packing of (32, ci<=32, 3, 3) fp32 conv weights into the MFMA
+    fragment layout; dgrad=True produces the transposed/flipped weights for
+    the data-gradient conv."""
+    return require().pack_torus_weights_hip(w.contiguous(), scale, dgrad)
+
+
+def bn_nhwc_fwd(x, res, weight, bias, running_mean, running_var,
+                momentum, eps, relu):
+    """NHWC bf16 BN training fwd (+ optional residual add + relu).
+    Returns (y, save_mean, save_rstd)."""
+    return require().bn_nhwc_fwd(x, res, weight, bias, running_mean,
+                                 running_var, momentum, eps, relu)
+
+
+def bn_nhwc_bwd(x, dy, y, weight, save_mean, save_rstd, had_relu, want_dres):
+    """Backward of bn_nhwc_fwd. Returns (dx, dweight, dbias, dres)."""
+    return require().bn_nhwc_bwd(x, dy.contiguous(), y, weight,
+                                 save_mean, save_rstd, had_relu, want_dres)
 
 
 def torus_neighbor_table(device=None):

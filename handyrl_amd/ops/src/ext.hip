@@ -290,6 +290,168 @@ __global__ __launch_bounds__(256) void bn_bwd_apply_kernel(
     }
 }
 
+// --- weight packing for the MFMA torus conv ------------------------------
+// Packs (co=32, ci<=32, 3, 3) fp32 conv weights into the B-fragment layout
+// frag[tap][cotile][khi][lane_lo][e] (bf16), optionally scaled per out-
+// channel (BN fold) and optionally in DGRAD form (flip taps, swap ci/co).
+// One kernel instead of ~10 torch ops per layer per refresh.
+__global__ void pack_torus_weights_kernel(
+    const float* __restrict__ w,      // (32, ci_in, 3, 3)
+    const float* __restrict__ scale,  // (32,) or nullptr
+    short* __restrict__ frag,         // (9, 2, 4, 16, 8) bf16
+    int ci_in, int dgrad)
+{
+    // flat output index over 9*2*4*16*8 = 9216
+    const int idx = blockIdx.x * blockDim.x + threadIdx.x;
+    if (idx >= 9 * 2 * 4 * 16 * 8) return;
+    const int e = idx & 7;
+    const int lo = (idx >> 3) & 15;
+    const int khi = (idx >> 7) & 3;
+    const int cotile = (idx >> 9) & 1;
+    const int tap = idx >> 10;
+    const int k = khi * 8 + e;            // ci index (0..31)
+    const int n = cotile * 16 + lo;       // co index (0..31)
+    float v = 0.f;
+    if (!dgrad) {
+        // frag[k=ci][n=co] at tap
+        if (k < ci_in)
+            v = w[((n * ci_in) + k) * 9 + tap];
+        if (scale) v *= scale[n];
+    } else {
+        // dgrad conv: W'[tap][k=co][n=ci] = W[co=k][ci=n][8-tap]
+        if (n < ci_in)
+            v = w[((k * ci_in) + n) * 9 + (8 - tap)];
+    }
+    frag[idx] = f2bf(v);
+}
+
+// --- NHWC BatchNorm (training) with fused residual-add + ReLU ------------
+// Activations (N, 77, 32) bf16: channels are innermost, so thread t
+// (stride 256, 256 % 32 == 0) always sees channel c = t & 31 — private
+// accumulators + perfectly coalesced loads.  apply: y = relu(bn(x) + res).
+template <typename T>
+__global__ __launch_bounds__(256) void bn_nhwc_stats_kernel(
+    const T* __restrict__ x, float* __restrict__ accum,   // (2C)
+    long total, int C)
+{
+    const int c = threadIdx.x & 31;
+    float acc = 0.f, acc2 = 0.f;
+    for (long i = (long)blockIdx.x * 256 + threadIdx.x; i < total;
+         i += (long)gridDim.x * 256) {
+        const float v = load_as_f32(x + i);
+        acc += v;
+        acc2 += v * v;
+    }
+    __shared__ float s1[256], s2[256];
+    s1[threadIdx.x] = acc; s2[threadIdx.x] = acc2;
+    __syncthreads();
+    // reduce the 8 threads per channel (stride 32 within the block)
+    if (threadIdx.x < 32) {
+        float t1 = 0.f, t2 = 0.f;
+        for (int j = threadIdx.x; j < 256; j += 32) { t1 += s1[j]; t2 += s2[j]; }
+        atomicAdd(&accum[c], t1);
+        atomicAdd(&accum[C + c], t2);
+    }
+}
+
+template <typename T>
+__global__ __launch_bounds__(256) void bn_nhwc_apply_kernel(
+    const T* __restrict__ x, const T* __restrict__ res,   // res nullable
+    T* __restrict__ y,
+    const float* __restrict__ accum,
+    const float* __restrict__ weight, const float* __restrict__ bias,
+    float* __restrict__ running_mean, float* __restrict__ running_var,
+    float* __restrict__ save_mean, float* __restrict__ save_rstd,
+    long total, int C, float momentum, float eps, int apply_relu)
+{
+    const long i0 = (long)blockIdx.x * 256 + threadIdx.x;
+    const float M = (float)(total / C);
+    if (i0 < C) {    // publish stats + running updates once
+        const int c = (int)i0;
+        const float mean = accum[c] / M;
+        const float var = fmaxf(accum[C + c] / M - mean * mean, 0.f);
+        save_mean[c] = mean;
+        save_rstd[c] = rsqrtf(var + eps);
+        const float unbiased = var * (M / fmaxf(M - 1.f, 1.f));
+        running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * mean;
+        running_var[c] = (1.f - momentum) * running_var[c] + momentum * unbiased;
+    }
+    const int c = threadIdx.x & 31;
+    const float mean = accum[c] / M;
+    const float var = fmaxf(accum[C + c] / M - mean * mean, 0.f);
+    const float rstd = rsqrtf(var + eps);
+    const float scale = weight[c] * rstd;
+    const float shift = bias[c] - mean * scale;
+    for (long i = i0; i < total; i += (long)gridDim.x * 256) {
+        float v = load_as_f32(x + i) * scale + shift;
+        if (res) v += load_as_f32(res + i);
+        if (apply_relu) v = fmaxf(v, 0.f);
+        store_f32(y + i, v);
+    }
+}
+
+// backward of y = relu(bn(x) + res): given dy and y, the relu mask is
+// y > 0; dz = dy * mask flows to BOTH the bn input grad and the residual
+// grad (dres = dz).  stats: sum(dz), sum(dz * xhat).
+template <typename T>
+__global__ __launch_bounds__(256) void bn_nhwc_bwd_stats_kernel(
+    const T* __restrict__ x, const T* __restrict__ dy, const T* __restrict__ y,
+    float* __restrict__ accum,
+    const float* __restrict__ save_mean, const float* __restrict__ save_rstd,
+    long total, int C, int had_relu)
+{
+    const int c = threadIdx.x & 31;
+    const float mean = save_mean[c], rstd = save_rstd[c];
+    float acc = 0.f, acc2 = 0.f;
+    for (long i = (long)blockIdx.x * 256 + threadIdx.x; i < total;
+         i += (long)gridDim.x * 256) {
+        float g = load_as_f32(dy + i);
+        if (had_relu && load_as_f32(y + i) <= 0.f) g = 0.f;
+        const float xh = (load_as_f32(x + i) - mean) * rstd;
+        acc += g;
+        acc2 += g * xh;
+    }
+    __shared__ float s1[256], s2[256];
+    s1[threadIdx.x] = acc; s2[threadIdx.x] = acc2;
+    __syncthreads();
+    if (threadIdx.x < 32) {
+        float t1 = 0.f, t2 = 0.f;
+        for (int j = threadIdx.x; j < 256; j += 32) { t1 += s1[j]; t2 += s2[j]; }
+        atomicAdd(&accum[c], t1);
+        atomicAdd(&accum[C + c], t2);
+    }
+}
+
+template <typename T>
+__global__ __launch_bounds__(256) void bn_nhwc_bwd_apply_kernel(
+    const T* __restrict__ x, const T* __restrict__ dy, const T* __restrict__ y,
+    T* __restrict__ dx, T* __restrict__ dres,            // dres nullable
+    const float* __restrict__ accum,
+    const float* __restrict__ weight,
+    const float* __restrict__ save_mean, const float* __restrict__ save_rstd,
+    float* __restrict__ dweight, float* __restrict__ dbias,
+    long total, int C, int had_relu)
+{
+    const long i0 = (long)blockIdx.x * 256 + threadIdx.x;
+    if (i0 < C) {
+        dbias[i0] = accum[i0];
+        dweight[i0] = accum[C + i0];
+    }
+    const int c = threadIdx.x & 31;
+    const float M = (float)(total / C);
+    const float mean = save_mean[c], rstd = save_rstd[c];
+    const float k_dy = accum[c] / M;
+    const float k_dyx = accum[C + c] / M;
+    const float wr = weight[c] * rstd;
+    for (long i = i0; i < total; i += (long)gridDim.x * 256) {
+        float g = load_as_f32(dy + i);
+        if (had_relu && load_as_f32(y + i) <= 0.f) g = 0.f;
+        if (dres) store_f32(dres + i, g);
+        const float xh = (load_as_f32(x + i) - mean) * rstd;
+        store_f32(dx + i, wr * (g - k_dy - xh * k_dyx));
+    }
+}
+
 // --- MFMA fragment-layout probe (test harness for the conv kernel) ------
 // Computes D(16x16) = A(16x32) @ B(32x16) with one v_mfma_f32_16x16x32_bf16
 // using the assumed lane->fragment mapping:
@@ -344,9 +506,10 @@ __global__ __launch_bounds__(256) void torus_conv_fused_kernel(
     const short* __restrict__ wfrag,    // (9,2,4,16,8) bf16, BN-folded
     const float* __restrict__ shift,    // (32,)
     const int* __restrict__ nbr,        // (77,9) wrap-around neighbor cells
+    const short* __restrict__ res,      // nullable residual, (N,77,32) bf16
     short* __restrict__ y,              // (N,77,32) bf16
     long total_pos,                     // N*77
-    int add_residual, int apply_relu) {
+    int apply_relu) {
     const int wid = threadIdx.x >> 6;
     const int lane = threadIdx.x & 63;
     const int khi = lane >> 4, lo = lane & 15;
@@ -378,9 +541,9 @@ __global__ __launch_bounds__(256) void torus_conv_fused_kernel(
         float v0 = acc0[r] + sh0;
         float v1 = acc1[r] + sh1;
         const long off = prow * 32;
-        if (add_residual) {
-            v0 += bf2f(x[off + lo]);
-            v1 += bf2f(x[off + 16 + lo]);
+        if (res) {
+            v0 += bf2f(res[off + lo]);
+            v1 += bf2f(res[off + 16 + lo]);
         }
         if (apply_relu) {
             v0 = fmaxf(v0, 0.f);
@@ -498,7 +661,7 @@ static torch::Tensor obs_to_nhwc(torch::Tensor obs) {
 
 static torch::Tensor torus_conv_fused(
     torch::Tensor x, torch::Tensor wfrag, torch::Tensor shift,
-    torch::Tensor nbr, bool add_residual, bool apply_relu) {
+    torch::Tensor nbr, c10::optional<torch::Tensor> res, bool apply_relu) {
     TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kBFloat16);
     TORCH_CHECK(x.dim() == 3 && x.size(1) == 77 && x.size(2) == 32,
                 "torus_conv_fused: x must be (N,77,32) bf16");
@@ -506,11 +669,87 @@ static torch::Tensor torus_conv_fused(
     auto y = torch::empty_like(x);
     const long grid = (total_pos + 63) / 64;
     auto stream = at::cuda::getCurrentCUDAStream();
+    const short* res_ptr = res.has_value() ? (const short*)res->data_ptr() : nullptr;
     hipLaunchKernelGGL(torus_conv_fused_kernel, dim3(grid), dim3(256), 0, stream,
         (const short*)x.data_ptr(), (const short*)wfrag.data_ptr(),
-        shift.data_ptr<float>(), nbr.data_ptr<int>(), (short*)y.data_ptr(),
-        total_pos, (int)add_residual, (int)apply_relu);
+        shift.data_ptr<float>(), nbr.data_ptr<int>(), res_ptr, (short*)y.data_ptr(),
+        total_pos, (int)apply_relu);
     return y;
+}
+
+static torch::Tensor pack_torus_weights_hip(
+    torch::Tensor w, c10::optional<torch::Tensor> scale, bool dgrad) {
+    TORCH_CHECK(w.is_cuda() && w.scalar_type() == torch::kFloat32 && w.dim() == 4);
+    const int ci_in = w.size(1);
+    TORCH_CHECK(!dgrad || ci_in == 32, "dgrad pack needs ci==32");
+    auto frag = torch::empty({9, 2, 4, 16, 8}, w.options().dtype(torch::kBFloat16));
+    auto stream = at::cuda::getCurrentCUDAStream();
+    const float* sc = scale.has_value() ? scale->data_ptr<float>() : nullptr;
+    hipLaunchKernelGGL(pack_torus_weights_kernel, dim3(36), dim3(256), 0, stream,
+        w.data_ptr<float>(), sc, (short*)frag.data_ptr(), ci_in, (int)dgrad);
+    return frag;
+}
+
+static int bn_grid(long total) {
+    const long blocks = (total + 255) / 256;
+    return (int)std::min<long>(blocks, 2048);
+}
+
+static std::vector<torch::Tensor> bn_nhwc_fwd(
+    torch::Tensor x, c10::optional<torch::Tensor> res,
+    torch::Tensor weight, torch::Tensor bias,
+    torch::Tensor running_mean, torch::Tensor running_var,
+    double momentum, double eps, bool apply_relu) {
+    TORCH_CHECK(x.is_cuda() && x.dim() == 3 && x.size(2) == 32,
+                "bn_nhwc_fwd: (N,77,32) expected");
+    const int C = x.size(2);
+    const long total = x.numel();
+    auto y = torch::empty_like(x);
+    auto accum = torch::empty({2 * C}, weight.options());
+    auto save_mean = torch::empty({C}, weight.options());
+    auto save_rstd = torch::empty({C}, weight.options());
+    auto stream = at::cuda::getCurrentCUDAStream();
+    hipMemsetAsync(accum.data_ptr(), 0, 2 * C * sizeof(float), stream);
+    TORCH_CHECK(x.scalar_type() == torch::kBFloat16, "bn_nhwc: bf16 only");
+    const short* res_ptr = res.has_value() ? (const short*)res->data_ptr() : nullptr;
+    hipLaunchKernelGGL(bn_nhwc_stats_kernel<short>, dim3(bn_grid(total)), dim3(256), 0,
+        stream, (const short*)x.data_ptr(), accum.data_ptr<float>(), total, C);
+    hipLaunchKernelGGL(bn_nhwc_apply_kernel<short>, dim3(bn_grid(total)), dim3(256), 0,
+        stream, (const short*)x.data_ptr(), res_ptr, (short*)y.data_ptr(),
+        accum.data_ptr<float>(), weight.data_ptr<float>(), bias.data_ptr<float>(),
+        running_mean.data_ptr<float>(), running_var.data_ptr<float>(),
+        save_mean.data_ptr<float>(), save_rstd.data_ptr<float>(),
+        total, C, (float)momentum, (float)eps, (int)apply_relu);
+    return {y, save_mean, save_rstd};
+}
+
+static std::vector<torch::Tensor> bn_nhwc_bwd(
+    torch::Tensor x, torch::Tensor dy, torch::Tensor y, torch::Tensor weight,
+    torch::Tensor save_mean, torch::Tensor save_rstd,
+    bool had_relu, bool want_dres) {
+    const int C = x.size(2);
+    const long total = x.numel();
+    auto dx = torch::empty_like(x);
+    auto dres = want_dres ? torch::empty_like(x) : torch::Tensor();
+    auto accum = torch::empty({2 * C}, weight.options());
+    auto dweight = torch::empty({C}, weight.options());
+    auto dbias = torch::empty({C}, weight.options());
+    auto stream = at::cuda::getCurrentCUDAStream();
+    hipMemsetAsync(accum.data_ptr(), 0, 2 * C * sizeof(float), stream);
+    hipLaunchKernelGGL(bn_nhwc_bwd_stats_kernel<short>, dim3(bn_grid(total)), dim3(256),
+        0, stream, (const short*)x.data_ptr(), (const short*)dy.data_ptr(),
+        (const short*)y.data_ptr(), accum.data_ptr<float>(),
+        save_mean.data_ptr<float>(), save_rstd.data_ptr<float>(),
+        total, C, (int)had_relu);
+    hipLaunchKernelGGL(bn_nhwc_bwd_apply_kernel<short>, dim3(bn_grid(total)), dim3(256),
+        0, stream, (const short*)x.data_ptr(), (const short*)dy.data_ptr(),
+        (const short*)y.data_ptr(), (short*)dx.data_ptr(),
+        want_dres ? (short*)dres.data_ptr() : nullptr,
+        accum.data_ptr<float>(), weight.data_ptr<float>(),
+        save_mean.data_ptr<float>(), save_rstd.data_ptr<float>(),
+        dweight.data_ptr<float>(), dbias.data_ptr<float>(),
+        total, C, (int)had_relu);
+    return {dx, dweight, dbias, dres};
 }
 
 static std::vector<torch::Tensor> target_scan(
@@ -572,6 +811,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("obs_to_nhwc", &obs_to_nhwc, "uint8 NCHW obs -> padded NHWC bf16");
     m.def("torus_conv_fused", &torus_conv_fused,
           "fused wrap-around conv3x3 + BN-fold + residual + relu (MFMA)");
+    m.def("pack_torus_weights_hip", &pack_torus_weights_hip,
+          "pack conv weights into MFMA fragment layout (fwd or dgrad)");
+    m.def("bn_nhwc_fwd", &bn_nhwc_fwd,
+          "NHWC BN training fwd + fused residual/relu");
+    m.def("bn_nhwc_bwd", &bn_nhwc_bwd,
+          "NHWC BN training bwd + fused relu mask / residual grad");
     m.def("bn_train_fwd", &bn_train_fwd,
           "BatchNorm2d training forward (y, save_mean, save_rstd)");
     m.def("bn_train_bwd", &bn_train_bwd,

@@ -96,7 +96,7 @@ def test_torus_conv_fused_matches_eager(residual):
     wfrag = ops.pack_torus_weights(layer.conv.weight, scale)
     nbr = ops.torus_neighbor_table('cuda')
     y = ops.torus_conv_fused(x_nhwc, wfrag, shift.contiguous(), nbr,
-                             residual, True)
+                             x_nhwc if residual else None, True)
     torch.cuda.synchronize()
     y_nchw = y.float().reshape(N, 7, 11, 32).permute(0, 3, 1, 2)
     # bf16 inputs/weights: tolerance at bf16 resolution of the accumulations
@@ -173,6 +173,52 @@ def test_fused_bn_train_matches_stock(dtype):
         else dict(rtol=5e-2, atol=1.0)
     torch.testing.assert_close(bn_mine.weight.grad, bn_ref.weight.grad, **wtol)
     torch.testing.assert_close(bn_mine.bias.grad, bn_ref.bias.grad, **wtol)
+
+
+@requires_gpu
+def test_custom_training_path_matches_eager():
+    """The hand-written NHWC training path (MFMA conv fwd/dgrad + NHWC BN +
+    gather/bmm wgrad) vs the eager torch path: outputs, every parameter
+    gradient, and BN running stats."""
+    import copy
+    import os
+    from handyrl_amd.models.geese_net import GeeseNet
+    torch.manual_seed(3)
+    net_ref = GeeseNet(layers=4).cuda()
+    net_mine = copy.deepcopy(net_ref)
+    net_ref.train(); net_mine.train()
+
+    obs = (torch.rand(512, 17, 7, 11, device='cuda') < 0.2).float()
+
+    os.environ['HANDYRL_NO_FUSED'] = '1'
+    out_ref = net_ref(obs, None)
+    os.environ.pop('HANDYRL_NO_FUSED')
+    out_mine = net_mine(obs, None)
+    torch.cuda.synchronize()
+    assert not torch.allclose(out_mine['policy'],
+                              torch.zeros_like(out_mine['policy']))
+    torch.testing.assert_close(out_mine['policy'], out_ref['policy'],
+                               rtol=5e-2, atol=5e-2)
+    torch.testing.assert_close(out_mine['value'], out_ref['value'],
+                               rtol=5e-2, atol=5e-2)
+
+    loss_ref = out_ref['policy'].square().sum() + out_ref['value'].square().sum()
+    loss_mine = out_mine['policy'].square().sum() + out_mine['value'].square().sum()
+    loss_ref.backward()
+    loss_mine.backward()
+    torch.cuda.synchronize()
+
+    for (name, p_ref), (_, p_mine) in zip(net_ref.named_parameters(),
+                                          net_mine.named_parameters()):
+        assert p_mine.grad is not None, name
+        scale = p_ref.grad.abs().mean().clamp(min=1e-6)
+        rel = (p_mine.grad - p_ref.grad).abs().max() / scale
+        assert rel < 0.25, '%s: rel grad err %.3f' % (name, rel)
+    for (name, b_ref), (_, b_mine) in zip(net_ref.named_buffers(),
+                                          net_mine.named_buffers()):
+        if b_ref.dtype.is_floating_point:
+            torch.testing.assert_close(b_mine, b_ref, rtol=2e-2, atol=2e-2,
+                                       msg=lambda m: '%s: %s' % (name, m))
 
 
 @requires_gpu
