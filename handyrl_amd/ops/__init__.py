@@ -116,6 +116,11 @@ def pack_weights_hip(w, scale=None, dgrad=False):
     return require().pack_torus_weights_hip(w.contiguous(), scale, dgrad)
 
 
+def torus_wgrad(x, dy, nbr):
+    """dW[tap][ci][co] for the torus conv, fp32 (9,32,32)."""
+    return require().torus_wgrad(x, dy, nbr)
+
+
 def bn_nhwc_fwd(x, res, weight, bias, running_mean, running_var,
                 momentum, eps, relu):
     """NHWC bf16 BN training fwd (+ optional residual add + relu).
