@@ -31,7 +31,7 @@ from handyrl_amd.batch import Batcher
 from handyrl_amd.models.geese_net import GeeseNet
 from handyrl_amd.train import Trainer
 
-N_ENVS = 512            # self-play games per GPU (measured best 1-GPU throughput)
+N_ENVS = 768            # self-play games per GPU (measured best 1-GPU throughput)
 ACTOR_VEC_STEPS = 16    # env transitions (per game) per learner step
 
 
@@ -135,10 +135,15 @@ def main():
         t1 = time.time()
         batch = batcher.batch()
         t2 = time.time()
-        losses, dcnt = trainer.train_step(batch)
+        graphed = getattr(trainer, 'graphed_step', None)
+        if graphed is not None:
+            # issue the whole train step asynchronously: its GPU time hides
+            # under the next actor phase (no host sync, no loss readback)
+            losses, _ = graphed.step(batch)
+            trainer.steps += 0   # steps counted inside graphed.step
+        else:
+            losses, dcnt = trainer.train_step(batch)
         pool.refresh_weights()          # re-fold BN into the MFMA actor path
-        if use_cuda:
-            torch.cuda.synchronize()
         t3 = time.time()
         phase_t['actor'] += t1 - t0
         phase_t['batch_wait'] += t2 - t1

@@ -193,9 +193,14 @@ class GeeseVecEnv:
     def observations(self):
         """uint8 (G, 4, 17, 7, 11): the 17-plane encoding for every seat."""
         G = self.G
-        head_grid = np.zeros((G, N_PLAYERS, N_CELLS), dtype=np.uint8)
-        tail_grid = np.zeros_like(head_grid)
-        prev_grid = np.zeros_like(head_grid)
+        if not hasattr(self, '_obs_buf'):
+            self._obs_buf = np.empty((G, N_PLAYERS, 17, N_CELLS), dtype=np.uint8)
+            self._scratch = np.zeros((3, G, N_PLAYERS, N_CELLS), dtype=np.uint8)
+            self._food_grid = np.zeros((G, N_CELLS), dtype=np.uint8)
+        head_grid, tail_grid, prev_grid = self._scratch
+        head_grid[:] = 0
+        tail_grid[:] = 0
+        prev_grid[:] = 0
         heads = self._head()
         tails = self._tail_cell()
         gi, pi = np.nonzero(self.alive)
@@ -204,22 +209,22 @@ class GeeseVecEnv:
         gi, pi = np.nonzero(self.prev_head >= 0)
         prev_grid[gi, pi, self.prev_head[gi, pi]] = 1
 
-        food_grid = np.zeros((G, N_CELLS), dtype=np.uint8)
+        food_grid = self._food_grid
+        food_grid[:] = 0
         for f in range(MIN_FOOD):
             ok = self.food[:, f] >= 0
             food_grid[np.nonzero(ok)[0], self.food[ok, f]] = 1
 
-        # groups stacked on a new axis: (G, group, goose, cell)
-        groups = np.stack([head_grid, tail_grid, self.body_grid, prev_grid], axis=1)
-        # seat-relative gather over the goose axis -> (G, seat, group, goose, cell)
-        rel = groups[:, :, _REL, :]                   # (G, group, seat, goose, cell)
-        rel = rel.transpose(0, 2, 1, 3, 4)            # (G, seat, group, goose, cell)
-        obs = np.empty((G, N_PLAYERS, 17, N_CELLS), dtype=np.uint8)
-        obs[:, :, 0:4] = rel[:, :, 0]
-        obs[:, :, 4:8] = rel[:, :, 1]
-        obs[:, :, 8:12] = rel[:, :, 2]
-        obs[:, :, 12:16] = rel[:, :, 3]
-        obs[:, :, 16] = food_grid[:, None, :]
+        # per-seat relative goose order, written straight into the output
+        # buffer (no (G,4,4,4,77) intermediate)
+        obs = self._obs_buf
+        for k in range(N_PLAYERS):
+            perm = _REL[k]
+            obs[:, k, 0:4] = head_grid[:, perm]
+            obs[:, k, 4:8] = tail_grid[:, perm]
+            obs[:, k, 8:12] = self.body_grid[:, perm]
+            obs[:, k, 12:16] = prev_grid[:, perm]
+            obs[:, k, 16] = food_grid
         return obs.reshape(G, N_PLAYERS, 17, ROWS, COLS)
 
     def outcomes(self, games):
