@@ -330,25 +330,25 @@ __global__ void pack_torus_weights_kernel(
 // (stride 256, 256 % 32 == 0) always sees channel c = t & 31 — private
 // accumulators + perfectly coalesced loads.  apply: y = relu(bn(x) + res).
 template <typename T>
-__global__ __launch_bounds__(256) void bn_nhwc_stats_kernel(
+__global__ __launch_bounds__(1024) void bn_nhwc_stats_kernel(
     const T* __restrict__ x, float* __restrict__ accum,   // (2C)
     long total, int C)
 {
     const int c = threadIdx.x & 31;
     float acc = 0.f, acc2 = 0.f;
-    for (long i = (long)blockIdx.x * 256 + threadIdx.x; i < total;
-         i += (long)gridDim.x * 256) {
+    for (long i = (long)blockIdx.x * 1024 + threadIdx.x; i < total;
+         i += (long)gridDim.x * 1024) {
         const float v = load_as_f32(x + i);
         acc += v;
         acc2 += v * v;
     }
-    __shared__ float s1[256], s2[256];
+    __shared__ float s1[1024], s2[1024];
     s1[threadIdx.x] = acc; s2[threadIdx.x] = acc2;
     __syncthreads();
-    // reduce the 8 threads per channel (stride 32 within the block)
+    // reduce the 32 threads per channel (stride 32 within the block)
     if (threadIdx.x < 32) {
         float t1 = 0.f, t2 = 0.f;
-        for (int j = threadIdx.x; j < 256; j += 32) { t1 += s1[j]; t2 += s2[j]; }
+        for (int j = threadIdx.x; j < 1024; j += 32) { t1 += s1[j]; t2 += s2[j]; }
         atomicAdd(&accum[c], t1);
         atomicAdd(&accum[C + c], t2);
     }
@@ -394,7 +394,7 @@ __global__ __launch_bounds__(256) void bn_nhwc_apply_kernel(
 // y > 0; dz = dy * mask flows to BOTH the bn input grad and the residual
 // grad (dres = dz).  stats: sum(dz), sum(dz * xhat).
 template <typename T>
-__global__ __launch_bounds__(256) void bn_nhwc_bwd_stats_kernel(
+__global__ __launch_bounds__(1024) void bn_nhwc_bwd_stats_kernel(
     const T* __restrict__ x, const T* __restrict__ dy, const T* __restrict__ y,
     float* __restrict__ accum,
     const float* __restrict__ save_mean, const float* __restrict__ save_rstd,
@@ -403,20 +403,20 @@ __global__ __launch_bounds__(256) void bn_nhwc_bwd_stats_kernel(
     const int c = threadIdx.x & 31;
     const float mean = save_mean[c], rstd = save_rstd[c];
     float acc = 0.f, acc2 = 0.f;
-    for (long i = (long)blockIdx.x * 256 + threadIdx.x; i < total;
-         i += (long)gridDim.x * 256) {
+    for (long i = (long)blockIdx.x * 1024 + threadIdx.x; i < total;
+         i += (long)gridDim.x * 1024) {
         float g = load_as_f32(dy + i);
         if (had_relu && load_as_f32(y + i) <= 0.f) g = 0.f;
         const float xh = (load_as_f32(x + i) - mean) * rstd;
         acc += g;
         acc2 += g * xh;
     }
-    __shared__ float s1[256], s2[256];
+    __shared__ float s1[1024], s2[1024];
     s1[threadIdx.x] = acc; s2[threadIdx.x] = acc2;
     __syncthreads();
     if (threadIdx.x < 32) {
         float t1 = 0.f, t2 = 0.f;
-        for (int j = threadIdx.x; j < 256; j += 32) { t1 += s1[j]; t2 += s2[j]; }
+        for (int j = threadIdx.x; j < 1024; j += 32) { t1 += s1[j]; t2 += s2[j]; }
         atomicAdd(&accum[c], t1);
         atomicAdd(&accum[C + c], t2);
     }
@@ -734,7 +734,7 @@ static torch::Tensor torus_wgrad(torch::Tensor x, torch::Tensor dy,
     auto dW = torch::empty({9, 32, 32}, x.options().dtype(torch::kFloat32));
     auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
     hipMemsetAsync(dW.data_ptr(), 0, 9 * 32 * 32 * sizeof(float), stream);
-    const int blocks = std::max(1, std::min(N, 128));
+    const int blocks = std::max(1, std::min(N, 256));
     hipLaunchKernelGGL(torus_wgrad_kernel, dim3(blocks), dim3(256), 0, stream,
         (const short*)x.data_ptr(), (const short*)dy.data_ptr(),
         nbr.data_ptr<int>(), dW.data_ptr<float>(), N);
@@ -803,7 +803,7 @@ static int bn_grid(long total) {
 // stats kernels end with 64 atomics per block on 64 addresses: keep the
 // block count low enough that contention stays off the critical path
 static int bn_stats_grid(long total) {
-    const long blocks = (total + 255) / 256;
+    const long blocks = (total + 1023) / 1024;
     return (int)std::min<long>(blocks, 240);
 }
 
@@ -824,7 +824,7 @@ static std::vector<torch::Tensor> bn_nhwc_fwd(
     hipMemsetAsync(accum.data_ptr(), 0, 2 * C * sizeof(float), stream);
     TORCH_CHECK(x.scalar_type() == torch::kBFloat16, "bn_nhwc: bf16 only");
     const short* res_ptr = res.has_value() ? (const short*)res->data_ptr() : nullptr;
-    hipLaunchKernelGGL(bn_nhwc_stats_kernel<short>, dim3(bn_stats_grid(total)), dim3(256), 0,
+    hipLaunchKernelGGL(bn_nhwc_stats_kernel<short>, dim3(bn_stats_grid(total)), dim3(1024), 0,
         stream, (const short*)x.data_ptr(), accum.data_ptr<float>(), total, C);
     hipLaunchKernelGGL(bn_nhwc_apply_kernel<short>, dim3(bn_grid(total)), dim3(256), 0,
         stream, (const short*)x.data_ptr(), res_ptr, (short*)y.data_ptr(),
@@ -848,7 +848,7 @@ static std::vector<torch::Tensor> bn_nhwc_bwd(
     auto dbias = torch::empty({C}, weight.options());
     auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
     hipMemsetAsync(accum.data_ptr(), 0, 2 * C * sizeof(float), stream);
-    hipLaunchKernelGGL(bn_nhwc_bwd_stats_kernel<short>, dim3(bn_stats_grid(total)), dim3(256),
+    hipLaunchKernelGGL(bn_nhwc_bwd_stats_kernel<short>, dim3(bn_stats_grid(total)), dim3(1024),
         0, stream, (const short*)x.data_ptr(), (const short*)dy.data_ptr(),
         (const short*)y.data_ptr(), accum.data_ptr<float>(),
         save_mean.data_ptr<float>(), save_rstd.data_ptr<float>(),
