@@ -77,11 +77,19 @@ def main():
 
     args = bench_args(cli.batch_size, cli.forward_steps)
 
-    # fork the batch-builder processes FIRST, before any HIP context or
-    # torch thread-pool exists in this process
+    # fork the batch-builder and actor-env processes FIRST, before any HIP
+    # context or torch thread-pool exists in this process
     from handyrl_amd.batch import EpisodeBuffer
     buffer = EpisodeBuffer(args)
     batcher = Batcher(args, buffer)
+
+    actor_procs = int(os.environ.get('HANDYRL_ACTOR_PROCS', '3'))
+    mpool = None
+    if actor_procs > 0:
+        from handyrl_amd.actor import MultiProcGeesePool
+        mpool = MultiProcGeesePool(args, n_games=cli.envs,
+                                   seed=1000 + hdist.env_rank() * 31,
+                                   workers=actor_procs)
 
     use_cuda = torch.cuda.is_available()
     device = torch.device('cuda', local_rank) if use_cuda else torch.device('cpu')
@@ -96,11 +104,15 @@ def main():
     if world > 1:
         hdist.broadcast_params(trainer.model)
 
-    # pipelined two-shard pool on GPU (CPU env work overlaps GPU forwards);
-    # plain pool on CPU
-    pool_cls = PipelinedGeesePool if use_cuda else GeeseActorPool
-    pool = pool_cls(trainer.model, args, n_games=cli.envs,
-                    device=device, seed=1000 + rank)
+    if mpool is not None:
+        # env work in child processes; parent runs the inference engine
+        mpool.attach(trainer.model, device)
+        pool = mpool
+    else:
+        # in-process fallback: pipelined two-shard pool on GPU, plain on CPU
+        pool_cls = PipelinedGeesePool if use_cuda else GeeseActorPool
+        pool = pool_cls(trainer.model, args, n_games=cli.envs,
+                        device=device, seed=1000 + rank)
     actor_calls = getattr(pool, 'calls_per_vec_step', 1)
 
     def pump_actor(n_vec_steps):

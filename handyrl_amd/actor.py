@@ -60,7 +60,7 @@ class GeeseActorPool:
         self.fused = None
         if engine is not None:
             self.fused, self.graphed = engine
-        else:
+        elif model is not None:
             if self.device.type == 'cuda' and os.environ.get('HANDYRL_NO_FUSED') != '1' \
                     and hasattr(model, 'stem') and ops.available():
                 from .models.geese_net import GeeseFusedEval
@@ -94,6 +94,40 @@ class GeeseActorPool:
     def step_once(self):
         """Advance every live game by one transition; returns #frames."""
         self._phase1()
+        return self._phase2()
+
+    # -- external-inference mode (env-worker processes) --------------------
+    def prepare_step(self, out_buf):
+        """Build observations, record obs/alive columns, and write the
+        selected live-seat observations into ``out_buf[:M]``.  Returns M.
+        The caller supplies (actions, probs, values) to complete_step."""
+        import time
+        tm = self.timing
+        vec = self.vec
+        t0 = time.time()
+        obs_u8 = vec.observations()
+        tm['obs'] += time.time() - t0
+        live = vec.alive & ~vec.over[:, None]
+        gi, pi = np.nonzero(live)
+        if len(gi) == 0:
+            self._pending = None
+            return 0
+        game_has_live = live.any(axis=1)
+        lg = np.nonzero(game_has_live)[0]
+        t_idx = self.rec_len[lg]
+        self.rec_obs[lg, t_idx] = obs_u8[lg]
+        self.rec_alive[lg, t_idx] = live[lg]
+        self._rec_slot = (lg, t_idx)
+        M = len(gi)
+        out_buf[:M] = obs_u8[gi, pi]
+        self._pending = ('ext', gi, pi, live, M)
+        return M
+
+    def complete_step(self, actions, probs, values):
+        """Apply externally computed inference results; returns #frames."""
+        assert self._pending is not None and self._pending[0] == 'ext'
+        _, gi, pi, live, M = self._pending
+        self._pending = ('done', gi, pi, live, M, actions, probs, values)
         return self._phase2()
 
     def _phase1(self):
@@ -306,3 +340,173 @@ class PipelinedGeesePool:
                 merged[k] += v
         merged['n'] = max(1, merged['n'] // 2)
         return merged
+
+
+def _geese_env_worker(conn, obs_name, res_name, n_games, args, seed):
+    """Env-side child process: vectorized stepping, columnar recording and
+    episode packaging on host cores; observations/results move through
+    shared memory, inference runs in the parent (GPU)."""
+    from multiprocessing import shared_memory
+    obs_shm = shared_memory.SharedMemory(name=obs_name)
+    res_shm = shared_memory.SharedMemory(name=res_name)
+    cap = n_games * N_PLAYERS
+    obs_view = np.ndarray((cap, 17, 7, 11), dtype=np.uint8, buffer=obs_shm.buf)
+    res_view = np.ndarray((cap, 3), dtype=np.float32, buffer=res_shm.buf)
+
+    pool = GeeseActorPool(None, args, n_games=n_games,
+                          device=torch.device('cpu'), use_graphs=False,
+                          seed=seed)
+    frames_prev = 0
+    eps_out = []
+    while True:
+        M = pool.prepare_step(obs_view)
+        conn.send(('obs', M, frames_prev, eps_out))
+        frames_prev, eps_out = 0, []
+        cmd = conn.recv()
+        if cmd == 'quit':
+            break
+        if M:
+            r = res_view[:M]
+            frames_prev = pool.complete_step(
+                r[:, 0].astype(np.int64), r[:, 1].copy(), r[:, 2].copy())
+            eps_out = pool.harvest()
+
+
+class MultiProcGeesePool:
+    """Actor pool with env work in W child processes: each child owns a
+    game shard (vectorized stepping + columnar recording + packaging on
+    its own core); the parent services children round-robin with the
+    shared hipGraph/MFMA inference engine, one child's GPU batch in flight
+    while the next child's staging overlaps.
+
+    Construction forks the children — call it BEFORE any HIP context
+    exists in the parent; attach(model, device) wires the engine after.
+    """
+
+    def __init__(self, args, n_games=768, seed=0, workers=3):
+        import multiprocessing as mp
+        from multiprocessing import shared_memory
+        self.args = args
+        self.workers = workers
+        self.calls_per_vec_step = workers
+        per = max(1, n_games // workers)
+        self.n_per = per
+        self.conns, self.procs, self.shms = [], [], []
+        self.obs_views, self.res_views = [], []
+        cap = per * N_PLAYERS
+        for w in range(workers):
+            obs_shm = shared_memory.SharedMemory(
+                create=True, size=cap * 17 * 7 * 11)
+            res_shm = shared_memory.SharedMemory(create=True, size=cap * 3 * 4)
+            self.shms += [obs_shm, res_shm]
+            self.obs_views.append(np.ndarray((cap, 17, 7, 11), dtype=np.uint8,
+                                             buffer=obs_shm.buf))
+            self.res_views.append(np.ndarray((cap, 3), dtype=np.float32,
+                                             buffer=res_shm.buf))
+            parent_conn, child_conn = mp.Pipe(duplex=True)
+            proc = mp.Process(target=_geese_env_worker,
+                              args=(child_conn, obs_shm.name, res_shm.name,
+                                    per, args, seed + 977 * w),
+                              daemon=True)
+            proc.start()
+            child_conn.close()
+            self.conns.append(parent_conn)
+            self.procs.append(proc)
+
+        self.device = None
+        self.model = None
+        self.graphed = None
+        self.fused = None
+        self.inflight = {}
+        self.rr = 0
+        self.completed = []
+        self.frames = 0
+        self.episodes_done = 0
+        self.timing = {k: 0.0 for k in
+                       ('obs', 'fwd', 'sample', 'record', 'env', 'package')}
+        self.timing['n'] = 1
+
+    def attach(self, model, device):
+        """Wire the inference engine (after CUDA init)."""
+        self.model = model
+        self.device = device
+        if device.type == 'cuda':
+            from .models.geese_net import GeeseFusedEval
+            from .hipgraph import GraphedActorForward
+            self.fused = GeeseFusedEval(model, device)
+            self.graphed = GraphedActorForward(model, device, fused=self.fused)
+            cap = self.n_per * N_PLAYERS
+            self._obs_pin = [torch.empty(cap, 17, 7, 11, dtype=torch.uint8,
+                                         pin_memory=True)
+                             for _ in range(self.workers)]
+            self._obs_pin_np = [t.numpy() for t in self._obs_pin]
+            self._out_pin = [torch.empty(cap, 3, dtype=torch.float32,
+                                         pin_memory=True)
+                             for _ in range(self.workers)]
+            self._out_pin_np = [t.numpy() for t in self._out_pin]
+            self._events = [torch.cuda.Event() for _ in range(self.workers)]
+
+    def _complete(self, wid):
+        M = self.inflight.pop(wid)
+        if M and self.graphed is not None:
+            self._events[wid].synchronize()
+            np.copyto(self.res_views[wid][:M], self._out_pin_np[wid][:M])
+        self.conns[wid].send('go')
+
+    def step_once(self):
+        """Service one child: collect its obs, issue its inference, and
+        complete the previous child's round.  Returns frames reported."""
+        import time
+        wid = self.rr
+        self.rr = (self.rr + 1) % self.workers
+        tag, M, frames, eps = self.conns[wid].recv()
+        assert tag == 'obs'
+        if eps:
+            self.completed.extend(eps)
+            self.episodes_done += len(eps)
+        self.frames += frames
+
+        t0 = time.time()
+        if M and self.graphed is not None:
+            np.copyto(self._obs_pin_np[wid][:M], self.obs_views[wid][:M])
+            self.graphed.run_async(self._obs_pin[wid], M,
+                                   self._out_pin[wid], self._events[wid])
+            self.inflight[wid] = M
+        elif M:
+            # CPU fallback (tests): synchronous eager inference
+            obs_t = torch.from_numpy(self.obs_views[wid][:M].copy()).float()
+            with torch.no_grad():
+                out = self.model(obs_t, None)
+            probs = torch.softmax(out['policy'].float(), dim=-1)
+            acts = torch.multinomial(probs, 1).squeeze(-1)
+            sel = probs.gather(-1, acts.unsqueeze(-1)).squeeze(-1)
+            self.res_views[wid][:M, 0] = acts.numpy()
+            self.res_views[wid][:M, 1] = sel.numpy()
+            self.res_views[wid][:M, 2] = out['value'].float().squeeze(-1).numpy()
+            self.inflight[wid] = 0       # results already in shm
+        else:
+            self.inflight[wid] = 0
+        self.timing['fwd'] += time.time() - t0
+
+        prev = (wid - 1) % self.workers
+        if prev in self.inflight:
+            self._complete(prev)
+        return frames
+
+    def refresh_weights(self):
+        if self.fused is not None:
+            self.fused.refresh()
+
+    def harvest(self):
+        out = self.completed
+        self.completed = []
+        return out
+
+    def shutdown(self):
+        for wid in list(self.inflight):
+            self._complete(wid)
+        for conn in self.conns:
+            try:
+                conn.send('quit')
+            except (BrokenPipeError, OSError):
+                pass

@@ -129,3 +129,33 @@ def test_actor_episodes_train():
     losses, dcnt = trainer.train_step(batch)
     assert dcnt > 0
     assert torch.isfinite(losses['total'])
+
+
+def test_multiproc_pool_cpu():
+    """Env-worker processes + shared-memory transport, CPU inference."""
+    from handyrl_amd.actor import MultiProcGeesePool
+    args = _args()
+    mpool = MultiProcGeesePool(args, n_games=12, seed=5, workers=2)
+    try:
+        model = GeeseNet(layers=1)
+        model.eval()
+        mpool.attach(model, torch.device('cpu'))
+        for _ in range(900):
+            mpool.step_once()
+            if mpool.episodes_done >= 6:
+                break
+        eps = mpool.harvest()
+        assert len(eps) >= 6
+        for ep in eps[:3]:
+            assert ep['columnar'] and ep['steps'] >= 1
+            assert abs(sum(ep['outcome'].values())) < 1e-6
+        assert mpool.frames > 0
+        # episodes train
+        buf = EpisodeBuffer(args)
+        buf.extend(eps)
+        trainer = Trainer(args, GeeseNet(layers=1), device=torch.device('cpu'))
+        batch = make_batch([buf.select_episode() for _ in range(4)], args)
+        losses, dcnt = trainer.train_step(batch)
+        assert torch.isfinite(losses['total'])
+    finally:
+        mpool.shutdown()
