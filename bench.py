@@ -31,7 +31,7 @@ from handyrl_amd.batch import Batcher
 from handyrl_amd.models.geese_net import GeeseNet
 from handyrl_amd.train import Trainer
 
-N_ENVS = 768            # self-play games per GPU (measured best 1-GPU throughput)
+N_ENVS = 2048           # self-play games per GPU (8 env-worker processes)
 ACTOR_VEC_STEPS = 16    # env transitions (per game) per learner step
 
 
@@ -83,7 +83,7 @@ def main():
     buffer = EpisodeBuffer(args)
     batcher = Batcher(args, buffer)
 
-    actor_procs = int(os.environ.get('HANDYRL_ACTOR_PROCS', '3'))
+    actor_procs = int(os.environ.get('HANDYRL_ACTOR_PROCS', '8'))
     mpool = None
     if actor_procs > 0:
         from handyrl_amd.actor import MultiProcGeesePool
