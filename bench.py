@@ -245,6 +245,9 @@ def main():
         }
         print(json.dumps(result), flush=True)
 
+    if hasattr(pool, 'shutdown'):
+        pool.shutdown()
+
 
 if __name__ == '__main__':
     main()

@@ -510,3 +510,12 @@ class MultiProcGeesePool:
                 conn.send('quit')
             except (BrokenPipeError, OSError):
                 pass
+        for proc in self.procs:
+            proc.join(timeout=5)
+        for shm in self.shms:
+            try:
+                shm.close()
+                shm.unlink()
+            except (FileNotFoundError, OSError):
+                pass
+        self.shms = []

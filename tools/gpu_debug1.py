@@ -1,3 +1,5 @@
+import sys, os
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import sys, time, torch
 mode = sys.argv[1]
 from handyrl_amd.actor import GeeseActorPool

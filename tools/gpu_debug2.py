@@ -1,3 +1,5 @@
+import sys, os
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 # Bisect which train-step stage breaks hipGraph capture.
 import torch, torch.nn as nn
 from handyrl_amd.models.geese_net import GeeseNet

@@ -1,3 +1,5 @@
+import sys, os
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 # Time train-step components: batch H2D copy, graph replay, weight refresh.
 import time, torch
 from handyrl_amd.actor import GeeseActorPool
