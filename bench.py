@@ -81,7 +81,11 @@ def main():
     # context or torch thread-pool exists in this process
     from handyrl_amd.batch import EpisodeBuffer
     buffer = EpisodeBuffer(args)
-    batcher = Batcher(args, buffer)
+    # with the device-resident replay the batch is gathered ON DEVICE inside
+    # the training graph: no batch-builder processes at all
+    device_replay = torch.cuda.is_available() and \
+        os.environ.get('HANDYRL_DEVICE_REPLAY', '1') == '1'
+    batcher = False if device_replay else Batcher(args, buffer)
 
     actor_procs = int(os.environ.get('HANDYRL_ACTOR_PROCS', '8'))
     mpool = None
@@ -104,6 +108,12 @@ def main():
     if world > 1:
         hdist.broadcast_params(trainer.model)
 
+    replay = None
+    if device_replay:
+        from handyrl_amd.replay import DeviceReplay
+        replay = DeviceReplay(args, device, bytes_budget=int(
+            os.environ.get('HANDYRL_REPLAY_BYTES', str(4 << 30))))
+
     if mpool is not None:
         # env work in child processes; parent runs the inference engine
         mpool.attach(trainer.model, device)
@@ -125,7 +135,7 @@ def main():
             trainer.model.train()
         eps = pool.harvest()
         if eps:
-            trainer.episodes.extend(eps)
+            (replay if replay is not None else trainer.episodes).extend(eps)
         return frames
 
     # ---- prefill: generate the minimum episode set (untimed) ----
@@ -133,7 +143,8 @@ def main():
     while pool.episodes_done < args['minimum_episodes']:
         pump_actor(8)
     trainer.episodes.trim(args['maximum_episodes'])
-    batcher.run()
+    if batcher:
+        batcher.run()
     if rank == 0:
         import sys
         print('# prefill: %d episodes in %.1fs' %
@@ -141,20 +152,25 @@ def main():
 
     phase_t = {'actor': 0.0, 'batch_wait': 0.0, 'train': 0.0, 'n': 0}
 
+    replay_step = [None]
+
     def one_step():
         t0 = time.time()
         frames = pump_actor(ACTOR_VEC_STEPS)
         t1 = time.time()
-        batch = batcher.batch()
-        t2 = time.time()
-        graphed = getattr(trainer, 'graphed_step', None)
-        if graphed is not None:
-            # issue the whole train step asynchronously: its GPU time hides
-            # under the next actor phase (no host sync, no loss readback)
-            losses, _ = graphed.step(batch)
-            trainer.steps += 0   # steps counted inside graphed.step
+        if replay_step[0] is not None:
+            t2 = time.time()
+            # sample-gather + train, one graph replay, issued async
+            losses, _ = replay_step[0].step()
         else:
-            losses, dcnt = trainer.train_step(batch)
+            batch = batcher.batch()
+            t2 = time.time()
+            graphed = getattr(trainer, 'graphed_step', None)
+            if graphed is not None:
+                # async issue: the GPU time hides under the next actor phase
+                losses, _ = graphed.step(batch)
+            else:
+                losses, dcnt = trainer.train_step(batch)
         pool.refresh_weights()          # re-fold BN into the MFMA actor path
         t3 = time.time()
         phase_t['actor'] += t1 - t0
@@ -164,7 +180,15 @@ def main():
         return frames, losses
 
     # ---- capture the train step as a hipGraph (fixed shapes) ----
-    if use_cuda and os.environ.get('HANDYRL_NO_GRAPHS') != '1':
+    if replay is not None:
+        from handyrl_amd.hipgraph import GraphedReplayTrainStep
+        replay_step[0] = GraphedReplayTrainStep(trainer, replay, cli.batch_size)
+        if rank == 0:
+            import sys
+            print('# replay train-step hipGraph: %s' %
+                  ('captured' if replay_step[0].graph is not None else 'EAGER'),
+                  file=sys.stderr, flush=True)
+    elif use_cuda and os.environ.get('HANDYRL_NO_GRAPHS') != '1':
         example = batcher.batch()
         ok = trainer.enable_cuda_graph(example)
         if rank == 0:

@@ -156,3 +156,86 @@ class GraphedActorForward:
         graph.replay()
         out_pinned[:M].copy_(packed[:M], non_blocking=True)
         event.record()
+
+
+class GraphedReplayTrainStep:
+    """Train-step capture with the batch GATHERED FROM DEVICE REPLAY inside
+    the graph: one replay executes sample-gather + forward + loss +
+    backward + all-reduce + grad-clip + Adam.  Per-step variability enters
+    through six tiny index tensors (B elements each)."""
+
+    def __init__(self, trainer, replay, batch_size, warmup_iters=3):
+        from .train import compute_loss
+        self._compute_loss = compute_loss
+        self.trainer = trainer
+        self.replay = replay
+        self.batch_size = batch_size
+        assert trainer.device.type == 'cuda'
+        dev = trainer.device
+        B = batch_size
+        self.idx = {
+            'pos0': torch.zeros(B, dtype=torch.int64, device=dev),
+            'start': torch.zeros(B, dtype=torch.int64, device=dev),
+            'length': torch.zeros(B, dtype=torch.int64, device=dev),
+            'seat': torch.zeros(B, dtype=torch.int64, device=dev),
+            'outcome': torch.zeros(B, 4, device=dev),
+            'inv_total': torch.zeros(B, device=dev),
+        }
+        for group in trainer.optimizer.param_groups:
+            group['capturable'] = True
+
+        trainer.model.train()
+        stream = torch.cuda.Stream()
+        stream.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(stream):
+            for _ in range(warmup_iters):
+                self._fill()
+                self._run()
+        torch.cuda.current_stream().wait_stream(stream)
+
+        try:
+            self.graph = torch.cuda.CUDAGraph()
+            self._fill()
+            with torch.cuda.graph(self.graph):
+                self.losses, self.dcnt = self._run()
+        except Exception as e:      # noqa: BLE001 - run eager if capture fails
+            import sys
+            print('replay train-step capture failed, running eager: %r' % (e,),
+                  file=sys.stderr)
+            self.graph = None
+
+    def _fill(self):
+        pos0, start, length, seat, outcome, inv_total = \
+            self.replay.sample_indices(self.batch_size)
+        for key, arr in (('pos0', pos0), ('start', start), ('length', length),
+                         ('seat', seat), ('outcome', outcome),
+                         ('inv_total', inv_total)):
+            self.idx[key].copy_(torch.from_numpy(arr), non_blocking=True)
+
+    def _run(self):
+        tr = self.trainer
+        batch = self.replay.gather_batch(
+            self.idx['pos0'], self.idx['start'], self.idx['length'],
+            self.idx['seat'], self.idx['outcome'], self.idx['inv_total'])
+        if tr.use_amp:
+            with torch.autocast('cuda', dtype=torch.bfloat16):
+                losses, dcnt = self._compute_loss(batch, tr.wrapped_model,
+                                                  None, tr.args)
+        else:
+            losses, dcnt = self._compute_loss(batch, tr.wrapped_model,
+                                              None, tr.args)
+        tr.optimizer.zero_grad(set_to_none=False)
+        losses['total'].backward()
+        tr.reducer.allreduce_()
+        nn.utils.clip_grad_norm_(tr.params, 4.0)
+        tr.optimizer.step()
+        return losses, dcnt
+
+    def step(self):
+        """Sample indices on the host, copy them in, replay the graph."""
+        self._fill()
+        self.trainer.steps += 1
+        if self.graph is not None:
+            self.graph.replay()
+            return self.losses, self.dcnt
+        return self._run()

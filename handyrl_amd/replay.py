@@ -1,0 +1,202 @@
+"""Device-resident trajectory replay for columnar episodes.
+
+The MI355X replacement for the host-side episode deque + multiprocess
+batch builders on the flagship (solo, feed-forward) path: trajectories
+live in a flat columnar ring in HBM3E (capacity is a byte budget — 288 GB
+per GPU makes very large replays practical), window sampling keeps the
+reference's recency-biased accept-loop semantics on the host (tiny index
+math), and the (B, T, 1, ...) batch is GATHERED ON DEVICE with ~20 torch
+ops that capture into the training hipGraph — so one graph replay does
+sample-gather + forward + loss + backward + all-reduce + Adam with no
+per-step host batch traffic.
+
+Semantics parity with batch.make_batch's columnar solo path is tested in
+tests/test_replay.py (CPU) and on GPU in tests/test_gpu.py.
+"""
+
+import random
+from collections import deque
+
+import numpy as np
+import torch
+
+OBS_SHAPE = (4, 17, 7, 11)
+ROW_OBS = int(np.prod(OBS_SHAPE))
+
+
+class DeviceReplay:
+    """Flat columnar ring of per-step rows + a host-side episode table."""
+
+    def __init__(self, args, device, bytes_budget=4 << 30):
+        assert args.get('burn_in_steps', 0) == 0, \
+            'DeviceReplay supports feed-forward (no burn-in) training'
+        self.args = args
+        self.device = device
+        row_bytes = ROW_OBS + 4 + 3 * 4 * 4       # obs u8, alive, act/prob/val
+        self.ring_T = max(1024, int(bytes_budget // row_bytes))
+        dev = device
+        self.obs = torch.empty((self.ring_T,) + OBS_SHAPE, dtype=torch.uint8,
+                               device=dev)
+        self.alive = torch.empty(self.ring_T, 4, dtype=torch.bool, device=dev)
+        self.action = torch.empty(self.ring_T, 4, dtype=torch.int32, device=dev)
+        self.prob = torch.empty(self.ring_T, 4, dtype=torch.float32, device=dev)
+        self.value = torch.empty(self.ring_T, 4, dtype=torch.float32, device=dev)
+        self.head = 0                  # monotonically increasing write cursor
+        self.table = deque()           # (pos0, steps, outcome(np[4]))
+        self.total_added = 0
+        self._pin = {}
+
+    def __len__(self):
+        return len(self.table)
+
+    # -- write path --------------------------------------------------------
+    def _stage(self, key, arr):
+        """Host->device through a growing pinned staging buffer."""
+        t = torch.from_numpy(np.ascontiguousarray(arr))
+        if self.device.type != 'cuda':
+            return t.to(self.device)
+        pin = self._pin.get(key)
+        if pin is None or pin.shape[0] < t.shape[0]:
+            cap = max(int(t.shape[0] * 1.5), 1024)
+            pin = torch.empty((cap,) + tuple(t.shape[1:]), dtype=t.dtype,
+                              pin_memory=True)
+            self._pin[key] = pin
+        pin[:t.shape[0]].copy_(t)
+        return pin[:t.shape[0]].to(self.device, non_blocking=True)
+
+    def _write_ring(self, dst, src):
+        n = src.shape[0]
+        p = self.head % self.ring_T
+        first = min(n, self.ring_T - p)
+        dst[p:p + first] = src[:first]
+        if n > first:
+            dst[:n - first] = src[first:]
+
+    def extend(self, episodes):
+        """Append columnar episodes (numpy fields) to the device ring."""
+        if not episodes:
+            return
+        obs = np.concatenate([ep['obs'] for ep in episodes])
+        alive = np.concatenate([ep['alive'] for ep in episodes])
+        action = np.concatenate([ep['action'] for ep in episodes])
+        prob = np.concatenate([ep['prob'] for ep in episodes])
+        value = np.concatenate([ep['value'] for ep in episodes])
+        n = obs.shape[0]
+        if n > self.ring_T:
+            raise ValueError('episode block larger than the replay ring')
+
+        self._write_ring(self.obs, self._stage('obs', obs))
+        self._write_ring(self.alive, self._stage('alive', alive))
+        self._write_ring(self.action, self._stage('action', action.astype(np.int32)))
+        self._write_ring(self.prob, self._stage('prob', prob))
+        self._write_ring(self.value, self._stage('value', value))
+
+        pos = self.head
+        for ep in episodes:
+            oc = np.array([ep['outcome'][p] for p in range(4)], dtype=np.float32)
+            self.table.append((pos, int(ep['steps']), oc))
+            pos += int(ep['steps'])
+        self.head = pos
+        self.total_added += len(episodes)
+
+        # drop episodes whose rows have been overwritten, and respect the
+        # configured episode cap
+        min_valid = self.head - self.ring_T
+        max_eps = self.args['maximum_episodes']
+        while self.table and (self.table[0][0] < min_valid or
+                              len(self.table) > max_eps):
+            self.table.popleft()
+
+    def trim(self, maximum):
+        while len(self.table) > maximum:
+            self.table.popleft()
+
+    # -- sample path ---------------------------------------------------------
+    def sample_indices(self, batch_size):
+        """Recency-biased episode picks + window cuts + solo-seat choice
+        (reference train.py:291-315 / make_batch solo semantics).
+        Returns int64/float32 numpy arrays of length B."""
+        args = self.args
+        fs = args['forward_steps']
+        n = len(self.table)
+        assert n > 0, 'empty replay'
+        pos0 = np.empty(batch_size, dtype=np.int64)
+        start = np.empty(batch_size, dtype=np.int64)
+        length = np.empty(batch_size, dtype=np.int64)
+        seat = np.empty(batch_size, dtype=np.int64)
+        outcome = np.empty((batch_size, 4), dtype=np.float32)
+        inv_total = np.empty(batch_size, dtype=np.float32)
+        m = min(n, args['maximum_episodes'])
+        for b in range(batch_size):
+            while True:
+                idx = random.randrange(m)
+                accept = 1 - (m - 1 - idx) / m
+                if random.random() < accept:
+                    break
+            p0, steps, oc = self.table[idx]
+            train_st = random.randrange(1 + max(0, steps - fs))
+            ed = min(train_st + fs, steps)
+            pos0[b] = p0 + train_st
+            start[b] = train_st
+            length[b] = ed - train_st
+            seat[b] = random.randrange(4)
+            outcome[b] = oc
+            inv_total[b] = 1.0 / steps
+        return pos0, start, length, seat, outcome, inv_total
+
+    def gather_batch(self, pos0, start, length, seat, outcome, inv_total):
+        """Build the (B, T, 1, ...) training batch on device.  All inputs
+        are device tensors of length B; every op here is hipGraph-capturable
+        (per-step variability flows through these index tensors)."""
+        args = self.args
+        B = pos0.shape[0]
+        T = args['forward_steps']
+        dev = self.device
+        t_range = torch.arange(T, device=dev)
+        rows = (pos0.unsqueeze(1) + t_range) % self.ring_T      # (B, T)
+        in_range = (t_range.unsqueeze(0) < length.unsqueeze(1))  # (B, T) bool
+
+        flat = rows.reshape(-1)
+        sel = seat.unsqueeze(1).expand(B, T).reshape(-1)
+
+        obs = self.obs[flat, sel]                                # (B*T, 17,7,11)
+        alive = self.alive[flat, sel]
+        act = self.action[flat, sel].long()
+        prob = self.prob[flat, sel]
+        val = self.value[flat, sel]
+
+        in_r = in_range.reshape(-1)
+        tmask = (alive & in_r).float()
+        obs = obs * tmask.to(torch.uint8).view(-1, 1, 1, 1)
+        oc_b = outcome.gather(1, seat.unsqueeze(1)).squeeze(1)   # (B,)
+        oc_bt = oc_b.unsqueeze(1).expand(B, T).reshape(-1)
+        v = torch.where(in_r, val * tmask, oc_bt)
+        prob = torch.where(tmask.bool(), prob, torch.ones_like(prob))
+        act = act * tmask.long()
+        amask = torch.where(tmask.bool(), torch.zeros_like(prob),
+                            torch.full_like(prob, 1e32))
+        amask4 = amask.unsqueeze(1).expand(B * T, 4).contiguous()
+        progress = torch.where(
+            in_r,
+            (start.unsqueeze(1) + t_range).reshape(-1).float() *
+            inv_total.unsqueeze(1).expand(B, T).reshape(-1),
+            torch.ones(1, device=dev))
+
+        def shp(x, *tail):
+            return x.reshape(B, T, 1, *tail)
+
+        zeros = torch.zeros(B, T, 1, 1, device=dev)
+        return {
+            'observation': shp(obs, 17, 7, 11),
+            'selected_prob': shp(prob, 1),
+            'value': shp(v, 1),
+            'action': shp(act, 1),
+            'outcome': oc_b.view(B, 1, 1, 1),
+            'reward': zeros,
+            'return': zeros.clone(),
+            'episode_mask': in_range.float().view(B, T, 1, 1),
+            'turn_mask': shp(tmask, 1),
+            'observation_mask': shp(tmask.clone(), 1),
+            'action_mask': shp(amask4, 4),
+            'progress': progress.view(B, T, 1),
+        }

@@ -254,6 +254,46 @@ def test_gpu_train_step_bf16():
 
 
 @requires_gpu
+def test_device_replay_graphed_training():
+    """Device-resident replay: sample-gather + whole train step as ONE
+    graph replay; weights change and losses stay finite across steps."""
+    from handyrl_amd.actor import GeeseActorPool
+    from handyrl_amd.models.geese_net import GeeseNet
+    from handyrl_amd.replay import DeviceReplay
+    from handyrl_amd.train import Trainer
+    from handyrl_amd.hipgraph import GraphedReplayTrainStep
+
+    args = {
+        'turn_based_training': False, 'observation': False, 'gamma': 0.8,
+        'forward_steps': 8, 'burn_in_steps': 0, 'compress_steps': 4,
+        'entropy_regularization': 0.1, 'entropy_regularization_decay': 0.1,
+        'batch_size': 16, 'minimum_episodes': 2, 'maximum_episodes': 500,
+        'num_batchers': 1, 'lambda': 0.7, 'policy_target': 'VTRACE',
+        'value_target': 'VTRACE', 'seed': 0, 'bf16': True,
+        'compress_episodes': False,
+    }
+    device = torch.device('cuda', 0)
+    trainer = Trainer(args, GeeseNet(), device=device, batcher=False)
+    pool = GeeseActorPool(trainer.model, args, n_games=64, device=device, seed=0)
+    trainer.model.eval()
+    while pool.episodes_done < 40:
+        pool.step_once()
+    replay = DeviceReplay(args, device, bytes_budget=256 << 20)
+    replay.extend(pool.harvest())
+
+    step = GraphedReplayTrainStep(trainer, replay, args['batch_size'])
+    assert step.graph is not None, 'capture must succeed at world_size 1'
+    before = [p.detach().clone() for p in trainer.params]
+    for _ in range(3):
+        losses, dcnt = step.step()
+        torch.cuda.synchronize()
+        assert torch.isfinite(losses['total']), losses
+        assert float(dcnt) > 0
+    assert any(not torch.equal(b, p.detach())
+               for b, p in zip(before, trainer.params))
+
+
+@requires_gpu
 def test_gpu_rnn_geister_step():
     from handyrl_amd.envs import geister
     from handyrl_amd.train import Trainer
