@@ -112,7 +112,8 @@ def main():
     if device_replay:
         from handyrl_amd.replay import DeviceReplay
         replay = DeviceReplay(args, device, bytes_budget=int(
-            os.environ.get('HANDYRL_REPLAY_BYTES', str(4 << 30))))
+            os.environ.get('HANDYRL_REPLAY_BYTES', str(4 << 30))),
+            ingest_thread=True)
 
     if mpool is not None:
         # env work in child processes; parent runs the inference engine
@@ -182,6 +183,7 @@ def main():
     # ---- capture the train step as a hipGraph (fixed shapes) ----
     if replay is not None:
         from handyrl_amd.hipgraph import GraphedReplayTrainStep
+        replay.flush()                 # ensure the prefill block is published
         replay_step[0] = GraphedReplayTrainStep(trainer, replay, cli.batch_size)
         if rank == 0:
             import sys

@@ -205,6 +205,7 @@ class GraphedReplayTrainStep:
             self.graph = None
 
     def _fill(self):
+        self.replay.publish()          # admit finished background ingests
         pos0, start, length, seat, outcome, inv_total = \
             self.replay.sample_indices(self.batch_size)
         for key, arr in (('pos0', pos0), ('start', start), ('length', length),

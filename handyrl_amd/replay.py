@@ -14,7 +14,9 @@ Semantics parity with batch.make_batch's columnar solo path is tested in
 tests/test_replay.py (CPU) and on GPU in tests/test_gpu.py.
 """
 
+import queue as queue_mod
 import random
+import threading
 from collections import deque
 
 import numpy as np
@@ -23,11 +25,23 @@ import torch
 OBS_SHAPE = (4, 17, 7, 11)
 ROW_OBS = int(np.prod(OBS_SHAPE))
 
+# sampleable rows keep this distance from the overwrite frontier so that
+# in-flight ring writes (background ingest stream) can never race a graph
+# replay's gathers of still-tabled episodes
+EVICT_MARGIN = 16384
+
 
 class DeviceReplay:
-    """Flat columnar ring of per-step rows + a host-side episode table."""
+    """Flat columnar ring of per-step rows + a host-side episode table.
 
-    def __init__(self, args, device, bytes_budget=4 << 30):
+    With ``ingest_thread=True`` (the bench default on GPU) episode blocks
+    are staged and copied on a background thread + side HIP stream; the
+    consumer calls ``publish()`` (done inside GraphedReplayTrainStep._fill)
+    to make completed writes sampleable, with a stream wait on the write
+    event so gathers are ordered after the copies.
+    """
+
+    def __init__(self, args, device, bytes_budget=4 << 30, ingest_thread=False):
         assert args.get('burn_in_steps', 0) == 0, \
             'DeviceReplay supports feed-forward (no burn-in) training'
         self.args = args
@@ -45,24 +59,70 @@ class DeviceReplay:
         self.table = deque()           # (pos0, steps, outcome(np[4]))
         self.total_added = 0
         self._pin = {}
+        self._ingest = None
+        if ingest_thread and device.type == 'cuda':
+            self._ingest_q = queue_mod.Queue(maxsize=64)
+            self._ready = deque()
+            self._ready_lock = threading.Lock()
+            self._ingest_stream = torch.cuda.Stream()
+            self._ingest = threading.Thread(target=self._ingest_loop,
+                                            daemon=True)
+            self._ingest.start()
+
+    def _ingest_loop(self):
+        torch.cuda.set_device(self.device)
+        while True:
+            episodes = self._ingest_q.get()
+            with torch.cuda.stream(self._ingest_stream):
+                entries, n_rows = self._copy_block(episodes)
+                event = torch.cuda.Event()
+                event.record(self._ingest_stream)
+            with self._ready_lock:
+                self._ready.append((entries, n_rows, event))
+
+    def publish(self, stream=None):
+        """Make finished background writes sampleable (consumer thread)."""
+        if self._ingest is None:
+            return
+        if stream is None:
+            stream = torch.cuda.current_stream()
+        with self._ready_lock:
+            ready, self._ready = self._ready, deque()
+        for entries, n_rows, event in ready:
+            stream.wait_event(event)
+            self.table.extend(entries)
+            self.total_added += len(entries)
+        self._evict()
+
+    def flush(self):
+        """Block until every queued episode block is published."""
+        if self._ingest is None:
+            return
+        while not self._ingest_q.empty():
+            threading.Event().wait(0.005)
+        torch.cuda.synchronize()
+        self.publish()
 
     def __len__(self):
         return len(self.table)
 
     # -- write path --------------------------------------------------------
-    def _stage(self, key, arr):
-        """Host->device through a growing pinned staging buffer."""
-        t = torch.from_numpy(np.ascontiguousarray(arr))
+    def _stage(self, key, parts, n, dtype, tail_shape):
+        """Fill a pinned staging buffer from per-episode arrays (one memcpy
+        each, GIL released) and return its device copy (async H2D)."""
         if self.device.type != 'cuda':
-            return t.to(self.device)
+            return torch.from_numpy(np.concatenate(parts)).to(self.device)
         pin = self._pin.get(key)
-        if pin is None or pin.shape[0] < t.shape[0]:
-            cap = max(int(t.shape[0] * 1.5), 1024)
-            pin = torch.empty((cap,) + tuple(t.shape[1:]), dtype=t.dtype,
-                              pin_memory=True)
+        if pin is None or pin.shape[0] < n:
+            cap = max(int(n * 1.5), 1024)
+            pin = torch.empty((cap,) + tail_shape, dtype=dtype, pin_memory=True)
             self._pin[key] = pin
-        pin[:t.shape[0]].copy_(t)
-        return pin[:t.shape[0]].to(self.device, non_blocking=True)
+        view = pin.numpy()
+        off = 0
+        for part in parts:
+            view[off:off + part.shape[0]] = part
+            off += part.shape[0]
+        return pin[:n].to(self.device, non_blocking=True)
 
     def _write_ring(self, dst, src):
         n = src.shape[0]
@@ -72,40 +132,49 @@ class DeviceReplay:
         if n > first:
             dst[:n - first] = src[first:]
 
-    def extend(self, episodes):
-        """Append columnar episodes (numpy fields) to the device ring."""
-        if not episodes:
-            return
-        obs = np.concatenate([ep['obs'] for ep in episodes])
-        alive = np.concatenate([ep['alive'] for ep in episodes])
-        action = np.concatenate([ep['action'] for ep in episodes])
-        prob = np.concatenate([ep['prob'] for ep in episodes])
-        value = np.concatenate([ep['value'] for ep in episodes])
-        n = obs.shape[0]
+    def _copy_block(self, episodes):
+        """Ring-write one block of episodes; returns (table entries, rows)."""
+        n = sum(int(ep['steps']) for ep in episodes)
         if n > self.ring_T:
             raise ValueError('episode block larger than the replay ring')
-
-        self._write_ring(self.obs, self._stage('obs', obs))
-        self._write_ring(self.alive, self._stage('alive', alive))
-        self._write_ring(self.action, self._stage('action', action.astype(np.int32)))
-        self._write_ring(self.prob, self._stage('prob', prob))
-        self._write_ring(self.value, self._stage('value', value))
-
+        self._write_ring(self.obs, self._stage(
+            'obs', [ep['obs'] for ep in episodes], n, torch.uint8, OBS_SHAPE))
+        self._write_ring(self.alive, self._stage(
+            'alive', [ep['alive'] for ep in episodes], n, torch.bool, (4,)))
+        self._write_ring(self.action, self._stage(
+            'action', [ep['action'] for ep in episodes], n, torch.int32, (4,)))
+        self._write_ring(self.prob, self._stage(
+            'prob', [ep['prob'] for ep in episodes], n, torch.float32, (4,)))
+        self._write_ring(self.value, self._stage(
+            'value', [ep['value'] for ep in episodes], n, torch.float32, (4,)))
+        entries = []
         pos = self.head
         for ep in episodes:
             oc = np.array([ep['outcome'][p] for p in range(4)], dtype=np.float32)
-            self.table.append((pos, int(ep['steps']), oc))
+            entries.append((pos, int(ep['steps']), oc))
             pos += int(ep['steps'])
         self.head = pos
-        self.total_added += len(episodes)
+        return entries, n
 
-        # drop episodes whose rows have been overwritten, and respect the
-        # configured episode cap
-        min_valid = self.head - self.ring_T
+    def _evict(self):
+        margin = EVICT_MARGIN if self._ingest is not None else 0
+        min_valid = self.head - self.ring_T + margin
         max_eps = self.args['maximum_episodes']
         while self.table and (self.table[0][0] < min_valid or
                               len(self.table) > max_eps):
             self.table.popleft()
+
+    def extend(self, episodes):
+        """Append columnar episodes (numpy fields) to the device ring."""
+        if not episodes:
+            return
+        if self._ingest is not None:
+            self._ingest_q.put(episodes)
+            return
+        entries, _ = self._copy_block(episodes)
+        self.table.extend(entries)
+        self.total_added += len(entries)
+        self._evict()
 
     def trim(self, maximum):
         while len(self.table) > maximum:
