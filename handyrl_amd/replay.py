@@ -79,6 +79,7 @@ class DeviceReplay:
                 event.record(self._ingest_stream)
             with self._ready_lock:
                 self._ready.append((entries, n_rows, event))
+            self._ingest_q.task_done()
 
     def publish(self, stream=None):
         """Make finished background writes sampleable (consumer thread)."""
@@ -98,8 +99,7 @@ class DeviceReplay:
         """Block until every queued episode block is published."""
         if self._ingest is None:
             return
-        while not self._ingest_q.empty():
-            threading.Event().wait(0.005)
+        self._ingest_q.join()
         torch.cuda.synchronize()
         self.publish()
 
