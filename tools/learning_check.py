@@ -1,0 +1,56 @@
+# Does the full pipeline LEARN? Track mean episode length + losses while
+# training Hungry Geese self-play for N steps on one GPU.
+import sys, os, time
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+import torch
+from handyrl_amd.actor import MultiProcGeesePool
+from handyrl_amd.models.geese_net import GeeseNet
+from handyrl_amd.train import Trainer
+from handyrl_amd.replay import DeviceReplay
+from handyrl_amd.hipgraph import GraphedReplayTrainStep
+from bench import bench_args
+
+torch.set_num_threads(1)
+args = bench_args(128, 16)
+mpool = MultiProcGeesePool(args, n_games=1024, seed=7, workers=4)
+device = torch.device('cuda', 0)
+torch.cuda.set_device(device)
+torch.manual_seed(0)
+trainer = Trainer(args, GeeseNet(), device=device, batcher=False)
+mpool.attach(trainer.model, device)
+replay = DeviceReplay(args, device, bytes_budget=2 << 30, ingest_thread=True)
+
+def pump(n):
+    frames = 0
+    for _ in range(n * mpool.calls_per_vec_step):
+        frames += mpool.step_once()
+    eps = mpool.harvest()
+    lens = [ep['steps'] for ep in eps]
+    replay.extend(eps)
+    return frames, lens
+
+while mpool.episodes_done < args['minimum_episodes']:
+    pump(8)
+replay.flush()
+step = GraphedReplayTrainStep(trainer, replay, args['batch_size'])
+assert step.graph is not None
+
+N = int(sys.argv[1]) if len(sys.argv) > 1 else 400
+window_lens, t0 = [], time.time()
+for i in range(N):
+    frames, lens = pump(16)
+    window_lens += lens
+    losses, dcnt = step.step()
+    mpool.refresh_weights()
+    if (i + 1) % 50 == 0:
+        torch.cuda.synchronize()
+        ml = sum(window_lens) / max(1, len(window_lens))
+        print('step %4d | mean_ep_len %.2f | eps %d | p %.4f v %.4f ent %.3f | %.1fs'
+              % (i + 1, ml, len(window_lens),
+                 float(losses['p']) / max(1.0, float(dcnt)),
+                 float(losses['v']) / max(1.0, float(dcnt)),
+                 float(losses['ent']) / max(1.0, float(dcnt)),
+                 time.time() - t0), flush=True)
+        window_lens = []
+mpool.shutdown()
+print('LEARNING_CHECK_DONE')
