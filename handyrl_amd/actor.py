@@ -7,9 +7,9 @@ network forward for all alive seats (bf16, inference mode) and one fused
 masked-softmax-sample kernel (handyrl_amd/ops), instead of per-env
 single-sample CPU inference.
 
-Finished games are packaged into reference-format episodes (uncompressed
-moment blocks by default — same interchange as the CPU worker path, minus
-the bz2 that a local learner does not need).
+Finished games are packaged as COLUMNAR episodes (struct-of-arrays) —
+the replay buffer and batch maker accept those alongside the reference
+moment-dict format used by remote CPU workers.
 """
 
 import os
@@ -24,25 +24,17 @@ from . import ops
 from .envs.vec_geese import GeeseVecEnv, N_PLAYERS
 from .envs.hungry_geese import MAX_STEPS
 
-MOMENT_KEYS = ('observation', 'selected_prob', 'action_mask', 'action',
-               'value', 'reward', 'return')
-
-
 class GeeseActorPool:
     """Self-play actor pool for Hungry Geese on one GPU."""
 
     def __init__(self, model, args, n_games=256, device=None, seed=0,
-                 store_uint8_obs=True, use_graphs=True, engine=None):
+                 use_graphs=True, engine=None):
         self.args = args
         self.device = device if device is not None else (
             torch.device('cuda') if torch.cuda.is_available() else torch.device('cpu'))
         self.model = model
         self.vec = GeeseVecEnv(n_games, seed=seed)
         self.n_games = n_games
-        self.store_uint8_obs = store_uint8_obs
-        self.gamma = args.get('gamma', 0.8)
-        self.compress = args.get('compress_episodes', False)
-        self.compress_steps = args.get('compress_steps', 4)
         # columnar trajectory recording: struct-of-arrays ring per game
         # (episodes stay columnar through the replay buffer and batch maker)
         G, CAPT = n_games, MAX_STEPS

@@ -24,11 +24,9 @@ import threading
 import time
 import warnings
 
-import numpy as np
 import torch
 import torch.nn as nn
 import torch.nn.functional as F
-import torch.distributions as torch_dist
 import torch.optim as optim
 
 try:
@@ -40,7 +38,7 @@ from .environment import prepare_env, make_env
 from .util import map_r, bimap_r, trimap_r
 from .model import to_torch, to_gpu, ModelWrapper
 from .losses import compute_target
-from .batch import EpisodeBuffer, Batcher, make_batch
+from .batch import EpisodeBuffer, Batcher
 from .worker import WorkerCluster, WorkerServer
 from . import dist as hdist
 
