@@ -1,0 +1,266 @@
+"""Batched Geister self-play: the recurrent-actor counterpart of
+handyrl_amd/actor.py.
+
+Geister is turn-based, partially observed and recurrent (GeisterNet's DRC
+Conv-LSTM core), so the reference runs one CPU process per environment
+with per-step single-sample CPU inference — the DRC makes that inference
+dominate (9 ConvLSTM cell evaluations per step).  Here:
+
+* W child processes each own a shard of python Environment instances and
+  do the game-logic side (legal actions, moves, moment recording, episode
+  packaging in the reference moment-dict format);
+* observations/legal-masks/turn-parity travel through shared memory;
+* the parent runs ONE batched GeisterNet forward per child round with the
+  per-(game, player) DRC hidden states RESIDENT ON THE GPU — gathered by
+  turn parity, scattered back after the step, zeroed on game reset — plus
+  the fused masked-softmax-sample kernel over the 214-action space.
+"""
+
+import numpy as np
+import torch
+
+from . import ops
+from .environment import make_env
+
+N_ACTIONS = 214
+SCALAR_DIM = 18
+BOARD_SHAPE = (7, 6, 6)
+
+
+def _geister_env_worker(conn, shm_names, n_games, args, seed):
+    from multiprocessing import shared_memory
+    from .batch import pack_moments
+
+    shms = {k: shared_memory.SharedMemory(name=v) for k, v in shm_names.items()}
+    G = n_games
+    scalar_v = np.ndarray((G, SCALAR_DIM), dtype=np.float32, buffer=shms['scalar'].buf)
+    board_v = np.ndarray((G,) + BOARD_SHAPE, dtype=np.float32, buffer=shms['board'].buf)
+    mask_v = np.ndarray((G, N_ACTIONS), dtype=np.float32, buffer=shms['mask'].buf)
+    parity_v = np.ndarray((G,), dtype=np.int8, buffer=shms['parity'].buf)
+    reset_v = np.ndarray((G,), dtype=np.uint8, buffer=shms['reset'].buf)
+    res_v = np.ndarray((G, 4), dtype=np.float32, buffer=shms['res'].buf)
+
+    import random
+    random.seed(seed)
+    envs = [make_env({'env': 'Geister', 'id': seed * 1000 + i}) for i in range(G)]
+    for env in envs:
+        env.reset()
+    moments = [[] for _ in range(G)]
+    gamma = args.get('gamma', 0.8)
+    compress = args.get('compress_episodes', False)
+    compress_steps = args.get('compress_steps', 4)
+    job_args = {'player': [0, 1], 'model_id': {0: -1, 1: -1}}
+
+    def package(g):
+        ms = moments[g]
+        if not ms:
+            return None
+        outcome = envs[g].outcome()
+        for p in (0, 1):
+            ret = 0.0
+            for m in reversed(ms):
+                ret = (m['reward'][p] or 0) + gamma * ret
+                m['return'][p] = ret
+        return {'args': job_args, 'steps': len(ms), 'outcome': outcome,
+                'moment': pack_moments(ms, compress_steps, compress=compress)}
+
+    frames_prev = 0
+    eps_out = []
+    while True:
+        # stage one observation round
+        for g, env in enumerate(envs):
+            reset_v[g] = 0
+            if env.terminal():
+                ep = package(g)
+                if ep is not None:
+                    eps_out.append(ep)
+                moments[g] = []
+                env.reset()
+                reset_v[g] = 1
+            p = env.turn()
+            obs = env.observation(p)
+            scalar_v[g] = obs['scalar']
+            board_v[g] = obs['board']
+            legal = env.legal_actions(p)
+            mask_v[g] = 1e32
+            mask_v[g, legal] = 0.0
+            parity_v[g] = p
+
+        conn.send(('obs', G, frames_prev, eps_out))
+        frames_prev, eps_out = 0, []
+        cmd = conn.recv()
+        if cmd == 'quit':
+            break
+
+        # apply sampled actions + record moments (reference generation.py
+        # semantics for the turn-based, observation=False configuration)
+        for g, env in enumerate(envs):
+            p = int(parity_v[g])
+            action = int(res_v[g, 0])
+            moment = {key: {0: None, 1: None} for key in
+                      ('observation', 'selected_prob', 'action_mask', 'action',
+                       'value', 'reward', 'return')}
+            moment['observation'][p] = {'scalar': scalar_v[g].copy(),
+                                        'board': board_v[g].copy()}
+            moment['selected_prob'][p] = float(res_v[g, 1])
+            moment['action_mask'][p] = mask_v[g].copy()
+            moment['action'][p] = action
+            moment['value'][p] = np.array([res_v[g, 2]], dtype=np.float32)
+            moment['turn'] = [p]
+            env.play(action)
+            reward = env.reward()
+            for q in (0, 1):
+                moment['reward'][q] = reward.get(q, None)
+            moments[g].append(moment)
+            frames_prev += 1
+
+
+class GeisterMultiProcPool:
+    """256-actor-style Geister self-play on one GPU: W env-worker processes
+    + batched recurrent inference with GPU-resident DRC hidden state."""
+
+    def __init__(self, args, n_games=256, seed=0, workers=8):
+        import multiprocessing as mp
+        from multiprocessing import shared_memory
+        self.args = args
+        self.workers = workers
+        per = max(1, n_games // workers)
+        self.n_per = per
+        self.conns, self.procs, self.shms = [], [], []
+        self.views = []
+        sizes = {
+            'scalar': per * SCALAR_DIM * 4,
+            'board': per * int(np.prod(BOARD_SHAPE)) * 4,
+            'mask': per * N_ACTIONS * 4,
+            'parity': per,
+            'reset': per,
+            'res': per * 4 * 4,
+        }
+        for w in range(workers):
+            shm = {k: shared_memory.SharedMemory(create=True, size=v)
+                   for k, v in sizes.items()}
+            self.shms.extend(shm.values())
+            views = {
+                'scalar': np.ndarray((per, SCALAR_DIM), dtype=np.float32,
+                                     buffer=shm['scalar'].buf),
+                'board': np.ndarray((per,) + BOARD_SHAPE, dtype=np.float32,
+                                    buffer=shm['board'].buf),
+                'mask': np.ndarray((per, N_ACTIONS), dtype=np.float32,
+                                   buffer=shm['mask'].buf),
+                'parity': np.ndarray((per,), dtype=np.int8,
+                                     buffer=shm['parity'].buf),
+                'reset': np.ndarray((per,), dtype=np.uint8,
+                                    buffer=shm['reset'].buf),
+                'res': np.ndarray((per, 4), dtype=np.float32,
+                                  buffer=shm['res'].buf),
+            }
+            self.views.append(views)
+            parent_conn, child_conn = mp.Pipe(duplex=True)
+            proc = mp.Process(
+                target=_geister_env_worker,
+                args=(child_conn, {k: s.name for k, s in shm.items()},
+                      per, args, seed + 131 * w),
+                daemon=True)
+            proc.start()
+            child_conn.close()
+            self.conns.append(parent_conn)
+            self.procs.append(proc)
+
+        self.model = None
+        self.device = None
+        self.hidden = None
+        self.completed = []
+        self.frames = 0
+        self.episodes_done = 0
+        self.rr = 0
+        self.calls_per_vec_step = workers
+
+    def attach(self, model, device):
+        self.model = model
+        self.device = device
+        per = self.n_per
+        # per-(game, player) DRC hidden state, flattened rows = game*2+player
+        self.hidden = []
+        for _ in range(self.workers):
+            hs, cs = model.init_hidden([per * 2])
+            self.hidden.append((
+                [h.to(device) for h in hs], [c.to(device) for c in cs]))
+        self._arange2 = torch.arange(per, device=device) * 2
+
+    @torch.no_grad()
+    def _infer(self, wid):
+        v = self.views[wid]
+        dev = self.device
+        per = self.n_per
+        scalar = torch.from_numpy(v['scalar']).to(dev, non_blocking=True)
+        board = torch.from_numpy(v['board']).to(dev, non_blocking=True)
+        mask = torch.from_numpy(v['mask']).to(dev, non_blocking=True)
+        parity = torch.from_numpy(v['parity'].astype(np.int64)).to(dev, non_blocking=True)
+        reset = torch.from_numpy(v['reset'].astype(np.float32)).to(dev, non_blocking=True)
+
+        hs, cs = self.hidden[wid]
+        # zero both players' hidden rows for games that just reset
+        keep = (1.0 - reset).repeat_interleave(2).view(per * 2, 1, 1, 1)
+        for i in range(len(hs)):
+            hs[i] = hs[i] * keep
+            cs[i] = cs[i] * keep
+
+        rows = self._arange2 + parity                  # (per,) turn rows
+        h_in = ([h.index_select(0, rows) for h in hs],
+                [c.index_select(0, rows) for c in cs])
+        out = self.model({'scalar': scalar, 'board': board}, h_in)
+        h_out, c_out = out['hidden']
+        for i in range(len(hs)):
+            hs[i] = hs[i].index_copy(0, rows, h_out[i])
+            cs[i] = cs[i].index_copy(0, rows, c_out[i])
+        self.hidden[wid] = (hs, cs)
+
+        policy = out['policy'].float()
+        uniform = torch.rand(per, device=dev)
+        if dev.type == 'cuda':
+            actions, probs = ops.masked_sample(policy, mask, uniform)
+        else:
+            pr = torch.softmax(policy - mask, dim=-1)
+            actions = torch.multinomial(pr, 1).squeeze(-1)
+            probs = pr.gather(-1, actions.unsqueeze(-1)).squeeze(-1)
+        packed = torch.stack([actions.float(), probs,
+                              out['value'].float().squeeze(-1),
+                              out['return'].float().squeeze(-1)], dim=1)
+        np.copyto(v['res'], packed.cpu().numpy())
+
+    def step_once(self):
+        wid = self.rr
+        self.rr = (self.rr + 1) % self.workers
+        tag, G, frames, eps = self.conns[wid].recv()
+        assert tag == 'obs'
+        if eps:
+            self.completed.extend(eps)
+            self.episodes_done += len(eps)
+        self.frames += frames
+        self._infer(wid)
+        self.conns[wid].send('go')
+        return frames
+
+    def harvest(self):
+        out = self.completed
+        self.completed = []
+        return out
+
+    def refresh_weights(self):
+        pass                       # shares the live module directly
+
+    def shutdown(self):
+        for conn in self.conns:
+            try:
+                conn.send('quit')
+            except (BrokenPipeError, OSError):
+                pass
+        for proc in self.procs:
+            proc.join(timeout=5)
+        for shm in self.shms:
+            try:
+                shm.close()
+                shm.unlink()
+            except (FileNotFoundError, OSError):
+                pass
+        self.shms = []

@@ -1,0 +1,110 @@
+# Geister self-play throughput (BASELINE.json config #5 shape): N actors
+# in env-worker processes + batched recurrent (DRC) GPU inference, with a
+# concurrent RNN learner (UPGO+TD, bf16).  Prints one JSON line.
+import json
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import torch
+
+
+def main():
+    import argparse
+    ap = argparse.ArgumentParser()
+    ap.add_argument('--actors', type=int, default=256)
+    ap.add_argument('--workers', type=int, default=8)
+    ap.add_argument('--steps', type=int, default=30)
+    ap.add_argument('--warmup', type=int, default=5)
+    ap.add_argument('--batch-size', type=int, default=32)
+    cli = ap.parse_args()
+
+    torch.set_num_threads(1)
+    args = {
+        'turn_based_training': True, 'observation': False, 'gamma': 0.8,
+        'forward_steps': 16, 'burn_in_steps': 0, 'compress_steps': 4,
+        'entropy_regularization': 0.1, 'entropy_regularization_decay': 0.1,
+        'batch_size': cli.batch_size, 'minimum_episodes': 80,
+        'maximum_episodes': 2000, 'num_batchers': 3, 'lambda': 0.7,
+        'policy_target': 'UPGO', 'value_target': 'TD', 'seed': 0,
+        'bf16': False,                    # RNN path trains fp32 (small net)
+        'compress_episodes': False,
+    }
+
+    from handyrl_amd.batch import EpisodeBuffer, Batcher
+    buffer = EpisodeBuffer(args)
+    batcher = Batcher(args, buffer)
+    from handyrl_amd.actor_geister import GeisterMultiProcPool
+    pool = GeisterMultiProcPool(args, n_games=cli.actors, seed=17,
+                                workers=cli.workers)
+
+    use_cuda = torch.cuda.is_available()
+    device = torch.device('cuda', 0) if use_cuda else torch.device('cpu')
+    from handyrl_amd.envs.geister import Environment
+    from handyrl_amd.train import Trainer
+    torch.manual_seed(0)
+    env = Environment()
+    trainer = Trainer(args, env.net(), device=device, episodes=buffer,
+                      batcher=batcher)
+    actor_model = env.net().to(device)
+    actor_model.load_state_dict(trainer.model.state_dict())
+    actor_model.eval()
+    pool.attach(actor_model, device)
+
+    def pump(n):
+        frames = 0
+        for _ in range(n * pool.calls_per_vec_step):
+            frames += pool.step_once()
+        eps = pool.harvest()
+        if eps:
+            buffer.extend(eps)
+        return frames
+
+    t0 = time.time()
+    while pool.episodes_done < args['minimum_episodes']:
+        pump(2)
+    batcher.run()
+    print('# prefill %d eps in %.1fs' % (len(buffer), time.time() - t0),
+          file=sys.stderr, flush=True)
+
+    def one_step():
+        frames = pump(4)
+        batch = batcher.batch()
+        losses, dcnt = trainer.train_step(batch)
+        # actors follow the trained weights (model push each step)
+        actor_model.load_state_dict(trainer.model.state_dict())
+        return frames
+
+    for _ in range(cli.warmup):
+        one_step()
+    if use_cuda:
+        torch.cuda.synchronize()
+    start = time.time()
+    total = 0
+    for _ in range(cli.steps):
+        total += one_step()
+    if use_cuda:
+        torch.cuda.synchronize()
+    el = time.time() - start
+    pool.shutdown()
+
+    print(json.dumps({
+        'metric': 'geister_selfplay_env_frames_per_sec',
+        'value': round(total / el, 1),
+        'unit': 'frames/s', 'n_gpus': 1, 'steps': cli.steps,
+        'warmup': cli.warmup, 'ms_per_step': round(1000 * el / cli.steps, 2),
+        'higher_is_better': True, 'scaling': 'weak', 'vs_baseline': None,
+        'dtype': 'fp32', 'data': 'synthetic',
+        'config': {'model': 'GeisterNet(DRC 3x3 ConvLSTM)',
+                   'actors': cli.actors, 'workers': cli.workers,
+                   'global_batch': cli.batch_size, 'seq_len': 16,
+                   'parallelism': 'dp1',
+                   'learner_steps_per_sec': round(cli.steps / el, 2),
+                   'loss': 'UPGO policy + TD(lambda) value/return'},
+    }), flush=True)
+
+
+if __name__ == '__main__':
+    main()
