@@ -186,46 +186,106 @@ class GeisterMultiProcPool:
             self.hidden.append((
                 [h.to(device) for h in hs], [c.to(device) for c in cs]))
         self._arange2 = torch.arange(per, device=device) * 2
+        self._graphs = [None] * self.workers
+        if device.type == 'cuda' and \
+                __import__('os').environ.get('HANDYRL_NO_GRAPHS') != '1':
+            self._capture_graphs()
+
+    def _static_in(self):
+        per = self.n_per
+        dev = self.device
+        return {
+            'scalar': torch.zeros(per, SCALAR_DIM, device=dev),
+            'board': torch.zeros((per,) + BOARD_SHAPE, device=dev),
+            'mask': torch.full((per, N_ACTIONS), 1e32, device=dev),
+            'parity': torch.zeros(per, dtype=torch.int64, device=dev),
+            'keep': torch.ones(per * 2, 1, 1, 1, device=dev),
+        }
+
+    def _infer_body(self, wid, st):
+        """One batched recurrent inference step on static tensors; every
+        hidden update is IN-PLACE so the storages stay fixed under graph
+        capture."""
+        hs, cs = self.hidden[wid]
+        rows = self._arange2 + st['parity']
+        for i in range(len(hs)):
+            hs[i].mul_(st['keep'])
+            cs[i].mul_(st['keep'])
+        h_in = ([h.index_select(0, rows) for h in hs],
+                [c.index_select(0, rows) for c in cs])
+        out = self.model({'scalar': st['scalar'], 'board': st['board']}, h_in)
+        h_out, c_out = out['hidden']
+        for i in range(len(hs)):
+            hs[i].index_copy_(0, rows, h_out[i])
+            cs[i].index_copy_(0, rows, c_out[i])
+        policy = out['policy'].float()
+        uniform = torch.rand(policy.shape[0], device=self.device)
+        if self.device.type == 'cuda':
+            actions, probs = ops.masked_sample(policy, st['mask'], uniform)
+        else:
+            pr = torch.softmax(policy - st['mask'], dim=-1)
+            actions = torch.multinomial(pr, 1).squeeze(-1)
+            probs = pr.gather(-1, actions.unsqueeze(-1)).squeeze(-1)
+        return torch.stack([actions.float(), probs,
+                            out['value'].float().squeeze(-1),
+                            out['return'].float().squeeze(-1)], dim=1)
+
+    @torch.no_grad()
+    def _capture_graphs(self):
+        try:
+            self._statics = []
+            for wid in range(self.workers):
+                st = self._static_in()
+                stream = torch.cuda.Stream()
+                stream.wait_stream(torch.cuda.current_stream())
+                with torch.cuda.stream(stream):
+                    for _ in range(2):
+                        self._infer_body(wid, st)
+                torch.cuda.current_stream().wait_stream(stream)
+                graph = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(graph):
+                    packed = self._infer_body(wid, st)
+                self._graphs[wid] = (graph, st, packed)
+            # captured warmups corrupted the hidden state: reset it
+            for wid in range(self.workers):
+                hs, cs = self.hidden[wid]
+                for t in hs + cs:
+                    t.zero_()
+        except Exception as e:     # noqa: BLE001 - eager fallback
+            import sys
+            print('geister actor graph capture failed, running eager: %r'
+                  % (e,), file=sys.stderr)
+            self._graphs = [None] * self.workers
 
     @torch.no_grad()
     def _infer(self, wid):
         v = self.views[wid]
         dev = self.device
         per = self.n_per
-        scalar = torch.from_numpy(v['scalar']).to(dev, non_blocking=True)
-        board = torch.from_numpy(v['board']).to(dev, non_blocking=True)
-        mask = torch.from_numpy(v['mask']).to(dev, non_blocking=True)
-        parity = torch.from_numpy(v['parity'].astype(np.int64)).to(dev, non_blocking=True)
-        reset = torch.from_numpy(v['reset'].astype(np.float32)).to(dev, non_blocking=True)
-
-        hs, cs = self.hidden[wid]
-        # zero both players' hidden rows for games that just reset
-        keep = (1.0 - reset).repeat_interleave(2).view(per * 2, 1, 1, 1)
-        for i in range(len(hs)):
-            hs[i] = hs[i] * keep
-            cs[i] = cs[i] * keep
-
-        rows = self._arange2 + parity                  # (per,) turn rows
-        h_in = ([h.index_select(0, rows) for h in hs],
-                [c.index_select(0, rows) for c in cs])
-        out = self.model({'scalar': scalar, 'board': board}, h_in)
-        h_out, c_out = out['hidden']
-        for i in range(len(hs)):
-            hs[i] = hs[i].index_copy(0, rows, h_out[i])
-            cs[i] = cs[i].index_copy(0, rows, c_out[i])
-        self.hidden[wid] = (hs, cs)
-
-        policy = out['policy'].float()
-        uniform = torch.rand(per, device=dev)
-        if dev.type == 'cuda':
-            actions, probs = ops.masked_sample(policy, mask, uniform)
-        else:
-            pr = torch.softmax(policy - mask, dim=-1)
-            actions = torch.multinomial(pr, 1).squeeze(-1)
-            probs = pr.gather(-1, actions.unsqueeze(-1)).squeeze(-1)
-        packed = torch.stack([actions.float(), probs,
-                              out['value'].float().squeeze(-1),
-                              out['return'].float().squeeze(-1)], dim=1)
+        graphed = self._graphs[wid]
+        if graphed is not None:
+            graph, st, packed = graphed
+            st['scalar'].copy_(torch.from_numpy(v['scalar']), non_blocking=True)
+            st['board'].copy_(torch.from_numpy(v['board']), non_blocking=True)
+            st['mask'].copy_(torch.from_numpy(v['mask']), non_blocking=True)
+            st['parity'].copy_(torch.from_numpy(v['parity'].astype(np.int64)),
+                               non_blocking=True)
+            st['keep'].copy_(torch.from_numpy(
+                (1.0 - v['reset'].astype(np.float32)).repeat(2)
+                .reshape(per * 2, 1, 1, 1)), non_blocking=True)
+            graph.replay()
+            np.copyto(v['res'], packed.cpu().numpy())
+            return
+        st = {
+            'scalar': torch.from_numpy(v['scalar']).to(dev, non_blocking=True),
+            'board': torch.from_numpy(v['board']).to(dev, non_blocking=True),
+            'mask': torch.from_numpy(v['mask']).to(dev, non_blocking=True),
+            'parity': torch.from_numpy(v['parity'].astype(np.int64)).to(dev),
+            'keep': torch.from_numpy(
+                (1.0 - v['reset'].astype(np.float32)).repeat(2)
+                .reshape(per * 2, 1, 1, 1)).to(dev),
+        }
+        packed = self._infer_body(wid, st)
         np.copyto(v['res'], packed.cpu().numpy())
 
     def step_once(self):
