@@ -294,6 +294,39 @@ def test_device_replay_graphed_training():
 
 
 @requires_gpu
+def test_geister_pool_graphed_gpu():
+    """Batched Geister actors on GPU: graphed recurrent inference with
+    resident hidden state produces valid episodes."""
+    from handyrl_amd.actor_geister import GeisterMultiProcPool
+    from handyrl_amd.envs.geister import Environment
+    args = {
+        'turn_based_training': True, 'observation': False, 'gamma': 0.8,
+        'forward_steps': 4, 'burn_in_steps': 0, 'compress_steps': 4,
+        'entropy_regularization': 0.1, 'entropy_regularization_decay': 0.1,
+        'batch_size': 4, 'minimum_episodes': 2, 'maximum_episodes': 50,
+        'num_batchers': 1, 'lambda': 0.7, 'policy_target': 'UPGO',
+        'value_target': 'TD', 'seed': 0, 'bf16': False,
+        'compress_episodes': False,
+    }
+    pool = GeisterMultiProcPool(args, n_games=16, seed=2, workers=2)
+    try:
+        model = Environment().net().cuda().eval()
+        pool.attach(model, torch.device('cuda', 0))
+        assert pool._graphs[0] is not None, 'geister actor graph must capture'
+        for _ in range(3000):
+            pool.step_once()
+            if pool.episodes_done >= 3:
+                break
+        eps = pool.harvest()
+        assert len(eps) >= 3
+        for ep in eps[:2]:
+            assert ep['steps'] >= 3
+            assert set(ep['outcome'].keys()) == {0, 1}
+    finally:
+        pool.shutdown()
+
+
+@requires_gpu
 def test_gpu_rnn_geister_step():
     from handyrl_amd.envs import geister
     from handyrl_amd.train import Trainer

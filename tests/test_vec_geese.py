@@ -137,3 +137,62 @@ def test_vec_grid_consistency():
         finished = np.nonzero(done)[0]
         if len(finished):
             vec.reset_games(finished)
+
+
+def test_hunger_tick_and_growth():
+    """Targeted rules: eating grows by 1; every 40th transition shrinks the
+    tail; a starved goose dies — checked against the single-game oracle."""
+    import random as _random
+    vec = GeeseVecEnv(1, seed=99)
+    st = hg.GeeseState(_random.Random(99))
+    st.reset()
+    st.geese = vec_to_lists(vec, 0)
+    st.food = set(int(c) for c in vec.food[0])
+    st.alive = [True] * 4
+    st.last_actions = [None] * 4
+    st.prev_heads = [None] * 4
+    st.step_count = 0
+    st.over = False
+
+    # drive one goose onto food: place food right of goose 0's head
+    head = vec_to_lists(vec, 0)[0][0]
+    target = hg.shift(head, 3)           # EAST
+    # ensure target is free
+    occupied = {c for g in vec_to_lists(vec, 0) for c in g}
+    if target not in occupied:
+        vec.food[0, 0] = target
+        st.food = {int(vec.food[0, 0]), int(vec.food[0, 1])}
+        len_before = vec.length[0, 0]
+        acts = np.array([[3, 0, 0, 0]], dtype=np.int32)
+        # other geese may die; only check goose 0's growth if it survives
+        vec.step(acts)
+        st.step({p: int(acts[0, p]) for p in range(4) if st.alive[p]})
+        assert vec_to_lists(vec, 0) == [list(g) for g in st.geese]
+        if vec.alive[0, 0]:
+            assert vec.length[0, 0] == len_before + 1
+
+    # hunger tick parity across the 40-step boundary with safe looped play
+    vec2 = GeeseVecEnv(1, seed=5)
+    st2 = hg.GeeseState(_random.Random(5))
+    st2.reset()
+    st2.geese = vec_to_lists(vec2, 0)
+    st2.food = set(int(c) for c in vec2.food[0])
+    st2.alive = [True] * 4
+    st2.last_actions = [None] * 4
+    st2.prev_heads = [None] * 4
+    st2.step_count = 0
+    st2.over = False
+    rng = _random.Random(1)
+    pattern = [0, 3, 1, 2]               # N,E,S,W loop: never reverses
+    for step in range(90):
+        if vec2.over[0]:
+            break
+        a = pattern[step % 4]
+        acts = np.array([[a, pattern[(step + 1) % 4],
+                          pattern[(step + 2) % 4], pattern[(step + 3) % 4]]],
+                        dtype=np.int32)
+        vec2.step(acts)
+        st2.step({p: int(acts[0, p]) for p in range(4) if st2.alive[p]})
+        assert list(vec2.alive[0]) == st2.alive, step
+        assert vec_to_lists(vec2, 0) == [list(g) for g in st2.geese], step
+        sync_food(vec2, 0, st2)
