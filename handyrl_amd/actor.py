@@ -410,6 +410,7 @@ class MultiProcGeesePool:
         self.graphed = None
         self.fused = None
         self.inflight = {}
+        self._fifo = []
         self.rr = 0
         self.completed = []
         self.frames = 0
@@ -446,12 +447,18 @@ class MultiProcGeesePool:
         self.conns[wid].send('go')
 
     def step_once(self):
-        """Service one child: collect its obs, issue its inference, and
-        complete the previous child's round.  Returns frames reported."""
+        """Service one child (whichever is ready first — a jittering
+        child never stalls the sweep): collect its obs, issue its
+        inference, and complete the oldest in-flight round.  Returns
+        frames reported."""
         import time
-        wid = self.rr
-        self.rr = (self.rr + 1) % self.workers
-        tag, M, frames, eps = self.conns[wid].recv()
+        import multiprocessing.connection as mpc
+        waitable = [c for i, c in enumerate(self.conns) if i not in self.inflight]
+        ready = mpc.wait(waitable)
+        conn = self.conns[self.rr] if self.conns[self.rr] in ready else ready[0]
+        wid = self.conns.index(conn)
+        self.rr = (wid + 1) % self.workers
+        tag, M, frames, eps = conn.recv()
         assert tag == 'obs'
         if eps:
             self.completed.extend(eps)
@@ -478,11 +485,11 @@ class MultiProcGeesePool:
             self.inflight[wid] = 0       # results already in shm
         else:
             self.inflight[wid] = 0
+        self._fifo.append(wid)
         self.timing['fwd'] += time.time() - t0
 
-        prev = (wid - 1) % self.workers
-        if prev in self.inflight:
-            self._complete(prev)
+        while len(self._fifo) > 1:
+            self._complete(self._fifo.pop(0))
         return frames
 
     def refresh_weights(self):
@@ -495,8 +502,8 @@ class MultiProcGeesePool:
         return out
 
     def shutdown(self):
-        for wid in list(self.inflight):
-            self._complete(wid)
+        while self._fifo:
+            self._complete(self._fifo.pop(0))
         for conn in self.conns:
             try:
                 conn.send('quit')

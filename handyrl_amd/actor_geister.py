@@ -289,16 +289,19 @@ class GeisterMultiProcPool:
         np.copyto(v['res'], packed.cpu().numpy())
 
     def step_once(self):
-        wid = self.rr
-        self.rr = (self.rr + 1) % self.workers
-        tag, G, frames, eps = self.conns[wid].recv()
+        import multiprocessing.connection as mpc
+        ready = mpc.wait(self.conns)
+        conn = self.conns[self.rr] if self.conns[self.rr] in ready else ready[0]
+        wid = self.conns.index(conn)
+        self.rr = (wid + 1) % self.workers
+        tag, G, frames, eps = conn.recv()
         assert tag == 'obs'
         if eps:
             self.completed.extend(eps)
             self.episodes_done += len(eps)
         self.frames += frames
         self._infer(wid)
-        self.conns[wid].send('go')
+        conn.send('go')
         return frames
 
     def harvest(self):
