@@ -334,7 +334,7 @@ class PipelinedGeesePool:
         return merged
 
 
-def _geese_env_worker(conn, obs_name, res_name, n_games, args, seed):
+def _geese_env_worker(conn, ep_conn, obs_name, res_name, n_games, args, seed):
     """Env-side child process: vectorized stepping, columnar recording and
     episode packaging on host cores; observations/results move through
     shared memory, inference runs in the parent (GPU)."""
@@ -349,11 +349,10 @@ def _geese_env_worker(conn, obs_name, res_name, n_games, args, seed):
                           device=torch.device('cpu'), use_graphs=False,
                           seed=seed)
     frames_prev = 0
-    eps_out = []
     while True:
         M = pool.prepare_step(obs_view)
-        conn.send(('obs', M, frames_prev, eps_out))
-        frames_prev, eps_out = 0, []
+        conn.send(('obs', M, frames_prev))
+        frames_prev = 0
         cmd = conn.recv()
         if cmd == 'quit':
             break
@@ -361,7 +360,11 @@ def _geese_env_worker(conn, obs_name, res_name, n_games, args, seed):
             r = res_view[:M]
             frames_prev = pool.complete_step(
                 r[:, 0].astype(np.int64), r[:, 1].copy(), r[:, 2].copy())
-            eps_out = pool.harvest()
+            eps = pool.harvest()
+            if eps:
+                # episodes travel on their own pipe, drained by a parent
+                # background thread: the service path never deserializes them
+                ep_conn.send(eps)
 
 
 class MultiProcGeesePool:
@@ -383,7 +386,7 @@ class MultiProcGeesePool:
         self.calls_per_vec_step = workers
         per = max(1, n_games // workers)
         self.n_per = per
-        self.conns, self.procs, self.shms = [], [], []
+        self.conns, self.ep_conns, self.procs, self.shms = [], [], [], []
         self.obs_views, self.res_views = [], []
         cap = per * N_PLAYERS
         for w in range(workers):
@@ -396,13 +399,16 @@ class MultiProcGeesePool:
             self.res_views.append(np.ndarray((cap, 3), dtype=np.float32,
                                              buffer=res_shm.buf))
             parent_conn, child_conn = mp.Pipe(duplex=True)
+            ep_parent, ep_child = mp.Pipe(duplex=False)
             proc = mp.Process(target=_geese_env_worker,
-                              args=(child_conn, obs_shm.name, res_shm.name,
-                                    per, args, seed + 977 * w),
+                              args=(child_conn, ep_child, obs_shm.name,
+                                    res_shm.name, per, args, seed + 977 * w),
                               daemon=True)
             proc.start()
             child_conn.close()
+            ep_child.close()
             self.conns.append(parent_conn)
+            self.ep_conns.append(ep_parent)
             self.procs.append(proc)
 
         self.device = None
@@ -413,8 +419,12 @@ class MultiProcGeesePool:
         self._fifo = []
         self.rr = 0
         self.completed = []
+        self._completed_lock = __import__('threading').Lock()
         self.frames = 0
         self.episodes_done = 0
+        drain = __import__('threading').Thread(target=self._drain_episodes,
+                                               daemon=True)
+        drain.start()
         self.timing = {k: 0.0 for k in
                        ('obs', 'fwd', 'sample', 'record', 'env', 'package')}
         self.timing['n'] = 1
@@ -439,6 +449,22 @@ class MultiProcGeesePool:
             self._out_pin_np = [t.numpy() for t in self._out_pin]
             self._events = [torch.cuda.Event() for _ in range(self.workers)]
 
+    def _drain_episodes(self):
+        import multiprocessing.connection as mpc
+        while True:
+            try:
+                ready = mpc.wait(self.ep_conns, timeout=1.0)
+            except OSError:
+                return
+            for conn in ready:
+                try:
+                    eps = conn.recv()
+                except (EOFError, OSError):
+                    return
+                with self._completed_lock:
+                    self.completed.extend(eps)
+                    self.episodes_done += len(eps)
+
     def _complete(self, wid):
         M = self.inflight.pop(wid)
         if M and self.graphed is not None:
@@ -458,11 +484,8 @@ class MultiProcGeesePool:
         conn = self.conns[self.rr] if self.conns[self.rr] in ready else ready[0]
         wid = self.conns.index(conn)
         self.rr = (wid + 1) % self.workers
-        tag, M, frames, eps = conn.recv()
+        tag, M, frames = conn.recv()
         assert tag == 'obs'
-        if eps:
-            self.completed.extend(eps)
-            self.episodes_done += len(eps)
         self.frames += frames
 
         t0 = time.time()
@@ -497,8 +520,9 @@ class MultiProcGeesePool:
             self.fused.refresh()
 
     def harvest(self):
-        out = self.completed
-        self.completed = []
+        with self._completed_lock:
+            out = self.completed
+            self.completed = []
         return out
 
     def shutdown(self):
