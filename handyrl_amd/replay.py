@@ -73,13 +73,37 @@ class DeviceReplay:
         torch.cuda.set_device(self.device)
         while True:
             episodes = self._ingest_q.get()
-            with torch.cuda.stream(self._ingest_stream):
-                entries, n_rows = self._copy_block(episodes)
-                event = torch.cuda.Event()
-                event.record(self._ingest_stream)
-            with self._ready_lock:
-                self._ready.append((entries, n_rows, event))
-            self._ingest_q.task_done()
+            try:
+                for chunk in self._chunk(episodes):
+                    with torch.cuda.stream(self._ingest_stream):
+                        entries, n_rows = self._copy_block(chunk)
+                        event = torch.cuda.Event()
+                        event.record(self._ingest_stream)
+                    with self._ready_lock:
+                        self._ready.append((entries, n_rows, event))
+            except Exception as e:   # noqa: BLE001 - never kill the ingester
+                import sys
+                print('replay ingest dropped a block: %r' % (e,),
+                      file=sys.stderr, flush=True)
+            finally:
+                self._ingest_q.task_done()
+
+    def _chunk(self, episodes):
+        """Split an episode list so no block exceeds a quarter of the ring
+        (bursts of long episodes can otherwise overflow a small ring)."""
+        limit = max(512, self.ring_T // 4)
+        chunk, rows = [], 0
+        for ep in episodes:
+            steps = int(ep['steps'])
+            if steps > limit:
+                continue                      # cannot ever fit: drop
+            if chunk and rows + steps > limit:
+                yield chunk
+                chunk, rows = [], 0
+            chunk.append(ep)
+            rows += steps
+        if chunk:
+            yield chunk
 
     def publish(self, stream=None):
         """Make finished background writes sampleable (consumer thread)."""
@@ -171,9 +195,10 @@ class DeviceReplay:
         if self._ingest is not None:
             self._ingest_q.put(episodes)
             return
-        entries, _ = self._copy_block(episodes)
-        self.table.extend(entries)
-        self.total_added += len(entries)
+        for chunk in self._chunk(episodes):
+            entries, _ = self._copy_block(chunk)
+            self.table.extend(entries)
+            self.total_added += len(entries)
         self._evict()
 
     def trim(self, maximum):
