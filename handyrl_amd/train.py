@@ -36,7 +36,7 @@ except ImportError:       # pragma: no cover - psutil is in the image
 
 from .environment import prepare_env, make_env
 from .util import map_r, bimap_r, trimap_r
-from .model import to_torch, to_gpu, ModelWrapper
+from .model import to_gpu, ModelWrapper
 from .losses import compute_target
 from .batch import EpisodeBuffer, Batcher
 from .worker import WorkerCluster, WorkerServer

@@ -4,7 +4,7 @@ multi-process recurrent path, trainable through the RNN learner."""
 import torch
 
 from handyrl_amd.actor_geister import GeisterMultiProcPool
-from handyrl_amd.batch import make_batch, EpisodeBuffer, unpack_moments
+from handyrl_amd.batch import make_batch, EpisodeBuffer, unpack_moments  # noqa: F401
 from handyrl_amd.envs.geister import Environment as GeisterEnv
 from handyrl_amd.train import Trainer
 
