@@ -175,3 +175,45 @@ def test_full_local_training_run():
         assert 'updated model(' in out
         assert os.path.exists(os.path.join(tmp, 'models', '1.pth'))
         assert os.path.exists(os.path.join(tmp, 'models', 'latest.pth'))
+
+
+def test_restart_epoch_resume():
+    """Checkpoint/resume: restart_epoch=N loads models/N.pth and continues
+    numbering at N+1 (reference train.py:420-423 semantics)."""
+    base_args = {
+        'env_args': {'env': 'TicTacToe'},
+        'train_args': {
+            'turn_based_training': True, 'observation': False,
+            'gamma': 0.8, 'forward_steps': 4, 'burn_in_steps': 0,
+            'compress_steps': 4, 'entropy_regularization': 0.1,
+            'entropy_regularization_decay': 0.1, 'update_episodes': 12,
+            'batch_size': 4, 'minimum_episodes': 5, 'maximum_episodes': 200,
+            'epochs': 1, 'num_batchers': 1, 'eval_rate': 0.1,
+            'worker': {'num_parallel': 2}, 'lambda': 0.7,
+            'policy_target': 'TD', 'value_target': 'TD',
+            'eval': {'opponent': ['random']}, 'seed': 0, 'restart_epoch': 0,
+            'bf16': False,
+        },
+    }
+    script = textwrap.dedent('''
+        import sys, json
+        sys.path.insert(0, %r)
+        from handyrl_amd.train import train_main
+        args = json.loads(%r)
+        args['train_args']['restart_epoch'] = int(sys.argv[1])
+        args['train_args']['epochs'] = int(sys.argv[2])
+        train_main(args)
+        print('TRAIN_DONE')
+    ''')
+    import json as _json
+    script = script % (REPO, _json.dumps(base_args))
+    with tempfile.TemporaryDirectory() as tmp:
+        r1 = subprocess.run([sys.executable, '-c', script, '0', '1'], cwd=tmp,
+                            capture_output=True, text=True, timeout=240)
+        assert 'TRAIN_DONE' in r1.stdout, (r1.stdout[-2000:], r1.stderr[-2000:])
+        assert os.path.exists(os.path.join(tmp, 'models', '1.pth'))
+        r2 = subprocess.run([sys.executable, '-c', script, '1', '2'], cwd=tmp,
+                            capture_output=True, text=True, timeout=240)
+        assert 'TRAIN_DONE' in r2.stdout, (r2.stdout[-2000:], r2.stderr[-2000:])
+        assert os.path.exists(os.path.join(tmp, 'models', '2.pth'))
+        assert 'epoch 1' in r2.stdout       # resumed at the loaded epoch
