@@ -13,6 +13,7 @@ moment-dict format used by remote CPU workers.
 """
 
 import os
+import sys
 
 import numpy as np
 import torch
@@ -334,37 +335,54 @@ class PipelinedGeesePool:
         return merged
 
 
-def _geese_env_worker(conn, ep_conn, obs_name, res_name, n_games, args, seed):
+def _geese_env_worker(conn, ep_conn, obs_name, res_name, n_games, args, seed,
+                      slots=2):
     """Env-side child process: vectorized stepping, columnar recording and
     episode packaging on host cores; observations/results move through
-    shared memory, inference runs in the parent (GPU)."""
+    shared memory, inference runs in the parent (GPU).
+
+    The shard is split into ``slots`` software-pipelined halves: while one
+    half's observations are away at the parent (GPU forward + service
+    latency), the worker steps/records the other half — the inference
+    round trip hides entirely behind env CPU work.
+    """
     from multiprocessing import shared_memory
     obs_shm = shared_memory.SharedMemory(name=obs_name)
     res_shm = shared_memory.SharedMemory(name=res_name)
-    cap = n_games * N_PLAYERS
-    obs_view = np.ndarray((cap, 17, 7, 11), dtype=np.uint8, buffer=obs_shm.buf)
-    res_view = np.ndarray((cap, 3), dtype=np.float32, buffer=res_shm.buf)
+    per = max(1, n_games // slots)
+    cap = per * N_PLAYERS
+    obs_views, res_views, pools = [], [], []
+    for s in range(slots):
+        obs_views.append(np.ndarray((cap, 17, 7, 11), dtype=np.uint8,
+                                    buffer=obs_shm.buf,
+                                    offset=s * cap * 17 * 7 * 11))
+        res_views.append(np.ndarray((cap, 3), dtype=np.float32,
+                                    buffer=res_shm.buf, offset=s * cap * 12))
+        pools.append(GeeseActorPool(None, args, n_games=per,
+                                    device=torch.device('cpu'),
+                                    use_graphs=False, seed=seed + 131 * s))
 
-    pool = GeeseActorPool(None, args, n_games=n_games,
-                          device=torch.device('cpu'), use_graphs=False,
-                          seed=seed)
-    frames_prev = 0
+    m_inflight = [0] * slots
+    for s in range(slots):                    # prime the pipeline
+        m_inflight[s] = pools[s].prepare_step(obs_views[s])
+        conn.send(('obs', s, m_inflight[s], 0))
     while True:
-        M = pool.prepare_step(obs_view)
-        conn.send(('obs', M, frames_prev))
-        frames_prev = 0
         cmd = conn.recv()
         if cmd == 'quit':
             break
+        s = cmd[1]                            # ('go', slot)
+        M, frames = m_inflight[s], 0
         if M:
-            r = res_view[:M]
-            frames_prev = pool.complete_step(
+            r = res_views[s][:M]
+            frames = pools[s].complete_step(
                 r[:, 0].astype(np.int64), r[:, 1].copy(), r[:, 2].copy())
-            eps = pool.harvest()
+            eps = pools[s].harvest()
             if eps:
                 # episodes travel on their own pipe, drained by a parent
                 # background thread: the service path never deserializes them
                 ep_conn.send(eps)
+        m_inflight[s] = pools[s].prepare_step(obs_views[s])
+        conn.send(('obs', s, m_inflight[s], frames))
 
 
 class MultiProcGeesePool:
@@ -378,31 +396,41 @@ class MultiProcGeesePool:
     exists in the parent; attach(model, device) wires the engine after.
     """
 
-    def __init__(self, args, n_games=768, seed=0, workers=3):
+    def __init__(self, args, n_games=768, seed=0, workers=3, slots=None):
         import multiprocessing as mp
         from multiprocessing import shared_memory
         self.args = args
         self.workers = workers
-        self.calls_per_vec_step = workers
+        if slots is None:
+            slots = int(os.environ.get('HANDYRL_ACTOR_SLOTS', '2'))
         per = max(1, n_games // workers)
-        self.n_per = per
+        slots = max(1, min(slots, per))
+        self.slots = slots
+        self.calls_per_vec_step = workers * slots
+        self.n_per = max(1, per // slots)       # games per slot
         self.conns, self.ep_conns, self.procs, self.shms = [], [], [], []
-        self.obs_views, self.res_views = [], []
-        cap = per * N_PLAYERS
+        self.obs_views, self.res_views = [], []  # [worker][slot]
+        cap = self.n_per * N_PLAYERS
         for w in range(workers):
             obs_shm = shared_memory.SharedMemory(
-                create=True, size=cap * 17 * 7 * 11)
-            res_shm = shared_memory.SharedMemory(create=True, size=cap * 3 * 4)
+                create=True, size=slots * cap * 17 * 7 * 11)
+            res_shm = shared_memory.SharedMemory(
+                create=True, size=slots * cap * 3 * 4)
             self.shms += [obs_shm, res_shm]
-            self.obs_views.append(np.ndarray((cap, 17, 7, 11), dtype=np.uint8,
-                                             buffer=obs_shm.buf))
-            self.res_views.append(np.ndarray((cap, 3), dtype=np.float32,
-                                             buffer=res_shm.buf))
+            self.obs_views.append([
+                np.ndarray((cap, 17, 7, 11), dtype=np.uint8,
+                           buffer=obs_shm.buf, offset=s * cap * 17 * 7 * 11)
+                for s in range(slots)])
+            self.res_views.append([
+                np.ndarray((cap, 3), dtype=np.float32,
+                           buffer=res_shm.buf, offset=s * cap * 12)
+                for s in range(slots)])
             parent_conn, child_conn = mp.Pipe(duplex=True)
             ep_parent, ep_child = mp.Pipe(duplex=False)
             proc = mp.Process(target=_geese_env_worker,
                               args=(child_conn, ep_child, obs_shm.name,
-                                    res_shm.name, per, args, seed + 977 * w),
+                                    res_shm.name, per, args, seed + 977 * w,
+                                    slots),
                               daemon=True)
             proc.start()
             child_conn.close()
@@ -416,6 +444,7 @@ class MultiProcGeesePool:
         self.graphed = None
         self.fused = None
         self.inflight = {}
+        self._use_registered = False
         self._fifo = []
         self.rr = 0
         self.completed = []
@@ -439,15 +468,54 @@ class MultiProcGeesePool:
             self.fused = GeeseFusedEval(model, device)
             self.graphed = GraphedActorForward(model, device, fused=self.fused)
             cap = self.n_per * N_PLAYERS
-            self._obs_pin = [torch.empty(cap, 17, 7, 11, dtype=torch.uint8,
-                                         pin_memory=True)
-                             for _ in range(self.workers)]
-            self._obs_pin_np = [t.numpy() for t in self._obs_pin]
-            self._out_pin = [torch.empty(cap, 3, dtype=torch.float32,
-                                         pin_memory=True)
-                             for _ in range(self.workers)]
-            self._out_pin_np = [t.numpy() for t in self._out_pin]
-            self._events = [torch.cuda.Event() for _ in range(self.workers)]
+            mk = lambda shape, dt: [[torch.empty(*shape, dtype=dt,
+                                                 pin_memory=True)
+                                     for _ in range(self.slots)]
+                                    for _ in range(self.workers)]
+            self._obs_pin = mk((cap, 17, 7, 11), torch.uint8)
+            self._obs_pin_np = [[t.numpy() for t in row] for row in self._obs_pin]
+            self._out_pin = mk((cap, 3), torch.float32)
+            self._out_pin_np = [[t.numpy() for t in row] for row in self._out_pin]
+            self._events = [[torch.cuda.Event() for _ in range(self.slots)]
+                            for _ in range(self.workers)]
+            self._register_shm()
+
+    def _register_shm(self):
+        """hipHostRegister the shared-memory obs/result buffers so the DMA
+        engines move them directly (no shm->pinned staging memcpy on the
+        service path).  Falls back to the staging copy if registration is
+        refused (e.g. exotic shm mounts)."""
+        self._use_registered = False
+        if os.environ.get('HANDYRL_NO_SHM_REGISTER'):
+            return
+        flat = [v for row in self.obs_views for v in row] + \
+               [v for row in self.res_views for v in row]
+        try:
+            cudart = torch.cuda.cudart()
+            registered = []
+            for view in flat:
+                rc = cudart.cudaHostRegister(view.ctypes.data, view.nbytes, 0)
+                if int(rc) != 0:
+                    break
+                registered.append(view)
+            else:
+                self._obs_src = [[torch.from_numpy(v) for v in row]
+                                 for row in self.obs_views]
+                self._res_dst = [[torch.from_numpy(v) for v in row]
+                                 for row in self.res_views]
+                self._use_registered = all(
+                    t.is_pinned()
+                    for row in self._obs_src + self._res_dst for t in row)
+            if not self._use_registered:
+                for view in registered:
+                    cudart.cudaHostUnregister(view.ctypes.data)
+            else:
+                print('# actor shm hipHostRegister: direct-DMA service path',
+                      file=sys.stderr)
+        except Exception as e:
+            print('shm register unavailable (%s); using staged copies' % e,
+                  file=sys.stderr)
+            self._use_registered = False
 
     def _drain_episodes(self):
         import multiprocessing.connection as mpc
@@ -465,54 +533,72 @@ class MultiProcGeesePool:
                     self.completed.extend(eps)
                     self.episodes_done += len(eps)
 
-    def _complete(self, wid):
-        M = self.inflight.pop(wid)
+    def _inflight_cnt(self, wid):
+        return sum(1 for (w, _s) in self.inflight if w == wid)
+
+    def _complete(self, wid, slot):
+        M = self.inflight.pop((wid, slot))
         if M and self.graphed is not None:
-            self._events[wid].synchronize()
-            np.copyto(self.res_views[wid][:M], self._out_pin_np[wid][:M])
-        self.conns[wid].send('go')
+            self._events[wid][slot].synchronize()
+            if not self._use_registered:
+                np.copyto(self.res_views[wid][slot][:M],
+                          self._out_pin_np[wid][slot][:M])
+        self.conns[wid].send(('go', slot))
 
     def step_once(self):
-        """Service one child (whichever is ready first — a jittering
-        child never stalls the sweep): collect its obs, issue its
-        inference, and complete the oldest in-flight round.  Returns
-        frames reported."""
+        """Service one child request (whichever is ready first — a
+        jittering child never stalls the sweep): collect its obs, issue
+        its inference, and complete the oldest in-flight round.  With
+        ``slots`` > 1 each child keeps another half-shard stepping while
+        this one's round trip is in flight.  Returns frames reported."""
         import time
         import multiprocessing.connection as mpc
-        waitable = [c for i, c in enumerate(self.conns) if i not in self.inflight]
+        waitable = [c for i, c in enumerate(self.conns)
+                    if self._inflight_cnt(i) < self.slots]
         ready = mpc.wait(waitable)
         conn = self.conns[self.rr] if self.conns[self.rr] in ready else ready[0]
         wid = self.conns.index(conn)
         self.rr = (wid + 1) % self.workers
-        tag, M, frames = conn.recv()
+        tag, slot, M, frames = conn.recv()
         assert tag == 'obs'
         self.frames += frames
 
         t0 = time.time()
         if M and self.graphed is not None:
-            np.copyto(self._obs_pin_np[wid][:M], self.obs_views[wid][:M])
-            self.graphed.run_async(self._obs_pin[wid], M,
-                                   self._out_pin[wid], self._events[wid])
-            self.inflight[wid] = M
+            if self._use_registered:
+                # shm is hipHostRegister'd: DMA straight from/to it, no
+                # staging memcpy on the service path
+                self.graphed.run_async(self._obs_src[wid][slot], M,
+                                       self._res_dst[wid][slot],
+                                       self._events[wid][slot])
+            else:
+                np.copyto(self._obs_pin_np[wid][slot][:M],
+                          self.obs_views[wid][slot][:M])
+                self.graphed.run_async(self._obs_pin[wid][slot], M,
+                                       self._out_pin[wid][slot],
+                                       self._events[wid][slot])
+            self.inflight[(wid, slot)] = M
         elif M:
             # CPU fallback (tests): synchronous eager inference
-            obs_t = torch.from_numpy(self.obs_views[wid][:M].copy()).float()
+            obs_t = torch.from_numpy(
+                self.obs_views[wid][slot][:M].copy()).float()
             with torch.no_grad():
                 out = self.model(obs_t, None)
             probs = torch.softmax(out['policy'].float(), dim=-1)
             acts = torch.multinomial(probs, 1).squeeze(-1)
             sel = probs.gather(-1, acts.unsqueeze(-1)).squeeze(-1)
-            self.res_views[wid][:M, 0] = acts.numpy()
-            self.res_views[wid][:M, 1] = sel.numpy()
-            self.res_views[wid][:M, 2] = out['value'].float().squeeze(-1).numpy()
-            self.inflight[wid] = 0       # results already in shm
+            res = self.res_views[wid][slot]
+            res[:M, 0] = acts.numpy()
+            res[:M, 1] = sel.numpy()
+            res[:M, 2] = out['value'].float().squeeze(-1).numpy()
+            self.inflight[(wid, slot)] = 0   # results already in shm
         else:
-            self.inflight[wid] = 0
-        self._fifo.append(wid)
+            self.inflight[(wid, slot)] = 0
+        self._fifo.append((wid, slot))
         self.timing['fwd'] += time.time() - t0
 
         while len(self._fifo) > 1:
-            self._complete(self._fifo.pop(0))
+            self._complete(*self._fifo.pop(0))
         return frames
 
     def refresh_weights(self):
@@ -527,7 +613,7 @@ class MultiProcGeesePool:
 
     def shutdown(self):
         while self._fifo:
-            self._complete(self._fifo.pop(0))
+            self._complete(*self._fifo.pop(0))
         for conn in self.conns:
             try:
                 conn.send('quit')

@@ -53,26 +53,32 @@ class GeeseVecEnv:
         self.over = np.zeros(G, dtype=bool)
         # incremental grids
         self.body_grid = np.zeros((G, N_PLAYERS, N_CELLS), dtype=np.uint8)
+        self._gp_arange = np.arange(G * N_PLAYERS)
         self.reset_games(np.arange(G))
 
     # -- helpers -----------------------------------------------------------
     def _head(self, gmask=None):
-        idx = self.start % CAP
-        return np.take_along_axis(self.body, idx[..., None], axis=2)[..., 0]
+        flat = self.body.reshape(self.G * N_PLAYERS, CAP)
+        return flat[self._gp_arange, (self.start % CAP).ravel()] \
+            .reshape(self.G, N_PLAYERS)
 
     def _tail_cell(self):
-        idx = (self.start + self.length - 1) % CAP
-        return np.take_along_axis(self.body, idx[..., None], axis=2)[..., 0]
+        flat = self.body.reshape(self.G * N_PLAYERS, CAP)
+        idx = ((self.start + self.length - 1) % CAP).ravel()
+        return flat[self._gp_arange, idx].reshape(self.G, N_PLAYERS)
 
     def reset_games(self, games):
         """Reset the given game indices to fresh initial states."""
         if len(games) == 0:
             return
-        for g in games:
-            cells = self.rng.choice(N_CELLS, size=N_PLAYERS + MIN_FOOD, replace=False)
-            self.body[g] = -1
-            self.body[g, :, 0] = cells[:N_PLAYERS]
-            self.food[g] = cells[N_PLAYERS:]
+        # distinct random cells per game: first K columns of an argsort of
+        # iid uniforms = a uniform sample without replacement
+        K = N_PLAYERS + MIN_FOOD
+        cells = np.argsort(self.rng.random((len(games), N_CELLS)),
+                           axis=1)[:, :K].astype(np.int32)
+        self.body[games] = -1
+        self.body[games, :, 0] = cells[:, :N_PLAYERS]
+        self.food[games] = cells[:, N_PLAYERS:]
         self.start[games] = 0
         self.length[games] = 1
         self.alive[games] = True

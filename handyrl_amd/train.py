@@ -210,6 +210,12 @@ class Trainer:
         self.default_lr = 3e-8
         self.data_cnt_ema = args['batch_size'] * args['forward_steps']
         lr = self.default_lr * self.data_cnt_ema
+        # On GPU the lr lives as a device tensor so a CAPTURED Adam step
+        # still sees each epoch's dynamic-lr update (reference
+        # train.py:382-384 semantics) without re-capturing the graph:
+        # the replayed kernels read the tensor, we fill_() it in place.
+        if self.device.type == 'cuda':
+            lr = torch.tensor(lr, dtype=torch.float32, device=self.device)
         self.optimizer = optim.Adam(self.params, lr=lr, weight_decay=1e-5) \
             if len(self.params) > 0 else None
         self.steps = 0
@@ -302,8 +308,12 @@ class Trainer:
 
         global_data_cnt = hdist.allreduce_scalar(data_cnt / (1e-2 + batch_cnt))
         self.data_cnt_ema = self.data_cnt_ema * 0.8 + global_data_cnt * 0.2
+        new_lr = self.default_lr * self.data_cnt_ema / (1 + self.steps * 1e-5)
         for param_group in self.optimizer.param_groups:
-            param_group['lr'] = self.default_lr * self.data_cnt_ema / (1 + self.steps * 1e-5)
+            if torch.is_tensor(param_group['lr']):
+                param_group['lr'].fill_(new_lr)
+            else:
+                param_group['lr'] = new_lr
         return self.snapshot()
 
     def run(self):
