@@ -88,6 +88,8 @@ def main():
     batcher = False if device_replay else Batcher(args, buffer)
 
     actor_procs = int(os.environ.get('HANDYRL_ACTOR_PROCS', '8'))
+    # 3 pipeline slots per worker measured best (8x3 sweep)
+    os.environ.setdefault('HANDYRL_ACTOR_SLOTS', '3')
     mpool = None
     if actor_procs > 0:
         from handyrl_amd.actor import MultiProcGeesePool

@@ -22,7 +22,7 @@ import torch
 os.environ.setdefault('MIOPEN_FIND_MODE', 'FAST')
 
 from . import ops
-from .envs.vec_geese import GeeseVecEnv, N_PLAYERS
+from .envs.vec_geese import GeeseVecEnv, N_PLAYERS, CHMAP
 from .envs.hungry_geese import MAX_STEPS
 
 class GeeseActorPool:
@@ -39,7 +39,9 @@ class GeeseActorPool:
         # columnar trajectory recording: struct-of-arrays ring per game
         # (episodes stay columnar through the replay buffer and batch maker)
         G, CAPT = n_games, MAX_STEPS
-        self.rec_obs = np.zeros((G, CAPT, N_PLAYERS, 17, 7, 11), dtype=np.uint8)
+        # obs are stored CANONICAL (per game, not per seat, 4x smaller);
+        # the seat channel-gather happens on the GPU (CHMAP)
+        self.rec_obs = np.zeros((G, CAPT, 17, 7, 11), dtype=np.uint8)
         self.rec_alive = np.zeros((G, CAPT, N_PLAYERS), dtype=bool)
         self.rec_act = np.zeros((G, CAPT, N_PLAYERS), dtype=np.int32)
         self.rec_prob = np.zeros((G, CAPT, N_PLAYERS), dtype=np.float32)
@@ -61,14 +63,14 @@ class GeeseActorPool:
             if use_graphs and self.device.type == 'cuda':
                 from .hipgraph import GraphedActorForward
                 self.graphed = GraphedActorForward(model, self.device, fused=self.fused)
-        # pipelined-transfer staging (graphed path)
+        # pipelined-transfer staging (graphed path): obs are canonical
+        # (one row per game), results are per-seat (4 rows per game)
         if self.graphed is not None:
-            cap = N_PLAYERS * n_games
-            self._obs_pin = torch.empty(cap, 17, 7, 11, dtype=torch.uint8,
+            self._obs_pin = torch.empty(n_games, 17, 7, 11, dtype=torch.uint8,
                                         pin_memory=True)
             self._obs_pin_np = self._obs_pin.numpy()
-            self._out_pin = torch.empty(cap, 3, dtype=torch.float32,
-                                        pin_memory=True)
+            self._out_pin = torch.empty(N_PLAYERS * n_games, 3,
+                                        dtype=torch.float32, pin_memory=True)
             self._out_pin_np = self._out_pin.numpy()
             self._event = torch.cuda.Event()
         self._pending = None
@@ -91,36 +93,36 @@ class GeeseActorPool:
 
     # -- external-inference mode (env-worker processes) --------------------
     def prepare_step(self, out_buf):
-        """Build observations, record obs/alive columns, and write the
-        selected live-seat observations into ``out_buf[:M]``.  Returns M.
-        The caller supplies (actions, probs, values) to complete_step."""
+        """Build CANONICAL observations, record obs/alive columns, and
+        write the live games' obs into ``out_buf[:M]`` (M = live games).
+        The caller supplies per-seat (actions, probs, values) of length
+        4*M in game-major seat order to complete_step."""
         import time
         tm = self.timing
         vec = self.vec
         t0 = time.time()
-        obs_u8 = vec.observations()
+        obs_u8 = vec.observations()          # (G, 17, 7, 11) canonical
         tm['obs'] += time.time() - t0
         live = vec.alive & ~vec.over[:, None]
-        gi, pi = np.nonzero(live)
-        if len(gi) == 0:
+        lg = np.nonzero(live.any(axis=1))[0]
+        if len(lg) == 0:
             self._pending = None
             return 0
-        game_has_live = live.any(axis=1)
-        lg = np.nonzero(game_has_live)[0]
         t_idx = self.rec_len[lg]
         self.rec_obs[lg, t_idx] = obs_u8[lg]
         self.rec_alive[lg, t_idx] = live[lg]
         self._rec_slot = (lg, t_idx)
-        M = len(gi)
-        out_buf[:M] = obs_u8[gi, pi]
-        self._pending = ('ext', gi, pi, live, M)
+        M = len(lg)
+        out_buf[:M] = obs_u8[lg]
+        self._pending = ('ext', lg, live, M)
         return M
 
     def complete_step(self, actions, probs, values):
-        """Apply externally computed inference results; returns #frames."""
+        """Apply externally computed inference results (4*M game-major
+        seat rows; dead seats' rows are ignored); returns #frames."""
         assert self._pending is not None and self._pending[0] == 'ext'
-        _, gi, pi, live, M = self._pending
-        self._pending = ('done', gi, pi, live, M, actions, probs, values)
+        _, lg, live, M = self._pending
+        self._pending = ('done', lg, live, M, actions, probs, values)
         return self._phase2()
 
     def _phase1(self):
@@ -130,50 +132,50 @@ class GeeseActorPool:
         tm = self.timing
         vec = self.vec
         t0 = time.time()
-        obs_u8 = vec.observations()                     # (G, 4, 17, 7, 11)
+        obs_u8 = vec.observations()                     # (G, 17, 7, 11)
         tm['obs'] += time.time() - t0
         live = vec.alive & ~vec.over[:, None]
-        gi, pi = np.nonzero(live)
-        if len(gi) == 0:
+        lg = np.nonzero(live.any(axis=1))[0]
+        if len(lg) == 0:
             self._pending = None
             return
         t0 = time.time()
-        obs_sel = obs_u8[gi, pi]                        # (M, 17, 7, 11)
-        M = len(gi)
+        M = len(lg)
 
         # record the observation/alive columns now (obs is in hand and this
         # CPU work overlaps the GPU forward on the pipelined path)
-        game_has_live = live.any(axis=1)
-        lg = np.nonzero(game_has_live)[0]
         t_idx = self.rec_len[lg]
         self.rec_obs[lg, t_idx] = obs_u8[lg]
         self.rec_alive[lg, t_idx] = live[lg]
         self._rec_slot = (lg, t_idx)
 
         if self.graphed is not None:
-            np.copyto(self._obs_pin_np[:M], obs_sel)
+            np.copyto(self._obs_pin_np[:M], obs_u8[lg])
             self.graphed.run_async(self._obs_pin, M, self._out_pin, self._event)
-            self._pending = ('async', gi, pi, live, M)
+            self._pending = ('async', lg, live, M)
             tm['fwd'] += time.time() - t0
             return
 
+        # eager fallback: expand canonical -> per-seat on host (tests)
+        obs_sel = obs_u8.reshape(self.n_games, 17, 77)[lg][:, CHMAP] \
+            .reshape(M * N_PLAYERS, 17, 7, 11)
         if self.device.type == 'cuda':
-            # eager GPU path: pad to a fixed bucket so MIOpen keeps one
-            # solution per shape
-            bucket = 256 * ((M + 255) // 256)
-            if bucket > M:
-                pad = np.zeros((bucket - M,) + obs_sel.shape[1:], dtype=obs_sel.dtype)
+            # pad to a fixed bucket so MIOpen keeps one solution per shape
+            R = M * N_PLAYERS
+            bucket = 256 * ((R + 255) // 256)
+            if bucket > R:
+                pad = np.zeros((bucket - R,) + obs_sel.shape[1:], dtype=obs_sel.dtype)
                 obs_in = np.concatenate([obs_sel, pad], axis=0)
             else:
                 obs_in = obs_sel
             obs_t = torch.from_numpy(obs_in).to(self.device, non_blocking=True)
             policy, value = self._policy_forward(obs_t.float())
-            policy, value = policy[:M], value[:M]
+            policy, value = policy[:R], value[:R]
             A = policy.shape[1]
-            if self._zero_mask is None or self._zero_mask.shape[0] < M:
-                self._zero_mask = torch.zeros(max(M, 1), A, device=self.device)
-            uniform = torch.rand(M, device=self.device)
-            actions_t, probs_t = ops.masked_sample(policy, self._zero_mask[:M], uniform)
+            if self._zero_mask is None or self._zero_mask.shape[0] < R:
+                self._zero_mask = torch.zeros(max(R, 1), A, device=self.device)
+            uniform = torch.rand(R, device=self.device)
+            actions_t, probs_t = ops.masked_sample(policy, self._zero_mask[:R], uniform)
             packed = torch.cat([actions_t.float().unsqueeze(1),
                                 probs_t.unsqueeze(1), value], dim=1).cpu().numpy()
             actions = packed[:, 0].astype(np.int64)
@@ -188,11 +190,13 @@ class GeeseActorPool:
             probs = probs_full.gather(-1, actions_t.unsqueeze(-1)).squeeze(-1).numpy()
             values = value.squeeze(-1).numpy()
         tm['fwd'] += time.time() - t0
-        self._pending = ('done', gi, pi, live, M, actions, probs, values)
+        self._pending = ('done', lg, live, M, actions, probs, values)
 
     def _phase2(self):
         """Wait for the issued forward, then apply actions: record, step
-        the vectorized env, package finished games.  Returns #frames."""
+        the vectorized env, package finished games.  Returns #frames.
+        Inference rows are game-major per-seat: row m*4+p is seat p of
+        live game lg[m]; dead seats' rows carry garbage and are masked."""
         if self._pending is None:
             return 0
         import time
@@ -201,23 +205,24 @@ class GeeseActorPool:
         pend, self._pending = self._pending, None
         t0 = time.time()
         if pend[0] == 'async':
-            _, gi, pi, live, M = pend
+            _, lg, live, M = pend
             self._event.synchronize()
-            packed = self._out_pin_np[:M]
+            packed = self._out_pin_np[:M * N_PLAYERS]
             actions = packed[:, 0].astype(np.int64)
             probs = packed[:, 1]
             values = packed[:, 2]
         else:
-            _, gi, pi, live, M, actions, probs, values = pend
+            _, lg, live, M, actions, probs, values = pend
         tm['sample'] += time.time() - t0
         t0 = time.time()
+        live_lg = live[lg]                               # (M, 4)
         act_grid = np.zeros((self.n_games, N_PLAYERS), dtype=np.int32)
-        act_grid[gi, pi] = actions
+        act_grid[lg] = np.where(live_lg, actions.reshape(M, N_PLAYERS), 0)
 
         prob_row = np.zeros((self.n_games, N_PLAYERS), dtype=np.float32)
         val_row = np.zeros((self.n_games, N_PLAYERS), dtype=np.float32)
-        prob_row[gi, pi] = probs
-        val_row[gi, pi] = values
+        prob_row[lg] = np.where(live_lg, probs.reshape(M, N_PLAYERS), 0.0)
+        val_row[lg] = np.where(live_lg, values.reshape(M, N_PLAYERS), 0.0)
 
         # finish the columnar step record started in _phase1
         lg, t_idx = self._rec_slot
@@ -255,6 +260,7 @@ class GeeseActorPool:
             'steps': S,
             'outcome': {p: float(outcome_row[p]) for p in range(N_PLAYERS)},
             'columnar': True,
+            'canonical_obs': True,     # (S, 17, 7, 11); seat views = CHMAP gather
             'n_actions': 4,
             'obs': self.rec_obs[g, :S].copy(),
             'alive': self.rec_alive[g, :S].copy(),
@@ -350,14 +356,14 @@ def _geese_env_worker(conn, ep_conn, obs_name, res_name, n_games, args, seed,
     obs_shm = shared_memory.SharedMemory(name=obs_name)
     res_shm = shared_memory.SharedMemory(name=res_name)
     per = max(1, n_games // slots)
-    cap = per * N_PLAYERS
     obs_views, res_views, pools = [], [], []
     for s in range(slots):
-        obs_views.append(np.ndarray((cap, 17, 7, 11), dtype=np.uint8,
+        obs_views.append(np.ndarray((per, 17, 7, 11), dtype=np.uint8,
                                     buffer=obs_shm.buf,
-                                    offset=s * cap * 17 * 7 * 11))
-        res_views.append(np.ndarray((cap, 3), dtype=np.float32,
-                                    buffer=res_shm.buf, offset=s * cap * 12))
+                                    offset=s * per * 17 * 7 * 11))
+        res_views.append(np.ndarray((per * N_PLAYERS, 3), dtype=np.float32,
+                                    buffer=res_shm.buf,
+                                    offset=s * per * N_PLAYERS * 12))
         pools.append(GeeseActorPool(None, args, n_games=per,
                                     device=torch.device('cpu'),
                                     use_graphs=False, seed=seed + 131 * s))
@@ -373,7 +379,7 @@ def _geese_env_worker(conn, ep_conn, obs_name, res_name, n_games, args, seed,
         s = cmd[1]                            # ('go', slot)
         M, frames = m_inflight[s], 0
         if M:
-            r = res_views[s][:M]
+            r = res_views[s][:M * N_PLAYERS]      # 4 seat rows per game
             frames = pools[s].complete_step(
                 r[:, 0].astype(np.int64), r[:, 1].copy(), r[:, 2].copy())
             eps = pools[s].harvest()
@@ -410,20 +416,21 @@ class MultiProcGeesePool:
         self.n_per = max(1, per // slots)       # games per slot
         self.conns, self.ep_conns, self.procs, self.shms = [], [], [], []
         self.obs_views, self.res_views = [], []  # [worker][slot]
-        cap = self.n_per * N_PLAYERS
+        capg = self.n_per                        # canonical obs: 1 row/game
+        capr = self.n_per * N_PLAYERS            # results: 4 seat rows/game
         for w in range(workers):
             obs_shm = shared_memory.SharedMemory(
-                create=True, size=slots * cap * 17 * 7 * 11)
+                create=True, size=slots * capg * 17 * 7 * 11)
             res_shm = shared_memory.SharedMemory(
-                create=True, size=slots * cap * 3 * 4)
+                create=True, size=slots * capr * 3 * 4)
             self.shms += [obs_shm, res_shm]
             self.obs_views.append([
-                np.ndarray((cap, 17, 7, 11), dtype=np.uint8,
-                           buffer=obs_shm.buf, offset=s * cap * 17 * 7 * 11)
+                np.ndarray((capg, 17, 7, 11), dtype=np.uint8,
+                           buffer=obs_shm.buf, offset=s * capg * 17 * 7 * 11)
                 for s in range(slots)])
             self.res_views.append([
-                np.ndarray((cap, 3), dtype=np.float32,
-                           buffer=res_shm.buf, offset=s * cap * 12)
+                np.ndarray((capr, 3), dtype=np.float32,
+                           buffer=res_shm.buf, offset=s * capr * 12)
                 for s in range(slots)])
             parent_conn, child_conn = mp.Pipe(duplex=True)
             ep_parent, ep_child = mp.Pipe(duplex=False)
@@ -472,7 +479,7 @@ class MultiProcGeesePool:
                                                  pin_memory=True)
                                      for _ in range(self.slots)]
                                     for _ in range(self.workers)]
-            self._obs_pin = mk((cap, 17, 7, 11), torch.uint8)
+            self._obs_pin = mk((self.n_per, 17, 7, 11), torch.uint8)
             self._obs_pin_np = [[t.numpy() for t in row] for row in self._obs_pin]
             self._out_pin = mk((cap, 3), torch.float32)
             self._out_pin_np = [[t.numpy() for t in row] for row in self._out_pin]
@@ -541,8 +548,9 @@ class MultiProcGeesePool:
         if M and self.graphed is not None:
             self._events[wid][slot].synchronize()
             if not self._use_registered:
-                np.copyto(self.res_views[wid][slot][:M],
-                          self._out_pin_np[wid][slot][:M])
+                R = M * N_PLAYERS
+                np.copyto(self.res_views[wid][slot][:R],
+                          self._out_pin_np[wid][slot][:R])
         self.conns[wid].send(('go', slot))
 
     def step_once(self):
@@ -579,18 +587,21 @@ class MultiProcGeesePool:
                                        self._events[wid][slot])
             self.inflight[(wid, slot)] = M
         elif M:
-            # CPU fallback (tests): synchronous eager inference
-            obs_t = torch.from_numpy(
-                self.obs_views[wid][slot][:M].copy()).float()
+            # CPU fallback (tests): synchronous eager inference on the
+            # seat-expanded view (CHMAP gather of the canonical obs)
+            canon = self.obs_views[wid][slot][:M].reshape(M, 17, 77)
+            obs_np = canon[:, CHMAP].reshape(M * N_PLAYERS, 17, 7, 11)
+            obs_t = torch.from_numpy(obs_np.copy()).float()
             with torch.no_grad():
                 out = self.model(obs_t, None)
             probs = torch.softmax(out['policy'].float(), dim=-1)
             acts = torch.multinomial(probs, 1).squeeze(-1)
             sel = probs.gather(-1, acts.unsqueeze(-1)).squeeze(-1)
+            R = M * N_PLAYERS
             res = self.res_views[wid][slot]
-            res[:M, 0] = acts.numpy()
-            res[:M, 1] = sel.numpy()
-            res[:M, 2] = out['value'].float().squeeze(-1).numpy()
+            res[:R, 0] = acts.numpy()
+            res[:R, 1] = sel.numpy()
+            res[:R, 2] = out['value'].float().squeeze(-1).numpy()
             self.inflight[(wid, slot)] = 0   # results already in shm
         else:
             self.inflight[(wid, slot)] = 0

@@ -71,7 +71,14 @@ def _episode_fields_columnar(ep, args):
     A = int(ep['n_actions'])
 
     aliveP = alive[:, pl]                                        # (T, P)
-    obs = ep['obs'][:, pl] * aliveP[..., None, None, None].astype(ep['obs'].dtype)
+    if ep.get('canonical_obs'):
+        # obs is (T, 17, 7, 11) canonical; seat views = CHMAP channel gather
+        from .envs.vec_geese import CHMAP
+        canon = ep['obs'].reshape(T, 17, 77)
+        obs = canon[:, CHMAP[pl]].reshape(T, NP, 17, 7, 11)
+    else:
+        obs = ep['obs'][:, pl]
+    obs = obs * aliveP[..., None, None, None].astype(obs.dtype)
     prob = np.where(aliveP, ep['prob'][:, pl], 1.0)[..., None].astype(np.float64)
     act = (ep['action'][:, pl] * aliveP).astype(np.int64)[..., None]
     amask = np.full((T, NP, A), 1e32, dtype=np.float32)
@@ -87,7 +94,7 @@ def _episode_fields_columnar(ep, args):
     omask = tmask.copy()
     progress = np.arange(ep['start'], ep['end'],
                          dtype=np.float32)[..., np.newaxis] / ep['total']
-    obs_zeros = np.zeros_like(ep['obs'][0, 0])
+    obs_zeros = np.zeros_like(obs[0, 0])
     return (obs, obs_zeros, prob, v, act, oc, rew, ret, emask, tmask, omask,
             amask, progress)
 
@@ -269,6 +276,7 @@ class EpisodeBuffer:
             # columnar episodes: the window is an array view, no block math
             return {
                 'args': ep['args'], 'outcome': ep['outcome'], 'columnar': True,
+                'canonical_obs': ep.get('canonical_obs', False),
                 'n_actions': ep['n_actions'],
                 'obs': ep['obs'][st:ed], 'alive': ep['alive'][st:ed],
                 'action': ep['action'][st:ed], 'prob': ep['prob'][st:ed],

@@ -33,6 +33,18 @@ _OPP = np.array(OPPOSITE, dtype=np.int32)
 # seat-relative channel gather: obs channel c of player k shows goose (c+k)%4
 _REL = (np.arange(N_PLAYERS)[None, :] + np.arange(N_PLAYERS)[:, None]) % N_PLAYERS
 
+# CHMAP[k, c] = canonical channel feeding seat k's obs channel c.  The
+# canonical per-game encoding is [heads 0-3, tails 4-7, bodies 8-11,
+# prev-heads 12-15, food 16] in GOOSE order; a seat view is the same data
+# with the four goose planes of each group rotated so the seat's own goose
+# comes first.  Seat expansion is a pure channel gather — done on the GPU
+# (ops.obs_to_nhwc_rot / an index_select in the replay gather), never on
+# the host hot path.
+_c = np.arange(17)
+CHMAP = np.where(_c[None, :] < 16,
+                 (_c[None, :] & 12) | ((_c[None, :] + np.arange(N_PLAYERS)[:, None]) & 3),
+                 16).astype(np.int64)          # (4, 17)
+
 
 class GeeseVecEnv:
     """G simultaneous Hungry Geese games."""
@@ -197,16 +209,21 @@ class GeeseVecEnv:
 
     # -- observation / outcome ----------------------------------------------
     def observations(self):
-        """uint8 (G, 4, 17, 7, 11): the 17-plane encoding for every seat."""
+        """uint8 (G, 17, 7, 11): CANONICAL per-game planes (goose order
+        0-3 per group; see CHMAP).  Seat views are channel gathers done on
+        the GPU; use observations_per_seat() for host-side parity checks."""
         G = self.G
         if not hasattr(self, '_obs_buf'):
-            self._obs_buf = np.empty((G, N_PLAYERS, 17, N_CELLS), dtype=np.uint8)
-            self._scratch = np.zeros((3, G, N_PLAYERS, N_CELLS), dtype=np.uint8)
-            self._food_grid = np.zeros((G, N_CELLS), dtype=np.uint8)
-        head_grid, tail_grid, prev_grid = self._scratch
+            self._obs_buf = np.zeros((G, 17, N_CELLS), dtype=np.uint8)
+        obs = self._obs_buf
+        head_grid = obs[:, 0:4]
+        tail_grid = obs[:, 4:8]
+        prev_grid = obs[:, 12:16]
+        food_grid = obs[:, 16]
         head_grid[:] = 0
         tail_grid[:] = 0
         prev_grid[:] = 0
+        food_grid[:] = 0
         heads = self._head()
         tails = self._tail_cell()
         gi, pi = np.nonzero(self.alive)
@@ -214,24 +231,17 @@ class GeeseVecEnv:
         tail_grid[gi, pi, tails[gi, pi]] = 1
         gi, pi = np.nonzero(self.prev_head >= 0)
         prev_grid[gi, pi, self.prev_head[gi, pi]] = 1
-
-        food_grid = self._food_grid
-        food_grid[:] = 0
+        obs[:, 8:12] = self.body_grid
         for f in range(MIN_FOOD):
             ok = self.food[:, f] >= 0
             food_grid[np.nonzero(ok)[0], self.food[ok, f]] = 1
+        return obs.reshape(G, 17, ROWS, COLS)
 
-        # per-seat relative goose order, written straight into the output
-        # buffer (no (G,4,4,4,77) intermediate)
-        obs = self._obs_buf
-        for k in range(N_PLAYERS):
-            perm = _REL[k]
-            obs[:, k, 0:4] = head_grid[:, perm]
-            obs[:, k, 4:8] = tail_grid[:, perm]
-            obs[:, k, 8:12] = self.body_grid[:, perm]
-            obs[:, k, 12:16] = prev_grid[:, perm]
-            obs[:, k, 16] = food_grid
-        return obs.reshape(G, N_PLAYERS, 17, ROWS, COLS)
+    def observations_per_seat(self):
+        """uint8 (G, 4, 17, 7, 11): seat-expanded views (CPU fallback /
+        tests) — CHMAP channel gather of the canonical encoding."""
+        canon = self.observations().reshape(self.G, 17, N_CELLS)
+        return canon[:, CHMAP].reshape(self.G, N_PLAYERS, 17, ROWS, COLS)
 
     def outcomes(self, games):
         """Pairwise rank outcome per seat for the given finished games."""

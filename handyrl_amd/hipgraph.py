@@ -86,17 +86,30 @@ class GraphedActorForward:
     """Per-bucket capture of {uint8 obs -> bf16 forward -> masked softmax
     sample -> packed (action, prob, value)} for the GPU actor pool."""
 
-    def __init__(self, model, device, warmup_iters=2, fused=None):
+    def __init__(self, model, device, warmup_iters=2, fused=None,
+                 canonical=True):
         from . import ops
+        from .envs.vec_geese import CHMAP
         self._ops = ops
         self.model = model
         self.device = device
         self.fused = fused        # GeeseFusedEval: hand-written MFMA path
+        self.canonical = canonical  # obs rows are per GAME; out rows x4
+        self._chmap = torch.from_numpy(CHMAP).to(device)
         self.graphs = {}
 
     def _fwd_sample(self, obs_u8, zero_mask):
         with torch.no_grad():
-            if self.fused is not None:
+            if self.canonical:
+                if self.fused is not None:
+                    out = self.fused.forward_canonical(obs_u8)
+                else:
+                    G = obs_u8.shape[0]
+                    obs_f = obs_u8.reshape(G, 17, 77).float()
+                    obs4 = obs_f[:, self._chmap].reshape(G * 4, 17, 7, 11)
+                    with torch.autocast('cuda', dtype=torch.bfloat16):
+                        out = self.model(obs4, None)
+            elif self.fused is not None:
                 out = self.fused.forward(obs_u8)
             else:
                 obs_f = obs_u8.float()
@@ -114,7 +127,8 @@ class GraphedActorForward:
         self.model.eval()
         static_obs = torch.zeros(bucket, 17, 7, 11, dtype=torch.uint8,
                                  device=self.device)
-        zero_mask = torch.zeros(bucket, n_actions, device=self.device)
+        rows = bucket * 4 if self.canonical else bucket
+        zero_mask = torch.zeros(rows, n_actions, device=self.device)
         stream = torch.cuda.Stream()
         stream.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(stream):
@@ -133,28 +147,35 @@ class GraphedActorForward:
         after caller padding decides the bucket). Returns packed (M, 3) on
         device."""
         M = obs_u8_cpu.shape[0]
-        bucket = 256 * ((M + 255) // 256)
+        bucket = self._bucket(M)
         if bucket not in self.graphs:
             self._capture(bucket, n_actions)
         graph, static_obs, packed = self.graphs[bucket]
         static_obs[:M].copy_(obs_u8_cpu, non_blocking=True)
         # rows [M:bucket) keep stale data; every op is row-independent and
-        # the outputs are sliced to [:M]
+        # the outputs are sliced to the live rows
         graph.replay()
-        return packed[:M]
+        return packed[:M * 4] if self.canonical else packed[:M]
+
+    def _bucket(self, M):
+        # canonical: M counts GAMES (4 net rows each) -> 64-game buckets
+        # keep the network batch at 256-row multiples either way
+        q = 64 if self.canonical else 256
+        return q * ((M + q - 1) // q)
 
     def run_async(self, obs_pinned, M, out_pinned, event, n_actions=4):
         """Pipelined variant: H2D from a pinned staging tensor, replay, and
         an async D2H of the packed result into ``out_pinned``; ``event``
         records completion.  No host sync — the caller overlaps CPU work
         and waits on the event."""
-        bucket = 256 * ((M + 255) // 256)
+        bucket = self._bucket(M)
         if bucket not in self.graphs:
             self._capture(bucket, n_actions)
         graph, static_obs, packed = self.graphs[bucket]
         static_obs[:M].copy_(obs_pinned[:M], non_blocking=True)
         graph.replay()
-        out_pinned[:M].copy_(packed[:M], non_blocking=True)
+        R = M * 4 if self.canonical else M
+        out_pinned[:R].copy_(packed[:R], non_blocking=True)
         event.record()
 
 

@@ -99,6 +99,13 @@ def obs_to_nhwc(obs_u8):
     return require().obs_to_nhwc(obs_u8.reshape(obs_u8.shape[0], 17, 77).contiguous())
 
 
+def obs_to_nhwc_rot(obs_u8):
+    """Canonical (G,17,7,11) uint8 -> seat-expanded (4G,77,32) bf16 NHWC
+    (row g*4+k = seat k of game g; CHMAP channel rotation done in-kernel)."""
+    return require().obs_to_nhwc_rot(
+        obs_u8.reshape(obs_u8.shape[0], 17, 77).contiguous())
+
+
 def torus_conv_fused(x, wfrag, shift, nbr, residual=None, relu=False):
     """y = act(torus_conv3x3(x) * scale + shift [+ residual]) on NHWC bf16.
 

@@ -578,6 +578,30 @@ __global__ void obs_to_nhwc_kernel(const unsigned char* __restrict__ obs,
     out[idx] = f2bf(v);
 }
 
+// --- canonical obs (N,17,77) u8 -> seat-expanded (N*4,77,32) bf16 NHWC --
+// Output row n = g*4 + k is seat k's view of game g: channel c < 16 reads
+// canonical channel (c & 12) | ((c + k) & 3) (the four goose planes of each
+// group rotated so the seat's own goose comes first), channel 16 = food.
+// Host code ships ONE board per game (4x less bytes end-to-end); the seat
+// rotation is free address math here.
+__global__ void obs_to_nhwc_rot_kernel(const unsigned char* __restrict__ obs,
+                                       short* __restrict__ out, long total) {
+    const long idx = (long)blockIdx.x * blockDim.x + threadIdx.x;  // N*4*77*32
+    if (idx >= total) return;
+    const int c = idx & 31;
+    const long pc = idx >> 5;
+    const long row = pc / 77;              // g*4 + k
+    const int cell = pc % 77;
+    const long g = row >> 2;
+    const int k = row & 3;
+    float v = 0.f;
+    if (c < 17) {
+        const int src = (c < 16) ? ((c & 12) | ((c + k) & 3)) : 16;
+        v = (float)obs[(g * 17 + src) * 77 + cell];
+    }
+    out[idx] = f2bf(v);
+}
+
 // --- fused torus conv block: y = act(conv3x3_wrap(x)*scale + shift [+ x]) -
 // Implicit GEMM on MFMA, NHWC bf16 activations, BN folded into the packed
 // weights (wfrag) and the per-channel shift.  The 3x3 wrap-around gather is
@@ -764,6 +788,20 @@ static torch::Tensor obs_to_nhwc(torch::Tensor obs) {
     return out;
 }
 
+static torch::Tensor obs_to_nhwc_rot(torch::Tensor obs) {
+    TORCH_CHECK(obs.is_cuda() && obs.scalar_type() == torch::kUInt8);
+    const long N = obs.size(0);
+    auto out = torch::empty({N * 4, 77, 32},
+                            obs.options().dtype(torch::kBFloat16));
+    const long total = N * 4 * 77 * 32;
+    const int block = 256;
+    const long grid = (total + block - 1) / block;
+    auto stream = at::cuda::getCurrentCUDAStream();
+    hipLaunchKernelGGL(obs_to_nhwc_rot_kernel, dim3(grid), dim3(block), 0,
+        stream, obs.data_ptr<unsigned char>(), (short*)out.data_ptr(), total);
+    return out;
+}
+
 static torch::Tensor torus_conv_fused(
     torch::Tensor x, torch::Tensor wfrag, torch::Tensor shift,
     torch::Tensor nbr, c10::optional<torch::Tensor> res, bool apply_relu) {
@@ -921,6 +959,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "batched masked-softmax action sampling (actions, probs)");
     m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 MFMA layout probe");
     m.def("obs_to_nhwc", &obs_to_nhwc, "uint8 NCHW obs -> padded NHWC bf16");
+    m.def("obs_to_nhwc_rot", &obs_to_nhwc_rot,
+          "canonical uint8 obs -> 4 seat-rotated NHWC bf16 rows per game");
     m.def("torus_conv_fused", &torus_conv_fused,
           "fused wrap-around conv3x3 + BN-fold + residual + relu (MFMA)");
     m.def("torus_wgrad", &torus_wgrad,

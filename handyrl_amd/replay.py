@@ -22,7 +22,10 @@ from collections import deque
 import numpy as np
 import torch
 
-OBS_SHAPE = (4, 17, 7, 11)
+# ring rows store CANONICAL per-game obs (one 17-plane board per step, not
+# four seat views): 4x the episode capacity per byte of HBM; the per-seat
+# channel rotation (vec_geese.CHMAP) happens inside gather_batch on device
+OBS_SHAPE = (17, 7, 11)
 ROW_OBS = int(np.prod(OBS_SHAPE))
 
 # sampleable rows keep this distance from the overwrite frontier so that
@@ -55,6 +58,9 @@ class DeviceReplay:
         self.action = torch.empty(self.ring_T, 4, dtype=torch.int32, device=dev)
         self.prob = torch.empty(self.ring_T, 4, dtype=torch.float32, device=dev)
         self.value = torch.empty(self.ring_T, 4, dtype=torch.float32, device=dev)
+        from .envs.vec_geese import CHMAP
+        self._chmap = torch.from_numpy(CHMAP).to(dev)     # (4, 17)
+        self._arange_cache = {}
         self.head = 0                  # monotonically increasing write cursor
         self.table = deque()           # (pos0, steps, outcome(np[4]))
         self.total_added = 0
@@ -253,7 +259,15 @@ class DeviceReplay:
         flat = rows.reshape(-1)
         sel = seat.unsqueeze(1).expand(B, T).reshape(-1)
 
-        obs = self.obs[flat, sel]                                # (B*T, 17,7,11)
+        # canonical row gather + per-seat channel rotation (CHMAP)
+        obs_c = self.obs[flat].reshape(-1, 17, 77)               # (B*T, 17, 77)
+        chmap_bt = self._chmap[sel]                              # (B*T, 17)
+        nbt = obs_c.shape[0]
+        ar = self._arange_cache.get(nbt)
+        if ar is None:
+            ar = torch.arange(nbt, device=dev).unsqueeze(1)
+            self._arange_cache[nbt] = ar
+        obs = obs_c[ar, chmap_bt].reshape(-1, 17, 7, 11)         # (B*T, 17,7,11)
         alive = self.alive[flat, sel]
         act = self.action[flat, sel].long()
         prob = self.prob[flat, sel]

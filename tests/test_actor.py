@@ -44,7 +44,8 @@ def test_actor_pool_generates_valid_episodes():
         assert ep['columnar']
         assert set(ep['outcome'].keys()) == {0, 1, 2, 3}
         assert abs(sum(ep['outcome'].values())) < 1e-6    # pairwise zero sum
-        assert ep['obs'].shape == (S, 4, 17, 7, 11) and ep['obs'].dtype == np.uint8
+        assert ep['canonical_obs']
+        assert ep['obs'].shape == (S, 17, 7, 11) and ep['obs'].dtype == np.uint8
         assert ep['alive'].shape == (S, 4)
         assert ep['alive'][0].all()                        # all alive at start
         live = ep['alive']
@@ -58,15 +59,18 @@ def test_actor_pool_generates_valid_episodes():
 
 def _dict_episode_from_columnar(ep):
     """Reference-format (moment dict) episode with identical content."""
+    from handyrl_amd.envs.vec_geese import CHMAP
     moments = []
     S = ep['steps']
+    # canonical obs -> per-seat views for the dict format
+    obs_seat = ep['obs'].reshape(S, 17, 77)[:, CHMAP].reshape(S, 4, 17, 7, 11)
     for t in range(S):
         keys = ('observation', 'selected_prob', 'action_mask', 'action',
                 'value', 'reward', 'return')
         moment = {k: {p: None for p in range(4)} for k in keys}
         turn = [p for p in range(4) if ep['alive'][t, p]]
         for p in turn:
-            moment['observation'][p] = ep['obs'][t, p]
+            moment['observation'][p] = obs_seat[t, p]
             moment['selected_prob'][p] = float(ep['prob'][t, p])
             moment['action_mask'][p] = np.zeros(4, dtype=np.float32)
             moment['action'][p] = int(ep['action'][t, p])

@@ -357,3 +357,30 @@ def test_gpu_rnn_geister_step():
     losses, dcnt = trainer.train_step(batch)
     assert torch.isfinite(losses['total'].detach())
     torch.cuda.synchronize()
+
+
+@requires_gpu
+def test_fused_eval_canonical_matches_seat_expanded():
+    """forward_canonical (in-kernel CHMAP seat rotation) must be bitwise
+    identical to forward() on host-expanded per-seat observations."""
+    from handyrl_amd.models.geese_net import GeeseNet, GeeseFusedEval
+    from handyrl_amd.envs.vec_geese import CHMAP
+    torch.manual_seed(4)
+    net = GeeseNet().cuda()
+    for layer in [net.stem] + list(net.blocks):
+        layer.bn.running_mean.uniform_(-0.2, 0.2)
+        layer.bn.running_var.uniform_(0.7, 1.4)
+    net.eval()
+
+    G = 65
+    canon = (torch.rand(G, 17, 7, 11, device='cuda') < 0.15).to(torch.uint8)
+    fused = GeeseFusedEval(net, torch.device('cuda'))
+    chmap = torch.from_numpy(CHMAP).cuda()
+    expanded = canon.reshape(G, 17, 77)[:, chmap] \
+        .reshape(G * 4, 17, 7, 11).contiguous()
+    with torch.no_grad():
+        out_c = fused.forward_canonical(canon)
+        out_e = fused.forward(expanded)
+    torch.cuda.synchronize()
+    torch.testing.assert_close(out_c['policy'], out_e['policy'], rtol=0, atol=0)
+    torch.testing.assert_close(out_c['value'], out_e['value'], rtol=0, atol=0)

@@ -67,7 +67,7 @@ def test_device_replay_matches_make_batch():
         ed = st + int(length[b])
         sels.append({
             'args': ep['args'], 'outcome': ep['outcome'], 'columnar': True,
-            'n_actions': 4,
+            'canonical_obs': ep.get('canonical_obs', False), 'n_actions': 4,
             'obs': ep['obs'][st:ed], 'alive': ep['alive'][st:ed],
             'action': ep['action'][st:ed], 'prob': ep['prob'][st:ed],
             'value': ep['value'][st:ed],

@@ -104,7 +104,7 @@ def test_vec_observation_matches_single():
     for step in range(60):
         if vec.over[0]:
             break
-        obs_vec = vec.observations()[0]         # (4, 17, 7, 11) uint8
+        obs_vec = vec.observations_per_seat()[0]    # (4, 17, 7, 11) uint8
         for p in range(4):
             ref = env.observation(p)
             np.testing.assert_array_equal(obs_vec[p].astype(np.float32), ref,
