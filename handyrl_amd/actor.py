@@ -544,6 +544,8 @@ class MultiProcGeesePool:
         return sum(1 for (w, _s) in self.inflight if w == wid)
 
     def _complete(self, wid, slot):
+        import time
+        t0 = time.time()
         M = self.inflight.pop((wid, slot))
         if M and self.graphed is not None:
             self._events[wid][slot].synchronize()
@@ -552,6 +554,7 @@ class MultiProcGeesePool:
                 np.copyto(self.res_views[wid][slot][:R],
                           self._out_pin_np[wid][slot][:R])
         self.conns[wid].send(('go', slot))
+        self.timing['sample'] += time.time() - t0    # event sync + go
 
     def step_once(self):
         """Service one child request (whichever is ready first — a
@@ -561,6 +564,7 @@ class MultiProcGeesePool:
         this one's round trip is in flight.  Returns frames reported."""
         import time
         import multiprocessing.connection as mpc
+        t0 = time.time()
         waitable = [c for i, c in enumerate(self.conns)
                     if self._inflight_cnt(i) < self.slots]
         ready = mpc.wait(waitable)
@@ -570,6 +574,8 @@ class MultiProcGeesePool:
         tag, slot, M, frames = conn.recv()
         assert tag == 'obs'
         self.frames += frames
+        self.timing['obs'] += time.time() - t0       # wait + recv
+        self.timing['n'] += 1
 
         t0 = time.time()
         if M and self.graphed is not None:

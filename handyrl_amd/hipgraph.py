@@ -158,9 +158,10 @@ class GraphedActorForward:
         return packed[:M * 4] if self.canonical else packed[:M]
 
     def _bucket(self, M):
-        # canonical: M counts GAMES (4 net rows each) -> 64-game buckets
-        # keep the network batch at 256-row multiples either way
-        q = 64 if self.canonical else 256
+        # canonical: M counts GAMES (4 net rows each); 32-game buckets =
+        # 128-row forwards (the fused MFMA kernels take any row count —
+        # the 256-row multiple was only ever a MIOpen find-cache guard)
+        q = 32 if self.canonical else 256
         return q * ((M + q - 1) // q)
 
     def run_async(self, obs_pinned, M, out_pinned, event, n_actions=4):
