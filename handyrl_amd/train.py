@@ -214,7 +214,8 @@ class Trainer:
         # still sees each epoch's dynamic-lr update (reference
         # train.py:382-384 semantics) without re-capturing the graph:
         # the replayed kernels read the tensor, we fill_() it in place.
-        if self.device.type == 'cuda':
+        if self.device.type == 'cuda' and \
+                os.environ.get('HANDYRL_TENSOR_LR', '1') == '1':
             lr = torch.tensor(lr, dtype=torch.float32, device=self.device)
         self.optimizer = optim.Adam(self.params, lr=lr, weight_decay=1e-5) \
             if len(self.params) > 0 else None
