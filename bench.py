@@ -92,9 +92,9 @@ def main():
     batcher = False if device_replay else Batcher(args, buffer)
 
     actor_procs = int(os.environ.get('HANDYRL_ACTOR_PROCS', '6'))
-    # 6 workers x 2 pipeline slots measured best with canonical obs
-    # (interleaved A/B sweeps)
-    os.environ.setdefault('HANDYRL_ACTOR_SLOTS', '2')
+    # single slot per worker: the faster 2-slot double-buffering is gated
+    # off pending a GPU transport-race fix (BASELINE.md learning sanity)
+    os.environ.setdefault('HANDYRL_ACTOR_SLOTS', '1')
     mpool = None
     if actor_procs > 0:
         from handyrl_amd.actor import MultiProcGeesePool

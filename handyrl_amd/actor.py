@@ -408,7 +408,11 @@ class MultiProcGeesePool:
         self.args = args
         self.workers = workers
         if slots is None:
-            slots = int(os.environ.get('HANDYRL_ACTOR_SLOTS', '2'))
+            # >1 slots (double-buffered half-shards) measures faster but a
+            # GPU-side transport race poisons recorded values (NaN losses
+            # after ~100 learner steps, see BASELINE.md learning-sanity
+            # section) — off until the race is found; opt in to reproduce.
+            slots = int(os.environ.get('HANDYRL_ACTOR_SLOTS', '1'))
         per = max(1, n_games // workers)
         slots = max(1, min(slots, per))
         self.slots = slots
@@ -493,7 +497,9 @@ class MultiProcGeesePool:
         service path).  Falls back to the staging copy if registration is
         refused (e.g. exotic shm mounts)."""
         self._use_registered = False
-        if os.environ.get('HANDYRL_NO_SHM_REGISTER'):
+        # direct-DMA from hipHostRegister'd shm also showed a slow-burning
+        # value-corruption race (NaN by ~400 steps) — opt-in only.
+        if os.environ.get('HANDYRL_SHM_REGISTER', '0') != '1':
             return
         flat = [v for row in self.obs_views for v in row] + \
                [v for row in self.res_views for v in row]
