@@ -620,7 +620,11 @@ class MultiProcGeesePool:
         self._fifo.append((wid, slot))
         self.timing['fwd'] += time.time() - t0
 
-        while len(self._fifo) > 1:
+        # HANDYRL_ACTOR_SYNC=1: complete every service immediately (no
+        # overlap at all) — a race-diagnosis probe for
+        # tools/repro_transport_race.py
+        depth = 0 if os.environ.get('HANDYRL_ACTOR_SYNC') == '1' else 1
+        while len(self._fifo) > depth:
             self._complete(*self._fifo.pop(0))
         return frames
 
