@@ -18,7 +18,13 @@ torch.cuda.set_device(device)
 torch.manual_seed(0)
 trainer = Trainer(args, GeeseNet(), device=device, batcher=False)
 mpool.attach(trainer.model, device)
-replay = DeviceReplay(args, device, bytes_budget=2 << 30, ingest_thread=True)
+# HANDYRL_INGEST_THREAD=0: synchronous main-stream ingest (race probe)
+ingest = os.environ.get('HANDYRL_INGEST_THREAD', '1') == '1'
+replay = DeviceReplay(args, device, bytes_budget=2 << 30,
+                      ingest_thread=ingest)
+print('# slots=%d sync=%s ingest_thread=%s register=%s' % (
+    mpool.slots, os.environ.get('HANDYRL_ACTOR_SYNC', '0'), ingest,
+    getattr(mpool, '_use_registered', False)), flush=True)
 
 def pump(n):
     frames = 0
@@ -37,11 +43,15 @@ assert step.graph is not None
 
 N = int(sys.argv[1]) if len(sys.argv) > 1 else 400
 window_lens, t0 = [], time.time()
+import math
 for i in range(N):
     frames, lens = pump(16)
     window_lens += lens
     losses, dcnt = step.step()
     mpool.refresh_weights()
+    if (i + 1) % 10 == 0 and not math.isfinite(float(losses['p'])):
+        print('NONFINITE_LOSS at step %d' % (i + 1), flush=True)
+        break
     if (i + 1) % 50 == 0:
         torch.cuda.synchronize()
         ml = sum(window_lens) / max(1, len(window_lens))
