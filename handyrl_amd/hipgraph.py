@@ -230,6 +230,12 @@ class GraphedReplayTrainStep:
         self.replay.publish()          # admit finished background ingests
         pos0, start, length, seat, outcome, inv_total = \
             self.replay.sample_indices(self.batch_size)
+        # keep the host arrays referenced until the next fill: the async
+        # H2D below reads them after this function returns (and the stash
+        # doubles as diagnosis ground truth — see tools/learning_check.py)
+        self._last_host_idx = {
+            'pos0': pos0, 'start': start, 'length': length, 'seat': seat,
+            'outcome': outcome, 'inv_total': inv_total}
         for key, arr in (('pos0', pos0), ('start', start), ('length', length),
                          ('seat', seat), ('outcome', outcome),
                          ('inv_total', inv_total)):

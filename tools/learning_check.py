@@ -51,6 +51,31 @@ for i in range(N):
     mpool.refresh_weights()
     if (i + 1) % 10 == 0 and not math.isfinite(float(losses['p'])):
         print('NONFINITE_LOSS at step %d' % (i + 1), flush=True)
+        if os.environ.get('HANDYRL_VALIDATE') == '1':
+            torch.cuda.synchronize()
+            import numpy as np
+            host = step._last_host_idx
+            for k in ('pos0', 'start', 'length', 'seat'):
+                dev = step.idx[k].cpu().numpy()
+                same = np.array_equal(dev, host[k])
+                print('  idx[%s] device==host: %s  dev[min,max]=[%d,%d]'
+                      % (k, same, dev.min(), dev.max()), flush=True)
+            with torch.no_grad():
+                batch = replay.gather_batch(
+                    step.idx['pos0'], step.idx['start'], step.idx['length'],
+                    step.idx['seat'], step.idx['outcome'],
+                    step.idx['inv_total'])
+                tm = batch['turn_mask'].bool().squeeze(-1)
+                pr = batch['selected_prob'].squeeze(-1)[tm]
+                vv = batch['value'].squeeze(-1)[tm]
+                print('  eager regather: prob[min,max]=[%g,%g] n_nonpos=%d '
+                      'n_nonfinite=%d | value[min,max]=[%g,%g] nonfinite=%d'
+                      % (pr.min(), pr.max(), (pr <= 0).sum(),
+                         (~torch.isfinite(pr)).sum(), vv.min(), vv.max(),
+                         (~torch.isfinite(vv)).sum()), flush=True)
+                p = next(trainer.model.parameters())
+                print('  model params finite: %s'
+                      % bool(torch.isfinite(p).all()), flush=True)
         break
     if (i + 1) % 50 == 0:
         torch.cuda.synchronize()
