@@ -1,0 +1,100 @@
+"""Parity of the vectorized Geister engine against the single-game oracle
+(handyrl_amd.envs.geister.Environment): legal masks, observations, win
+detection and outcomes under shared random play."""
+
+import random
+
+import numpy as np
+
+from handyrl_amd.envs import geister as gz
+from handyrl_amd.envs.vec_geister import (GeisterVecEnv, N_ACTIONS,
+                                          N_MOVE_ACTIONS)
+
+
+def oracle_mask(env):
+    mask = np.full(N_ACTIONS, 1e32, dtype=np.float32)
+    mask[env.legal_actions()] = 0.0
+    return mask
+
+
+def oracle_obs(env):
+    obs = env.observation(env.turn())
+    return obs['scalar'], obs['board']
+
+
+def test_vec_matches_single_game_engine():
+    G = 16
+    rng = random.Random(11)
+    vec = GeisterVecEnv(G, seed=11)
+    vec.reset_games(np.arange(G))
+    envs = [gz.Environment() for _ in range(G)]
+    finished = 0
+
+    for step in range(450):
+        masks = vec.legal_masks()
+        scalar, board = vec.observations()
+        actions = np.zeros(G, dtype=np.int64)
+        for g, env in enumerate(envs):
+            legal = env.legal_actions()
+            vec_legal = np.nonzero(masks[g] == 0.0)[0]
+            assert sorted(legal) == vec_legal.tolist(), \
+                'legal mismatch g=%d step=%d' % (g, step)
+            o_scalar, o_board = oracle_obs(env)
+            np.testing.assert_array_equal(scalar[g], o_scalar)
+            np.testing.assert_array_equal(board[g], o_board)
+            assert int(vec.turn()[g]) == env.turn()
+            actions[g] = rng.choice(legal)
+            env.play(actions[g])
+        done = vec.step(actions)
+        for g, env in enumerate(envs):
+            assert bool(done[g]) == env.terminal(), \
+                'terminal mismatch g=%d step=%d' % (g, step)
+            if done[g]:
+                oc = vec.outcomes(np.array([g]))[0]
+                ref = env.outcome()
+                assert oc[0] == ref[0] and oc[1] == ref[1]
+                finished += 1
+                env.reset()
+                vec.reset_games(np.array([g]))
+    assert finished >= 4, 'too few finished games to trust the parity sweep'
+
+
+def test_vec_piece_invariants():
+    G = 8
+    rng = random.Random(3)
+    vec = GeisterVecEnv(G, seed=3)
+    vec.reset_games(np.arange(G))
+    for step in range(300):
+        masks = vec.legal_masks()
+        actions = np.array([rng.choice(np.nonzero(masks[g] == 0.0)[0])
+                            for g in range(G)], dtype=np.int64)
+        vec.step(actions)
+        done_idx = np.nonzero(vec.over)[0]
+        started = vec.turn_count >= 0
+        # counts match the board contents for every running game
+        for g in np.nonzero(started & ~vec.over)[0]:
+            b = vec.board[g]
+            for code in range(4):
+                assert int(vec.piece_cnt[g, code]) == int((b == code).sum())
+            # slot table round-trips
+            for cell in np.nonzero(b >= 0)[0]:
+                slot = int(vec.slot_of[g, cell])
+                assert int(vec.piece_pos[g, slot]) == cell
+        if len(done_idx):
+            vec.reset_games(done_idx)
+
+
+def test_layout_turns_then_moves():
+    vec = GeisterVecEnv(2, seed=0)
+    vec.reset_games(np.arange(2))
+    m = vec.legal_masks()
+    assert (m[:, :N_MOVE_ACTIONS] == 1e32).all()
+    assert (m[:, N_MOVE_ACTIONS:] == 0.0).all()
+    vec.step(np.array([N_MOVE_ACTIONS, N_MOVE_ACTIONS + 69]))
+    m = vec.legal_masks()
+    assert (m[:, :N_MOVE_ACTIONS] == 1e32).all()     # second layout turn
+    vec.step(np.array([N_MOVE_ACTIONS + 5, N_MOVE_ACTIONS + 17]))
+    m = vec.legal_masks()
+    assert (m[:, N_MOVE_ACTIONS:] == 1e32).all()
+    assert (m[:, :N_MOVE_ACTIONS] == 0.0).any(axis=1).all()
+    assert (vec.piece_cnt[:, :] == 4).all()
