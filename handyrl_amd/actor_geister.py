@@ -115,15 +115,104 @@ def _geister_env_worker(conn, shm_names, n_games, args, seed):
             frames_prev += 1
 
 
+def _geister_vec_worker(conn, shm_names, n_games, args, seed):
+    """Vectorized variant of _geister_env_worker: one GeisterVecEnv steps
+    the whole shard (legality, observations, moves, captures and win
+    detection as batched numpy ops — envs/vec_geister.py, parity-tested
+    against the single-game oracle); only the per-step moment recording
+    stays per-game python, in the reference episode format."""
+    from multiprocessing import shared_memory
+    from .batch import pack_moments
+    from .envs.vec_geister import GeisterVecEnv
+
+    shms = {k: shared_memory.SharedMemory(name=v) for k, v in shm_names.items()}
+    G = n_games
+    scalar_v = np.ndarray((G, SCALAR_DIM), dtype=np.float32, buffer=shms['scalar'].buf)
+    board_v = np.ndarray((G,) + BOARD_SHAPE, dtype=np.float32, buffer=shms['board'].buf)
+    mask_v = np.ndarray((G, N_ACTIONS), dtype=np.float32, buffer=shms['mask'].buf)
+    parity_v = np.ndarray((G,), dtype=np.int8, buffer=shms['parity'].buf)
+    reset_v = np.ndarray((G,), dtype=np.uint8, buffer=shms['reset'].buf)
+    res_v = np.ndarray((G, 4), dtype=np.float32, buffer=shms['res'].buf)
+
+    vec = GeisterVecEnv(G, seed=seed)
+    vec.reset_games(np.arange(G))
+    moments = [[] for _ in range(G)]
+    gamma = args.get('gamma', 0.8)
+    compress = args.get('compress_episodes', False)
+    compress_steps = args.get('compress_steps', 4)
+    job_args = {'player': [0, 1], 'model_id': {0: -1, 1: -1}}
+    reward = {0: -0.01, 1: -0.01}            # geister.py reward(): constant
+
+    def package(g, outcome):
+        ms = moments[g]
+        if not ms:
+            return None
+        for p in (0, 1):
+            ret = 0.0
+            for m in reversed(ms):
+                ret = (m['reward'][p] or 0) + gamma * ret
+                m['return'][p] = ret
+        return {'args': job_args, 'steps': len(ms), 'outcome': outcome,
+                'moment': pack_moments(ms, compress_steps, compress=compress)}
+
+    frames_prev = 0
+    eps_out = []
+    while True:
+        reset_v[:] = 0
+        done_idx = np.nonzero(vec.over)[0]
+        if len(done_idx):
+            ocs = vec.outcomes(done_idx)
+            for k, g in enumerate(done_idx):
+                ep = package(g, {0: float(ocs[k, 0]), 1: float(ocs[k, 1])})
+                if ep is not None:
+                    eps_out.append(ep)
+                moments[g] = []
+            vec.reset_games(done_idx)
+            reset_v[done_idx] = 1
+        scalar, board = vec.observations()
+        scalar_v[:] = scalar
+        board_v[:] = board
+        vec.legal_masks(out=mask_v)
+        parity_v[:] = vec.turn()
+
+        conn.send(('obs', G, frames_prev, eps_out))
+        frames_prev, eps_out = 0, []
+        cmd = conn.recv()
+        if cmd == 'quit':
+            break
+
+        actions = res_v[:, 0].astype(np.int64)
+        for g in range(G):
+            p = int(parity_v[g])
+            moment = {key: {0: None, 1: None} for key in
+                      ('observation', 'selected_prob', 'action_mask', 'action',
+                       'value', 'reward', 'return')}
+            moment['observation'][p] = {'scalar': scalar_v[g].copy(),
+                                        'board': board_v[g].copy()}
+            moment['selected_prob'][p] = float(res_v[g, 1])
+            moment['action_mask'][p] = mask_v[g].copy()
+            moment['action'][p] = int(actions[g])
+            moment['value'][p] = np.array([res_v[g, 2]], dtype=np.float32)
+            moment['turn'] = [p]
+            moment['reward'] = dict(reward)
+            moments[g].append(moment)
+        vec.step(actions)
+        frames_prev += G
+
+
 class GeisterMultiProcPool:
     """256-actor-style Geister self-play on one GPU: W env-worker processes
     + batched recurrent inference with GPU-resident DRC hidden state."""
 
-    def __init__(self, args, n_games=256, seed=0, workers=8):
+    def __init__(self, args, n_games=256, seed=0, workers=8, vec=None):
         import multiprocessing as mp
+        import os
         from multiprocessing import shared_memory
         self.args = args
         self.workers = workers
+        if vec is None:
+            vec = os.environ.get('HANDYRL_GEISTER_VEC', '1') == '1'
+        worker_fn = _geister_vec_worker if vec else _geister_env_worker
         per = max(1, n_games // workers)
         self.n_per = per
         self.conns, self.procs, self.shms = [], [], []
@@ -157,7 +246,7 @@ class GeisterMultiProcPool:
             self.views.append(views)
             parent_conn, child_conn = mp.Pipe(duplex=True)
             proc = mp.Process(
-                target=_geister_env_worker,
+                target=worker_fn,
                 args=(child_conn, {k: s.name for k, s in shm.items()},
                       per, args, seed + 131 * w),
                 daemon=True)

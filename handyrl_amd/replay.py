@@ -62,6 +62,12 @@ class DeviceReplay:
         self._chmap = torch.from_numpy(CHMAP).to(dev)     # (4, 17)
         self._arange_cache = {}
         self.head = 0                  # monotonically increasing write cursor
+        self.head_planned = 0          # head + queued-but-unwritten rows
+        # oldest logical row any in-flight gather may read; None until the
+        # first sample (pre-training fill), only ever rises (stale reads by
+        # the ingest thread are conservative)
+        self._reader_floor = None
+        self._floor_prev = None
         self.table = deque()           # (pos0, steps, outcome(np[4]))
         self.total_added = 0
         self._pin = {}
@@ -76,11 +82,22 @@ class DeviceReplay:
             self._ingest.start()
 
     def _ingest_loop(self):
+        import time
         torch.cuda.set_device(self.device)
         while True:
             episodes = self._ingest_q.get()
             try:
                 for chunk in self._chunk(episodes):
+                    # back-pressure: never write ring positions that alias
+                    # rows at or above the consumer's reader floor (covers
+                    # entries an in-flight captured gather may still read)
+                    n_rows = sum(int(ep['steps']) for ep in chunk)
+                    while True:
+                        floor = self._reader_floor
+                        if floor is None or \
+                                self.head + n_rows - self.ring_T <= floor:
+                            break
+                        time.sleep(0.001)
                     with torch.cuda.stream(self._ingest_stream):
                         entries, n_rows = self._copy_block(chunk)
                         event = torch.cuda.Event()
@@ -139,20 +156,33 @@ class DeviceReplay:
     # -- write path --------------------------------------------------------
     def _stage(self, key, parts, n, dtype, tail_shape):
         """Fill a pinned staging buffer from per-episode arrays (one memcpy
-        each, GIL released) and return its device copy (async H2D)."""
+        each, GIL released) and return its device copy (async H2D).
+
+        The previous block's H2D from this buffer may still be in flight
+        (the ingest stream's copies queue behind the actor pool's DMA
+        traffic under load), so refilling must wait on the event recorded
+        after that copy — without it the in-flight DMA reads torn bytes
+        and poisons the ring with garbage floats (NaN losses; diagnosed
+        in gpurun_out/learn_race*.log)."""
         if self.device.type != 'cuda':
             return torch.from_numpy(np.concatenate(parts)).to(self.device)
-        pin = self._pin.get(key)
+        pin, ev = self._pin.get(key, (None, None))
         if pin is None or pin.shape[0] < n:
             cap = max(int(n * 1.5), 1024)
             pin = torch.empty((cap,) + tail_shape, dtype=dtype, pin_memory=True)
-            self._pin[key] = pin
+            ev = None
+        if ev is not None:
+            ev.synchronize()
         view = pin.numpy()
         off = 0
         for part in parts:
             view[off:off + part.shape[0]] = part
             off += part.shape[0]
-        return pin[:n].to(self.device, non_blocking=True)
+        dev = pin[:n].to(self.device, non_blocking=True)
+        ev = torch.cuda.Event()
+        ev.record()                    # ingest stream inside the ingest loop
+        self._pin[key] = (pin, ev)
+        return dev
 
     def _write_ring(self, dst, src):
         n = src.shape[0]
@@ -187,8 +217,16 @@ class DeviceReplay:
         return entries, n
 
     def _evict(self):
-        margin = EVICT_MARGIN if self._ingest is not None else 0
-        min_valid = self.head - self.ring_T + margin
+        # With background ingest, evict against the PLANNED write frontier
+        # (every queued-but-unwritten row counted): entries stay at least
+        # EVICT_MARGIN rows clear of any ring position an in-flight or
+        # queued block can touch, so concurrent ingest writes never alias
+        # rows a captured gather may still read.  self.head alone lags the
+        # queue and is advanced by the ingest thread (racy to read here).
+        if self._ingest is not None:
+            min_valid = self.head_planned - self.ring_T + EVICT_MARGIN
+        else:
+            min_valid = self.head - self.ring_T
         max_eps = self.args['maximum_episodes']
         while self.table and (self.table[0][0] < min_valid or
                               len(self.table) > max_eps):
@@ -198,8 +236,12 @@ class DeviceReplay:
         """Append columnar episodes (numpy fields) to the device ring."""
         if not episodes:
             return
+        limit = max(512, self.ring_T // 4)
+        self.head_planned += sum(int(ep['steps']) for ep in episodes
+                                 if int(ep['steps']) <= limit)
         if self._ingest is not None:
             self._ingest_q.put(episodes)
+            self._evict()              # clear the planned frontier's alias zone
             return
         for chunk in self._chunk(episodes):
             entries, _ = self._copy_block(chunk)
@@ -242,6 +284,12 @@ class DeviceReplay:
             seat[b] = random.randrange(4)
             outcome[b] = oc
             inv_total[b] = 1.0 / steps
+        # reader floor covers this fill AND the previous one (the previous
+        # graph replay may still be in flight when this one is sampled)
+        floor_now = self.table[0][0]
+        self._reader_floor = floor_now if self._floor_prev is None \
+            else min(floor_now, self._floor_prev)
+        self._floor_prev = floor_now
         return pos0, start, length, seat, outcome, inv_total
 
     def gather_batch(self, pos0, start, length, seat, outcome, inv_total):
