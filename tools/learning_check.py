@@ -26,12 +26,34 @@ print('# slots=%d sync=%s ingest_thread=%s register=%s' % (
     mpool.slots, os.environ.get('HANDYRL_ACTOR_SYNC', '0'), ingest,
     getattr(mpool, '_use_registered', False)), flush=True)
 
+VALIDATE = os.environ.get('HANDYRL_VALIDATE') == '1'
+min_prob_seen = [1.0]
+bad_eps = [0]
+
+
+def _check_eps(eps):
+    """Cheap host-side validation of harvested episodes (pre-ingest)."""
+    import numpy as np
+    for ep in eps:
+        alive = ep['alive']
+        pr, v = ep['prob'][alive], ep['value'][alive]
+        if pr.size:
+            m = float(pr.min())
+            if m < min_prob_seen[0]:
+                min_prob_seen[0] = m
+        if (not np.isfinite(pr).all() or (pr.size and pr.min() <= 0)
+                or not np.isfinite(v).all()):
+            bad_eps[0] += 1
+
+
 def pump(n):
     frames = 0
     for _ in range(n * mpool.calls_per_vec_step):
         frames += mpool.step_once()
     eps = mpool.harvest()
     lens = [ep['steps'] for ep in eps]
+    if VALIDATE:
+        _check_eps(eps)
     replay.extend(eps)
     return frames, lens
 
@@ -49,6 +71,11 @@ for i in range(N):
     window_lens += lens
     losses, dcnt = step.step()
     mpool.refresh_weights()
+    if VALIDATE and (i + 1) % 10 == 0:
+        p = next(trainer.model.parameters())
+        pf = bool(torch.isfinite(p).all())
+        print('  probe step %d: params_finite=%s bad_eps=%d min_prob=%.3e'
+              % (i + 1, pf, bad_eps[0], min_prob_seen[0]), flush=True)
     if (i + 1) % 10 == 0 and not math.isfinite(float(losses['p'])):
         print('NONFINITE_LOSS at step %d' % (i + 1), flush=True)
         if os.environ.get('HANDYRL_VALIDATE') == '1':
