@@ -71,11 +71,17 @@ for i in range(N):
     window_lens += lens
     losses, dcnt = step.step()
     mpool.refresh_weights()
-    if VALIDATE and (i + 1) % 10 == 0:
+    if VALIDATE:
         p = next(trainer.model.parameters())
         pf = bool(torch.isfinite(p).all())
-        print('  probe step %d: params_finite=%s bad_eps=%d min_prob=%.3e'
-              % (i + 1, pf, bad_eps[0], min_prob_seen[0]), flush=True)
+        lf = math.isfinite(float(losses['p'])) and \
+            math.isfinite(float(losses['v']))
+        if not pf or not lf or bad_eps[0]:
+            print('  probe step %d: params_finite=%s loss_finite=%s '
+                  'bad_eps=%d min_prob=%.3e'
+                  % (i + 1, pf, lf, bad_eps[0], min_prob_seen[0]), flush=True)
+            if bad_eps[0] > 500 or ((not pf) and i > 25):
+                break
     if (i + 1) % 10 == 0 and not math.isfinite(float(losses['p'])):
         print('NONFINITE_LOSS at step %d' % (i + 1), flush=True)
         if os.environ.get('HANDYRL_VALIDATE') == '1':
