@@ -66,6 +66,14 @@ class GraphedTrainStep:
                                               None, tr.args)
         tr.optimizer.zero_grad(set_to_none=False)
         losses['total'].backward()
+        if tr.grad_guard:
+            # capturable finite-guard: one spike step (inf grads with a
+            # still-finite loss) otherwise turns the clip scale into NaN
+            # and poisons the weights permanently (BASELINE.md diagnosis)
+            for p in tr.params:
+                if p.grad is not None:
+                    torch.nan_to_num_(p.grad, nan=0.0, posinf=1e6,
+                                      neginf=-1e6)
         tr.reducer.allreduce_()
         nn.utils.clip_grad_norm_(tr.params, 4.0)
         tr.optimizer.step()
@@ -255,6 +263,14 @@ class GraphedReplayTrainStep:
                                               None, tr.args)
         tr.optimizer.zero_grad(set_to_none=False)
         losses['total'].backward()
+        if tr.grad_guard:
+            # capturable finite-guard: one spike step (inf grads with a
+            # still-finite loss) otherwise turns the clip scale into NaN
+            # and poisons the weights permanently (BASELINE.md diagnosis)
+            for p in tr.params:
+                if p.grad is not None:
+                    torch.nan_to_num_(p.grad, nan=0.0, posinf=1e6,
+                                      neginf=-1e6)
         tr.reducer.allreduce_()
         nn.utils.clip_grad_norm_(tr.params, 4.0)
         tr.optimizer.step()

@@ -206,6 +206,13 @@ class Trainer:
             self.model = self.model.to(memory_format=torch.channels_last)
         self.params = list(self.model.parameters())
         self.reducer = hdist.GradReducer(self.params)
+        # Finite-guard the gradients before clipping: a single spike step
+        # (observed under policy-saturation trajectories: loss still finite,
+        # grads inf -> clip_grad_norm scales inf*0 -> NaN weights, after
+        # which the whole self-play loop is poisoned — diagnosis in
+        # BASELINE.md "learning sanity") becomes a survivable clipped
+        # update instead of permanent NaN.  HANDYRL_GRAD_GUARD=0 disables.
+        self.grad_guard = os.environ.get('HANDYRL_GRAD_GUARD', '1') == '1'
 
         self.default_lr = 3e-8
         self.data_cnt_ema = args['batch_size'] * args['forward_steps']
@@ -282,6 +289,11 @@ class Trainer:
 
         self.optimizer.zero_grad(set_to_none=False)
         losses['total'].backward()
+        if self.grad_guard:
+            for p in self.params:
+                if p.grad is not None:
+                    torch.nan_to_num_(p.grad, nan=0.0, posinf=1e6,
+                                      neginf=-1e6)
         self.reducer.allreduce_()                 # fused RCCL all-reduce (DP)
         nn.utils.clip_grad_norm_(self.params, 4.0)
         self.optimizer.step()
