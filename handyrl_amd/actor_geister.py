@@ -134,16 +134,51 @@ def _geister_vec_worker(conn, shm_names, n_games, args, seed):
     reset_v = np.ndarray((G,), dtype=np.uint8, buffer=shms['reset'].buf)
     res_v = np.ndarray((G, 4), dtype=np.float32, buffer=shms['res'].buf)
 
+    import os
     vec = GeisterVecEnv(G, seed=seed)
     vec.reset_games(np.arange(G))
-    moments = [[] for _ in range(G)]
     gamma = args.get('gamma', 0.8)
     compress = args.get('compress_episodes', False)
     compress_steps = args.get('compress_steps', 4)
     job_args = {'player': [0, 1], 'model_id': {0: -1, 1: -1}}
     reward = {0: -0.01, 1: -0.01}            # geister.py reward(): constant
+    # columnar episodes (plain arrays consumed by make_batch's turn-based
+    # columnar fast path; parity-tested against the moment-dict format in
+    # tests/test_columnar_turn.py) skip all per-step dict building —
+    # HANDYRL_GEISTER_COLUMNAR=0 restores reference-format moments
+    columnar = os.environ.get('HANDYRL_GEISTER_COLUMNAR', '1') == '1'
+    moments = [[] for _ in range(G)]
+    CAP = 202                                # 2 layout turns + 200 moves
+    gar = np.arange(G)
+    if columnar:
+        rec = {
+            'scalar': np.zeros((G, CAP, SCALAR_DIM), np.uint8),
+            'board': np.zeros((G, CAP) + BOARD_SHAPE, np.uint8),
+            'mask': np.zeros((G, CAP, N_ACTIONS), bool),
+            'turn': np.zeros((G, CAP), np.int8),
+            'action': np.zeros((G, CAP), np.int16),
+            'prob': np.zeros((G, CAP), np.float32),
+            'value': np.zeros((G, CAP), np.float32),
+        }
+        rec_len = np.zeros(G, np.int32)
 
     def package(g, outcome):
+        if columnar:
+            S = int(rec_len[g])
+            if S == 0:
+                return None
+            acc, rets = 0.0, np.empty(S, np.float32)
+            for t in range(S - 1, -1, -1):   # constant -0.01 both players
+                acc = -0.01 + gamma * acc
+                rets[t] = acc
+            ep = {'args': job_args, 'steps': S, 'outcome': outcome,
+                  'columnar': True, 'turn_based': True,
+                  'n_actions': N_ACTIONS, 'n_players': 2,
+                  'reward': np.full((S, 2), -0.01, np.float32),
+                  'return': np.stack([rets, rets], axis=1)}
+            for k, buf in rec.items():
+                ep[k] = buf[g, :S].copy()
+            return ep
         ms = moments[g]
         if not ms:
             return None
@@ -166,7 +201,10 @@ def _geister_vec_worker(conn, shm_names, n_games, args, seed):
                 ep = package(g, {0: float(ocs[k, 0]), 1: float(ocs[k, 1])})
                 if ep is not None:
                     eps_out.append(ep)
-                moments[g] = []
+                if columnar:
+                    rec_len[g] = 0
+                else:
+                    moments[g] = []
             vec.reset_games(done_idx)
             reset_v[done_idx] = 1
         scalar, board = vec.observations()
@@ -182,20 +220,31 @@ def _geister_vec_worker(conn, shm_names, n_games, args, seed):
             break
 
         actions = res_v[:, 0].astype(np.int64)
-        for g in range(G):
-            p = int(parity_v[g])
-            moment = {key: {0: None, 1: None} for key in
-                      ('observation', 'selected_prob', 'action_mask', 'action',
-                       'value', 'reward', 'return')}
-            moment['observation'][p] = {'scalar': scalar_v[g].copy(),
-                                        'board': board_v[g].copy()}
-            moment['selected_prob'][p] = float(res_v[g, 1])
-            moment['action_mask'][p] = mask_v[g].copy()
-            moment['action'][p] = int(actions[g])
-            moment['value'][p] = np.array([res_v[g, 2]], dtype=np.float32)
-            moment['turn'] = [p]
-            moment['reward'] = dict(reward)
-            moments[g].append(moment)
+        if columnar:
+            rows = np.minimum(rec_len, CAP - 1)
+            rec['scalar'][gar, rows] = scalar_v.astype(np.uint8)
+            rec['board'][gar, rows] = board_v.astype(np.uint8)
+            rec['mask'][gar, rows] = mask_v == 0.0
+            rec['turn'][gar, rows] = parity_v
+            rec['action'][gar, rows] = actions.astype(np.int16)
+            rec['prob'][gar, rows] = res_v[:, 1]
+            rec['value'][gar, rows] = res_v[:, 2]
+            rec_len += 1
+        else:
+            for g in range(G):
+                p = int(parity_v[g])
+                moment = {key: {0: None, 1: None} for key in
+                          ('observation', 'selected_prob', 'action_mask',
+                           'action', 'value', 'reward', 'return')}
+                moment['observation'][p] = {'scalar': scalar_v[g].copy(),
+                                            'board': board_v[g].copy()}
+                moment['selected_prob'][p] = float(res_v[g, 1])
+                moment['action_mask'][p] = mask_v[g].copy()
+                moment['action'][p] = int(actions[g])
+                moment['value'][p] = np.array([res_v[g, 2]], dtype=np.float32)
+                moment['turn'] = [p]
+                moment['reward'] = dict(reward)
+                moments[g].append(moment)
         vec.step(actions)
         frames_prev += G
 

@@ -99,6 +99,50 @@ def _episode_fields_columnar(ep, args):
             amask, progress)
 
 
+def _episode_fields_columnar_turn(ep, args):
+    """Per-episode batch fields from a COLUMNAR TURN-BASED episode window
+    (observation=False configs, e.g. Geister): one trained seat per step
+    (the mover), value/reward/return rows for every player.
+
+    Shapes and dtypes are bit-identical to the moment-dict path (see
+    tests/test_columnar_turn.py): obs/prob/action/action_mask carry the
+    mover axis (T, 1, ...), value-side fields carry the player axis
+    (T, P, ...)."""
+    assert args['turn_based_training'] and not args['observation'], \
+        'turn-based columnar episodes need turn_based_training w/o observation'
+    T = ep['turn'].shape[0]
+    P = int(ep.get('n_players', 2))
+    A = int(ep['n_actions'])
+    turn = ep['turn'].astype(np.int64)
+    tr = np.arange(T)
+
+    obs = {'scalar': ep['scalar'].astype(np.float32)[:, None, :],
+           'board': ep['board'].astype(np.float32)[:, None]}
+    prob = ep['prob'].astype(np.float64)[:, None, None]
+    act = ep['action'].astype(np.int64)[:, None, None]
+    if ep['mask'].dtype == np.bool_:          # legality bools -> additive mask
+        amask = np.where(ep['mask'], np.float32(0), np.float32(1e32))[:, None]
+    else:
+        amask = ep['mask'].astype(np.float32)[:, None]
+    v = np.zeros((T, P, 1), dtype=np.float32)
+    v[tr, turn, 0] = ep['value']
+    rew = ep['reward'].astype(np.float32).reshape(T, P, 1)
+    ret = ep['return'].astype(np.float32).reshape(T, P, 1)
+    oc = np.array([ep['outcome'][p] for p in range(P)],
+                  dtype=np.float32).reshape(1, P, -1)
+    emask = np.ones((T, 1, 1), dtype=np.float32)
+    tmask = np.zeros((T, P, 1), dtype=np.float32)
+    tmask[tr, turn, 0] = 1.0
+    omask = tmask.copy()
+    progress = np.arange(ep['start'], ep['end'],
+                         dtype=np.float32)[..., np.newaxis] / ep['total']
+    obs_zeros = {'scalar': np.zeros(ep['scalar'].shape[-1], dtype=np.float32),
+                 'board': np.zeros(ep['board'].shape[2:], dtype=np.float32)}
+    del A
+    return (obs, obs_zeros, prob, v, act, oc, rew, ret, emask, tmask, omask,
+            amask, progress)
+
+
 def make_batch(episodes, args):
     """Assemble a (B, T, P, ...) training batch from sampled episode windows."""
     obss, datum = [], []
@@ -108,8 +152,10 @@ def make_batch(episodes, args):
 
     for ep in episodes:
         if ep.get('columnar'):
+            fields_fn = _episode_fields_columnar_turn if ep.get('turn_based') \
+                else _episode_fields_columnar
             (obs, obs_zeros, prob, v, act, oc, rew, ret, emask, tmask, omask,
-             amask, progress) = _episode_fields_columnar(ep, args)
+             amask, progress) = fields_fn(ep, args)
             T = emask.shape[0]
             batch_steps = args['burn_in_steps'] + args['forward_steps']
             if T < batch_steps:
@@ -118,7 +164,7 @@ def make_batch(episodes, args):
                 pad3 = lambda a, val=0: np.pad(
                     a, [(pad_b, pad_a)] + [(0, 0)] * (a.ndim - 1),
                     'constant', constant_values=val)
-                obs = pad3(obs)
+                obs = map_r(obs, pad3)      # plain array or {scalar, board}
                 prob = pad3(prob, val=1)
                 v = np.concatenate(
                     [np.pad(v, [(pad_b, 0), (0, 0), (0, 0)], 'constant'),
@@ -274,16 +320,23 @@ class EpisodeBuffer:
         ed = min(train_st + args['forward_steps'], ep['steps'])
         if ep.get('columnar'):
             # columnar episodes: the window is an array view, no block math
-            return {
+            if ep.get('turn_based'):
+                step_keys = ('scalar', 'board', 'mask', 'turn', 'action',
+                             'prob', 'value', 'reward', 'return')
+            else:
+                step_keys = ('obs', 'alive', 'action', 'prob', 'value')
+            out = {
                 'args': ep['args'], 'outcome': ep['outcome'], 'columnar': True,
+                'turn_based': ep.get('turn_based', False),
                 'canonical_obs': ep.get('canonical_obs', False),
                 'n_actions': ep['n_actions'],
-                'obs': ep['obs'][st:ed], 'alive': ep['alive'][st:ed],
-                'action': ep['action'][st:ed], 'prob': ep['prob'][st:ed],
-                'value': ep['value'][st:ed],
+                'n_players': ep.get('n_players', 4),
                 'start': st, 'end': ed, 'train_start': train_st,
                 'total': ep['steps'],
             }
+            for k in step_keys:
+                out[k] = ep[k][st:ed]
+            return out
         st_block = st // args['compress_steps']
         ed_block = (ed - 1) // args['compress_steps'] + 1
         return {

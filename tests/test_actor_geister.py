@@ -1,6 +1,7 @@
 """Batched Geister actor pool: valid reference-format episodes from the
 multi-process recurrent path, trainable through the RNN learner."""
 
+import numpy as np
 import torch
 
 from handyrl_amd.actor_geister import GeisterMultiProcPool
@@ -23,20 +24,59 @@ def _args(**over):
     return args
 
 
-def test_geister_pool_generates_and_trains():
+def _pump_episodes(pool, env, n=3):
+    model = env.net()
+    model.eval()
+    pool.attach(model, torch.device('cpu'))
+    for _ in range(4000):
+        pool.step_once()
+        if pool.episodes_done >= n:
+            break
+    eps = pool.harvest()
+    assert len(eps) >= n
+    return eps
+
+
+def _train_on(eps, env, args):
+    buf = EpisodeBuffer(args)
+    buf.extend(eps)
+    trainer = Trainer(args, env.net(), device=torch.device('cpu'))
+    batch = make_batch([buf.select_episode() for _ in range(4)], args)
+    losses, dcnt = trainer.train_step(batch)
+    assert torch.isfinite(losses['total'])
+    assert dcnt > 0
+
+
+def test_geister_pool_generates_and_trains_columnar():
+    """Default path: vec engine + columnar turn-based episodes."""
     args = _args()
     pool = GeisterMultiProcPool(args, n_games=8, seed=3, workers=2)
     try:
         env = GeisterEnv()
-        model = env.net()
-        model.eval()
-        pool.attach(model, torch.device('cpu'))
-        for _ in range(4000):
-            pool.step_once()
-            if pool.episodes_done >= 3:
-                break
-        eps = pool.harvest()
-        assert len(eps) >= 3
+        eps = _pump_episodes(pool, env)
+        for ep in eps[:2]:
+            assert ep['columnar'] and ep['turn_based']
+            assert ep['steps'] >= 3
+            assert set(ep['outcome'].keys()) == {0, 1}
+            S = ep['steps']
+            assert ep['board'].shape == (S, 7, 6, 6)
+            assert ep['action'][0] >= 144         # layout turn comes first
+            assert ep['turn'][0] == 0 and ep['turn'][1] == 1
+            assert (ep['reward'] == np.float32(-0.01)).all()
+            assert ep['return'].shape == (S, 2)
+        _train_on(eps, env, args)
+    finally:
+        pool.shutdown()
+
+
+def test_geister_pool_generates_and_trains_moment_dicts(monkeypatch):
+    """Reference moment-dict format (HANDYRL_GEISTER_COLUMNAR=0)."""
+    monkeypatch.setenv('HANDYRL_GEISTER_COLUMNAR', '0')
+    args = _args()
+    pool = GeisterMultiProcPool(args, n_games=8, seed=3, workers=2)
+    try:
+        env = GeisterEnv()
+        eps = _pump_episodes(pool, env)
         for ep in eps[:2]:
             assert ep['steps'] >= 3
             assert set(ep['outcome'].keys()) == {0, 1}
@@ -50,13 +90,6 @@ def test_geister_pool_generates_and_trains():
             assert m0['reward'][0] == -0.01
             # returns backfilled for both seats
             assert moments[-1]['return'][0] is not None
-
-        buf = EpisodeBuffer(args)
-        buf.extend(eps)
-        trainer = Trainer(args, env.net(), device=torch.device('cpu'))
-        batch = make_batch([buf.select_episode() for _ in range(4)], args)
-        losses, dcnt = trainer.train_step(batch)
-        assert torch.isfinite(losses['total'])
-        assert dcnt > 0
+        _train_on(eps, env, args)
     finally:
         pool.shutdown()
