@@ -141,6 +141,18 @@ def test_columnar_turn_padding_matches_dict_path():
     _assert_batches_equal(bd, bc)
 
 
+def test_columnar_turn_burn_in_matches_dict_path():
+    """Recurrent configs: a burn-in prefix precedes the trained window."""
+    args = _args(forward_steps=6, burn_in_steps=4)
+    ep_dict, ep_col, S = _record_trajectory(steps=20)
+    for st, ed, train_st in ((6, 16, 10),     # full burn-in prefix
+                             (0, 8, 2),       # clipped prefix at episode start
+                             (14, 20, 18)):   # short tail -> padding too
+        bd = make_batch([_window_dict(ep_dict, st, ed, train_st, args)], args)
+        bc = make_batch([_window_col(ep_col, st, ed, train_st, args)], args)
+        _assert_batches_equal(bd, bc)
+
+
 def test_columnar_turn_trains():
     from handyrl_amd.train import Trainer
     from handyrl_amd.envs.geister import Environment
