@@ -29,7 +29,13 @@ class GeeseActorPool:
     """Self-play actor pool for Hungry Geese on one GPU."""
 
     def __init__(self, model, args, n_games=256, device=None, seed=0,
-                 use_graphs=True, engine=None):
+                 use_graphs=True, engine=None, block_episodes=False):
+        # block_episodes: package ALL games finishing in a step as ONE
+        # concatenated columnar block (env-worker processes; the parent's
+        # drain thread splits it back into zero-copy per-episode views via
+        # split_episode_block) — replaces per-episode dict building and
+        # slice copies on the worker hot loop
+        self.block_episodes = block_episodes
         self.args = args
         self.device = device if device is not None else (
             torch.device('cuda') if torch.cuda.is_available() else torch.device('cpu'))
@@ -241,9 +247,13 @@ class GeeseActorPool:
         finished = np.nonzero(done)[0]
         if len(finished):
             outcomes = vec.outcomes(finished)
-            for k, g in enumerate(finished):
-                self.completed.append(self._package(g, outcomes[k]))
-                self.rec_len[g] = 0
+            if self.block_episodes:
+                self.completed.append(self._package_block(finished, outcomes))
+                self.rec_len[finished] = 0
+            else:
+                for k, g in enumerate(finished):
+                    self.completed.append(self._package(g, outcomes[k]))
+                    self.rec_len[g] = 0
             self.episodes_done += len(finished)
             vec.reset_games(finished)
         tm['package'] += time.time() - t0
@@ -267,6 +277,22 @@ class GeeseActorPool:
             'action': self.rec_act[g, :S].copy(),
             'prob': self.rec_prob[g, :S].copy(),
             'value': self.rec_val[g, :S].copy(),
+        }
+
+    def _package_block(self, finished, outcome_rows):
+        """One concatenated columnar block for every game finishing this
+        step (worker hot path: 5 array concats + one dict instead of
+        per-episode copies and dicts)."""
+        lens = self.rec_len[finished].astype(np.int32)
+        pairs = [(int(g), int(s)) for g, s in zip(finished, lens)]
+        cat = lambda buf: np.concatenate([buf[g, :s] for g, s in pairs])
+        return {
+            'block': True, 'columnar': True, 'canonical_obs': True,
+            'n_actions': 4, 'lens': lens,
+            'outcome': outcome_rows.astype(np.float32),
+            'obs': cat(self.rec_obs), 'alive': cat(self.rec_alive),
+            'action': cat(self.rec_act), 'prob': cat(self.rec_prob),
+            'value': cat(self.rec_val),
         }
 
     def refresh_weights(self):
@@ -341,6 +367,33 @@ class PipelinedGeesePool:
         return merged
 
 
+def split_episode_block(item):
+    """Expand a concatenated episode block (GeeseActorPool._package_block)
+    into per-episode dicts whose arrays are VIEWS into the block — the
+    standard columnar episode format, no copies.  Non-block items pass
+    through unchanged."""
+    if not isinstance(item, dict) or not item.get('block'):
+        return [item]
+    job_args = {'player': list(range(N_PLAYERS)),
+                'model_id': {p: -1 for p in range(N_PLAYERS)}}
+    eps, off = [], 0
+    oc = item['outcome']
+    for i, S in enumerate(item['lens']):
+        S = int(S)
+        sl = slice(off, off + S)
+        off += S
+        eps.append({
+            'args': job_args, 'steps': S,
+            'outcome': {p: float(oc[i, p]) for p in range(N_PLAYERS)},
+            'columnar': True, 'canonical_obs': item['canonical_obs'],
+            'n_actions': item['n_actions'],
+            'obs': item['obs'][sl], 'alive': item['alive'][sl],
+            'action': item['action'][sl], 'prob': item['prob'][sl],
+            'value': item['value'][sl],
+        })
+    return eps
+
+
 def _geese_env_worker(conn, ep_conn, obs_name, res_name, n_games, args, seed,
                       slots=2):
     """Env-side child process: vectorized stepping, columnar recording and
@@ -364,9 +417,11 @@ def _geese_env_worker(conn, ep_conn, obs_name, res_name, n_games, args, seed,
         res_views.append(np.ndarray((per * N_PLAYERS, 3), dtype=np.float32,
                                     buffer=res_shm.buf,
                                     offset=s * per * N_PLAYERS * 12))
-        pools.append(GeeseActorPool(None, args, n_games=per,
-                                    device=torch.device('cpu'),
-                                    use_graphs=False, seed=seed + 131 * s))
+        pools.append(GeeseActorPool(
+            None, args, n_games=per, device=torch.device('cpu'),
+            use_graphs=False, seed=seed + 131 * s,
+            block_episodes=os.environ.get('HANDYRL_BLOCK_EPISODES',
+                                          '1') == '1'))
 
     m_inflight = [0] * slots
     for s in range(slots):                    # prime the pipeline
@@ -539,9 +594,15 @@ class MultiProcGeesePool:
                 return
             for conn in ready:
                 try:
-                    eps = conn.recv()
+                    payload = conn.recv()
                 except (EOFError, OSError):
                     return
+                # workers ship concatenated blocks; split them into
+                # zero-copy per-episode views OFF the service path (this
+                # background thread), keeping the downstream format
+                eps = []
+                for item in payload:
+                    eps.extend(split_episode_block(item))
                 with self._completed_lock:
                     self.completed.extend(eps)
                     self.episodes_done += len(eps)
