@@ -83,11 +83,11 @@ class GeeseVecEnv:
         """Reset the given game indices to fresh initial states."""
         if len(games) == 0:
             return
-        # distinct random cells per game: first K columns of an argsort of
-        # iid uniforms = a uniform sample without replacement
+        # distinct random cells per game: the K smallest of iid uniforms
+        # (argpartition) = a uniform sample without replacement
         K = N_PLAYERS + MIN_FOOD
-        cells = np.argsort(self.rng.random((len(games), N_CELLS)),
-                           axis=1)[:, :K].astype(np.int32)
+        cells = np.argpartition(self.rng.random((len(games), N_CELLS)),
+                                K, axis=1)[:, :K].astype(np.int32)
         self.body[games] = -1
         self.body[games, :, 0] = cells[:, :N_PLAYERS]
         self.food[games] = cells[:, N_PLAYERS:]
@@ -180,20 +180,27 @@ class GeeseVecEnv:
             self._kill(gi[crash], pi[crash])
         live = self.alive & ~self.over[:, None]
 
-        # 5) food replenishment onto random free cells
+        # 5) food replenishment onto random free cells (vectorized over the
+        # needy games: argmax of iid uniforms over the free set = a uniform
+        # free-cell draw, matching the oracle's shuffle-and-take semantics)
         need_mask = (self.food < 0) & ~self.over[:, None]
-        for g in np.nonzero(need_mask.any(axis=1))[0]:
-            occ = self.body_grid[g].any(axis=0)
+        ng = np.nonzero(need_mask.any(axis=1))[0]
+        if len(ng):
+            occ = self.body_grid[ng].any(axis=1)            # (K, cells)
             for f in range(MIN_FOOD):
-                if self.food[g, f] >= 0:
-                    occ[self.food[g, f]] = True
-            free = np.nonzero(~occ)[0]
-            self.rng.shuffle(free)
-            k = 0
+                have = self.food[ng, f] >= 0
+                occ[np.nonzero(have)[0], self.food[ng[have], f]] = True
             for f in range(MIN_FOOD):
-                if self.food[g, f] < 0 and k < len(free):
-                    self.food[g, f] = free[k]
-                    k += 1
+                miss = self.food[ng, f] < 0
+                if not miss.any():
+                    continue
+                idx = np.nonzero(miss)[0]
+                r = self.rng.random((len(idx), N_CELLS))
+                r[occ[idx]] = -1.0
+                cell = r.argmax(axis=1)
+                ok = r[np.arange(len(idx)), cell] >= 0      # any free cell
+                self.food[ng[idx[ok]], f] = cell[ok].astype(np.int32)
+                occ[idx[ok], cell[ok]] = True
 
         stepped = ~self.over
         self.step_count[stepped] += 1
