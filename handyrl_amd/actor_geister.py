@@ -249,6 +249,222 @@ def _geister_vec_worker(conn, shm_names, n_games, args, seed):
         frames_prev += G
 
 
+class BatchedDRCEngine:
+    """Batched recurrent (DRC Conv-LSTM) inference for one shard of games:
+    per-(game, player) hidden state RESIDENT on the device, gathered by
+    turn parity per step, scattered back after, zeroed on game reset.
+    hipGraph-captured on GPU (in-place hidden updates keep the storages
+    fixed under capture); eager on CPU / when capture fails."""
+
+    def __init__(self, model, device, n_games, use_graphs=True):
+        self.model = model
+        self.device = device
+        self.n_per = n_games
+        hs, cs = model.init_hidden([n_games * 2])
+        self.hidden = ([h.to(device) for h in hs], [c.to(device) for c in cs])
+        self._arange2 = torch.arange(n_games, device=device) * 2
+        self._graph = None
+        if use_graphs and device.type == 'cuda' and \
+                __import__('os').environ.get('HANDYRL_NO_GRAPHS') != '1':
+            self._capture()
+
+    def _static_in(self):
+        per, dev = self.n_per, self.device
+        return {
+            'scalar': torch.zeros(per, SCALAR_DIM, device=dev),
+            'board': torch.zeros((per,) + BOARD_SHAPE, device=dev),
+            'mask': torch.full((per, N_ACTIONS), 1e32, device=dev),
+            'parity': torch.zeros(per, dtype=torch.int64, device=dev),
+            'keep': torch.ones(per * 2, 1, 1, 1, device=dev),
+        }
+
+    def _infer_body(self, st):
+        """One batched recurrent step on static tensors; hidden updates are
+        IN-PLACE so the storages stay fixed under graph capture."""
+        hs, cs = self.hidden
+        rows = self._arange2 + st['parity']
+        for i in range(len(hs)):
+            hs[i].mul_(st['keep'])
+            cs[i].mul_(st['keep'])
+        h_in = ([h.index_select(0, rows) for h in hs],
+                [c.index_select(0, rows) for c in cs])
+        out = self.model({'scalar': st['scalar'], 'board': st['board']}, h_in)
+        h_out, c_out = out['hidden']
+        for i in range(len(hs)):
+            hs[i].index_copy_(0, rows, h_out[i])
+            cs[i].index_copy_(0, rows, c_out[i])
+        policy = out['policy'].float()
+        uniform = torch.rand(policy.shape[0], device=self.device)
+        if self.device.type == 'cuda':
+            actions, probs = ops.masked_sample(policy, st['mask'], uniform)
+        else:
+            pr = torch.softmax(policy - st['mask'], dim=-1)
+            actions = torch.multinomial(pr, 1).squeeze(-1)
+            probs = pr.gather(-1, actions.unsqueeze(-1)).squeeze(-1)
+        return torch.stack([actions.float(), probs,
+                            out['value'].float().squeeze(-1),
+                            out['return'].float().squeeze(-1)], dim=1)
+
+    @torch.no_grad()
+    def _capture(self):
+        try:
+            st = self._static_in()
+            stream = torch.cuda.Stream()
+            stream.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(stream):
+                for _ in range(2):
+                    self._infer_body(st)
+            torch.cuda.current_stream().wait_stream(stream)
+            graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(graph):
+                packed = self._infer_body(st)
+            self._graph = (graph, st, packed)
+            # captured warmups corrupted the hidden state: reset it
+            hs, cs = self.hidden
+            for t in hs + cs:
+                t.zero_()
+        except Exception as e:     # noqa: BLE001 - eager fallback
+            import sys
+            print('geister actor graph capture failed, running eager: %r'
+                  % (e,), file=sys.stderr)
+            self._graph = None
+
+    @torch.no_grad()
+    def infer(self, scalar, board, mask, parity, reset, out):
+        """numpy in (shard-sized arrays), numpy out: fills out[:, 0:4] with
+        (action, prob, value, return) rows."""
+        per, dev = self.n_per, self.device
+        keep_np = (1.0 - reset.astype(np.float32)).repeat(2) \
+            .reshape(per * 2, 1, 1, 1)
+        if self._graph is not None:
+            graph, st, packed = self._graph
+            st['scalar'].copy_(torch.from_numpy(scalar), non_blocking=True)
+            st['board'].copy_(torch.from_numpy(board), non_blocking=True)
+            st['mask'].copy_(torch.from_numpy(mask), non_blocking=True)
+            st['parity'].copy_(torch.from_numpy(parity.astype(np.int64)),
+                               non_blocking=True)
+            st['keep'].copy_(torch.from_numpy(keep_np), non_blocking=True)
+            graph.replay()
+            np.copyto(out, packed.cpu().numpy())
+            return
+        st = {
+            'scalar': torch.from_numpy(scalar).to(dev, non_blocking=True),
+            'board': torch.from_numpy(board).to(dev, non_blocking=True),
+            'mask': torch.from_numpy(mask).to(dev, non_blocking=True),
+            'parity': torch.from_numpy(parity.astype(np.int64)).to(dev),
+            'keep': torch.from_numpy(keep_np).to(dev),
+        }
+        np.copyto(out, self._infer_body(st).cpu().numpy())
+
+
+class GeisterActorPool:
+    """In-process Geister self-play (the `worker: {type: 'gpu'}` pool for
+    recurrent envs): the vectorized rules engine steps the whole shard in
+    the calling thread, batched DRC inference runs with device-resident
+    hidden state (BatchedDRCEngine), and episodes are recorded columnar
+    turn-based.  No worker processes — safe to start from a thread of an
+    already-CUDA-initialized learner (unlike a forking pool)."""
+
+    CAP = 202                                # 2 layout turns + 200 moves
+
+    def __init__(self, model, args, n_games=256, device=None, seed=0):
+        self.args = args
+        self.device = device if device is not None else (
+            torch.device('cuda') if torch.cuda.is_available()
+            else torch.device('cpu'))
+        self.model = model
+        from .envs.vec_geister import GeisterVecEnv
+        self.vec = GeisterVecEnv(n_games, seed=seed)
+        self.vec.reset_games(np.arange(n_games))
+        G, CAP = n_games, self.CAP
+        self.n_games = G
+        self.gamma = args.get('gamma', 0.8)
+        self.rec = {
+            'scalar': np.zeros((G, CAP, SCALAR_DIM), np.uint8),
+            'board': np.zeros((G, CAP) + BOARD_SHAPE, np.uint8),
+            'mask': np.zeros((G, CAP, N_ACTIONS), bool),
+            'turn': np.zeros((G, CAP), np.int8),
+            'action': np.zeros((G, CAP), np.int16),
+            'prob': np.zeros((G, CAP), np.float32),
+            'value': np.zeros((G, CAP), np.float32),
+        }
+        self.rec_len = np.zeros(G, np.int32)
+        self._gar = np.arange(G)
+        self._mask_v = np.empty((G, N_ACTIONS), np.float32)
+        self._res_v = np.empty((G, 4), np.float32)
+        self._job_args = {'player': [0, 1], 'model_id': {0: -1, 1: -1}}
+        self.engine = BatchedDRCEngine(model, self.device, G)
+        self.completed = []
+        self.frames = 0
+        self.episodes_done = 0
+        self.calls_per_vec_step = 1
+
+    def _package(self, g, outcome):
+        S = int(self.rec_len[g])
+        if S == 0:
+            return None
+        acc, rets = 0.0, np.empty(S, np.float32)
+        for t in range(S - 1, -1, -1):       # constant -0.01 both players
+            acc = -0.01 + self.gamma * acc
+            rets[t] = acc
+        ep = {'args': self._job_args, 'steps': S, 'outcome': outcome,
+              'columnar': True, 'turn_based': True,
+              'n_actions': N_ACTIONS, 'n_players': 2,
+              'reward': np.full((S, 2), -0.01, np.float32),
+              'return': np.stack([rets, rets], axis=1)}
+        for k, buf in self.rec.items():
+            ep[k] = buf[g, :S].copy()
+        return ep
+
+    @torch.no_grad()
+    def step_once(self):
+        vec = self.vec
+        G = self.n_games
+        reset_flags = np.zeros(G, np.uint8)
+        done_idx = np.nonzero(vec.over)[0]
+        if len(done_idx):
+            ocs = vec.outcomes(done_idx)
+            for k, g in enumerate(done_idx):
+                ep = self._package(g, {0: float(ocs[k, 0]),
+                                       1: float(ocs[k, 1])})
+                if ep is not None:
+                    self.completed.append(ep)
+                    self.episodes_done += 1
+                self.rec_len[g] = 0
+            vec.reset_games(done_idx)
+            reset_flags[done_idx] = 1
+        scalar, board = vec.observations()
+        vec.legal_masks(out=self._mask_v)
+        parity = vec.turn().astype(np.int8)
+        self.engine.infer(scalar, board, self._mask_v, parity, reset_flags,
+                          self._res_v)
+        actions = self._res_v[:, 0].astype(np.int64)
+        rows = np.minimum(self.rec_len, self.CAP - 1)
+        gar = self._gar
+        self.rec['scalar'][gar, rows] = scalar.astype(np.uint8)
+        self.rec['board'][gar, rows] = board.astype(np.uint8)
+        self.rec['mask'][gar, rows] = self._mask_v == 0.0
+        self.rec['turn'][gar, rows] = parity
+        self.rec['action'][gar, rows] = actions.astype(np.int16)
+        self.rec['prob'][gar, rows] = self._res_v[:, 1]
+        self.rec['value'][gar, rows] = self._res_v[:, 2]
+        self.rec_len += 1
+        vec.step(actions)
+        self.frames += G
+        return G
+
+    def harvest(self):
+        out = self.completed
+        self.completed = []
+        return out
+
+    def refresh_weights(self):
+        pass                       # shares the live module directly
+
+    def shutdown(self):
+        pass
+
+
 class GeisterMultiProcPool:
     """256-actor-style Geister self-play on one GPU: W env-worker processes
     + batched recurrent inference with GPU-resident DRC hidden state."""
@@ -316,115 +532,15 @@ class GeisterMultiProcPool:
     def attach(self, model, device):
         self.model = model
         self.device = device
-        per = self.n_per
-        # per-(game, player) DRC hidden state, flattened rows = game*2+player
-        self.hidden = []
-        for _ in range(self.workers):
-            hs, cs = model.init_hidden([per * 2])
-            self.hidden.append((
-                [h.to(device) for h in hs], [c.to(device) for c in cs]))
-        self._arange2 = torch.arange(per, device=device) * 2
-        self._graphs = [None] * self.workers
-        if device.type == 'cuda' and \
-                __import__('os').environ.get('HANDYRL_NO_GRAPHS') != '1':
-            self._capture_graphs()
-
-    def _static_in(self):
-        per = self.n_per
-        dev = self.device
-        return {
-            'scalar': torch.zeros(per, SCALAR_DIM, device=dev),
-            'board': torch.zeros((per,) + BOARD_SHAPE, device=dev),
-            'mask': torch.full((per, N_ACTIONS), 1e32, device=dev),
-            'parity': torch.zeros(per, dtype=torch.int64, device=dev),
-            'keep': torch.ones(per * 2, 1, 1, 1, device=dev),
-        }
-
-    def _infer_body(self, wid, st):
-        """One batched recurrent inference step on static tensors; every
-        hidden update is IN-PLACE so the storages stay fixed under graph
-        capture."""
-        hs, cs = self.hidden[wid]
-        rows = self._arange2 + st['parity']
-        for i in range(len(hs)):
-            hs[i].mul_(st['keep'])
-            cs[i].mul_(st['keep'])
-        h_in = ([h.index_select(0, rows) for h in hs],
-                [c.index_select(0, rows) for c in cs])
-        out = self.model({'scalar': st['scalar'], 'board': st['board']}, h_in)
-        h_out, c_out = out['hidden']
-        for i in range(len(hs)):
-            hs[i].index_copy_(0, rows, h_out[i])
-            cs[i].index_copy_(0, rows, c_out[i])
-        policy = out['policy'].float()
-        uniform = torch.rand(policy.shape[0], device=self.device)
-        if self.device.type == 'cuda':
-            actions, probs = ops.masked_sample(policy, st['mask'], uniform)
-        else:
-            pr = torch.softmax(policy - st['mask'], dim=-1)
-            actions = torch.multinomial(pr, 1).squeeze(-1)
-            probs = pr.gather(-1, actions.unsqueeze(-1)).squeeze(-1)
-        return torch.stack([actions.float(), probs,
-                            out['value'].float().squeeze(-1),
-                            out['return'].float().squeeze(-1)], dim=1)
-
-    @torch.no_grad()
-    def _capture_graphs(self):
-        try:
-            self._statics = []
-            for wid in range(self.workers):
-                st = self._static_in()
-                stream = torch.cuda.Stream()
-                stream.wait_stream(torch.cuda.current_stream())
-                with torch.cuda.stream(stream):
-                    for _ in range(2):
-                        self._infer_body(wid, st)
-                torch.cuda.current_stream().wait_stream(stream)
-                graph = torch.cuda.CUDAGraph()
-                with torch.cuda.graph(graph):
-                    packed = self._infer_body(wid, st)
-                self._graphs[wid] = (graph, st, packed)
-            # captured warmups corrupted the hidden state: reset it
-            for wid in range(self.workers):
-                hs, cs = self.hidden[wid]
-                for t in hs + cs:
-                    t.zero_()
-        except Exception as e:     # noqa: BLE001 - eager fallback
-            import sys
-            print('geister actor graph capture failed, running eager: %r'
-                  % (e,), file=sys.stderr)
-            self._graphs = [None] * self.workers
+        # one engine (resident hidden + captured graph) per worker shard
+        self.engines = [BatchedDRCEngine(model, device, self.n_per)
+                        for _ in range(self.workers)]
 
     @torch.no_grad()
     def _infer(self, wid):
         v = self.views[wid]
-        dev = self.device
-        per = self.n_per
-        graphed = self._graphs[wid]
-        if graphed is not None:
-            graph, st, packed = graphed
-            st['scalar'].copy_(torch.from_numpy(v['scalar']), non_blocking=True)
-            st['board'].copy_(torch.from_numpy(v['board']), non_blocking=True)
-            st['mask'].copy_(torch.from_numpy(v['mask']), non_blocking=True)
-            st['parity'].copy_(torch.from_numpy(v['parity'].astype(np.int64)),
-                               non_blocking=True)
-            st['keep'].copy_(torch.from_numpy(
-                (1.0 - v['reset'].astype(np.float32)).repeat(2)
-                .reshape(per * 2, 1, 1, 1)), non_blocking=True)
-            graph.replay()
-            np.copyto(v['res'], packed.cpu().numpy())
-            return
-        st = {
-            'scalar': torch.from_numpy(v['scalar']).to(dev, non_blocking=True),
-            'board': torch.from_numpy(v['board']).to(dev, non_blocking=True),
-            'mask': torch.from_numpy(v['mask']).to(dev, non_blocking=True),
-            'parity': torch.from_numpy(v['parity'].astype(np.int64)).to(dev),
-            'keep': torch.from_numpy(
-                (1.0 - v['reset'].astype(np.float32)).repeat(2)
-                .reshape(per * 2, 1, 1, 1)).to(dev),
-        }
-        packed = self._infer_body(wid, st)
-        np.copyto(v['res'], packed.cpu().numpy())
+        self.engines[wid].infer(v['scalar'], v['board'], v['mask'],
+                                v['parity'], v['reset'], v['res'])
 
     def step_once(self):
         import multiprocessing.connection as mpc

@@ -433,11 +433,24 @@ class Learner:
         a vectorized env pool stepped by batched (graphed) inference on the
         trainer's live model, feeding columnar episodes straight into the
         replay buffer — no per-env worker processes, no pickled models."""
-        from .actor import GeeseActorPool
         n_envs = self.args['worker'].get('num_envs', 256)
-        pool = GeeseActorPool(self.trainer.model, self.args, n_games=n_envs,
-                              device=self.trainer.device,
-                              seed=self.args['seed'] + 1)
+        env_name = str(self.args.get('env', {}).get('env', ''))
+        if env_name == 'Geister':
+            from .actor_geister import GeisterActorPool
+            pool = GeisterActorPool(self.trainer.model, self.args,
+                                    n_games=n_envs,
+                                    device=self.trainer.device,
+                                    seed=self.args['seed'] + 1)
+        elif env_name == 'HungryGeese':
+            from .actor import GeeseActorPool
+            pool = GeeseActorPool(self.trainer.model, self.args,
+                                  n_games=n_envs,
+                                  device=self.trainer.device,
+                                  seed=self.args['seed'] + 1)
+        else:
+            raise ValueError(
+                "worker type 'gpu' supports HungryGeese and Geister "
+                "(got env=%r); use CPU workers for other envs" % env_name)
         last_epoch = -1
         print('started gpu actor pool (%d envs)' % n_envs)
         while not self.shutdown_flag:
