@@ -34,8 +34,14 @@ ROW_OBS = int(np.prod(OBS_SHAPE))
 EVICT_MARGIN = 16384
 
 
-class DeviceReplay:
+class ColumnRingReplay:
     """Flat columnar ring of per-step rows + a host-side episode table.
+
+    Subclasses declare ``COLUMNS`` ({attr: (torch dtype, tail shape,
+    episode key)}) and ``OUTCOME_P`` and provide env-shape-specific
+    ``sample_indices`` / ``gather_batch``; everything else — staging,
+    ring writes, the background-ingest thread + event-ordered publish,
+    planned-frontier eviction and reader-floor back-pressure — is shared.
 
     With ``ingest_thread=True`` (the bench default on GPU) episode blocks
     are staged and copied on a background thread + side HIP stream; the
@@ -44,23 +50,21 @@ class DeviceReplay:
     event so gathers are ordered after the copies.
     """
 
+    COLUMNS = {}
+    OUTCOME_P = 2
+
     def __init__(self, args, device, bytes_budget=4 << 30, ingest_thread=False):
         assert args.get('burn_in_steps', 0) == 0, \
-            'DeviceReplay supports feed-forward (no burn-in) training'
+            'device replay supports no-burn-in training windows'
         self.args = args
         self.device = device
-        row_bytes = ROW_OBS + 4 + 3 * 4 * 4       # obs u8, alive, act/prob/val
+        row_bytes = sum(
+            int(np.prod(tail or (1,))) * torch.empty((), dtype=dt).element_size()
+            for dt, tail, _k in self.COLUMNS.values())
         self.ring_T = max(1024, int(bytes_budget // row_bytes))
-        dev = device
-        self.obs = torch.empty((self.ring_T,) + OBS_SHAPE, dtype=torch.uint8,
-                               device=dev)
-        self.alive = torch.empty(self.ring_T, 4, dtype=torch.bool, device=dev)
-        self.action = torch.empty(self.ring_T, 4, dtype=torch.int32, device=dev)
-        self.prob = torch.empty(self.ring_T, 4, dtype=torch.float32, device=dev)
-        self.value = torch.empty(self.ring_T, 4, dtype=torch.float32, device=dev)
-        from .envs.vec_geese import CHMAP
-        self._chmap = torch.from_numpy(CHMAP).to(dev)     # (4, 17)
-        self._arange_cache = {}
+        for name, (dt, tail, _k) in self.COLUMNS.items():
+            setattr(self, name, torch.empty((self.ring_T,) + tail, dtype=dt,
+                                            device=device))
         self.head = 0                  # monotonically increasing write cursor
         self.head_planned = 0          # head + queued-but-unwritten rows
         # oldest logical row any in-flight gather may read; None until the
@@ -197,24 +201,39 @@ class DeviceReplay:
         n = sum(int(ep['steps']) for ep in episodes)
         if n > self.ring_T:
             raise ValueError('episode block larger than the replay ring')
-        self._write_ring(self.obs, self._stage(
-            'obs', [ep['obs'] for ep in episodes], n, torch.uint8, OBS_SHAPE))
-        self._write_ring(self.alive, self._stage(
-            'alive', [ep['alive'] for ep in episodes], n, torch.bool, (4,)))
-        self._write_ring(self.action, self._stage(
-            'action', [ep['action'] for ep in episodes], n, torch.int32, (4,)))
-        self._write_ring(self.prob, self._stage(
-            'prob', [ep['prob'] for ep in episodes], n, torch.float32, (4,)))
-        self._write_ring(self.value, self._stage(
-            'value', [ep['value'] for ep in episodes], n, torch.float32, (4,)))
+        for name, (dt, tail, key) in self.COLUMNS.items():
+            self._write_ring(getattr(self, name), self._stage(
+                name, [ep[key or name] for ep in episodes], n, dt, tail))
         entries = []
         pos = self.head
         for ep in episodes:
-            oc = np.array([ep['outcome'][p] for p in range(4)], dtype=np.float32)
+            oc = np.array([ep['outcome'][p] for p in range(self.OUTCOME_P)],
+                          dtype=np.float32)
             entries.append((pos, int(ep['steps']), oc))
             pos += int(ep['steps'])
         self.head = pos
         return entries, n
+
+    def _pick_window(self, m, fs):
+        """One recency-biased episode pick + window cut (reference
+        train.py:291-315 accept-loop semantics)."""
+        while True:
+            idx = random.randrange(m)
+            accept = 1 - (m - 1 - idx) / m
+            if random.random() < accept:
+                break
+        p0, steps, oc = self.table[idx]
+        train_st = random.randrange(1 + max(0, steps - fs))
+        ed = min(train_st + fs, steps)
+        return p0, steps, oc, train_st, ed
+
+    def _update_reader_floor(self):
+        # reader floor covers this fill AND the previous one (the previous
+        # graph replay may still be in flight when this one is sampled)
+        floor_now = self.table[0][0]
+        self._reader_floor = floor_now if self._floor_prev is None \
+            else min(floor_now, self._floor_prev)
+        self._floor_prev = floor_now
 
     def _evict(self):
         # With background ingest, evict against the PLANNED write frontier
@@ -253,6 +272,26 @@ class DeviceReplay:
         while len(self.table) > maximum:
             self.table.popleft()
 
+class DeviceReplay(ColumnRingReplay):
+    """Solo feed-forward replay (the flagship Hungry Geese path): rows
+    store CANONICAL per-game obs + per-seat alive/action/prob/value; the
+    per-seat channel rotation happens inside gather_batch on device."""
+
+    COLUMNS = {
+        'obs': (torch.uint8, OBS_SHAPE, None),
+        'alive': (torch.bool, (4,), None),
+        'action': (torch.int32, (4,), None),
+        'prob': (torch.float32, (4,), None),
+        'value': (torch.float32, (4,), None),
+    }
+    OUTCOME_P = 4
+
+    def __init__(self, args, device, bytes_budget=4 << 30, ingest_thread=False):
+        super().__init__(args, device, bytes_budget, ingest_thread)
+        from .envs.vec_geese import CHMAP
+        self._chmap = torch.from_numpy(CHMAP).to(device)     # (4, 17)
+        self._arange_cache = {}
+
     # -- sample path ---------------------------------------------------------
     def sample_indices(self, batch_size):
         """Recency-biased episode picks + window cuts + solo-seat choice
@@ -270,26 +309,14 @@ class DeviceReplay:
         inv_total = np.empty(batch_size, dtype=np.float32)
         m = min(n, args['maximum_episodes'])
         for b in range(batch_size):
-            while True:
-                idx = random.randrange(m)
-                accept = 1 - (m - 1 - idx) / m
-                if random.random() < accept:
-                    break
-            p0, steps, oc = self.table[idx]
-            train_st = random.randrange(1 + max(0, steps - fs))
-            ed = min(train_st + fs, steps)
+            p0, steps, oc, train_st, ed = self._pick_window(m, fs)
             pos0[b] = p0 + train_st
             start[b] = train_st
             length[b] = ed - train_st
             seat[b] = random.randrange(4)
             outcome[b] = oc
             inv_total[b] = 1.0 / steps
-        # reader floor covers this fill AND the previous one (the previous
-        # graph replay may still be in flight when this one is sampled)
-        floor_now = self.table[0][0]
-        self._reader_floor = floor_now if self._floor_prev is None \
-            else min(floor_now, self._floor_prev)
-        self._floor_prev = floor_now
+        self._update_reader_floor()
         return pos0, start, length, seat, outcome, inv_total
 
     def gather_batch(self, pos0, start, length, seat, outcome, inv_total):
@@ -354,5 +381,114 @@ class DeviceReplay:
             'turn_mask': shp(tmask, 1),
             'observation_mask': shp(tmask.clone(), 1),
             'action_mask': shp(amask4, 4),
+            'progress': progress.view(B, T, 1),
+        }
+
+
+class TurnDeviceReplay(ColumnRingReplay):
+    """Turn-based (observation=False) replay for recurrent envs (Geister):
+    rows store the MOVER's observation/legality plus per-player value-side
+    columns; gather_batch reproduces make_batch's turn-based columnar
+    fields on device (parity-tested in tests/test_turn_replay.py).
+
+    Episodes are the columnar turn-based format produced by the Geister
+    actor pools (actor_geister.py)."""
+
+    COLUMNS = {
+        'scalar': (torch.uint8, (18,), None),
+        'board': (torch.uint8, (7, 6, 6), None),
+        'mask': (torch.bool, (214,), None),          # legality bools
+        'turn': (torch.int8, (), None),
+        'action': (torch.int16, (), None),
+        'prob': (torch.float32, (), None),
+        'value': (torch.float32, (), None),
+        'reward': (torch.float32, (2,), None),
+        'ret': (torch.float32, (2,), 'return'),
+    }
+    OUTCOME_P = 2
+
+    def sample_indices(self, batch_size):
+        """Recency-biased picks + window cuts (no seat choice: the mover
+        axis is decided per step by the recorded turn column)."""
+        args = self.args
+        fs = args['forward_steps']
+        n = len(self.table)
+        assert n > 0, 'empty replay'
+        pos0 = np.empty(batch_size, dtype=np.int64)
+        start = np.empty(batch_size, dtype=np.int64)
+        length = np.empty(batch_size, dtype=np.int64)
+        outcome = np.empty((batch_size, 2), dtype=np.float32)
+        inv_total = np.empty(batch_size, dtype=np.float32)
+        m = min(n, args['maximum_episodes'])
+        for b in range(batch_size):
+            p0, steps, oc, train_st, ed = self._pick_window(m, fs)
+            pos0[b] = p0 + train_st
+            start[b] = train_st
+            length[b] = ed - train_st
+            outcome[b] = oc
+            inv_total[b] = 1.0 / steps
+        self._update_reader_floor()
+        return pos0, start, length, outcome, inv_total
+
+    def gather_batch(self, pos0, start, length, outcome, inv_total):
+        """Build the turn-based (B, T, P, ...) batch on device: obs/prob/
+        action/action_mask on the mover axis (P=1), value-side fields on
+        the player axis (P=2).  All inputs are device tensors of length B;
+        every op is hipGraph-capturable.  Out-of-window rows read garbage
+        ring memory, so every float path uses torch.where (never
+        multiply-by-mask, which leaks NaN*0)."""
+        args = self.args
+        B = pos0.shape[0]
+        T = args['forward_steps']
+        dev = self.device
+        t_range = torch.arange(T, device=dev)
+        rows = (pos0.unsqueeze(1) + t_range) % self.ring_T      # (B, T)
+        in_range = t_range.unsqueeze(0) < length.unsqueeze(1)   # (B, T)
+        flat = rows.reshape(-1)
+        in_r = in_range.reshape(-1)                             # (B*T,)
+
+        scalar = self.scalar[flat].float() * in_r.unsqueeze(1)  # u8: mul safe
+        board = self.board[flat].float() * in_r.view(-1, 1, 1, 1)
+        legal = self.mask[flat] & in_r.unsqueeze(1)
+        amask = torch.where(legal, torch.zeros((), device=dev),
+                            torch.full((), 1e32, device=dev))
+        turn = self.turn[flat].long().clamp(0, 1)               # (B*T,)
+        act = (self.action[flat].long() * in_r.long())
+        prob = torch.where(in_r, self.prob[flat],
+                           torch.ones((), device=dev))
+        oc_bt = outcome.unsqueeze(1).expand(B, T, 2).reshape(-1, 2)
+        tmask = torch.zeros(B * T, 2, device=dev)
+        tmask.scatter_(1, turn.unsqueeze(1), in_r.float().unsqueeze(1))
+        v_m = torch.zeros(B * T, 2, device=dev)
+        v_m.scatter_(1, turn.unsqueeze(1),
+                     torch.where(in_r, self.value[flat],
+                                 torch.zeros((), device=dev)).unsqueeze(1))
+        v = torch.where(in_r.unsqueeze(1), v_m, oc_bt)
+        rew = torch.where(in_r.unsqueeze(1), self.reward[flat],
+                          torch.zeros((), device=dev))
+        ret = torch.where(in_r.unsqueeze(1), self.ret[flat],
+                          torch.zeros((), device=dev))
+        progress = torch.where(
+            in_r,
+            (start.unsqueeze(1) + t_range).reshape(-1).float() *
+            inv_total.unsqueeze(1).expand(B, T).reshape(-1),
+            torch.ones((), device=dev))
+
+        def mover(x, *tail):                                    # (B, T, 1, ...)
+            return x.reshape(B, T, 1, *tail)
+
+        return {
+            'observation': {'scalar': mover(scalar, 18),
+                            'board': mover(board, 7, 6, 6)},
+            'selected_prob': mover(prob, 1),
+            'value': v.reshape(B, T, 2, 1),
+            'action': mover(act, 1),
+            'outcome': outcome.view(B, 1, 2, 1),
+            'reward': rew.reshape(B, T, 2, 1),
+            'return': ret.reshape(B, T, 2, 1),
+            'episode_mask': in_range.float().view(B, T, 1, 1),
+            'turn_mask': tmask.reshape(B, T, 2, 1),
+            'observation_mask': tmask.reshape(B, T, 2, 1).clone(),
+            'action_mask': mover(amask, 214),
             'progress': progress.view(B, T, 1),
         }
