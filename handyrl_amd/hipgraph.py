@@ -284,3 +284,99 @@ class GraphedReplayTrainStep:
             self.graph.replay()
             return self.losses, self.dcnt
         return self._run()
+
+
+class GraphedRecurrentTrainStep:
+    """Turn-based/recurrent counterpart of GraphedReplayTrainStep: the
+    batch is gathered from a TurnDeviceReplay ring inside the step, the
+    DRC forward runs the per-timestep RNN loop (train.forward_prediction)
+    from a STATIC zero initial hidden, and on GPU the whole step attempts
+    hipGraph capture (eager fallback otherwise — the capture of the
+    T-step ConvLSTM loop is exercised on hardware, not assumed)."""
+
+    def __init__(self, trainer, replay, batch_size, n_players=2,
+                 warmup_iters=2):
+        from .train import compute_loss
+        self._compute_loss = compute_loss
+        self.trainer = trainer
+        self.replay = replay
+        self.batch_size = batch_size
+        dev = trainer.device
+        B = batch_size
+        self.idx = {
+            'pos0': torch.zeros(B, dtype=torch.int64, device=dev),
+            'start': torch.zeros(B, dtype=torch.int64, device=dev),
+            'length': torch.zeros(B, dtype=torch.int64, device=dev),
+            'outcome': torch.zeros(B, n_players, device=dev),
+            'inv_total': torch.zeros(B, device=dev),
+        }
+        # initial hidden is always zeros: allocate once, reuse every step
+        self.hidden0 = trainer.wrapped_model.init_hidden([B, n_players])
+        if self.hidden0 is not None:
+            from .util import map_r
+            self.hidden0 = map_r(self.hidden0, lambda h: h.to(dev))
+        self.graph = None
+        if dev.type == 'cuda':
+            for group in trainer.optimizer.param_groups:
+                group['capturable'] = True
+            trainer.model.train()
+            stream = torch.cuda.Stream()
+            stream.wait_stream(torch.cuda.current_stream())
+            try:
+                with torch.cuda.stream(stream):
+                    for _ in range(warmup_iters):
+                        self._fill()
+                        self._run()
+                torch.cuda.current_stream().wait_stream(stream)
+                self.graph = torch.cuda.CUDAGraph()
+                self._fill()
+                with torch.cuda.graph(self.graph):
+                    self.losses, self.dcnt = self._run()
+            except Exception as e:  # noqa: BLE001 - run eager if capture fails
+                import sys
+                print('recurrent train-step capture failed, running eager: %r'
+                      % (e,), file=sys.stderr)
+                self.graph = None
+
+    def _fill(self):
+        self.replay.publish()
+        pos0, start, length, outcome, inv_total = \
+            self.replay.sample_indices(self.batch_size)
+        self._last_host_idx = {
+            'pos0': pos0, 'start': start, 'length': length,
+            'outcome': outcome, 'inv_total': inv_total}
+        for key, arr in self._last_host_idx.items():
+            self.idx[key].copy_(torch.from_numpy(arr), non_blocking=True)
+
+    def _run(self):
+        tr = self.trainer
+        batch = self.replay.gather_batch(
+            self.idx['pos0'], self.idx['start'], self.idx['length'],
+            self.idx['outcome'], self.idx['inv_total'])
+        if tr.use_amp:
+            with torch.autocast('cuda', dtype=torch.bfloat16):
+                losses, dcnt = self._compute_loss(batch, tr.wrapped_model,
+                                                  self.hidden0, tr.args)
+        else:
+            losses, dcnt = self._compute_loss(batch, tr.wrapped_model,
+                                              self.hidden0, tr.args)
+        tr.optimizer.zero_grad(set_to_none=False)
+        losses['total'].backward()
+        if tr.grad_guard:
+            for p in tr.params:
+                if p.grad is not None:
+                    torch.nan_to_num_(p.grad, nan=0.0, posinf=1e6,
+                                      neginf=-1e6)
+        tr.reducer.allreduce_()
+        nn.utils.clip_grad_norm_(tr.params, 4.0)
+        tr.optimizer.step()
+        return losses, dcnt
+
+    def step(self):
+        """Sample indices on the host, copy them in, replay (or run)."""
+        self._fill()
+        self.trainer.steps += 1
+        if self.graph is not None:
+            self.graph.replay()
+            return self.losses, self.dcnt
+        return self._run()

@@ -384,3 +384,46 @@ def test_fused_eval_canonical_matches_seat_expanded():
     torch.cuda.synchronize()
     torch.testing.assert_close(out_c['policy'], out_e['policy'], rtol=0, atol=0)
     torch.testing.assert_close(out_c['value'], out_e['value'], rtol=0, atol=0)
+
+
+@requires_gpu
+def test_turn_device_replay_recurrent_training():
+    """Turn-based device replay (Geister): episodes from the in-process
+    GPU actor pool into the HBM ring, then GraphedRecurrentTrainStep
+    (capture attempted; eager fallback allowed) trains with finite losses
+    and changing weights."""
+    from handyrl_amd.actor_geister import GeisterActorPool
+    from handyrl_amd.envs.geister import Environment as GeisterEnv
+    from handyrl_amd.replay import TurnDeviceReplay
+    from handyrl_amd.train import Trainer
+    from handyrl_amd.hipgraph import GraphedRecurrentTrainStep
+
+    args = {
+        'turn_based_training': True, 'observation': False, 'gamma': 0.8,
+        'forward_steps': 8, 'burn_in_steps': 0, 'compress_steps': 4,
+        'entropy_regularization': 0.1, 'entropy_regularization_decay': 0.1,
+        'batch_size': 8, 'minimum_episodes': 2, 'maximum_episodes': 500,
+        'num_batchers': 1, 'lambda': 0.7, 'policy_target': 'UPGO',
+        'value_target': 'TD', 'seed': 0, 'bf16': False,
+        'compress_episodes': False,
+    }
+    device = torch.device('cuda', 0)
+    trainer = Trainer(args, GeisterEnv().net(), device=device, batcher=False)
+    trainer.model.eval()
+    pool = GeisterActorPool(trainer.model, args, n_games=64, device=device,
+                            seed=0)
+    while pool.episodes_done < 30:
+        pool.step_once()
+    replay = TurnDeviceReplay(args, device, bytes_budget=256 << 20)
+    replay.extend(pool.harvest())
+
+    step = GraphedRecurrentTrainStep(trainer, replay, args['batch_size'])
+    print('recurrent train-step captured: %s' % (step.graph is not None))
+    before = [p.detach().clone() for p in trainer.params]
+    for _ in range(3):
+        losses, dcnt = step.step()
+        torch.cuda.synchronize()
+        assert torch.isfinite(losses['total']), losses
+        assert float(dcnt) > 0
+    assert any(not torch.equal(b, p.detach())
+               for b, p in zip(before, trainer.params))

@@ -19,6 +19,10 @@ def main():
     ap.add_argument('--steps', type=int, default=30)
     ap.add_argument('--warmup', type=int, default=5)
     ap.add_argument('--batch-size', type=int, default=32)
+    ap.add_argument('--device-replay', action='store_true',
+                    help='HBM-resident TurnDeviceReplay + (attempted) '
+                         'captured recurrent train step instead of the '
+                         'host EpisodeBuffer + Batcher')
     cli = ap.parse_args()
 
     torch.set_num_threads(1)
@@ -35,7 +39,7 @@ def main():
 
     from handyrl_amd.batch import EpisodeBuffer, Batcher
     buffer = EpisodeBuffer(args)
-    batcher = Batcher(args, buffer)
+    batcher = None if cli.device_replay else Batcher(args, buffer)
     from handyrl_amd.actor_geister import GeisterMultiProcPool
     pool = GeisterMultiProcPool(args, n_games=cli.actors, seed=17,
                                 workers=cli.workers)
@@ -47,11 +51,18 @@ def main():
     torch.manual_seed(0)
     env = Environment()
     trainer = Trainer(args, env.net(), device=device, episodes=buffer,
-                      batcher=batcher)
+                      batcher=batcher if batcher is not None else False)
     actor_model = env.net().to(device)
     actor_model.load_state_dict(trainer.model.state_dict())
     actor_model.eval()
     pool.attach(actor_model, device)
+
+    dreplay = dstep = None
+    if cli.device_replay:
+        from handyrl_amd.replay import TurnDeviceReplay
+        from handyrl_amd.hipgraph import GraphedRecurrentTrainStep
+        dreplay = TurnDeviceReplay(args, device, bytes_budget=2 << 30,
+                                   ingest_thread=use_cuda)
 
     def pump(n):
         frames = 0
@@ -59,20 +70,32 @@ def main():
             frames += pool.step_once()
         eps = pool.harvest()
         if eps:
-            buffer.extend(eps)
+            (dreplay if dreplay is not None else buffer).extend(eps)
         return frames
 
     t0 = time.time()
     while pool.episodes_done < args['minimum_episodes']:
         pump(2)
-    batcher.run()
-    print('# prefill %d eps in %.1fs' % (len(buffer), time.time() - t0),
+    if dreplay is not None:
+        dreplay.flush()
+        dstep = GraphedRecurrentTrainStep(trainer, dreplay,
+                                          args['batch_size'])
+        print('# recurrent train-step captured: %s' % (dstep.graph is not None),
+              file=sys.stderr, flush=True)
+        n_prefill = len(dreplay)
+    else:
+        batcher.run()
+        n_prefill = len(buffer)
+    print('# prefill %d eps in %.1fs' % (n_prefill, time.time() - t0),
           file=sys.stderr, flush=True)
 
     def one_step():
         frames = pump(4)
-        batch = batcher.batch()
-        losses, dcnt = trainer.train_step(batch)
+        if dstep is not None:
+            losses, dcnt = dstep.step()
+        else:
+            batch = batcher.batch()
+            losses, dcnt = trainer.train_step(batch)
         # actors follow the trained weights (model push each step)
         actor_model.load_state_dict(trainer.model.state_dict())
         return frames
