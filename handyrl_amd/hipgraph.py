@@ -200,7 +200,6 @@ class GraphedReplayTrainStep:
         self.trainer = trainer
         self.replay = replay
         self.batch_size = batch_size
-        assert trainer.device.type == 'cuda'
         dev = trainer.device
         B = batch_size
         self.idx = {
@@ -211,10 +210,13 @@ class GraphedReplayTrainStep:
             'outcome': torch.zeros(B, 4, device=dev),
             'inv_total': torch.zeros(B, device=dev),
         }
+        trainer.model.train()
+        self.graph = None
+        if dev.type != 'cuda':
+            return                    # eager path (CPU tests)
         for group in trainer.optimizer.param_groups:
             group['capturable'] = True
 
-        trainer.model.train()
         stream = torch.cuda.Stream()
         stream.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(stream):

@@ -72,7 +72,10 @@ class ColumnRingReplay:
         # the ingest thread are conservative)
         self._reader_floor = None
         self._floor_prev = None
-        self.table = deque()           # (pos0, steps, outcome(np[4]))
+        self.table = deque()           # (pos0, steps, outcome(np[P]))
+        # guards table/eviction state: the Learner path extends from its
+        # episode-feeder thread while the trainer thread publishes/samples
+        self._table_lock = threading.Lock()
         self.total_added = 0
         self._pin = {}
         self._ingest = None
@@ -140,11 +143,12 @@ class ColumnRingReplay:
             stream = torch.cuda.current_stream()
         with self._ready_lock:
             ready, self._ready = self._ready, deque()
-        for entries, n_rows, event in ready:
-            stream.wait_event(event)
-            self.table.extend(entries)
-            self.total_added += len(entries)
-        self._evict()
+        with self._table_lock:
+            for entries, n_rows, event in ready:
+                stream.wait_event(event)
+                self.table.extend(entries)
+                self.total_added += len(entries)
+            self._evict()
 
     def flush(self):
         """Block until every queued episode block is published."""
@@ -155,7 +159,8 @@ class ColumnRingReplay:
         self.publish()
 
     def __len__(self):
-        return len(self.table)
+        with self._table_lock:
+            return len(self.table)
 
     # -- write path --------------------------------------------------------
     def _stage(self, key, parts, n, dtype, tail_shape):
@@ -222,7 +227,8 @@ class ColumnRingReplay:
             accept = 1 - (m - 1 - idx) / m
             if random.random() < accept:
                 break
-        p0, steps, oc = self.table[idx]
+        with self._table_lock:
+            p0, steps, oc = self.table[min(idx, len(self.table) - 1)]
         train_st = random.randrange(1 + max(0, steps - fs))
         ed = min(train_st + fs, steps)
         return p0, steps, oc, train_st, ed
@@ -230,7 +236,10 @@ class ColumnRingReplay:
     def _update_reader_floor(self):
         # reader floor covers this fill AND the previous one (the previous
         # graph replay may still be in flight when this one is sampled)
-        floor_now = self.table[0][0]
+        with self._table_lock:
+            if not self.table:
+                return
+            floor_now = self.table[0][0]
         self._reader_floor = floor_now if self._floor_prev is None \
             else min(floor_now, self._floor_prev)
         self._floor_prev = floor_now
@@ -260,17 +269,21 @@ class ColumnRingReplay:
                                  if int(ep['steps']) <= limit)
         if self._ingest is not None:
             self._ingest_q.put(episodes)
-            self._evict()              # clear the planned frontier's alias zone
+            with self._table_lock:
+                self._evict()          # clear the planned frontier's alias zone
             return
         for chunk in self._chunk(episodes):
             entries, _ = self._copy_block(chunk)
-            self.table.extend(entries)
-            self.total_added += len(entries)
-        self._evict()
+            with self._table_lock:
+                self.table.extend(entries)
+                self.total_added += len(entries)
+        with self._table_lock:
+            self._evict()
 
     def trim(self, maximum):
-        while len(self.table) > maximum:
-            self.table.popleft()
+        with self._table_lock:
+            while len(self.table) > maximum:
+                self.table.popleft()
 
 class DeviceReplay(ColumnRingReplay):
     """Solo feed-forward replay (the flagship Hungry Geese path): rows

@@ -232,7 +232,38 @@ class Trainer:
         # process risks a child deadlock on inherited locks.  Callers that
         # need the fork even earlier (before any HIP context, e.g. bench.py)
         # pass a pre-built Batcher in.
-        self.batcher = batcher if batcher is not None else Batcher(args, self.episodes)
+        # replay: 'device' — an HBM-resident replay ring replaces BOTH the
+        # host episode buffer and the multiprocess batch builders; train()
+        # then runs the (captured) sample-gather train step.  Supported for
+        # feed-forward solo configs (DeviceReplay) and turn-based
+        # observation=False recurrent configs (TurnDeviceReplay).
+        self.device_replay = None
+        self._replay_step = None
+        if args.get('replay') == 'device':
+            from .replay import DeviceReplay, TurnDeviceReplay
+            if args.get('burn_in_steps', 0):
+                raise ValueError("replay: 'device' requires burn_in_steps 0")
+            budget = int(args.get('replay_bytes',
+                                  (4 << 30) if self.device.type == 'cuda'
+                                  else (64 << 20)))
+            ingest = self.device.type == 'cuda'
+            if args['turn_based_training']:
+                if args['observation']:
+                    raise ValueError(
+                        "replay: 'device' supports solo or turn-based "
+                        "observation=False configs")
+                self.device_replay = TurnDeviceReplay(
+                    args, self.device, bytes_budget=budget,
+                    ingest_thread=ingest)
+            else:
+                self.device_replay = DeviceReplay(
+                    args, self.device, bytes_budget=budget,
+                    ingest_thread=ingest)
+            self.episodes = self.device_replay
+            self.batcher = False
+        else:
+            self.batcher = batcher if batcher is not None \
+                else Batcher(args, self.episodes)
         self.update_flag = threading.Event()
         self.update_queue = queue.Queue(maxsize=1)
         self.wrapped_model = ModelWrapper(self.model)
@@ -309,8 +340,12 @@ class Trainer:
         self.model.train()
 
         while data_cnt == 0 or not self.update_flag.is_set():
-            batch = self.batcher.batch()
-            losses, dcnt = self.train_step(batch)
+            if self.device_replay is not None:
+                losses, dcnt = self._device_replay_step()
+                dcnt = float(dcnt)
+            else:
+                batch = self.batcher.batch()
+                losses, dcnt = self.train_step(batch)
             batch_cnt += 1
             data_cnt += dcnt
             for k, l in losses.items():
@@ -329,12 +364,30 @@ class Trainer:
                 param_group['lr'] = new_lr
         return self.snapshot()
 
+    def _device_replay_step(self):
+        """One step on the device-replay path (lazy build: the captured
+        gather-train step needs a non-empty ring)."""
+        if self._replay_step is None:
+            from .hipgraph import (GraphedReplayTrainStep,
+                                   GraphedRecurrentTrainStep)
+            if self.args['turn_based_training']:
+                self._replay_step = GraphedRecurrentTrainStep(
+                    self, self.device_replay, self.args['batch_size'])
+            else:
+                self._replay_step = GraphedReplayTrainStep(
+                    self, self.device_replay, self.args['batch_size'])
+        return self._replay_step.step()
+
     def run(self):
         print('waiting training')
         while len(self.episodes) < self.args['minimum_episodes']:
+            if self.device_replay is not None:
+                # admit background-ingested blocks so the count advances
+                self.device_replay.publish()
             time.sleep(1)
         if self.optimizer is not None:
-            self.batcher.run()
+            if self.batcher:
+                self.batcher.run()
             print('started training')
         while True:
             model = self.train()

@@ -253,3 +253,71 @@ def test_full_gpu_actor_mode_training_run_geister():
         assert 'started gpu actor pool' in out
         assert 'updated model(' in out
         assert os.path.exists(os.path.join(tmp, 'models', '1.pth'))
+
+
+def test_train_with_device_replay_solo():
+    """--train with replay: 'device' (HungryGeese solo): the HBM-ring
+    replay replaces the episode buffer + batchers; eager gather-train on
+    CPU here (captured on GPU)."""
+    script = textwrap.dedent('''
+        import sys
+        sys.path.insert(0, %r)
+        from handyrl_amd.train import train_main
+        args = {
+            'env_args': {'env': 'HungryGeese'},
+            'train_args': {
+                'turn_based_training': False, 'observation': False,
+                'gamma': 0.8, 'forward_steps': 8, 'burn_in_steps': 0,
+                'compress_steps': 4, 'entropy_regularization': 0.1,
+                'entropy_regularization_decay': 0.1, 'update_episodes': 30,
+                'batch_size': 4, 'minimum_episodes': 10, 'maximum_episodes': 300,
+                'epochs': 1, 'num_batchers': 1, 'eval_rate': 0.1,
+                'worker': {'type': 'gpu', 'num_parallel': 1, 'num_envs': 8},
+                'lambda': 0.7, 'policy_target': 'VTRACE', 'value_target': 'VTRACE',
+                'eval': {'opponent': ['random']}, 'seed': 0, 'restart_epoch': 0,
+                'bf16': False, 'replay': 'device',
+            },
+        }
+        train_main(args)
+        print('TRAIN_DONE')
+    ''') % REPO
+    with tempfile.TemporaryDirectory() as tmp:
+        res = subprocess.run([sys.executable, '-c', script], cwd=tmp,
+                             capture_output=True, text=True, timeout=300)
+        out = res.stdout
+        assert 'TRAIN_DONE' in out, (out[-3000:], res.stderr[-3000:])
+        assert 'updated model(' in out
+        assert os.path.exists(os.path.join(tmp, 'models', '1.pth'))
+
+
+def test_train_with_device_replay_turn_based():
+    """--train with replay: 'device' on the recurrent Geister config."""
+    script = textwrap.dedent('''
+        import sys
+        sys.path.insert(0, %r)
+        from handyrl_amd.train import train_main
+        args = {
+            'env_args': {'env': 'Geister'},
+            'train_args': {
+                'turn_based_training': True, 'observation': False,
+                'gamma': 0.8, 'forward_steps': 8, 'burn_in_steps': 0,
+                'compress_steps': 4, 'entropy_regularization': 0.1,
+                'entropy_regularization_decay': 0.1, 'update_episodes': 20,
+                'batch_size': 2, 'minimum_episodes': 4, 'maximum_episodes': 200,
+                'epochs': 1, 'num_batchers': 1, 'eval_rate': 0.1,
+                'worker': {'type': 'gpu', 'num_parallel': 1, 'num_envs': 8},
+                'lambda': 0.7, 'policy_target': 'UPGO', 'value_target': 'TD',
+                'eval': {'opponent': ['random']}, 'seed': 0, 'restart_epoch': 0,
+                'bf16': False, 'replay': 'device',
+            },
+        }
+        train_main(args)
+        print('TRAIN_DONE')
+    ''') % REPO
+    with tempfile.TemporaryDirectory() as tmp:
+        res = subprocess.run([sys.executable, '-c', script], cwd=tmp,
+                             capture_output=True, text=True, timeout=300)
+        out = res.stdout
+        assert 'TRAIN_DONE' in out, (out[-3000:], res.stderr[-3000:])
+        assert 'updated model(' in out
+        assert os.path.exists(os.path.join(tmp, 'models', '1.pth'))
