@@ -79,6 +79,12 @@ class GeeseVecEnv:
         idx = ((self.start + self.length - 1) % CAP).ravel()
         return flat[self._gp_arange, idx].reshape(self.G, N_PLAYERS)
 
+    def _tail_at(self, gi, pi):
+        """Tail cells for the (gi, pi) subset only (avoids the full-grid
+        gather on the hot path)."""
+        idx = (self.start[gi, pi] + self.length[gi, pi] - 1) % CAP
+        return self.body[gi, pi, idx]
+
     def reset_games(self, games):
         """Reset the given game indices to fresh initial states."""
         if len(games) == 0:
@@ -137,33 +143,34 @@ class GeeseVecEnv:
             self.food[hit.any(axis=1), f] = -1
 
         # pop tail unless the goose ate (bodies never self-overlap, so the
-        # grid bit of the vacated cell can be cleared directly)
-        popping = live & ~ate
-        gi, pi = np.nonzero(popping)
+        # grid bit of the vacated cell can be cleared directly); the live
+        # index set is reused for the push below
+        gl, pl = np.nonzero(live)
+        pop = ~ate[gl, pl]
+        gi, pi = gl[pop], pl[pop]
         if len(gi):
-            tail = self._tail_cell()[gi, pi]
+            tail = self._tail_at(gi, pi)
             self.length[gi, pi] -= 1
             self.body_grid[gi, pi, tail] = 0
 
         # push new head; a head landing on this goose's own remaining body is
         # a self-collision the 0/1 grid can't count — flag it explicitly
         self_crash = np.zeros((G, N_PLAYERS), dtype=bool)
-        gi, pi = np.nonzero(live)
-        if len(gi):
-            nh = new_head[gi, pi]
-            self_crash[gi, pi] = self.body_grid[gi, pi, nh] == 1
-            new_start = (self.start[gi, pi] - 1) % CAP
-            self.start[gi, pi] = new_start
-            self.body[gi, pi, new_start] = nh
-            self.length[gi, pi] += 1
-            self.body_grid[gi, pi, nh] = 1
+        if len(gl):
+            nh = new_head[gl, pl]
+            self_crash[gl, pl] = self.body_grid[gl, pl, nh] == 1
+            new_start = (self.start[gl, pl] - 1) % CAP
+            self.start[gl, pl] = new_start
+            self.body[gl, pl, new_start] = nh
+            self.length[gl, pl] += 1
+            self.body_grid[gl, pl, nh] = 1
 
         # 3) hunger shrink every HUNGER_RATE transitions
         hungry_games = (self.step_count + 1) % HUNGER_RATE == 0
         shrink = live & hungry_games[:, None]
         gi, pi = np.nonzero(shrink)
         if len(gi):
-            tail = self._tail_cell()[gi, pi]
+            tail = self._tail_at(gi, pi)
             self.length[gi, pi] -= 1
             keep = self_crash[gi, pi] & (tail == new_head[gi, pi])  # looped head
             self.body_grid[gi[~keep], pi[~keep], tail[~keep]] = 0
@@ -171,12 +178,14 @@ class GeeseVecEnv:
             self._kill(gi[starved], pi[starved])
         live = self.alive & ~self.over[:, None]
 
-        # 4) collision deaths: live head on a cell with >1 segment
-        counts = self.body_grid.sum(axis=1, dtype=np.int16)     # (G, cells)
+        # 4) collision deaths: live head on a cell with >1 segment (gather
+        # the 4 per-player grid bits at each live head instead of reducing
+        # the whole grid)
         heads = self._head()
         gi, pi = np.nonzero(live)
         if len(gi):
-            crash = (counts[gi, heads[gi, pi]] > 1) | self_crash[gi, pi]
+            seg = self.body_grid[gi, :, heads[gi, pi]]          # (n, 4)
+            crash = (seg.sum(axis=1) > 1) | self_crash[gi, pi]
             self._kill(gi[crash], pi[crash])
         live = self.alive & ~self.over[:, None]
 
