@@ -93,3 +93,20 @@ def test_geister_pool_generates_and_trains_moment_dicts(monkeypatch):
         _train_on(eps, env, args)
     finally:
         pool.shutdown()
+
+
+def test_geister_pool_python_env_fallback(monkeypatch):
+    """HANDYRL_GEISTER_VEC=0: the original python-Environment worker
+    shards still generate trainable reference-format episodes."""
+    monkeypatch.setenv('HANDYRL_GEISTER_VEC', '0')
+    args = _args()
+    pool = GeisterMultiProcPool(args, n_games=4, seed=6, workers=1)
+    try:
+        env = GeisterEnv()
+        eps = _pump_episodes(pool, env, n=2)
+        moments = unpack_moments({'moment': eps[0]['moment'], 'base': 0},
+                                 0, eps[0]['steps'])
+        assert len(moments) == eps[0]['steps']
+        _train_on(eps, env, args)
+    finally:
+        pool.shutdown()
