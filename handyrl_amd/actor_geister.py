@@ -25,6 +25,54 @@ from .environment import make_env
 N_ACTIONS = 214
 SCALAR_DIM = 18
 BOARD_SHAPE = (7, 6, 6)
+REC_CAP = 202                            # 2 layout turns + 200 moves
+_JOB_ARGS = {'player': [0, 1], 'model_id': {0: -1, 1: -1}}
+
+
+def _make_rec_buffers(G):
+    """Columnar per-game recording buffers (turn-based episode format)."""
+    rec = {
+        'scalar': np.zeros((G, REC_CAP, SCALAR_DIM), np.uint8),
+        'board': np.zeros((G, REC_CAP) + BOARD_SHAPE, np.uint8),
+        'mask': np.zeros((G, REC_CAP, N_ACTIONS), bool),
+        'turn': np.zeros((G, REC_CAP), np.int8),
+        'action': np.zeros((G, REC_CAP), np.int16),
+        'prob': np.zeros((G, REC_CAP), np.float32),
+        'value': np.zeros((G, REC_CAP), np.float32),
+    }
+    return rec, np.zeros(G, np.int32)
+
+
+def _record_round(rec, rec_len, gar, scalar, board, mask, parity, actions,
+                  probs, values):
+    """Scatter one step of every game into the columnar buffers."""
+    rows = np.minimum(rec_len, REC_CAP - 1)
+    rec['scalar'][gar, rows] = scalar.astype(np.uint8)
+    rec['board'][gar, rows] = board.astype(np.uint8)
+    rec['mask'][gar, rows] = mask == 0.0
+    rec['turn'][gar, rows] = parity
+    rec['action'][gar, rows] = actions.astype(np.int16)
+    rec['prob'][gar, rows] = probs
+    rec['value'][gar, rows] = values
+    rec_len += 1
+
+
+def _package_columnar(rec, g, S, outcome, gamma):
+    """One game's recording -> a columnar turn-based episode dict."""
+    if S == 0:
+        return None
+    acc, rets = 0.0, np.empty(S, np.float32)
+    for t in range(S - 1, -1, -1):       # constant -0.01 both players
+        acc = -0.01 + gamma * acc
+        rets[t] = acc
+    ep = {'args': _JOB_ARGS, 'steps': S, 'outcome': outcome,
+          'columnar': True, 'turn_based': True,
+          'n_actions': N_ACTIONS, 'n_players': 2,
+          'reward': np.full((S, 2), -0.01, np.float32),
+          'return': np.stack([rets, rets], axis=1)}
+    for k, buf in rec.items():
+        ep[k] = buf[g, :S].copy()
+    return ep
 
 
 def _geister_env_worker(conn, shm_names, n_games, args, seed):
@@ -148,37 +196,13 @@ def _geister_vec_worker(conn, shm_names, n_games, args, seed):
     # HANDYRL_GEISTER_COLUMNAR=0 restores reference-format moments
     columnar = os.environ.get('HANDYRL_GEISTER_COLUMNAR', '1') == '1'
     moments = [[] for _ in range(G)]
-    CAP = 202                                # 2 layout turns + 200 moves
     gar = np.arange(G)
     if columnar:
-        rec = {
-            'scalar': np.zeros((G, CAP, SCALAR_DIM), np.uint8),
-            'board': np.zeros((G, CAP) + BOARD_SHAPE, np.uint8),
-            'mask': np.zeros((G, CAP, N_ACTIONS), bool),
-            'turn': np.zeros((G, CAP), np.int8),
-            'action': np.zeros((G, CAP), np.int16),
-            'prob': np.zeros((G, CAP), np.float32),
-            'value': np.zeros((G, CAP), np.float32),
-        }
-        rec_len = np.zeros(G, np.int32)
+        rec, rec_len = _make_rec_buffers(G)
 
     def package(g, outcome):
         if columnar:
-            S = int(rec_len[g])
-            if S == 0:
-                return None
-            acc, rets = 0.0, np.empty(S, np.float32)
-            for t in range(S - 1, -1, -1):   # constant -0.01 both players
-                acc = -0.01 + gamma * acc
-                rets[t] = acc
-            ep = {'args': job_args, 'steps': S, 'outcome': outcome,
-                  'columnar': True, 'turn_based': True,
-                  'n_actions': N_ACTIONS, 'n_players': 2,
-                  'reward': np.full((S, 2), -0.01, np.float32),
-                  'return': np.stack([rets, rets], axis=1)}
-            for k, buf in rec.items():
-                ep[k] = buf[g, :S].copy()
-            return ep
+            return _package_columnar(rec, g, int(rec_len[g]), outcome, gamma)
         ms = moments[g]
         if not ms:
             return None
@@ -221,15 +245,8 @@ def _geister_vec_worker(conn, shm_names, n_games, args, seed):
 
         actions = res_v[:, 0].astype(np.int64)
         if columnar:
-            rows = np.minimum(rec_len, CAP - 1)
-            rec['scalar'][gar, rows] = scalar_v.astype(np.uint8)
-            rec['board'][gar, rows] = board_v.astype(np.uint8)
-            rec['mask'][gar, rows] = mask_v == 0.0
-            rec['turn'][gar, rows] = parity_v
-            rec['action'][gar, rows] = actions.astype(np.int16)
-            rec['prob'][gar, rows] = res_v[:, 1]
-            rec['value'][gar, rows] = res_v[:, 2]
-            rec_len += 1
+            _record_round(rec, rec_len, gar, scalar_v, board_v, mask_v,
+                          parity_v, actions, res_v[:, 1], res_v[:, 2])
         else:
             for g in range(G):
                 p = int(parity_v[g])
@@ -365,8 +382,6 @@ class GeisterActorPool:
     turn-based.  No worker processes — safe to start from a thread of an
     already-CUDA-initialized learner (unlike a forking pool)."""
 
-    CAP = 202                                # 2 layout turns + 200 moves
-
     def __init__(self, model, args, n_games=256, device=None, seed=0):
         self.args = args
         self.device = device if device is not None else (
@@ -376,45 +391,18 @@ class GeisterActorPool:
         from .envs.vec_geister import GeisterVecEnv
         self.vec = GeisterVecEnv(n_games, seed=seed)
         self.vec.reset_games(np.arange(n_games))
-        G, CAP = n_games, self.CAP
+        G = n_games
         self.n_games = G
         self.gamma = args.get('gamma', 0.8)
-        self.rec = {
-            'scalar': np.zeros((G, CAP, SCALAR_DIM), np.uint8),
-            'board': np.zeros((G, CAP) + BOARD_SHAPE, np.uint8),
-            'mask': np.zeros((G, CAP, N_ACTIONS), bool),
-            'turn': np.zeros((G, CAP), np.int8),
-            'action': np.zeros((G, CAP), np.int16),
-            'prob': np.zeros((G, CAP), np.float32),
-            'value': np.zeros((G, CAP), np.float32),
-        }
-        self.rec_len = np.zeros(G, np.int32)
+        self.rec, self.rec_len = _make_rec_buffers(G)
         self._gar = np.arange(G)
         self._mask_v = np.empty((G, N_ACTIONS), np.float32)
         self._res_v = np.empty((G, 4), np.float32)
-        self._job_args = {'player': [0, 1], 'model_id': {0: -1, 1: -1}}
         self.engine = BatchedDRCEngine(model, self.device, G)
         self.completed = []
         self.frames = 0
         self.episodes_done = 0
         self.calls_per_vec_step = 1
-
-    def _package(self, g, outcome):
-        S = int(self.rec_len[g])
-        if S == 0:
-            return None
-        acc, rets = 0.0, np.empty(S, np.float32)
-        for t in range(S - 1, -1, -1):       # constant -0.01 both players
-            acc = -0.01 + self.gamma * acc
-            rets[t] = acc
-        ep = {'args': self._job_args, 'steps': S, 'outcome': outcome,
-              'columnar': True, 'turn_based': True,
-              'n_actions': N_ACTIONS, 'n_players': 2,
-              'reward': np.full((S, 2), -0.01, np.float32),
-              'return': np.stack([rets, rets], axis=1)}
-        for k, buf in self.rec.items():
-            ep[k] = buf[g, :S].copy()
-        return ep
 
     @torch.no_grad()
     def step_once(self):
@@ -425,8 +413,9 @@ class GeisterActorPool:
         if len(done_idx):
             ocs = vec.outcomes(done_idx)
             for k, g in enumerate(done_idx):
-                ep = self._package(g, {0: float(ocs[k, 0]),
-                                       1: float(ocs[k, 1])})
+                ep = _package_columnar(self.rec, g, int(self.rec_len[g]),
+                                       {0: float(ocs[k, 0]),
+                                        1: float(ocs[k, 1])}, self.gamma)
                 if ep is not None:
                     self.completed.append(ep)
                     self.episodes_done += 1
@@ -439,16 +428,9 @@ class GeisterActorPool:
         self.engine.infer(scalar, board, self._mask_v, parity, reset_flags,
                           self._res_v)
         actions = self._res_v[:, 0].astype(np.int64)
-        rows = np.minimum(self.rec_len, self.CAP - 1)
-        gar = self._gar
-        self.rec['scalar'][gar, rows] = scalar.astype(np.uint8)
-        self.rec['board'][gar, rows] = board.astype(np.uint8)
-        self.rec['mask'][gar, rows] = self._mask_v == 0.0
-        self.rec['turn'][gar, rows] = parity
-        self.rec['action'][gar, rows] = actions.astype(np.int16)
-        self.rec['prob'][gar, rows] = self._res_v[:, 1]
-        self.rec['value'][gar, rows] = self._res_v[:, 2]
-        self.rec_len += 1
+        _record_round(self.rec, self.rec_len, self._gar, scalar, board,
+                      self._mask_v, parity, actions, self._res_v[:, 1],
+                      self._res_v[:, 2])
         vec.step(actions)
         self.frames += G
         return G
@@ -522,7 +504,7 @@ class GeisterMultiProcPool:
 
         self.model = None
         self.device = None
-        self.hidden = None
+        self.engines = None
         self.completed = []
         self.frames = 0
         self.episodes_done = 0
