@@ -312,7 +312,8 @@ def test_geister_pool_graphed_gpu():
     try:
         model = Environment().net().cuda().eval()
         pool.attach(model, torch.device('cuda', 0))
-        assert pool._graphs[0] is not None, 'geister actor graph must capture'
+        assert pool.engines[0]._graph is not None, \
+            'geister actor graph must capture'
         for _ in range(3000):
             pool.step_once()
             if pool.episodes_done >= 3:
