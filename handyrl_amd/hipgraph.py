@@ -312,6 +312,9 @@ class GraphedRecurrentTrainStep:
             'outcome': torch.zeros(B, n_players, device=dev),
             'inv_total': torch.zeros(B, device=dev),
         }
+        self.burn_in = int(trainer.args.get('burn_in_steps', 0))
+        if self.burn_in:
+            self.idx['lead'] = torch.zeros(B, dtype=torch.int64, device=dev)
         # initial hidden is always zeros: allocate once, reuse every step
         self.hidden0 = trainer.wrapped_model.init_hidden([B, n_players])
         if self.hidden0 is not None:
@@ -342,11 +345,10 @@ class GraphedRecurrentTrainStep:
 
     def _fill(self):
         self.replay.publish()
-        pos0, start, length, outcome, inv_total = \
-            self.replay.sample_indices(self.batch_size)
-        self._last_host_idx = {
-            'pos0': pos0, 'start': start, 'length': length,
-            'outcome': outcome, 'inv_total': inv_total}
+        out = self.replay.sample_indices(self.batch_size)
+        keys = ('pos0', 'start', 'length', 'outcome', 'inv_total') + \
+            (('lead',) if self.burn_in else ())
+        self._last_host_idx = dict(zip(keys, out))
         for key, arr in self._last_host_idx.items():
             self.idx[key].copy_(torch.from_numpy(arr), non_blocking=True)
 
@@ -354,7 +356,8 @@ class GraphedRecurrentTrainStep:
         tr = self.trainer
         batch = self.replay.gather_batch(
             self.idx['pos0'], self.idx['start'], self.idx['length'],
-            self.idx['outcome'], self.idx['inv_total'])
+            self.idx['outcome'], self.idx['inv_total'],
+            lead=self.idx.get('lead'))
         if tr.use_amp:
             with torch.autocast('cuda', dtype=torch.bfloat16):
                 losses, dcnt = self._compute_loss(batch, tr.wrapped_model,

@@ -54,8 +54,6 @@ class ColumnRingReplay:
     OUTCOME_P = 2
 
     def __init__(self, args, device, bytes_budget=4 << 30, ingest_thread=False):
-        assert args.get('burn_in_steps', 0) == 0, \
-            'device replay supports no-burn-in training windows'
         self.args = args
         self.device = device
         row_bytes = sum(
@@ -300,6 +298,8 @@ class DeviceReplay(ColumnRingReplay):
     OUTCOME_P = 4
 
     def __init__(self, args, device, bytes_budget=4 << 30, ingest_thread=False):
+        assert args.get('burn_in_steps', 0) == 0, \
+            'DeviceReplay supports feed-forward (no burn-in) training'
         super().__init__(args, device, bytes_budget, ingest_thread)
         from .envs.vec_geese import CHMAP
         self._chmap = torch.from_numpy(CHMAP).to(device)     # (4, 17)
@@ -422,34 +422,52 @@ class TurnDeviceReplay(ColumnRingReplay):
 
     def sample_indices(self, batch_size):
         """Recency-biased picks + window cuts (no seat choice: the mover
-        axis is decided per step by the recorded turn column)."""
+        axis is decided per step by the recorded turn column).  With
+        burn_in_steps > 0 the window is extended backwards by up to
+        burn_in rows and a per-sample ``lead`` pad (the shortfall at the
+        episode start) is returned as a sixth array."""
         args = self.args
         fs = args['forward_steps']
+        burn_in = args.get('burn_in_steps', 0)
         n = len(self.table)
         assert n > 0, 'empty replay'
         pos0 = np.empty(batch_size, dtype=np.int64)
         start = np.empty(batch_size, dtype=np.int64)
         length = np.empty(batch_size, dtype=np.int64)
+        lead = np.empty(batch_size, dtype=np.int64)
         outcome = np.empty((batch_size, 2), dtype=np.float32)
         inv_total = np.empty(batch_size, dtype=np.float32)
         m = min(n, args['maximum_episodes'])
         for b in range(batch_size):
             p0, steps, oc, train_st, ed = self._pick_window(m, fs)
-            pos0[b] = p0 + train_st
-            start[b] = train_st
-            length[b] = ed - train_st
+            st = max(0, train_st - burn_in)
+            pos0[b] = p0 + st
+            start[b] = st
+            length[b] = ed - st
+            lead[b] = burn_in - (train_st - st)
             outcome[b] = oc
             inv_total[b] = 1.0 / steps
         self._update_reader_floor()
-        return pos0, start, length, outcome, inv_total
+        if burn_in == 0:
+            return pos0, start, length, outcome, inv_total
+        return pos0, start, length, outcome, inv_total, lead
 
-    def gather_batch(self, pos0, start, length, outcome, inv_total):
+    def gather_batch(self, pos0, start, length, outcome, inv_total,
+                     lead=None):
         """Build the turn-based (B, T, P, ...) batch on device: obs/prob/
         action/action_mask on the mover axis (P=1), value-side fields on
         the player axis (P=2).  All inputs are device tensors of length B;
         every op is hipGraph-capturable.  Out-of-window rows read garbage
         ring memory, so every float path uses torch.where (never
-        multiply-by-mask, which leaks NaN*0)."""
+        multiply-by-mask, which leaks NaN*0).
+
+        With ``lead`` (burn_in configs) the batch spans burn_in +
+        forward_steps rows: column t maps to window row t - lead[b], rows
+        before the window (burn-in shortfall at the episode start) pad
+        like make_batch's prefix padding (v -> 0, not outcome)."""
+        if lead is not None:
+            return self._gather_burnin(pos0, start, length, outcome,
+                                       inv_total, lead)
         args = self.args
         B = pos0.shape[0]
         T = args['forward_steps']
@@ -500,6 +518,71 @@ class TurnDeviceReplay(ColumnRingReplay):
             'reward': rew.reshape(B, T, 2, 1),
             'return': ret.reshape(B, T, 2, 1),
             'episode_mask': in_range.float().view(B, T, 1, 1),
+            'turn_mask': tmask.reshape(B, T, 2, 1),
+            'observation_mask': tmask.reshape(B, T, 2, 1).clone(),
+            'action_mask': mover(amask, 214),
+            'progress': progress.view(B, T, 1),
+        }
+
+    def _gather_burnin(self, pos0, start, length, outcome, inv_total, lead):
+        """Burn-in variant of gather_batch: T = burn_in + forward_steps,
+        with a per-sample leading pad."""
+        args = self.args
+        B = pos0.shape[0]
+        T = args['burn_in_steps'] + args['forward_steps']
+        dev = self.device
+        t_range = torch.arange(T, device=dev)
+        k = t_range.unsqueeze(0) - lead.unsqueeze(1)            # (B, T)
+        valid = (k >= 0) & (k < length.unsqueeze(1))
+        after = k >= length.unsqueeze(1)                        # tail pad
+        rows = (pos0.unsqueeze(1) + k.clamp(min=0)) % self.ring_T
+        flat = rows.reshape(-1)
+        in_r = valid.reshape(-1)                                # (B*T,)
+        aft = after.reshape(-1)
+
+        scalar = self.scalar[flat].float() * in_r.unsqueeze(1)
+        board = self.board[flat].float() * in_r.view(-1, 1, 1, 1)
+        legal = self.mask[flat] & in_r.unsqueeze(1)
+        amask = torch.where(legal, torch.zeros((), device=dev),
+                            torch.full((), 1e32, device=dev))
+        turn = self.turn[flat].long().clamp(0, 1)
+        act = (self.action[flat].long() * in_r.long())
+        prob = torch.where(in_r, self.prob[flat],
+                           torch.ones((), device=dev))
+        oc_bt = outcome.unsqueeze(1).expand(B, T, 2).reshape(-1, 2)
+        tmask = torch.zeros(B * T, 2, device=dev)
+        tmask.scatter_(1, turn.unsqueeze(1), in_r.float().unsqueeze(1))
+        v_m = torch.zeros(B * T, 2, device=dev)
+        v_m.scatter_(1, turn.unsqueeze(1),
+                     torch.where(in_r, self.value[flat],
+                                 torch.zeros((), device=dev)).unsqueeze(1))
+        # prefix pads stay 0; only rows PAST the episode splice the outcome
+        v = torch.where(in_r.unsqueeze(1), v_m,
+                        torch.where(aft.unsqueeze(1), oc_bt,
+                                    torch.zeros((), device=dev)))
+        rew = torch.where(in_r.unsqueeze(1), self.reward[flat],
+                          torch.zeros((), device=dev))
+        ret = torch.where(in_r.unsqueeze(1), self.ret[flat],
+                          torch.zeros((), device=dev))
+        progress = torch.where(
+            in_r,
+            (start.unsqueeze(1) + k).reshape(-1).float() *
+            inv_total.unsqueeze(1).expand(B, T).reshape(-1),
+            torch.ones((), device=dev))
+
+        def mover(x, *tail):                                    # (B, T, 1, ...)
+            return x.reshape(B, T, 1, *tail)
+
+        return {
+            'observation': {'scalar': mover(scalar, 18),
+                            'board': mover(board, 7, 6, 6)},
+            'selected_prob': mover(prob, 1),
+            'value': v.reshape(B, T, 2, 1),
+            'action': mover(act, 1),
+            'outcome': outcome.view(B, 1, 2, 1),
+            'reward': rew.reshape(B, T, 2, 1),
+            'return': ret.reshape(B, T, 2, 1),
+            'episode_mask': valid.float().view(B, T, 1, 1),
             'turn_mask': tmask.reshape(B, T, 2, 1),
             'observation_mask': tmask.reshape(B, T, 2, 1).clone(),
             'action_mask': mover(amask, 214),

@@ -241,8 +241,11 @@ class Trainer:
         self._replay_step = None
         if args.get('replay') == 'device':
             from .replay import DeviceReplay, TurnDeviceReplay
-            if args.get('burn_in_steps', 0):
-                raise ValueError("replay: 'device' requires burn_in_steps 0")
+            if args.get('burn_in_steps', 0) and \
+                    not args['turn_based_training']:
+                raise ValueError(
+                    "replay: 'device' supports burn_in_steps only on the "
+                    "turn-based path")
             budget = int(args.get('replay_bytes',
                                   (4 << 30) if self.device.type == 'cuda'
                                   else (64 << 20)))

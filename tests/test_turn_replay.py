@@ -124,3 +124,71 @@ def test_recurrent_replay_train_step_runs():
         assert torch.isfinite(losses['total'])
         assert float(dcnt) > 0
     assert trainer.steps == 3
+
+
+def test_turn_device_replay_burnin_matches_make_batch():
+    """burn_in_steps > 0: the device gather spans burn_in+forward_steps
+    with per-sample leading pads, matching make_batch's prefix padding."""
+    args = _args(forward_steps=6, burn_in_steps=4)
+    episodes = _episodes(8, seed=13)
+    replay = TurnDeviceReplay(args, torch.device('cpu'), bytes_budget=64 << 20)
+    replay.extend(episodes)
+
+    B = 6
+    random.seed(31)
+    pos0, start, length, outcome, inv_total, lead = replay.sample_indices(B)
+    dev = torch.device('cpu')
+    batch_dev = replay.gather_batch(
+        torch.from_numpy(pos0).to(dev), torch.from_numpy(start).to(dev),
+        torch.from_numpy(length).to(dev), torch.from_numpy(outcome).to(dev),
+        torch.from_numpy(inv_total).to(dev), torch.from_numpy(lead).to(dev))
+
+    table = list(replay.table)
+    sels = []
+    for b in range(B):
+        ep_i = max(i for i, (p0, _st, _oc) in enumerate(table)
+                   if p0 <= pos0[b])
+        ep = episodes[ep_i]
+        st = int(start[b])
+        ed = st + int(length[b])
+        train_st = st + args['burn_in_steps'] - int(lead[b])
+        sel = {'args': ep['args'], 'outcome': ep['outcome'], 'columnar': True,
+               'turn_based': True, 'n_actions': ep['n_actions'],
+               'n_players': 2, 'start': st, 'end': ed,
+               'train_start': train_st, 'total': ep['steps']}
+        for k in ('scalar', 'board', 'mask', 'turn', 'action', 'prob',
+                  'value', 'reward', 'return'):
+            sel[k] = ep[k][st:ed]
+        sels.append(sel)
+    batch_ref = make_batch(sels, args)
+
+    for key in batch_ref:
+        if isinstance(batch_ref[key], dict):
+            for kk in batch_ref[key]:
+                td, tr = batch_dev[key][kk], batch_ref[key][kk]
+                assert tuple(td.shape) == tuple(tr.shape), (key, kk, td.shape,
+                                                            tr.shape)
+                torch.testing.assert_close(td.double(), tr.double(),
+                                           rtol=1e-5, atol=1e-5)
+        else:
+            td, tr = batch_dev[key], batch_ref[key]
+            assert tuple(td.shape) == tuple(tr.shape), (key, td.shape, tr.shape)
+            torch.testing.assert_close(td.double(), tr.double(), rtol=1e-5,
+                                       atol=1e-5,
+                                       msg=lambda m, k=key: '%s: %s' % (k, m))
+
+
+def test_recurrent_replay_train_step_with_burnin():
+    from handyrl_amd.hipgraph import GraphedRecurrentTrainStep
+    from handyrl_amd.train import Trainer
+    args = _args(batch_size=3, forward_steps=6, burn_in_steps=3)
+    episodes = _episodes(8, seed=21)
+    replay = TurnDeviceReplay(args, torch.device('cpu'), bytes_budget=64 << 20)
+    replay.extend(episodes)
+    trainer = Trainer(args, GeisterEnv().net(), device=torch.device('cpu'))
+    step = GraphedRecurrentTrainStep(trainer, replay, args['batch_size'])
+    random.seed(5)
+    for _ in range(2):
+        losses, dcnt = step.step()
+        assert torch.isfinite(losses['total'])
+        assert float(dcnt) > 0
