@@ -107,7 +107,9 @@ def _episode_fields_columnar_turn(ep, args):
     Shapes and dtypes are bit-identical to the moment-dict path (see
     tests/test_columnar_turn.py): obs/prob/action/action_mask carry the
     mover axis (T, 1, ...), value-side fields carry the player axis
-    (T, P, ...)."""
+    (T, P, ...) — reference train.py:57-110 make_batch semantics for the
+    turn_based_training & not observation seat selection
+    (train.py:65-68)."""
     assert args['turn_based_training'] and not args['observation'], \
         'turn-based columnar episodes need turn_based_training w/o observation'
     T = ep['turn'].shape[0]

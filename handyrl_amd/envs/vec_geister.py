@@ -5,8 +5,11 @@ oracle; parity is tested in tests/test_vec_geister.py), restructured as
 struct-of-arrays so the Geister actor pool (handyrl_amd/actor_geister.py)
 can run hundreds of self-play games against ONE batched DRC forward per
 turn — the MI355X replacement for the reference's one-process-per
--environment workers (reference worker.py / generation.py; rules from
-reference handyrl/envs/geister.py).
+-environment workers (reference worker.py / generation.py).  Rule
+semantics follow reference handyrl/envs/geister.py: play()
+geister.py:361-439, legal_actions() :476-490, observation() :495-522,
+outcome() :440-447 — via the local single-game oracle
+handyrl_amd/envs/geister.py, which those parity tests compare against.
 
 State per game: a flat 36-cell board of piece codes (-1 empty, color*2 +
 type), a 16-slot piece position table, per-code piece counts, mover color

@@ -402,7 +402,9 @@ class TurnDeviceReplay(ColumnRingReplay):
     """Turn-based (observation=False) replay for recurrent envs (Geister):
     rows store the MOVER's observation/legality plus per-player value-side
     columns; gather_batch reproduces make_batch's turn-based columnar
-    fields on device (parity-tested in tests/test_turn_replay.py).
+    fields on device (parity-tested in tests/test_turn_replay.py;
+    reference semantics: replay window cut train.py:291-315, batch
+    padding train.py:92-110, turn seat selection train.py:65-68).
 
     Episodes are the columnar turn-based format produced by the Geister
     actor pools (actor_geister.py)."""
