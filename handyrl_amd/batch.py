@@ -114,7 +114,6 @@ def _episode_fields_columnar_turn(ep, args):
         'turn-based columnar episodes need turn_based_training w/o observation'
     T = ep['turn'].shape[0]
     P = int(ep.get('n_players', 2))
-    A = int(ep['n_actions'])
     turn = ep['turn'].astype(np.int64)
     tr = np.arange(T)
 
@@ -140,7 +139,6 @@ def _episode_fields_columnar_turn(ep, args):
                          dtype=np.float32)[..., np.newaxis] / ep['total']
     obs_zeros = {'scalar': np.zeros(ep['scalar'].shape[-1], dtype=np.float32),
                  'board': np.zeros(ep['board'].shape[2:], dtype=np.float32)}
-    del A
     return (obs, obs_zeros, prob, v, act, oc, rew, ret, emask, tmask, omask,
             amask, progress)
 
