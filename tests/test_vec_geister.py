@@ -98,3 +98,71 @@ def test_layout_turns_then_moves():
     assert (m[:, N_MOVE_ACTIONS:] == 1e32).all()
     assert (m[:, :N_MOVE_ACTIONS] == 0.0).any(axis=1).all()
     assert (vec.piece_cnt[:, :] == 4).all()
+
+
+def _paired_after_layouts(layout0=0, layout1=0):
+    """A vec game and an oracle env advanced through identical layouts."""
+    vec = GeisterVecEnv(1, seed=0)
+    vec.reset_games(np.arange(1))
+    env = gz.Environment()
+    for a in (N_MOVE_ACTIONS + layout0, N_MOVE_ACTIONS + layout1):
+        env.play(a)
+        vec.step(np.array([a]))
+    return vec, env
+
+
+def _teleport(vec, env, fx, fy, tx, ty):
+    """Move a piece in BOTH engines (white-box test setup)."""
+    env._relocate(fx, fy, tx, ty)
+    f, t = fx * 6 + fy, tx * 6 + ty
+    code, slot = vec.board[0, f], vec.slot_of[0, f]
+    vec.board[0, f] = -1
+    vec.slot_of[0, f] = -1
+    vec.board[0, t] = code
+    vec.slot_of[0, t] = slot
+    vec.piece_pos[0, slot] = t
+
+
+def _remove(vec, env, x, y):
+    env._remove(x, y)
+    c = x * 6 + y
+    code, slot = int(vec.board[0, c]), int(vec.slot_of[0, c])
+    vec.board[0, c] = -1
+    vec.slot_of[0, c] = -1
+    vec.piece_pos[0, slot] = -1
+    vec.piece_cnt[0, code] -= 1
+
+
+def test_blue_goal_exit_wins():
+    vec, env = _paired_after_layouts()
+    # layout 0 puts BLACK blues on B2,C2,D2,E2; walk the B2 blue to A6
+    _teleport(vec, env, 1, 1, 0, 5)          # B2 -> A6 (goal-adjacent)
+    a = env._compose_action(0, 5, 0, gz.BLACK)   # step off through (-1, 5)
+    assert env.legal(a)
+    assert vec.legal_masks()[0, a] == 0.0
+    env.play(a)
+    vec.step(np.array([a]))
+    assert env.terminal() and bool(vec.over[0])
+    assert env.outcome() == {0: 1, 1: -1}
+    oc = vec.outcomes(np.array([0]))[0]
+    assert oc[0] == 1.0 and oc[1] == -1.0
+
+
+def test_capturing_all_reds_loses():
+    vec, env = _paired_after_layouts()
+    # layout 0: WHITE reds sit on the last four start squares (E6,D6,C6,B6)
+    for sq in ('E6', 'D6', 'C6'):
+        x, y = env._sq_parse(sq)
+        _remove(vec, env, x, y)
+    # move a BLACK piece next to white's last red at B6 (x=1, y=5)
+    _teleport(vec, env, 1, 1, 1, 4)          # B2 -> B5
+    a = env._compose_action(1, 4, 2, gz.BLACK)   # B5 -> B6 capture
+    assert env.legal(a)
+    assert vec.legal_masks()[0, a] == 0.0
+    env.play(a)
+    vec.step(np.array([a]))
+    assert env.terminal() and bool(vec.over[0])
+    # black captured ALL of white's reds -> black LOSES
+    assert env.outcome() == {0: -1, 1: 1}
+    oc = vec.outcomes(np.array([0]))[0]
+    assert oc[0] == -1.0 and oc[1] == 1.0
