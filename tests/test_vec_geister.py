@@ -166,3 +166,23 @@ def test_capturing_all_reds_loses():
     assert env.outcome() == {0: -1, 1: 1}
     oc = vec.outcomes(np.array([0]))[0]
     assert oc[0] == -1.0 and oc[1] == 1.0
+
+
+def test_capturing_all_blues_wins():
+    vec, env = _paired_after_layouts()
+    # layout 0: WHITE blues sit on E5,D5,C5,B5
+    for sq in ('E5', 'D5', 'C5'):
+        x, y = env._sq_parse(sq)
+        _remove(vec, env, x, y)
+    # move a BLACK piece next to white's last blue at B5 (x=1, y=4)
+    _teleport(vec, env, 1, 1, 1, 3)          # B2 -> B4
+    a = env._compose_action(1, 3, 2, gz.BLACK)   # B4 -> B5 capture
+    assert env.legal(a)
+    assert vec.legal_masks()[0, a] == 0.0
+    env.play(a)
+    vec.step(np.array([a]))
+    assert env.terminal() and bool(vec.over[0])
+    # black captured ALL of white's blues -> black WINS
+    assert env.outcome() == {0: 1, 1: -1}
+    oc = vec.outcomes(np.array([0]))[0]
+    assert oc[0] == 1.0 and oc[1] == -1.0
