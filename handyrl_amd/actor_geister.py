@@ -75,7 +75,7 @@ def _package_columnar(rec, g, S, outcome, gamma):
     return ep
 
 
-def _geister_env_worker(conn, shm_names, n_games, args, seed):
+def _geister_env_worker(conn, ep_conn, shm_names, n_games, args, seed):
     from multiprocessing import shared_memory
     from .batch import pack_moments
 
@@ -134,7 +134,11 @@ def _geister_env_worker(conn, shm_names, n_games, args, seed):
             mask_v[g, legal] = 0.0
             parity_v[g] = p
 
-        conn.send(('obs', G, frames_prev, eps_out))
+        if eps_out:
+            # episodes travel on their own pipe, drained by a parent
+            # background thread: the service path never deserializes them
+            ep_conn.send(eps_out)
+        conn.send(('obs', G, frames_prev))
         frames_prev, eps_out = 0, []
         cmd = conn.recv()
         if cmd == 'quit':
@@ -163,7 +167,7 @@ def _geister_env_worker(conn, shm_names, n_games, args, seed):
             frames_prev += 1
 
 
-def _geister_vec_worker(conn, shm_names, n_games, args, seed):
+def _geister_vec_worker(conn, ep_conn, shm_names, n_games, args, seed):
     """Vectorized variant of _geister_env_worker: one GeisterVecEnv steps
     the whole shard (legality, observations, moves, captures and win
     detection as batched numpy ops — envs/vec_geister.py, parity-tested
@@ -237,7 +241,11 @@ def _geister_vec_worker(conn, shm_names, n_games, args, seed):
         vec.legal_masks(out=mask_v)
         parity_v[:] = vec.turn()
 
-        conn.send(('obs', G, frames_prev, eps_out))
+        if eps_out:
+            # episodes travel on their own pipe, drained by a parent
+            # background thread: the service path never deserializes them
+            ep_conn.send(eps_out)
+        conn.send(('obs', G, frames_prev))
         frames_prev, eps_out = 0, []
         cmd = conn.recv()
         if cmd == 'quit':
@@ -462,7 +470,7 @@ class GeisterMultiProcPool:
         worker_fn = _geister_vec_worker if vec else _geister_env_worker
         per = max(1, n_games // workers)
         self.n_per = per
-        self.conns, self.procs, self.shms = [], [], []
+        self.conns, self.ep_conns, self.procs, self.shms = [], [], [], []
         self.views = []
         sizes = {
             'scalar': per * SCALAR_DIM * 4,
@@ -492,24 +500,32 @@ class GeisterMultiProcPool:
             }
             self.views.append(views)
             parent_conn, child_conn = mp.Pipe(duplex=True)
+            ep_parent, ep_child = mp.Pipe(duplex=False)
             proc = mp.Process(
                 target=worker_fn,
-                args=(child_conn, {k: s.name for k, s in shm.items()},
+                args=(child_conn, ep_child,
+                      {k: s.name for k, s in shm.items()},
                       per, args, seed + 131 * w),
                 daemon=True)
             proc.start()
             child_conn.close()
+            ep_child.close()
             self.conns.append(parent_conn)
+            self.ep_conns.append(ep_parent)
             self.procs.append(proc)
 
         self.model = None
         self.device = None
         self.engines = None
         self.completed = []
+        self._completed_lock = __import__('threading').Lock()
         self.frames = 0
         self.episodes_done = 0
         self.rr = 0
         self.calls_per_vec_step = workers
+        drain = __import__('threading').Thread(target=self._drain_episodes,
+                                               daemon=True)
+        drain.start()
 
     def attach(self, model, device):
         self.model = model
@@ -530,19 +546,33 @@ class GeisterMultiProcPool:
         conn = self.conns[self.rr] if self.conns[self.rr] in ready else ready[0]
         wid = self.conns.index(conn)
         self.rr = (wid + 1) % self.workers
-        tag, G, frames, eps = conn.recv()
+        tag, G, frames = conn.recv()
         assert tag == 'obs'
-        if eps:
-            self.completed.extend(eps)
-            self.episodes_done += len(eps)
         self.frames += frames
         self._infer(wid)
         conn.send('go')
         return frames
 
+    def _drain_episodes(self):
+        import multiprocessing.connection as mpc
+        while True:
+            try:
+                ready = mpc.wait(self.ep_conns, timeout=1.0)
+            except OSError:
+                return
+            for conn in ready:
+                try:
+                    eps = conn.recv()
+                except (EOFError, OSError):
+                    return
+                with self._completed_lock:
+                    self.completed.extend(eps)
+                    self.episodes_done += len(eps)
+
     def harvest(self):
-        out = self.completed
-        self.completed = []
+        with self._completed_lock:
+            out = self.completed
+            self.completed = []
         return out
 
     def refresh_weights(self):
