@@ -13,10 +13,11 @@ def test_bench_two_rank_gloo():
     env = dict(os.environ)
     env.pop('RANK', None)
     env.pop('WORLD_SIZE', None)
+    port = str(20000 + os.getpid() % 20000)
     res = subprocess.run(
         [sys.executable, '-m', 'torch.distributed.run', '--nnodes=1',
          '--nproc-per-node', '2', '--master-addr', '127.0.0.1',
-         '--master-port', '29611', 'bench.py', '--gpus', '2',
+         '--master-port', port, 'bench.py', '--gpus', '2',
          '--steps', '2', '--warmup', '1', '--envs', '48',
          '--batch-size', '4', '--forward-steps', '8'],
         cwd=REPO, env=env, capture_output=True, text=True, timeout=420)
