@@ -121,3 +121,65 @@ def test_make_batch_matches_reference():
         torch.testing.assert_close(ours[key].double(), theirs[key].double(),
                                    rtol=0, atol=0,
                                    msg=lambda m, key=key: '%s: %s' % (key, m))
+
+
+def test_recurrent_compute_loss_matches_reference():
+    """RNN path parity: Geister episodes (DRC hidden state, turn-based,
+    partial observation) through our compute_loss/forward_prediction and
+    the reference's own — identical losses and gradients."""
+    sys.path.insert(0, REFERENCE)
+    try:
+        from handyrl.train import compute_loss as ref_compute_loss
+        from handyrl.model import ModelWrapper as RefWrapper
+    finally:
+        sys.path.remove(REFERENCE)
+
+    from handyrl_amd.batch import make_batch, EpisodeBuffer
+    from handyrl_amd.generation import Generator
+    from handyrl_amd.model import ModelWrapper
+    from handyrl_amd.train import compute_loss
+    from handyrl_amd.envs import geister
+
+    args = _args(policy_target='UPGO', value_target='TD',
+                 forward_steps=6, burn_in_steps=2)
+    env = geister.Environment()
+    torch.manual_seed(4)
+    net = env.net()
+    gen = Generator(env, args)
+    models = {p: ModelWrapper(copy.deepcopy(net)) for p in env.players()}
+    job = {'player': env.players(), 'model_id': {p: 1 for p in env.players()}}
+    buf = EpisodeBuffer(args)
+    random.seed(9)
+    buf.extend([gen.generate(models, job) for _ in range(4)])
+    B = 4
+    batch = make_batch([buf.select_episode() for _ in range(B)], args)
+
+    net_ours = copy.deepcopy(net)
+    net_ref = copy.deepcopy(net)
+    wrap_ours = ModelWrapper(net_ours)
+    wrap_ref = RefWrapper(net_ref)
+    P = batch['value'].size(2)
+    torch.manual_seed(2)
+    losses_ours, dcnt_ours = compute_loss(
+        batch, wrap_ours, wrap_ours.init_hidden([B, P]), args)
+    torch.manual_seed(2)
+    losses_ref, dcnt_ref = ref_compute_loss(
+        {k: (v.clone() if torch.is_tensor(v) else
+             {kk: vv.clone() for kk, vv in v.items()})
+         for k, v in batch.items()},
+        wrap_ref, wrap_ref.init_hidden([B, P]), args)
+
+    assert float(dcnt_ours) == pytest.approx(float(dcnt_ref))
+    for key in losses_ref:
+        torch.testing.assert_close(
+            losses_ours[key].double(), losses_ref[key].double(),
+            rtol=1e-4, atol=1e-5,
+            msg=lambda m, key=key: '%s: %s' % (key, m))
+    losses_ours['total'].backward()
+    losses_ref['total'].backward()
+    for (n, p_o), (_, p_r) in zip(net_ours.named_parameters(),
+                                  net_ref.named_parameters()):
+        if p_o.grad is None and p_r.grad is None:
+            continue
+        torch.testing.assert_close(p_o.grad, p_r.grad, rtol=1e-4, atol=1e-6,
+                                   msg=lambda m, n=n: '%s: %s' % (n, m))
