@@ -183,3 +183,54 @@ def test_recurrent_compute_loss_matches_reference():
             continue
         torch.testing.assert_close(p_o.grad, p_r.grad, rtol=1e-4, atol=1e-6,
                                    msg=lambda m, n=n: '%s: %s' % (n, m))
+
+
+@pytest.mark.parametrize('env_name,over', [
+    ('parallel_tictactoe', {'turn_based_training': False}),   # simultaneous
+    ('tictactoe', {'observation': True}),                     # observer rows
+])
+def test_other_training_families_match_reference(env_name, over):
+    """Simultaneous-move (ParallelTicTacToe) and observation=True
+    (observer-trained) configs produce reference-identical losses."""
+    sys.path.insert(0, REFERENCE)
+    try:
+        from handyrl.train import compute_loss as ref_compute_loss
+        from handyrl.model import ModelWrapper as RefWrapper
+    finally:
+        sys.path.remove(REFERENCE)
+
+    from handyrl_amd.batch import make_batch, EpisodeBuffer
+    from handyrl_amd.generation import Generator
+    from handyrl_amd.model import ModelWrapper
+    from handyrl_amd.train import compute_loss
+    import importlib
+    env_mod = importlib.import_module('handyrl_amd.envs.%s' % env_name)
+
+    args = _args(policy_target='TD', value_target='TD', **over)
+    env = env_mod.Environment()
+    torch.manual_seed(7)
+    net = env.net()
+    gen = Generator(env, args)
+    models = {p: ModelWrapper(copy.deepcopy(net)) for p in env.players()}
+    job = {'player': env.players(), 'model_id': {p: 1 for p in env.players()}}
+    buf = EpisodeBuffer(args)
+    random.seed(17)
+    buf.extend([gen.generate(models, job) for _ in range(5)])
+    B = 4
+    batch = make_batch([buf.select_episode() for _ in range(B)], args)
+
+    net_ours = copy.deepcopy(net)
+    net_ref = copy.deepcopy(net)
+    torch.manual_seed(3)
+    losses_ours, dcnt_ours = compute_loss(batch, ModelWrapper(net_ours),
+                                          None, args)
+    torch.manual_seed(3)
+    losses_ref, dcnt_ref = ref_compute_loss(
+        {k: v.clone() for k, v in batch.items()}, RefWrapper(net_ref),
+        None, args)
+    assert float(dcnt_ours) == pytest.approx(float(dcnt_ref))
+    for key in losses_ref:
+        torch.testing.assert_close(
+            losses_ours[key].double(), losses_ref[key].double(),
+            rtol=1e-4, atol=1e-5,
+            msg=lambda m, key=key: '%s: %s' % (key, m))
