@@ -63,7 +63,7 @@ class GeeseActorPool:
             self.fused, self.graphed = engine
         elif model is not None:
             if self.device.type == 'cuda' and os.environ.get('HANDYRL_NO_FUSED') != '1' \
-                    and hasattr(model, 'stem') and ops.available():
+                    and hasattr(model, 'conv0') and ops.available():
                 from .models.geese_net import GeeseFusedEval
                 self.fused = GeeseFusedEval(model, self.device)
             if use_graphs and self.device.type == 'cuda':

@@ -36,12 +36,15 @@ class ConvBN(nn.Module):
 
 
 class ConvHead(nn.Module):
-    """1x1 conv -> LeakyReLU -> fully-connected head (reference tictactoe.py:35-49)."""
+    """1x1 conv -> LeakyReLU -> fully-connected head (reference
+    tictactoe.py:35-49).  The 1x1 conv nests as ``conv.conv`` so the
+    state_dict layout matches reference checkpoints
+    (tests/test_checkpoint_compat.py)."""
 
     def __init__(self, shape, mid_filters, outputs):
         super().__init__()
         ch, hh, ww = shape
-        self.conv = nn.Conv2d(ch, mid_filters, 1)
+        self.conv = ConvBN(ch, mid_filters, ksize=1, bn=False)
         self.fc = nn.Linear(mid_filters * hh * ww, outputs, bias=False)
 
     def forward(self, x):
@@ -54,13 +57,14 @@ class SimpleConv2dModel(nn.Module):
 
     def __init__(self, ch_in=3, filters=32, blocks=3, actions=9):
         super().__init__()
-        self.stem = nn.Conv2d(ch_in, filters, 3, padding=1)
+        # attribute named 'conv' for reference-checkpoint compatibility
+        self.conv = nn.Conv2d(ch_in, filters, 3, padding=1)
         self.blocks = nn.ModuleList(ConvBN(filters, filters) for _ in range(blocks))
         self.head_p = ConvHead((filters, 3, 3), 2, actions)
         self.head_v = ConvHead((filters, 3, 3), 1, 1)
 
     def forward(self, x, hidden=None):
-        h = F.relu(self.stem(x))
+        h = F.relu(self.conv(x))
         for blk in self.blocks:
             h = F.relu(blk(h))
         return {'policy': self.head_p(h), 'value': torch.tanh(self.head_v(h))}

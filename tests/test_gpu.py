@@ -110,7 +110,7 @@ def test_geese_fused_eval_matches_eager():
     torch.manual_seed(2)
     net = GeeseNet().cuda()
     # non-trivial BN stats
-    for layer in [net.stem] + list(net.blocks):
+    for layer in [net.conv0] + list(net.blocks):
         layer.bn.running_mean.uniform_(-0.2, 0.2)
         layer.bn.running_var.uniform_(0.7, 1.4)
     net.eval()
@@ -368,7 +368,7 @@ def test_fused_eval_canonical_matches_seat_expanded():
     from handyrl_amd.envs.vec_geese import CHMAP
     torch.manual_seed(4)
     net = GeeseNet().cuda()
-    for layer in [net.stem] + list(net.blocks):
+    for layer in [net.conv0] + list(net.blocks):
         layer.bn.running_mean.uniform_(-0.2, 0.2)
         layer.bn.running_var.uniform_(0.7, 1.4)
     net.eval()
