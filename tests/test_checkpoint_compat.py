@@ -88,3 +88,67 @@ def test_geese_net_checkpoint_layout():
                                rtol=1e-5, atol=1e-5)
     torch.testing.assert_close(o_ours['value'], o_ref['value'],
                                rtol=1e-5, atol=1e-5)
+
+
+def test_geister_net_forward_equivalence():
+    """Same weights -> same outputs: our GeisterNet computes the
+    reference's function (policy, value, return and hidden carry)."""
+    sys.path.insert(0, REFERENCE)
+    try:
+        from handyrl.envs.geister import GeisterNet as RefNet
+    finally:
+        sys.path.remove(REFERENCE)
+    from handyrl_amd.models.geister_net import GeisterNet
+
+    torch.manual_seed(11)
+    ref = RefNet()
+    for p in ref.parameters():
+        p.data.uniform_(-0.3, 0.3)
+    for m in ref.modules():
+        if isinstance(m, torch.nn.BatchNorm2d):
+            m.running_mean.uniform_(-0.2, 0.2)
+            m.running_var.uniform_(0.6, 1.4)
+    ours = GeisterNet()
+    ours.load_state_dict(ref.state_dict(), strict=True)
+    ref.eval(); ours.eval()
+
+    B = 9
+    x = {'scalar': torch.rand(B, 18), 'board': torch.rand(B, 7, 6, 6)}
+    h_ref = ref.init_hidden([B])
+    h_ours = ours.init_hidden([B])
+    with torch.no_grad():
+        o_ref = ref({'scalar': x['scalar'].clone(),
+                     'board': x['board'].clone()}, h_ref)
+        o_ours = ours({'scalar': x['scalar'].clone(),
+                       'board': x['board'].clone()}, h_ours)
+    for k in ('policy', 'value', 'return'):
+        torch.testing.assert_close(o_ours[k], o_ref[k], rtol=1e-5, atol=1e-5,
+                                   msg=lambda m, k=k: '%s: %s' % (k, m))
+    ho, co = o_ours['hidden']
+    hr, cr = o_ref['hidden']
+    for a, b in zip(list(ho) + list(co), list(hr) + list(cr)):
+        torch.testing.assert_close(a, b, rtol=1e-5, atol=1e-5)
+
+
+def test_tictactoe_net_forward_equivalence():
+    sys.path.insert(0, REFERENCE)
+    try:
+        from handyrl.envs.tictactoe import SimpleConv2dModel as RefNet
+    finally:
+        sys.path.remove(REFERENCE)
+    from handyrl_amd.envs.tictactoe import SimpleConv2dModel
+    torch.manual_seed(13)
+    ref = RefNet()
+    for p in ref.parameters():
+        p.data.uniform_(-0.4, 0.4)
+    ours = SimpleConv2dModel()
+    ours.load_state_dict(ref.state_dict(), strict=True)
+    ref.eval(); ours.eval()
+    x = torch.rand(7, 3, 3, 3)
+    with torch.no_grad():
+        o_ref = ref(x)
+        o_ours = ours(x, None)
+    torch.testing.assert_close(o_ours['policy'], o_ref['policy'],
+                               rtol=1e-5, atol=1e-5)
+    torch.testing.assert_close(o_ours['value'], o_ref['value'],
+                               rtol=1e-5, atol=1e-5)
