@@ -138,3 +138,89 @@ def test_gloo_dp_training_replicas_stay_synced():
         out, _ = p.communicate(timeout=240)
         assert p.returncode == 0, (rank, out[-3000:])
         assert 'DP_TRAIN_OK' in out
+
+
+DP_EQUIV = textwrap.dedent('''
+    import os, pickle, sys
+    sys.path.insert(0, %r)
+    import random
+    import torch
+    import torch.distributed as dist
+    from handyrl_amd.batch import make_batch, EpisodeBuffer
+    from handyrl_amd.generation import Generator
+    from handyrl_amd.model import ModelWrapper
+    from handyrl_amd.train import Trainer
+    from handyrl_amd.envs import tictactoe
+
+    rank = int(os.environ['RANK'])
+    dist.init_process_group('gloo')
+    args = {
+        'turn_based_training': True, 'observation': False, 'gamma': 0.8,
+        'forward_steps': 8, 'burn_in_steps': 0, 'compress_steps': 4,
+        'entropy_regularization': 0.1, 'entropy_regularization_decay': 0.1,
+        'batch_size': 4, 'minimum_episodes': 2, 'maximum_episodes': 100,
+        'num_batchers': 1, 'lambda': 0.7, 'policy_target': 'TD',
+        'value_target': 'TD', 'seed': 0, 'bf16': False,
+        'compress_episodes': False,
+    }
+    env = tictactoe.Environment()
+    torch.manual_seed(0)
+    net = env.net()
+
+    # identical episode pool on both ranks
+    gen = Generator(env, args)
+    models = {p: ModelWrapper(net) for p in env.players()}
+    job = {'player': env.players(), 'model_id': {p: 1 for p in env.players()}}
+    random.seed(7)
+    eps = [gen.generate(models, job) for _ in range(6)]
+    buf = EpisodeBuffer(args)
+    buf.extend(eps)
+    random.seed(11)
+    sels = [buf.select_episode() for _ in range(8)]
+
+    # single-process oracle on the FULL batch of 8 windows (rank 0 only)
+    import copy
+    from handyrl_amd.train import compute_loss
+    if rank == 0:
+        net_full = copy.deepcopy(net)
+        full = make_batch(sels, args)
+        losses, dcnt = compute_loss(full, ModelWrapper(net_full), None, args)
+        losses['total'].backward()
+        oracle = [p.grad.clone() for p in net_full.parameters()]
+
+    # DP: each rank computes loss/backward on HALF the windows, then the
+    # fused all-reduce sums gradients (pre-clip comparison point)
+    mine = sels[rank * 4:(rank + 1) * 4]
+    batch = make_batch(mine, args)
+    trainer2 = Trainer(args, copy.deepcopy(net), device=torch.device('cpu'),
+                       batcher=False)
+    l2, _ = compute_loss(batch, trainer2.wrapped_model, None, args)
+    trainer2.optimizer.zero_grad(set_to_none=False)
+    l2['total'].backward()
+    trainer2.reducer.allreduce_()
+    if rank == 0:
+        for p, og in zip(trainer2.model.parameters(), oracle):
+            assert torch.allclose(p.grad, og, rtol=1e-5, atol=1e-6), \
+                'DP summed grads != single-process full-batch grads'
+        print('DP_EQUIV_OK')
+    dist.barrier()
+''') % REPO
+
+
+def test_dp_grads_equal_single_process():
+    """Two DP ranks on half-batches produce EXACTLY the single-process
+    full-batch gradients (the reference's summed-loss DataParallel
+    semantics, reference train.py:339-371)."""
+    env = dict(os.environ)
+    env.update({'MASTER_ADDR': '127.0.0.1', 'MASTER_PORT': '29557',
+                'WORLD_SIZE': '2'})
+    procs = []
+    for r in range(2):
+        e = dict(env)
+        e['RANK'] = str(r)
+        procs.append(subprocess.Popen(
+            [sys.executable, '-c', DP_EQUIV], env=e,
+            stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True))
+    outs = [p.communicate(timeout=240)[0] for p in procs]
+    assert all(p.returncode == 0 for p in procs), outs
+    assert 'DP_EQUIV_OK' in outs[0], outs[0][-2000:]
