@@ -226,6 +226,8 @@ class ColumnRingReplay:
             if random.random() < accept:
                 break
         with self._table_lock:
+            if not self.table:
+                raise RuntimeError('replay table emptied during sampling')
             p0, steps, oc = self.table[min(idx, len(self.table) - 1)]
         train_st = random.randrange(1 + max(0, steps - fs))
         ed = min(train_st + fs, steps)
@@ -282,6 +284,7 @@ class ColumnRingReplay:
         with self._table_lock:
             while len(self.table) > maximum:
                 self.table.popleft()
+
 
 class DeviceReplay(ColumnRingReplay):
     """Solo feed-forward replay (the flagship Hungry Geese path): rows
