@@ -60,3 +60,20 @@ def test_primitive_bn_momentum_sequence():
                                rtol=1e-6, atol=1e-6)
     torch.testing.assert_close(bn_mine.running_var, bn_ref.running_var,
                                rtol=1e-6, atol=1e-6)
+
+
+def test_convlstm_split_matches_cell():
+    """The concat-free weight-split reformulation (the fused DRC kernel's
+    math, docs/drc_kernel_plan.md) is exactly ConvLSTMCell.forward."""
+    import torch
+    from handyrl_amd.models.geister_net import (ConvLSTMCell,
+                                                convlstm_cell_split)
+    torch.manual_seed(3)
+    cell = ConvLSTMCell(32, 32)
+    x = torch.randn(5, 32, 6, 6)
+    h = torch.randn(5, 32, 6, 6)
+    c = torch.randn(5, 32, 6, 6)
+    h1, c1 = cell(x, (h, c))
+    h2, c2 = convlstm_cell_split(cell, x, (h, c))
+    torch.testing.assert_close(h1, h2, rtol=1e-6, atol=1e-6)
+    torch.testing.assert_close(c1, c2, rtol=1e-6, atol=1e-6)
