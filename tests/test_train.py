@@ -291,7 +291,8 @@ def test_train_with_device_replay_solo():
 
 
 def test_train_with_device_replay_turn_based():
-    """--train with replay: 'device' on the recurrent Geister config."""
+    """--train with replay: 'device' on the recurrent Geister config,
+    including a burn-in prefix (device-gather lead pads)."""
     script = textwrap.dedent('''
         import sys
         sys.path.insert(0, %r)
@@ -300,7 +301,7 @@ def test_train_with_device_replay_turn_based():
             'env_args': {'env': 'Geister'},
             'train_args': {
                 'turn_based_training': True, 'observation': False,
-                'gamma': 0.8, 'forward_steps': 8, 'burn_in_steps': 0,
+                'gamma': 0.8, 'forward_steps': 8, 'burn_in_steps': 2,
                 'compress_steps': 4, 'entropy_regularization': 0.1,
                 'entropy_regularization_decay': 0.1, 'update_episodes': 20,
                 'batch_size': 2, 'minimum_episodes': 4, 'maximum_episodes': 200,
