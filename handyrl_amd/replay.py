@@ -10,8 +10,11 @@ ops that capture into the training hipGraph — so one graph replay does
 sample-gather + forward + loss + backward + all-reduce + Adam with no
 per-step host batch traffic.
 
-Semantics parity with batch.make_batch's columnar solo path is tested in
-tests/test_replay.py (CPU) and on GPU in tests/test_gpu.py.
+Two ring flavors share the machinery (ColumnRingReplay): DeviceReplay
+(feed-forward solo, the flagship Hungry Geese path) and TurnDeviceReplay
+(turn-based/recurrent, Geister, incl. burn-in windows).  Semantics parity
+with batch.make_batch is tested in tests/test_replay.py and
+tests/test_turn_replay.py (CPU) and on GPU in tests/test_gpu.py.
 """
 
 import queue as queue_mod
