@@ -433,6 +433,17 @@ class Learner:
 
         self.worker = WorkerServer(args) if remote else WorkerCluster(args)
         self.trainer = Trainer(args, copy.deepcopy(self.model))
+        # [amd] save_optimizer: restore the Adam state / step counters
+        # saved alongside restart_epoch (exact resume; the reference
+        # restarts the optimizer cold, train.py:420-423)
+        if self.model_epoch > 0 and self.trainer.optimizer is not None:
+            opt_path = self.optimizer_path(self.model_epoch)
+            if self.args.get('save_optimizer') and os.path.exists(opt_path):
+                st = torch.load(opt_path)
+                self.trainer.optimizer.load_state_dict(st['optimizer'])
+                self.trainer.steps = st['steps']
+                self.trainer.data_cnt_ema = st['data_cnt_ema']
+                print('restored optimizer state at epoch %d' % self.model_epoch)
 
         # GPU actor pool (worker: {type: 'gpu'}): self-play generation runs
         # as batched inference on the learner's GPU instead of CPU worker
@@ -446,6 +457,11 @@ class Learner:
     def latest_model_path(self):
         return os.path.join('models', 'latest.pth')
 
+    def optimizer_path(self, model_id):
+        # [amd] extension sidecar; the .pth files stay plain state_dicts
+        # (reference train.py:441-454 layout is a compatibility contract)
+        return os.path.join('models', str(model_id) + '.opt.pth')
+
     def update_model(self, model, steps):
         print('updated model(%d)' % steps)
         self.model_epoch += 1
@@ -453,6 +469,12 @@ class Learner:
         os.makedirs('models', exist_ok=True)
         torch.save(model.state_dict(), self.model_path(self.model_epoch))
         torch.save(model.state_dict(), self.latest_model_path())
+        if self.args.get('save_optimizer') and \
+                self.trainer.optimizer is not None:
+            torch.save({'optimizer': self.trainer.optimizer.state_dict(),
+                        'steps': self.trainer.steps,
+                        'data_cnt_ema': self.trainer.data_cnt_ema},
+                       self.optimizer_path(self.model_epoch))
 
     def feed_episodes(self, episodes):
         with self.feed_lock:

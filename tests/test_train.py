@@ -192,7 +192,7 @@ def test_restart_epoch_resume():
             'worker': {'num_parallel': 2}, 'lambda': 0.7,
             'policy_target': 'TD', 'value_target': 'TD',
             'eval': {'opponent': ['random']}, 'seed': 0, 'restart_epoch': 0,
-            'bf16': False,
+            'bf16': False, 'save_optimizer': True,
         },
     }
     script = textwrap.dedent('''
@@ -212,11 +212,14 @@ def test_restart_epoch_resume():
                             capture_output=True, text=True, timeout=240)
         assert 'TRAIN_DONE' in r1.stdout, (r1.stdout[-2000:], r1.stderr[-2000:])
         assert os.path.exists(os.path.join(tmp, 'models', '1.pth'))
+        # [amd] save_optimizer sidecar written next to the checkpoint
+        assert os.path.exists(os.path.join(tmp, 'models', '1.opt.pth'))
         r2 = subprocess.run([sys.executable, '-c', script, '1', '2'], cwd=tmp,
                             capture_output=True, text=True, timeout=240)
         assert 'TRAIN_DONE' in r2.stdout, (r2.stdout[-2000:], r2.stderr[-2000:])
         assert os.path.exists(os.path.join(tmp, 'models', '2.pth'))
         assert 'epoch 1' in r2.stdout       # resumed at the loaded epoch
+        assert 'restored optimizer state at epoch 1' in r2.stdout
 
 
 def test_full_gpu_actor_mode_training_run_geister():
