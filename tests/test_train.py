@@ -325,3 +325,28 @@ def test_train_with_device_replay_turn_based():
         assert 'TRAIN_DONE' in out, (out[-3000:], res.stderr[-3000:])
         assert 'updated model(' in out
         assert os.path.exists(os.path.join(tmp, 'models', '1.pth'))
+
+
+def test_device_replay_config_validation():
+    """replay: 'device' rejects unsupported configs with clear errors."""
+    import pytest as _pytest
+    import torch
+    from handyrl_amd.train import Trainer
+    from handyrl_amd.envs.tictactoe import SimpleConv2dModel
+    base = {
+        'turn_based_training': True, 'observation': True, 'gamma': 0.8,
+        'forward_steps': 4, 'burn_in_steps': 0, 'compress_steps': 4,
+        'entropy_regularization': 0.1, 'entropy_regularization_decay': 0.1,
+        'batch_size': 2, 'minimum_episodes': 1, 'maximum_episodes': 10,
+        'num_batchers': 1, 'lambda': 0.7, 'policy_target': 'TD',
+        'value_target': 'TD', 'seed': 0, 'bf16': False,
+        'replay': 'device',
+    }
+    with _pytest.raises(ValueError, match='observation'):
+        Trainer(dict(base), SimpleConv2dModel(), device=torch.device('cpu'))
+    bad = dict(base)
+    bad['turn_based_training'] = False
+    bad['observation'] = False
+    bad['burn_in_steps'] = 2
+    with _pytest.raises(ValueError, match='burn_in'):
+        Trainer(bad, SimpleConv2dModel(), device=torch.device('cpu'))
