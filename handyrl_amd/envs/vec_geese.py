@@ -12,10 +12,38 @@ at (start + i) % CAP), so head-push/tail-pop are O(1) index updates and
 board grids are maintained incrementally.
 """
 
+import os
+
 import numpy as np
 
 from .hungry_geese import (ROWS, COLS, N_CELLS, N_PLAYERS, HUNGER_RATE,
                            MIN_FOOD, MAX_LEN, MAX_STEPS, OPPOSITE)
+
+
+def _load_native_core():
+    """In-tree C++ env core (envs/src/vec_geese_core.cpp): the RNG-free
+    step phases and observation build as native loops.  Bit-equal to the
+    numpy engine (tests/test_vec_geese_native.py); HANDYRL_NO_NATIVE_ENV=1
+    or a missing .so falls back to pure numpy."""
+    if os.environ.get('HANDYRL_NO_NATIVE_ENV') == '1':
+        return None
+    so = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                      '_vec_geese_core.so')
+    if not os.path.exists(so):
+        return None
+    try:
+        import importlib.machinery
+        import importlib.util
+        loader = importlib.machinery.ExtensionFileLoader('_vec_geese_core', so)
+        spec = importlib.util.spec_from_loader('_vec_geese_core', loader)
+        mod = importlib.util.module_from_spec(spec)
+        loader.exec_module(mod)
+        return mod
+    except Exception:            # noqa: BLE001 - numpy fallback
+        return None
+
+
+_CORE = _load_native_core()
 
 CAP = N_CELLS  # a goose can never exceed the board
 
@@ -119,6 +147,12 @@ class GeeseVecEnv:
         """actions: (G, 4) int32; entries for dead seats/finished games ignored."""
         G = self.G
         act = np.asarray(actions, dtype=np.int32)
+        if _CORE is not None:
+            _CORE.step_core(self.body, self.start, self.length, self.alive,
+                            self.last_action, self.prev_head, self.food,
+                            self.step_count, self.over, self.body_grid,
+                            np.ascontiguousarray(act))
+            return self._step_tail()
         live = self.alive & ~self.over[:, None]
 
         self.prev_head = np.where(self.alive, self._head(), -1)
@@ -189,6 +223,13 @@ class GeeseVecEnv:
             self._kill(gi[crash], pi[crash])
         live = self.alive & ~self.over[:, None]
 
+        return self._step_tail()
+
+    def _step_tail(self):
+        """Post-move phases shared by the numpy and native cores: food
+        replenishment (the ONLY rng consumer in step — kept in python so
+        both cores read the identical stream), step counting, scores and
+        termination."""
         # 5) food replenishment onto random free cells (vectorized over the
         # needy games: argmax of iid uniforms over the free set = a uniform
         # free-cell draw, matching the oracle's shuffle-and-take semantics)
@@ -232,6 +273,11 @@ class GeeseVecEnv:
         if not hasattr(self, '_obs_buf'):
             self._obs_buf = np.zeros((G, 17, N_CELLS), dtype=np.uint8)
         obs = self._obs_buf
+        if _CORE is not None:
+            _CORE.observations_core(self.body, self.start, self.length,
+                                    self.alive, self.prev_head, self.food,
+                                    self.body_grid, obs)
+            return obs.reshape(G, 17, ROWS, COLS)
         head_grid = obs[:, 0:4]
         tail_grid = obs[:, 4:8]
         prev_grid = obs[:, 12:16]
