@@ -173,7 +173,7 @@ class ColumnRingReplay:
         traffic under load), so refilling must wait on the event recorded
         after that copy — without it the in-flight DMA reads torn bytes
         and poisons the ring with garbage floats (NaN losses; diagnosed
-        in gpurun_out/learn_race*.log)."""
+        in profiles/learn_race*.log)."""
         if self.device.type != 'cuda':
             return torch.from_numpy(np.concatenate(parts)).to(self.device)
         pin, ev = self._pin.get(key, (None, None))
