@@ -92,8 +92,9 @@ def main():
     batcher = False if device_replay else Batcher(args, buffer)
 
     actor_procs = int(os.environ.get('HANDYRL_ACTOR_PROCS', '6'))
-    # single slot per worker: the faster 2-slot double-buffering is gated
-    # off pending a GPU transport-race fix (BASELINE.md learning sanity)
+    # single slot per worker: 2-slot double-buffering measured SLOWER
+    # end-to-end (it steers self-play into the short-episode regime;
+    # BASELINE.md post-fix slots comparison)
     os.environ.setdefault('HANDYRL_ACTOR_SLOTS', '1')
     mpool = None
     if actor_procs > 0:
