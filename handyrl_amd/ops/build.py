@@ -20,8 +20,10 @@ def build(verbose=False):
     build_dir = os.path.join(here, '_build')
     os.makedirs(build_dir, exist_ok=True)
 
-    sources = sorted(
-        os.path.join(src_dir, f) for f in os.listdir(src_dir) if f.endswith('.hip'))
+    # List sources explicitly: torch's hipify pass writes generated
+    # `*_hip.hip` siblings next to the source, which a bare glob would pick
+    # up on the next build (duplicate kernels / double op registration).
+    sources = [os.path.join(src_dir, 'ext.hip')]
 
     mod = cpp_extension.load(
         name='handyrl_amd_C',
