@@ -21,6 +21,11 @@ import torch.nn as nn
 from .util import map_r, bimap_r
 
 
+def apply_grad_guard(params, counter=None):
+    from .train import apply_grad_guard as _g
+    _g(params, counter)
+
+
 def _copy_into(static, src, non_blocking=True):
     bimap_r(static, src, lambda dst, s: dst.copy_(s, non_blocking=non_blocking))
 
@@ -67,13 +72,11 @@ class GraphedTrainStep:
         tr.optimizer.zero_grad(set_to_none=False)
         losses['total'].backward()
         if tr.grad_guard:
-            # capturable finite-guard: one spike step (inf grads with a
-            # still-finite loss) otherwise turns the clip scale into NaN
-            # and poisons the weights permanently (BASELINE.md diagnosis)
-            for p in tr.params:
-                if p.grad is not None:
-                    torch.nan_to_num_(p.grad, nan=0.0, posinf=1e6,
-                                      neginf=-1e6)
+            # capturable finite-guard + activation counter: one spike step
+            # (inf grads with a still-finite loss) otherwise turns the clip
+            # scale into NaN and poisons the weights permanently
+            # (BASELINE.md diagnosis); fires are reported per epoch
+            apply_grad_guard(tr.params, tr.guard_fires)
         tr.reducer.allreduce_()
         nn.utils.clip_grad_norm_(tr.params, 4.0)
         tr.optimizer.step()
@@ -266,13 +269,11 @@ class GraphedReplayTrainStep:
         tr.optimizer.zero_grad(set_to_none=False)
         losses['total'].backward()
         if tr.grad_guard:
-            # capturable finite-guard: one spike step (inf grads with a
-            # still-finite loss) otherwise turns the clip scale into NaN
-            # and poisons the weights permanently (BASELINE.md diagnosis)
-            for p in tr.params:
-                if p.grad is not None:
-                    torch.nan_to_num_(p.grad, nan=0.0, posinf=1e6,
-                                      neginf=-1e6)
+            # capturable finite-guard + activation counter: one spike step
+            # (inf grads with a still-finite loss) otherwise turns the clip
+            # scale into NaN and poisons the weights permanently
+            # (BASELINE.md diagnosis); fires are reported per epoch
+            apply_grad_guard(tr.params, tr.guard_fires)
         tr.reducer.allreduce_()
         nn.utils.clip_grad_norm_(tr.params, 4.0)
         tr.optimizer.step()
@@ -368,10 +369,7 @@ class GraphedRecurrentTrainStep:
         tr.optimizer.zero_grad(set_to_none=False)
         losses['total'].backward()
         if tr.grad_guard:
-            for p in tr.params:
-                if p.grad is not None:
-                    torch.nan_to_num_(p.grad, nan=0.0, posinf=1e6,
-                                      neginf=-1e6)
+            apply_grad_guard(tr.params, tr.guard_fires)
         tr.reducer.allreduce_()
         nn.utils.clip_grad_norm_(tr.params, 4.0)
         tr.optimizer.step()

@@ -236,13 +236,22 @@ class ColumnRingReplay:
         ed = min(train_st + fs, steps)
         return p0, steps, oc, train_st, ed
 
-    def _update_reader_floor(self):
+    def _update_reader_floor(self, sampled_min=None):
         # reader floor covers this fill AND the previous one (the previous
-        # graph replay may still be in flight when this one is sampled)
+        # graph replay may still be in flight when this one is sampled).
+        # `sampled_min` is the smallest episode base position actually
+        # drawn this fill: the feeder thread may evict (and later overwrite)
+        # the table head between our picks and this call, so table[0] alone
+        # can fail to cover an entry that is already in the in-flight gather.
         with self._table_lock:
             if not self.table:
-                return
-            floor_now = self.table[0][0]
+                floor_now = sampled_min
+            else:
+                floor_now = self.table[0][0]
+                if sampled_min is not None:
+                    floor_now = min(floor_now, sampled_min)
+        if floor_now is None:
+            return
         self._reader_floor = floor_now if self._floor_prev is None \
             else min(floor_now, self._floor_prev)
         self._floor_prev = floor_now
@@ -335,7 +344,7 @@ class DeviceReplay(ColumnRingReplay):
             seat[b] = random.randrange(4)
             outcome[b] = oc
             inv_total[b] = 1.0 / steps
-        self._update_reader_floor()
+        self._update_reader_floor(int((pos0 - start).min()))
         return pos0, start, length, seat, outcome, inv_total
 
     def gather_batch(self, pos0, start, length, seat, outcome, inv_total):
@@ -455,7 +464,7 @@ class TurnDeviceReplay(ColumnRingReplay):
             lead[b] = burn_in - (train_st - st)
             outcome[b] = oc
             inv_total[b] = 1.0 / steps
-        self._update_reader_floor()
+        self._update_reader_floor(int((pos0 - start).min()))
         if burn_in == 0:
             return pos0, start, length, outcome, inv_total
         return pos0, start, length, outcome, inv_total, lead

@@ -184,6 +184,26 @@ def compute_loss(batch, model, hidden, args):
                           targets, batch, args)
 
 
+def apply_grad_guard(params, counter=None):
+    """Finite-guard gradients in place, counting activations.
+
+    Pure tensor ops (hipGraph-capturable): when any gradient holds a
+    non-finite value, `counter` (a device int64 scalar) is incremented by
+    one for the step, so silently-absorbed spike steps are observable
+    (the Trainer reports the count each epoch)."""
+    flag = None
+    for p in params:
+        g = p.grad
+        if g is None:
+            continue
+        if counter is not None:
+            nf = (~torch.isfinite(g)).sum()
+            flag = nf if flag is None else flag + nf
+        torch.nan_to_num_(g, nan=0.0, posinf=1e6, neginf=-1e6)
+    if counter is not None and flag is not None:
+        counter.add_((flag > 0).to(counter.dtype))
+
+
 class Trainer:
     """SGD loop over batches from the episode buffer.
 
@@ -213,6 +233,11 @@ class Trainer:
         # BASELINE.md "learning sanity") becomes a survivable clipped
         # update instead of permanent NaN.  HANDYRL_GRAD_GUARD=0 disables.
         self.grad_guard = os.environ.get('HANDYRL_GRAD_GUARD', '1') == '1'
+        # device-resident activation counter (readable without breaking
+        # graph capture; reported per epoch in train())
+        self.guard_fires = torch.zeros((), dtype=torch.int64,
+                                       device=self.device)
+        self._guard_reported = 0
 
         self.default_lr = 3e-8
         self.data_cnt_ema = args['batch_size'] * args['forward_steps']
@@ -324,10 +349,7 @@ class Trainer:
         self.optimizer.zero_grad(set_to_none=False)
         losses['total'].backward()
         if self.grad_guard:
-            for p in self.params:
-                if p.grad is not None:
-                    torch.nan_to_num_(p.grad, nan=0.0, posinf=1e6,
-                                      neginf=-1e6)
+            apply_grad_guard(self.params, self.guard_fires)
         self.reducer.allreduce_()                 # fused RCCL all-reduce (DP)
         nn.utils.clip_grad_norm_(self.params, 4.0)
         self.optimizer.step()
@@ -356,6 +378,12 @@ class Trainer:
 
         print('loss = %s' % ' '.join(
             [k + ':' + '%.3f' % (l / data_cnt) for k, l in loss_sum.items()]))
+        if self.grad_guard:
+            fires = int(self.guard_fires.item())
+            if fires > self._guard_reported:
+                print('grad guard fired on %d step(s) this epoch '
+                      '(%d total)' % (fires - self._guard_reported, fires))
+                self._guard_reported = fires
 
         global_data_cnt = hdist.allreduce_scalar(data_cnt / (1e-2 + batch_cnt))
         self.data_cnt_ema = self.data_cnt_ema * 0.8 + global_data_cnt * 0.2
