@@ -495,8 +495,16 @@ class Learner:
         self.model_epoch += 1
         self.model = model
         os.makedirs('models', exist_ok=True)
-        torch.save(model.state_dict(), self.model_path(self.model_epoch))
-        torch.save(model.state_dict(), self.latest_model_path())
+        # nets that differ structurally from their reference counterpart
+        # (e.g. GeeseNet drops the redundant under-BN conv biases) export a
+        # reference-layout state dict so saved .pth files load into the
+        # reference net unchanged (the .pth layout is a compat contract)
+        if hasattr(model, 'reference_state_dict'):
+            sd = model.reference_state_dict()
+        else:
+            sd = model.state_dict()
+        torch.save(sd, self.model_path(self.model_epoch))
+        torch.save(sd, self.latest_model_path())
         if self.args.get('save_optimizer') and \
                 self.trainer.optimizer is not None:
             torch.save({'optimizer': self.trainer.optimizer.state_dict(),

@@ -152,3 +152,35 @@ def test_tictactoe_net_forward_equivalence():
                                rtol=1e-5, atol=1e-5)
     torch.testing.assert_close(o_ours['value'], o_ref['value'],
                                rtol=1e-5, atol=1e-5)
+
+
+def test_geese_net_export_to_reference():
+    """The ours -> reference direction: a checkpoint saved by this repo
+    (reference_state_dict) loads into the reference GeeseNet with every
+    key present (no silently-kept random conv biases) and produces
+    identical outputs."""
+    ns = _load_ref_module('handyrl/envs/kaggle/hungry_geese.py',
+                          {'TorusConv2d', 'GeeseNet'})
+    from handyrl_amd.models.geese_net import GeeseNet
+    torch.manual_seed(7)
+    ours = GeeseNet()
+    for p in ours.parameters():
+        p.data.uniform_(-0.5, 0.5)
+    for m in ours.modules():
+        if isinstance(m, torch.nn.BatchNorm2d):
+            m.running_mean.uniform_(-0.3, 0.3)
+            m.running_var.uniform_(0.6, 1.5)
+    sd = ours.reference_state_dict()
+    ref = ns['GeeseNet']()
+    # full key coverage: strict load must succeed (this is what the
+    # reference's strict=False load would silently skip without the export)
+    ref.load_state_dict(sd, strict=True)
+    ours.eval(); ref.eval()
+    x = (torch.rand(23, 17, 7, 11) < 0.2).float()
+    with torch.no_grad():
+        o_ref = ref(x)
+        o_ours = ours(x, None)
+    torch.testing.assert_close(o_ours['policy'], o_ref['policy'],
+                               rtol=1e-5, atol=1e-5)
+    torch.testing.assert_close(o_ours['value'], o_ref['value'],
+                               rtol=1e-5, atol=1e-5)
