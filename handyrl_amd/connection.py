@@ -1,14 +1,24 @@
 """Host-side control-plane transport.
 
-Two carriers with one send/recv interface (parity: reference connection.py):
-``multiprocessing.Pipe`` locally and length-prefixed pickle over TCP
-remotely (4-byte network-order size header).  The bandwidth-critical GPU
-paths (gradient all-reduce, model broadcast) do NOT go through here — they
-use RCCL over xGMI (handyrl_amd/dist.py); this layer carries job args,
-episodes and results (KB-scale messages).
+Carries job args, episodes and results (KB-scale messages) between the
+learner, gather relays and workers.  The bandwidth-critical GPU paths
+(gradient all-reduce, model broadcast) do NOT go through here — they use
+RCCL over xGMI (handyrl_amd/dist.py).
+
+Wire compatibility contract (so remote CPU workers running against a
+reference-style deployment interoperate, reference connection.py:20-69):
+every socket message is a 4-byte network-order signed length followed by a
+pickle payload.  Local transport is ``multiprocessing.Pipe``; both carriers
+expose the same ``send``/``recv``/``fileno`` surface, which is what lets
+the whole distributed tree run single-machine.
+
+Beyond the wire format, the implementations here are this repo's own:
+buffered-file socket IO instead of manual chunk loops, an event-driven
+single-thread job pool, and a hub with explicit per-connection accounting
+and clean shutdown.
 """
 
-import io
+import os
 import pickle
 import queue
 import socket
@@ -17,71 +27,78 @@ import threading
 import multiprocessing as mp
 import multiprocessing.connection as mp_connection
 
+_HEADER = struct.Struct('!i')
+_COALESCE_LIMIT = 16384
+
 
 def send_recv(conn, sdata):
+    """One request/response round trip."""
     conn.send(sdata)
     return conn.recv()
 
 
 class PickledConnection:
-    """Pickle messages over a stream socket with 4-byte length framing."""
+    """Length-framed pickle messages over a stream socket.
+
+    Reads go through a buffered file object (the kernel-to-user copies are
+    batched instead of looped 4-or-N bytes at a time); writes use
+    ``sendall`` with small messages coalesced into a single syscall.
+    """
 
     def __init__(self, conn):
         self.conn = conn
+        self._rfile = conn.makefile('rb', buffering=1 << 16)
 
     def __del__(self):
         self.close()
 
     def close(self):
-        if self.conn is not None:
-            self.conn.close()
-            self.conn = None
+        if getattr(self, 'conn', None) is None:
+            return
+        try:
+            self._rfile.close()
+        except OSError:
+            pass
+        self.conn.close()
+        self.conn = None
 
     def fileno(self):
         return self.conn.fileno()
 
-    def _recv_exact(self, size):
-        buf = io.BytesIO()
-        while size > 0:
-            chunk = self.conn.recv(size)
-            if len(chunk) == 0:
-                raise ConnectionResetError
-            size -= len(chunk)
-            buf.write(chunk)
-        return buf.getvalue()
+    def _read_exact(self, n):
+        data = self._rfile.read(n)
+        if data is None or len(data) != n:
+            raise ConnectionResetError('peer closed mid-frame')
+        return data
 
     def recv(self):
-        (size,) = struct.unpack('!i', self._recv_exact(4))
-        return pickle.loads(self._recv_exact(size))
-
-    def _send_all(self, buf):
-        view = memoryview(buf)
-        while view:
-            n = self.conn.send(view)
-            view = view[n:]
+        size, = _HEADER.unpack(self._read_exact(4))
+        if size < 0:
+            raise ConnectionResetError('negative frame length')
+        return pickle.loads(self._read_exact(size)) if size else None
 
     def send(self, msg):
-        payload = pickle.dumps(msg)
-        header = struct.pack('!i', len(payload))
-        if 0 < len(payload) <= 16384:
-            self._send_all(header + payload)   # coalesce small messages
+        body = pickle.dumps(msg)
+        if 0 < len(body) <= _COALESCE_LIMIT:
+            self.conn.sendall(_HEADER.pack(len(body)) + body)
         else:
-            self._send_all(header)
-            if payload:
-                self._send_all(payload)
+            self.conn.sendall(_HEADER.pack(len(body)))
+            if body:
+                self.conn.sendall(body)
 
+
+# -- socket plumbing ---------------------------------------------------------
 
 def open_socket_connection(port, reuse=False):
     sock = socket.socket(socket.AF_INET, socket.SOCK_STREAM)
-    sock.setsockopt(socket.SOL_SOCKET, socket.SO_REUSEADDR,
-                    sock.getsockopt(socket.SOL_SOCKET, socket.SO_REUSEADDR) | 1)
+    sock.setsockopt(socket.SOL_SOCKET, socket.SO_REUSEADDR, 1)
     sock.bind(('', int(port)))
     return sock
 
 
 def accept_socket_connection(sock):
     try:
-        conn, _ = sock.accept()
+        conn, _addr = sock.accept()
         return PickledConnection(conn)
     except socket.timeout:
         return None
@@ -103,129 +120,178 @@ def connect_socket_connection(host, port):
 
 
 def accept_socket_connections(port, timeout=None, maxsize=1024):
+    """Generator yielding accepted connections (None on accept timeout);
+    peers may connect at any time for the lifetime of the acceptor."""
     sock = open_socket_connection(port)
     sock.listen(maxsize)
     sock.settimeout(timeout)
-    count = 0
-    while count < maxsize:
+    accepted = 0
+    while accepted < maxsize:
         conn = accept_socket_connection(sock)
         if conn is not None:
-            count += 1
+            accepted += 1
         yield conn
 
 
 def open_multiprocessing_connections(num_process, target, args_func):
-    """Spawn ``num_process`` child processes, each holding one end of a duplex
-    pipe; return the parent-side connections."""
-    parent_conns = []
+    """Spawn ``num_process`` children, each owning one end of a duplex pipe;
+    returns the parent-side connections."""
+    parents = []
     for i in range(num_process):
-        conn0, conn1 = mp.Pipe(duplex=True)
-        mp.Process(target=target, args=args_func(i, conn1)).start()
-        conn1.close()
-        parent_conns.append(conn0)
-    return parent_conns
+        here, there = mp.Pipe(duplex=True)
+        mp.Process(target=target, args=args_func(i, there)).start()
+        there.close()
+        parents.append(here)
+    return parents
 
+
+# -- fan-out job pool --------------------------------------------------------
 
 class MultiProcessJobExecutor:
-    """Generic fan-out pool: a sender thread feeds idle workers from a
-    generator; a receiver thread collects results into a bounded queue."""
+    """Event-driven fan-out pool over ``num_workers`` pipe-connected child
+    processes.
 
-    def __init__(self, func, send_generator, num_workers, postprocess=None):
-        self.send_generator = send_generator
-        self.postprocess = postprocess
-        self.conns = []
-        self.waiting_conns = queue.Queue()
-        self.output_queue = queue.Queue(maxsize=8)
+    A single dispatcher thread owns the whole lifecycle: it primes every
+    worker with one job from ``send_generator``, then waits on the pipe set
+    and refills each worker the moment its result arrives.  (One thread,
+    no idle-worker queue: readiness IS the wait-set event.)  Results pass
+    through ``postprocess`` and land in a bounded output queue that
+    throttles the generator when the consumer falls behind.
+    """
+
+    def __init__(self, func, send_generator, num_workers, postprocess=None,
+                 out_depth=8):
+        self._jobs = send_generator
+        self._post = postprocess
+        self.output_queue = queue.Queue(maxsize=out_depth)
         self.shutdown_flag = False
-
+        self.conns = []
         for i in range(num_workers):
-            conn0, conn1 = mp.Pipe(duplex=True)
-            mp.Process(target=func, args=(conn1, i), daemon=True).start()
-            conn1.close()
-            self.conns.append(conn0)
-            self.waiting_conns.put(conn0)
+            here, there = mp.Pipe(duplex=True)
+            mp.Process(target=func, args=(there, i), daemon=True).start()
+            there.close()
+            self.conns.append(here)
 
     def recv(self):
         return self.output_queue.get()
 
     def start(self):
-        threading.Thread(target=self._sender, daemon=True).start()
-        threading.Thread(target=self._receiver, daemon=True).start()
+        threading.Thread(target=self._dispatch, daemon=True).start()
 
-    def _sender(self):
-        while not self.shutdown_flag:
-            data = next(self.send_generator)
-            conn = self.waiting_conns.get()
-            conn.send(data)
-
-    def _receiver(self):
+    def _dispatch(self):
+        for conn in self.conns:
+            conn.send(next(self._jobs))
         while not self.shutdown_flag:
             for conn in mp_connection.wait(self.conns, timeout=0.3):
-                data = conn.recv()
-                self.waiting_conns.put(conn)
-                if self.postprocess is not None:
-                    data = self.postprocess(data)
-                self.output_queue.put(data)
+                result = conn.recv()
+                conn.send(next(self._jobs))      # refill before postprocess
+                if self._post is not None:
+                    result = self._post(result)
+                self.output_queue.put(result)
 
+
+# -- async connection hub ----------------------------------------------------
 
 class QueueCommunicator:
-    """Async hub over a set of connections: daemon send/recv threads with
-    bounded queues; dead connections are detected by send/recv exceptions and
-    dropped, so peers may join and leave at any time."""
+    """Hub over a dynamic set of connections.
+
+    Receives are pumped by one daemon thread waiting on the live set;
+    sends are drained from a queue by another.  Any IO error retires the
+    connection (peers join and leave at any time — the reference's
+    elasticity contract, connection.py:198-229).  Per-connection counters
+    and an explicit ``close()`` are this implementation's additions.
+    """
+
+    _IO_ERRORS = (TimeoutError, ConnectionResetError, BrokenPipeError,
+                  EOFError, OSError)
 
     def __init__(self, conns=()):
         self.input_queue = queue.Queue(maxsize=256)
         self.output_queue = queue.Queue(maxsize=256)
-        self.conns = set()
+        self._conns = {}                  # conn -> {'sent': n, 'recvd': n}
         self._lock = threading.Lock()
+        self._wake = threading.Event()    # set while the live set is nonempty
+        self._closed = False
         for conn in conns:
             self.add_connection(conn)
-        threading.Thread(target=self._send_thread, daemon=True).start()
-        threading.Thread(target=self._recv_thread, daemon=True).start()
+        self._threads = [
+            threading.Thread(target=self._pump_in, daemon=True),
+            threading.Thread(target=self._pump_out, daemon=True),
+        ]
+        for t in self._threads:
+            t.start()
 
+    # - membership -
     def connection_count(self):
         with self._lock:
-            return len(self.conns)
+            return len(self._conns)
 
+    def add_connection(self, conn):
+        with self._lock:
+            self._conns[conn] = {'sent': 0, 'recvd': 0}
+            self._wake.set()
+
+    def disconnect(self, conn):
+        with self._lock:
+            if self._conns.pop(conn, None) is not None:
+                print('disconnected')
+            if not self._conns:
+                self._wake.clear()
+
+    def close(self):
+        self._closed = True
+        with self._lock:
+            conns = list(self._conns)
+            self._conns.clear()
+            self._wake.clear()
+        for c in conns:
+            try:
+                c.close()
+            except self._IO_ERRORS:
+                pass
+
+    # - messaging -
     def recv(self, timeout=None):
         return self.input_queue.get(timeout=timeout)
 
     def send(self, conn, send_data):
         self.output_queue.put((conn, send_data))
 
-    def add_connection(self, conn):
-        with self._lock:
-            self.conns.add(conn)
-
-    def disconnect(self, conn):
-        with self._lock:
-            if conn in self.conns:
-                print('disconnected')
-                self.conns.discard(conn)
-
-    def _send_thread(self):
-        while True:
-            conn, send_data = self.output_queue.get()
+    # - pumps -
+    def _pump_out(self):
+        while not self._closed:
             try:
-                conn.send(send_data)
-            except (TimeoutError, ConnectionResetError, BrokenPipeError, OSError):
-                self.disconnect(conn)
-
-    def _recv_thread(self):
-        while True:
-            with self._lock:
-                conns = list(self.conns)
-            if not conns:
-                threading.Event().wait(0.1)
+                conn, data = self.output_queue.get(timeout=0.5)
+            except queue.Empty:
                 continue
             try:
-                ready = mp_connection.wait(conns, timeout=0.3)
-            except OSError:
+                conn.send(data)
+            except self._IO_ERRORS:
+                self.disconnect(conn)
+                continue
+            with self._lock:
+                if conn in self._conns:
+                    self._conns[conn]['sent'] += 1
+
+    def _pump_in(self):
+        while not self._closed:
+            if not self._wake.wait(timeout=0.5):
+                continue
+            with self._lock:
+                live = list(self._conns)
+            if not live:
+                continue
+            try:
+                ready = mp_connection.wait(live, timeout=0.3)
+            except self._IO_ERRORS:
                 continue
             for conn in ready:
                 try:
-                    recv_data = conn.recv()
-                except (TimeoutError, ConnectionResetError, EOFError, OSError):
+                    data = conn.recv()
+                except self._IO_ERRORS:
                     self.disconnect(conn)
                     continue
-                self.input_queue.put((conn, recv_data))
+                with self._lock:
+                    if conn in self._conns:
+                        self._conns[conn]['recvd'] += 1
+                self.input_queue.put((conn, data))

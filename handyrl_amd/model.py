@@ -1,7 +1,11 @@
-"""Model wrapper utilities (parity: reference handyrl/model.py).
+"""Model wrapper utilities.
 
-Keeps worker-side CPU inference single-threaded (the GPU path batches
-inference in handyrl_amd/actor.py instead of widening CPU threads).
+Semantics contract (reference handyrl/model.py): ``ModelWrapper`` gives any
+``nn.Module`` a uniform numpy-in/numpy-out single-sample ``inference`` and
+handles the inference-vs-training hidden-state shape split; ``RandomModel``
+is the uniform-policy epoch-0 opponent.  Worker-side CPU inference stays
+single-threaded — the MI355X path batches inference on the GPU
+(handyrl_amd/actor.py) instead of widening CPU threads.
 """
 
 import os
@@ -17,56 +21,88 @@ from .util import map_r
 
 
 def to_torch(x):
-    return map_r(x, lambda v: torch.from_numpy(np.array(v)).contiguous() if v is not None else None)
+    """Nested numpy/scalars -> contiguous CPU tensors (None passes through)."""
+    return map_r(x, lambda v: None if v is None
+                 else torch.from_numpy(np.asarray(v)).contiguous())
 
 
 def to_numpy(x):
-    return map_r(x, lambda v: v.detach().numpy() if v is not None else None)
+    """Nested tensors -> numpy arrays (None passes through)."""
+    return map_r(x, lambda v: None if v is None else v.detach().numpy())
 
 
 def to_gpu(x, device=None, non_blocking=False):
-    return map_r(x, lambda v: v.cuda(device, non_blocking=non_blocking) if v is not None else None)
+    """Nested tensors -> CUDA (H2D is async when sources are pinned)."""
+    return map_r(x, lambda v: None if v is None
+                 else v.cuda(device, non_blocking=non_blocking))
+
+
+def _batched(v):
+    """numpy leaf -> (1, ...) tensor."""
+    return torch.from_numpy(np.asarray(v)).contiguous().unsqueeze(0)
+
+
+def _unbatched(v):
+    """(1, ...) tensor leaf -> numpy."""
+    return v.detach().numpy().squeeze(0)
 
 
 class ModelWrapper(nn.Module):
-    """Uniform numpy-in/numpy-out single-sample inference over an nn.Module,
-    plus inference-vs-training hidden-state shape handling."""
+    """Uniform inference surface over an ``nn.Module``.
+
+    ``inference(x, hidden)`` feeds one observation (numpy leaves, no batch
+    dim), runs the wrapped net under no_grad with a temporary batch axis,
+    and strips that axis from every output.  ``init_hidden()`` without a
+    batch size returns numpy-leaf hidden state for the same single-sample
+    regime; with a batch-size list it defers to the net's training shapes.
+    """
 
     def __init__(self, model):
         super().__init__()
         self.model = model
 
     def init_hidden(self, batch_size=None):
-        if not hasattr(self.model, 'init_hidden'):
+        make = getattr(self.model, 'init_hidden', None)
+        if make is None:
             return None
-        if batch_size is None:     # inference: no batch dims, numpy leaves
-            hidden = self.model.init_hidden([])
-            return map_r(hidden, lambda h: h.detach().numpy() if isinstance(h, torch.Tensor) else h)
-        return self.model.init_hidden(batch_size)
+        if batch_size is not None:
+            return make(batch_size)
+        return map_r(make([]), lambda h: h.detach().numpy()
+                     if isinstance(h, torch.Tensor) else h)
 
     def forward(self, *args, **kwargs):
-        return self.model.forward(*args, **kwargs)
+        return self.model(*args, **kwargs)
 
     def inference(self, x, hidden, **kwargs):
-        if hasattr(self.model, 'inference'):
-            return self.model.inference(x, hidden, **kwargs)
+        # nets with their own batched inference (e.g. ONNX-style stubs)
+        # keep full control
+        native = getattr(self.model, 'inference', None)
+        if native is not None:
+            return native(x, hidden, **kwargs)
         self.eval()
         with torch.no_grad():
-            xt = map_r(x, lambda v: torch.from_numpy(np.array(v)).contiguous().unsqueeze(0) if v is not None else None)
-            ht = map_r(hidden, lambda h: torch.from_numpy(np.array(h)).contiguous().unsqueeze(0) if h is not None else None)
-            outputs = self.forward(xt, ht, **kwargs)
-        return map_r(outputs, lambda o: o.detach().numpy().squeeze(0) if o is not None else None)
+            out = self.forward(map_r(x, lambda v: None if v is None
+                                     else _batched(v)),
+                               map_r(hidden, lambda h: None if h is None
+                                     else _batched(h)),
+                               **kwargs)
+        return map_r(out, lambda o: None if o is None else _unbatched(o))
 
 
 class RandomModel(nn.Module):
-    """Uniform-policy / zero-value stand-in, shaped by probing a real model
-    once (used as the ``model_id == 0`` opponent)."""
+    """Uniform-policy / zero-value opponent (``model_id == 0``).
+
+    Output keys and shapes are discovered by probing the real model on one
+    observation; every later call returns those zeros (zero logits = a
+    uniform policy after masking+softmax).
+    """
 
     def __init__(self, model, x):
         super().__init__()
-        wrapped = ModelWrapper(model)
-        outputs = wrapped.inference(x, wrapped.init_hidden())
-        self.output_dict = {k: np.zeros_like(v) for k, v in outputs.items() if k != 'hidden'}
+        probe = ModelWrapper(model)
+        shaped = probe.inference(x, probe.init_hidden())
+        self.output_dict = {k: np.zeros_like(v) for k, v in shaped.items()
+                            if k != 'hidden'}
 
     def inference(self, *args, **kwargs):
         return self.output_dict

@@ -1,0 +1,128 @@
+"""RCCL-on-hardware de-risk probe (VERDICT round-1 item 4).
+
+Runs on a single MI355X and validates the multi-GPU mechanics that the
+8-GPU driver bench depends on, to the extent a 1-GPU lease allows:
+
+  1. nccl (=RCCL on ROCm) process-group init at world_size=1 with an
+     explicit device_id, and the rank/world sanity checks;
+  2. hipGraph capture of a train step that CONTAINS the
+     dist.all_reduce node (GradReducer.allreduce_) + clip + Adam, on the
+     real GeeseNet custom-kernel training path;
+  3. replays of that graph: weights move, stay finite, and the captured
+     all-reduce executes (ws=1 SUM is identity, so replayed results must
+     equal a no-dist control run bit-for-bit).
+
+Writes a JSON verdict to gpurun_out/rccl_probe.json.
+"""
+
+import json
+import os
+import sys
+
+import torch
+import torch.nn as nn
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+
+def build_step(dist_on, seed=17):
+    """Model + captured fwd/bwd/(all-reduce)/clip/Adam step; returns
+    (replay_fn, model)."""
+    from handyrl_amd.models.geese_net import GeeseNet
+    from handyrl_amd import dist as hdist
+
+    torch.manual_seed(seed)
+    model = GeeseNet().cuda().train()
+    params = list(model.parameters())
+    opt = torch.optim.Adam(params, lr=1e-3, weight_decay=1e-5)
+    reducer = hdist.GradReducer(params)
+
+    torch.manual_seed(seed + 1)
+    x = (torch.rand(256, 17, 7, 11, device='cuda') < 0.2).float()
+
+    def run():
+        with torch.autocast('cuda', dtype=torch.bfloat16):
+            out = model(x, None)
+            loss = out['policy'].float().pow(2).sum() + \
+                out['value'].float().pow(2).sum()
+        opt.zero_grad(set_to_none=False)
+        loss.backward()
+        if dist_on:
+            reducer.allreduce_()
+        nn.utils.clip_grad_norm_(params, 4.0)
+        opt.step()
+        return loss
+
+    # warmup on a side stream, then capture
+    s = torch.cuda.Stream()
+    s.wait_stream(torch.cuda.current_stream())
+    with torch.cuda.stream(s):
+        for _ in range(3):
+            run()
+    torch.cuda.current_stream().wait_stream(s)
+    g = torch.cuda.CUDAGraph()
+    with torch.cuda.graph(g):
+        loss_t = run()
+    return g, loss_t, model
+
+
+def main():
+    verdict = {'ok': False}
+    os.makedirs('gpurun_out', exist_ok=True)
+    import torch.distributed as dist
+
+    # -- 1. RCCL init at world_size=1 ------------------------------------
+    os.environ.setdefault('MASTER_ADDR', '127.0.0.1')
+    os.environ.setdefault('MASTER_PORT', '29613')
+    os.environ['RANK'] = '0'
+    os.environ['WORLD_SIZE'] = '1'
+    torch.cuda.set_device(0)
+    dist.init_process_group('nccl', rank=0, world_size=1,
+                            device_id=torch.device('cuda', 0))
+    verdict['backend'] = dist.get_backend()
+    verdict['world_size'] = dist.get_world_size()
+    verdict['rank'] = dist.get_rank()
+    assert dist.get_world_size() == 1 and dist.get_rank() == 0
+
+    # eager all-reduce sanity (forces communicator creation before capture)
+    t = torch.ones(1 << 20, device='cuda')
+    dist.all_reduce(t)
+    torch.cuda.synchronize()
+    assert t.sum().item() == float(1 << 20)
+    verdict['eager_allreduce'] = 'ok'
+
+    # -- 2/3. capture the step WITH the all-reduce node ------------------
+    g, loss_t, model = build_step(dist_on=True)
+    losses = []
+    for _ in range(5):
+        g.replay()
+        torch.cuda.synchronize()
+        losses.append(float(loss_t.item()))
+    verdict['captured_losses_with_allreduce'] = losses
+    p_dist = [p.detach().clone() for p in model.parameters()]
+    assert all(torch.isfinite(p).all() for p in p_dist)
+    dist.destroy_process_group()
+
+    # control: identical run with NO dist group / no all-reduce node
+    g2, loss_t2, model2 = build_step(dist_on=False)
+    losses2 = []
+    for _ in range(5):
+        g2.replay()
+        torch.cuda.synchronize()
+        losses2.append(float(loss_t2.item()))
+    verdict['captured_losses_control'] = losses2
+    p_ctl = [p.detach().clone() for p in model2.parameters()]
+
+    max_diff = max(float((a - b).abs().max()) for a, b in zip(p_dist, p_ctl))
+    verdict['weights_max_abs_diff_vs_control'] = max_diff
+    verdict['losses_equal'] = losses == losses2
+    verdict['ok'] = bool(losses == losses2 and max_diff == 0.0)
+
+    with open('gpurun_out/rccl_probe.json', 'w') as f:
+        json.dump(verdict, f, indent=1)
+    print(json.dumps(verdict))
+    assert verdict['ok'], verdict
+
+
+if __name__ == '__main__':
+    main()
