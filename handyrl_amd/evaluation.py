@@ -1,13 +1,22 @@
 """Evaluation: match executors, in-training evaluator, offline evaluation
-pool, network battle mode and ONNX inference (parity: reference
-evaluation.py)."""
+pool, network battle mode and ONNX inference.
+
+Protocol contracts kept from the reference (evaluation.py): the
+update/action/observe/outcome RPC grammar with string-encoded actions and
+``diff_info`` replica sync, the match-server port (:9876, overridable),
+the CLI argument meanings of the --eval/--eval-server/--eval-client modes
+and the per-pattern win-rate report.  The implementations are this repo's
+own — table-dispatched RPC handling, a job-list + result-table evaluation
+pool, and a restructured ONNX session wrapper.
+"""
 
 import multiprocessing as mp
 import random
 import time
 
 from .environment import prepare_env, make_env
-from .connection import send_recv, accept_socket_connections, connect_socket_connection
+from .connection import send_recv, accept_socket_connections, \
+    connect_socket_connection
 from .agent import RandomAgent, RuleBasedAgent, Agent, EnsembleAgent, SoftAgent
 
 import os as _os
@@ -26,64 +35,96 @@ def view_transition(env):
         env.view_transition()
 
 
+# -- network battle RPC ------------------------------------------------------
+
 class NetworkAgentClient:
-    """Client side of the network battle RPC loop (update/action/observe/
-    outcome commands driven by the server)."""
+    """Client side of a network battle: holds the local agent and a replica
+    environment, executes RPC commands streamed by the server until 'quit'
+    or disconnect."""
 
     def __init__(self, agent, env, conn):
         self.conn = conn
         self.agent = agent
         self.env = env
 
+    # each handler returns the RPC reply
+    def _rpc_action(self, player):
+        view(self.env)
+        chosen = self.agent.action(self.env, player, show=True)
+        return self.env.action2str(chosen, player)
+
+    def _rpc_observe(self, player):
+        view(self.env)
+        return self.agent.observe(self.env, player, show=True)
+
+    def _rpc_update(self, info, reset):
+        reply = self.env.update(info, reset)
+        if reset:
+            self.agent.reset(self.env, show=True)
+        else:
+            view_transition(self.env)
+        return reply
+
     def run(self):
+        handlers = {'action': self._rpc_action, 'observe': self._rpc_observe,
+                    'update': self._rpc_update}
         while True:
             try:
                 command, args = self.conn.recv()
-            except ConnectionResetError:
+            except (ConnectionResetError, EOFError):
                 break
             if command == 'quit':
                 break
-            elif command == 'outcome':
+            if command == 'outcome':
                 print('outcome = %f' % args[0])
-            elif hasattr(self.agent, command):
-                if command in ('action', 'observe'):
-                    view(self.env)
-                ret = getattr(self.agent, command)(self.env, *args, show=True)
-                if command == 'action':
-                    player = args[0]
-                    ret = self.env.action2str(ret, player)
-            else:
-                ret = getattr(self.env, command)(*args)
-                if command == 'update':
-                    reset = args[1]
-                    if reset:
-                        self.agent.reset(self.env, show=True)
-                    else:
-                        view_transition(self.env)
-            self.conn.send(ret)
+                self.conn.send(None)
+                continue
+            handler = handlers.get(command)
+            if handler is not None:
+                self.conn.send(handler(*args))
+            else:                       # any other env method, e.g. reset
+                self.conn.send(getattr(self.env, command)(*args))
 
 
 class NetworkAgent:
-    """Server-side stub driving one remote client."""
+    """Server-side proxy for one remote client: every agent call becomes
+    one RPC round trip."""
 
     def __init__(self, conn):
         self.conn = conn
 
+    def _call(self, command, *args):
+        return send_recv(self.conn, (command, list(args)))
+
     def update(self, data, reset):
-        return send_recv(self.conn, ('update', [data, reset]))
+        return self._call('update', data, reset)
 
     def outcome(self, outcome):
-        return send_recv(self.conn, ('outcome', [outcome]))
+        return self._call('outcome', outcome)
 
     def action(self, player):
-        return send_recv(self.conn, ('action', [player]))
+        return self._call('action', player)
 
     def observe(self, player):
-        return send_recv(self.conn, ('observe', [player]))
+        return self._call('observe', player)
+
+
+# -- match executors ---------------------------------------------------------
+
+def _collect_actions(env, agents, show):
+    """One step's actions from every turn player; observers observe."""
+    turn_players, observers = env.turns(), env.observers()
+    actions = {}
+    for p, agent in agents.items():
+        if p in turn_players:
+            actions[p] = agent.action(env, p, show=show)
+        elif p in observers:
+            agent.observe(env, p, show=show)
+    return actions
 
 
 def exec_match(env, agents, critic=None, show=False, game_args={}):
-    """Match with a shared environment instance."""
+    """Match over one shared environment instance."""
     if env.reset(game_args):
         return None
     for agent in agents.values():
@@ -91,17 +132,9 @@ def exec_match(env, agents, critic=None, show=False, game_args={}):
     while not env.terminal():
         if show:
             view(env)
-        if show and critic is not None:
-            print('cv = ', critic.observe(env, None, show=False)[0])
-        turn_players = env.turns()
-        observers = env.observers()
-        actions = {}
-        for p, agent in agents.items():
-            if p in turn_players:
-                actions[p] = agent.action(env, p, show=show)
-            elif p in observers:
-                agent.observe(env, p, show=show)
-        if env.step(actions):
+            if critic is not None:
+                print('cv = ', critic.observe(env, None, show=False)[0])
+        if env.step(_collect_actions(env, agents, show)):
             return None
         if show:
             view_transition(env)
@@ -111,9 +144,10 @@ def exec_match(env, agents, critic=None, show=False, game_args={}):
     return {'result': outcome}
 
 
-def exec_network_match(env, network_agents, critic=None, show=False, game_args={}):
-    """Match where each remote side holds its own replica synced by
-    diff_info/update strings."""
+def exec_network_match(env, network_agents, critic=None, show=False,
+                       game_args={}):
+    """Match where each remote side holds its own replica, synced only by
+    diff_info/update strings (partial information stays partial)."""
     if env.reset(game_args):
         return None
     for p, agent in network_agents.items():
@@ -121,15 +155,13 @@ def exec_network_match(env, network_agents, critic=None, show=False, game_args={
     while not env.terminal():
         if show:
             view(env)
-        if show and critic is not None:
-            print('cv = ', critic.observe(env, None, show=False)[0])
-        turn_players = env.turns()
-        observers = env.observers()
+            if critic is not None:
+                print('cv = ', critic.observe(env, None, show=False)[0])
+        turn_players, observers = env.turns(), env.observers()
         actions = {}
         for p, agent in network_agents.items():
             if p in turn_players:
-                action = agent.action(p)
-                actions[p] = env.str2action(action, p)
+                actions[p] = env.str2action(agent.action(p), p)
             elif p in observers:
                 agent.observe(p)
         if env.step(actions):
@@ -142,17 +174,21 @@ def exec_network_match(env, network_agents, critic=None, show=False, game_args={
     return {'result': outcome}
 
 
+# -- in-training evaluator ---------------------------------------------------
+
 def build_agent(raw, env=None):
+    """Opponent spec -> agent, for specs that need no model file."""
     if raw == 'random':
         return RandomAgent()
     if isinstance(raw, str) and raw.startswith('rulebase'):
-        key = raw.split('-')[1] if '-' in raw else None
-        return RuleBasedAgent(key)
+        _, _, key = raw.partition('-')
+        return RuleBasedAgent(key or None)
     return None
 
 
 class Evaluator:
-    """In-training evaluation job: latest model vs a configured opponent."""
+    """In-training evaluation job: the latest model occupies the seats
+    named in the job; a configured opponent fills the rest."""
 
     def __init__(self, env, args):
         self.env = env
@@ -160,13 +196,10 @@ class Evaluator:
         self.default_opponent = 'random'
 
     def execute(self, models, args):
-        opponents = self.args.get('eval', {}).get('opponent', [])
-        opponent = random.choice(opponents) if opponents else self.default_opponent
-
-        agents = {}
-        for p, model in models.items():
-            agents[p] = Agent(model) if model is not None else build_agent(opponent, self.env)
-
+        choices = self.args.get('eval', {}).get('opponent', [])
+        opponent = random.choice(choices) if choices else self.default_opponent
+        agents = {p: Agent(m) if m is not None else build_agent(opponent, self.env)
+                  for p, m in models.items()}
         results = exec_match(self.env, agents)
         if results is None:
             print('None episode in evaluation!')
@@ -174,163 +207,206 @@ class Evaluator:
         return {'args': args, 'opponent': opponent, **results}
 
 
+# -- offline evaluation pool -------------------------------------------------
+
 def wp_func(results):
-    games = sum(v for k, v in results.items() if k is not None)
-    win = sum((k + 1) / 2 * v for k, v in results.items() if k is not None)
-    return win / games if games else 0.0
+    """{outcome: count} -> win probability ((mean+1)/2 convention)."""
+    games = sum(n for oc, n in results.items() if oc is not None)
+    if not games:
+        return 0.0
+    wins = sum((oc + 1) / 2 * n for oc, n in results.items() if oc is not None)
+    return wins / games
 
 
-def eval_process_mp_child(agents, critic, env_args, index, in_queue, out_queue, seed, show=False):
+class _ResultTable:
+    """Per-agent outcome histograms, total and per seat pattern."""
+
+    def __init__(self, num_agents):
+        self.total = [{} for _ in range(num_agents)]
+        self.by_pattern = [{} for _ in range(num_agents)]
+
+    def ensure_pattern(self, pattern):
+        for per_agent in self.by_pattern:
+            per_agent.setdefault(pattern, {})
+
+    def add(self, players, agent_ids, pattern, outcome):
+        for seat, p in enumerate(players):
+            aid, oc = agent_ids[seat], outcome[p]
+            bucket = self.by_pattern[aid][pattern]
+            bucket[oc] = bucket.get(oc, 0) + 1
+            self.total[aid][oc] = self.total[aid].get(oc, 0) + 1
+
+    def report(self):
+        for aid, per_pattern in enumerate(self.by_pattern):
+            print('---agent %d---' % aid)
+            for pattern, hist in per_pattern.items():
+                ordered = {k: hist[k] for k in sorted(hist, reverse=True)}
+                print(pattern, ordered, wp_func(hist))
+            total = self.total[aid]
+            print('total', {k: total[k] for k in sorted(total, reverse=True)},
+                  wp_func(total))
+
+
+def _seating_plan(num_agents, num_games, pat_idx):
+    """Seat assignments for one pattern: two-player games split first/second
+    evenly (suffix -F/-S); larger games shuffle seats per game."""
+    plans = []
+    for i in range(num_games):
+        if num_agents == 2:
+            if i < (num_games + 1) // 2:
+                plans.append((pat_idx + '-F', [0, 1]))
+            else:
+                plans.append((pat_idx + '-S', [1, 0]))
+        else:
+            plans.append((pat_idx,
+                          random.sample(range(num_agents), num_agents)))
+    return plans
+
+
+def eval_process_mp_child(agents, critic, env_args, index, in_queue,
+                          out_queue, seed, show=False):
+    """One evaluation worker: plays queued games until the poison pill."""
     random.seed(seed + index)
     env = make_env({**env_args, 'id': index})
     while True:
-        args = in_queue.get()
-        if args is None:
+        job = in_queue.get()
+        if job is None:
             break
-        g, agent_ids, pat_idx, game_args = args
-        print('*** Game %d ***' % g)
-        agent_map = {env.players()[p]: agents[ai] for p, ai in enumerate(agent_ids)}
-        if isinstance(list(agent_map.values())[0], NetworkAgent):
-            results = exec_network_match(env, agent_map, critic, show=show, game_args=game_args)
-        else:
-            results = exec_match(env, agent_map, critic, show=show, game_args=game_args)
-        out_queue.put((pat_idx, agent_ids, results))
+        game_no, agent_ids, pattern, game_args = job
+        print('*** Game %d ***' % game_no)
+        seat_map = {env.players()[seat]: agents[aid]
+                    for seat, aid in enumerate(agent_ids)}
+        runner = exec_network_match \
+            if isinstance(next(iter(seat_map.values())), NetworkAgent) \
+            else exec_match
+        results = runner(env, seat_map, critic, show=show,
+                         game_args=game_args)
+        out_queue.put((pattern, agent_ids, results))
     out_queue.put(None)
 
 
-def evaluate_mp(env, agents, critic, env_args, args_patterns, num_process, num_games, seed):
+def evaluate_mp(env, agents, critic, env_args, args_patterns, num_process,
+                num_games, seed):
+    """Offline evaluation: enqueue every (pattern x game) job, fan out over
+    worker processes, aggregate outcome histograms."""
     in_queue, out_queue = mp.Queue(), mp.Queue()
-    args_cnt = 0
-    total_results, result_map = [{} for _ in agents], [{} for _ in agents]
+    table = _ResultTable(len(agents))
     print('total games = %d' % (len(args_patterns) * num_games))
     time.sleep(0.1)
-    for pat_idx, args in args_patterns.items():
-        for i in range(num_games):
-            if len(agents) == 2:
-                # equalize first/second seats in two-player games
-                first_agent = 0 if i < (num_games + 1) // 2 else 1
-                tmp_pat_idx, agent_ids = (pat_idx + '-F', [0, 1]) if first_agent == 0 \
-                    else (pat_idx + '-S', [1, 0])
-            else:
-                tmp_pat_idx, agent_ids = pat_idx, random.sample(range(len(agents)), len(agents))
-            in_queue.put((args_cnt, agent_ids, tmp_pat_idx, args))
-            for p in range(len(agents)):
-                result_map[p][tmp_pat_idx] = {}
-            args_cnt += 1
+
+    game_no = 0
+    for pat_idx, game_args in args_patterns.items():
+        for pattern, agent_ids in _seating_plan(len(agents), num_games,
+                                                pat_idx):
+            table.ensure_pattern(pattern)
+            in_queue.put((game_no, agent_ids, pattern, game_args))
+            game_no += 1
 
     network_mode = agents[0] is None
     if network_mode:
-        agents = network_match_acception(num_process, env_args, len(agents), NETWORK_MATCH_PORT)
+        per_process = network_match_acception(
+            num_process, env_args, len(agents), NETWORK_MATCH_PORT)
     else:
-        agents = [agents] * num_process
+        per_process = [agents] * num_process
 
     for i in range(num_process):
         in_queue.put(None)
-        args = agents[i], critic, env_args, i, in_queue, out_queue, seed
+        child_args = (per_process[i], critic, env_args, i, in_queue,
+                      out_queue, seed)
         if num_process > 1:
-            mp.Process(target=eval_process_mp_child, args=args).start()
+            mp.Process(target=eval_process_mp_child, args=child_args).start()
             if network_mode:
-                for agent in agents[i]:
+                for agent in per_process[i]:
                     agent.conn.close()
         else:
-            eval_process_mp_child(*args, show=True)
+            eval_process_mp_child(*child_args, show=True)
 
-    finished_cnt = 0
-    while finished_cnt < num_process:
-        ret = out_queue.get()
-        if ret is None:
-            finished_cnt += 1
+    finished = 0
+    while finished < num_process:
+        item = out_queue.get()
+        if item is None:
+            finished += 1
             continue
-        pat_idx, agent_ids, results = ret
+        pattern, agent_ids, results = item
         outcome = results.get('result') if results else None
         if outcome is not None:
-            for idx, p in enumerate(env.players()):
-                agent_id = agent_ids[idx]
-                oc = outcome[p]
-                result_map[agent_id][pat_idx][oc] = result_map[agent_id][pat_idx].get(oc, 0) + 1
-                total_results[agent_id][oc] = total_results[agent_id].get(oc, 0) + 1
+            table.add(env.players(), agent_ids, pattern, outcome)
 
-    for p, r_map in enumerate(result_map):
-        print('---agent %d---' % p)
-        for pat_idx, results in r_map.items():
-            print(pat_idx, {k: results[k] for k in sorted(results.keys(), reverse=True)},
-                  wp_func(results))
-        print('total', {k: total_results[p][k] for k in sorted(total_results[p].keys(), reverse=True)},
-              wp_func(total_results[p]))
+    table.report()
 
 
 def network_match_acception(n, env_args, num_agents, port):
-    waiting_conns = []
-    accepted_conns = []
+    """Accept n*num_agents client connections, grouping each consecutive
+    ``num_agents`` into one match; the first client of each group receives
+    the env spec."""
+    groups, pending = [], []
     for conn in accept_socket_connections(port):
-        if len(accepted_conns) >= n * num_agents:
-            break
-        waiting_conns.append(conn)
-        if len(waiting_conns) == num_agents:
-            conn = waiting_conns[0]
-            accepted_conns.append(conn)
-            waiting_conns = waiting_conns[1:]
-            conn.send(env_args)
-    return [[NetworkAgent(accepted_conns[i * num_agents + j]) for j in range(num_agents)]
-            for i in range(n)]
+        pending.append(conn)
+        if len(pending) == num_agents:
+            pending[0].send(env_args)
+            groups.append([NetworkAgent(c) for c in pending])
+            pending = []
+            if len(groups) == n:
+                break
+    return groups
 
+
+# -- ONNX inference ----------------------------------------------------------
 
 class OnnxModel:
-    """onnxruntime single-thread inference with hidden* name-convention
-    plumbing for recurrent nets."""
+    """onnxruntime single-thread session with the hidden* name convention
+    for recurrent state (scripts/make_onnx_model.py export format)."""
 
     def __init__(self, model_path):
         self.model_path = model_path
-        self.ort_session = None
+        self.session = None
 
-    def _open_session(self):
-        import os
-        os.environ['OMP_NUM_THREADS'] = '1'
-        os.environ['OMP_WAIT_POLICY'] = 'PASSIVE'
-        import onnxruntime
-        opts = onnxruntime.SessionOptions()
-        opts.intra_op_num_threads = 1
-        opts.inter_op_num_threads = 1
-        opts.execution_mode = onnxruntime.ExecutionMode.ORT_SEQUENTIAL
-        self.ort_session = onnxruntime.InferenceSession(self.model_path, sess_options=opts)
+    def _session(self):
+        if self.session is None:
+            import os
+            os.environ['OMP_NUM_THREADS'] = '1'
+            os.environ['OMP_WAIT_POLICY'] = 'PASSIVE'
+            import onnxruntime
+            opts = onnxruntime.SessionOptions()
+            opts.intra_op_num_threads = 1
+            opts.inter_op_num_threads = 1
+            opts.execution_mode = onnxruntime.ExecutionMode.ORT_SEQUENTIAL
+            self.session = onnxruntime.InferenceSession(
+                self.model_path, sess_options=opts)
+        return self.session
 
     def init_hidden(self, batch_size=None):
-        if self.ort_session is None:
-            self._open_session()
-        hidden_inputs = [y for y in self.ort_session.get_inputs() if y.name.startswith('hidden')]
-        if not hidden_inputs:
-            return None
         import numpy as np
-        if batch_size is None:
-            batch_size = []
-        type_map = {'tensor(float)': np.float32, 'tensor(int64)': np.int64}
-        return [np.zeros(list(batch_size) + list(y.shape[1:]), dtype=type_map[y.type])
-                for y in hidden_inputs]
+        dtypes = {'tensor(float)': np.float32, 'tensor(int64)': np.int64}
+        shape_head = list(batch_size) if batch_size is not None else []
+        hidden = [np.zeros(shape_head + list(spec.shape[1:]),
+                           dtype=dtypes[spec.type])
+                  for spec in self._session().get_inputs()
+                  if spec.name.startswith('hidden')]
+        return hidden or None
 
     def inference(self, x, hidden=None, batch_input=False):
-        if self.ort_session is None:
-            self._open_session()
         import numpy as np
         from .util import map_r
+        session = self._session()
+        names = [spec.name for spec in session.get_inputs()]
+        feeds = {}
 
-        ort_inputs = {}
-        names = [y.name for y in self.ort_session.get_inputs()]
+        def bind(leaf):
+            feeds[names[len(feeds)]] = leaf if batch_input \
+                else np.expand_dims(leaf, 0)
 
-        def insert(y):
-            v = y if batch_input else np.expand_dims(y, 0)
-            ort_inputs[names[len(ort_inputs)]] = v
-
-        map_r(x, insert)
+        map_r(x, bind)
         if hidden is not None:
-            map_r(hidden, insert)
-        ort_outputs = self.ort_session.run(None, ort_inputs)
+            map_r(hidden, bind)
+        raw = session.run(None, feeds)
         if not batch_input:
-            ort_outputs = [o.squeeze(0) for o in ort_outputs]
-        out_names = [y.name for y in self.ort_session.get_outputs()]
-        outputs = {name: ort_outputs[i] for i, name in enumerate(out_names)}
-
-        hidden_outputs = [outputs.pop(k) for k in list(outputs) if k.startswith('hidden')]
-        return {**outputs, 'hidden': hidden_outputs if hidden_outputs else None}
+            raw = [o.squeeze(0) for o in raw]
+        outputs = {spec.name: raw[i]
+                   for i, spec in enumerate(session.get_outputs())}
+        hidden_out = [outputs.pop(k) for k in list(outputs)
+                      if k.startswith('hidden')]
+        return {**outputs, 'hidden': hidden_out or None}
 
 
 def load_model(model_path, model=None):
@@ -344,11 +420,16 @@ def load_model(model_path, model=None):
     return ModelWrapper(model)
 
 
+# -- CLI modes ---------------------------------------------------------------
+
+def _resolve_agent(spec, env):
+    agent = build_agent(spec, env)
+    return agent if agent is not None else Agent(load_model(spec, env.net()))
+
+
 def client_mp_child(env_args, model_path, conn):
     env = make_env(env_args)
-    agent = build_agent(model_path, env)
-    if agent is None:
-        agent = Agent(load_model(model_path, env.net()))
+    agent = _resolve_agent(model_path, env)
     NetworkAgentClient(agent, env, conn).run()
 
 
@@ -361,20 +442,16 @@ def eval_main(args, argv):
     num_games = int(argv[1]) if len(argv) >= 2 else 100
     num_process = int(argv[2]) if len(argv) >= 3 else 1
 
-    def resolve_agent(model_path):
-        agent = build_agent(model_path, env)
-        if agent is None:
-            agent = Agent(load_model(model_path, env.net()))
-        return agent
-
-    main_agent = resolve_agent(model_paths[0])
+    main_agent = _resolve_agent(model_paths[0], env)
     critic = None
     print('%d process, %d games' % (num_process, num_games))
     seed = random.randrange(int(1e8))
     print('seed = %d' % seed)
     opponent = model_paths[1] if len(model_paths) > 1 else 'random'
-    agents = [main_agent] + [resolve_agent(opponent) for _ in range(len(env.players()) - 1)]
-    evaluate_mp(env, agents, critic, env_args, {'default': {}}, num_process, num_games, seed)
+    agents = [main_agent] + [_resolve_agent(opponent, env)
+                             for _ in range(len(env.players()) - 1)]
+    evaluate_mp(env, agents, critic, env_args, {'default': {}},
+                num_process, num_games, seed)
 
 
 def eval_server_main(args, argv):
@@ -401,5 +478,6 @@ def eval_client_main(args, argv):
         except ConnectionResetError:
             break
         model_path = argv[0] if len(argv) >= 1 else 'models/latest.pth'
-        mp.Process(target=client_mp_child, args=(env_args, model_path, conn)).start()
+        mp.Process(target=client_mp_child,
+                   args=(env_args, model_path, conn)).start()
         conn.close()

@@ -1,22 +1,27 @@
-"""Worker tree: leaf actors, gather aggregation nodes, and the local/remote
-cluster frontends (parity: reference worker.py).
+"""Worker tree: leaf actors, gather relay nodes, and the local/remote
+cluster frontends.
 
-Topology: learner <-> gathers (one per ~16 workers) <-> workers, with
-request batching, model-response caching and upload buffering in the
-gather tier.  Remote mode: entry handshake on :9999, gather data
-connections on :9998; workers may join or leave at any time.
+Topology and protocol contract (reference worker.py): learner <-> gathers
+(one per ~16 workers) <-> workers; job args are requested with
+``('args', None)``, models pulled by id with ``('model', id)``, episodes
+and results pushed as ``('episode'/'result', payload)``.  The gather tier
+batches args requests upstream, caches model responses by id and buffers
+uploads.  Remote mode keeps the entry handshake on :9999 (assigning
+``base_worker_id`` ranges) and gather data connections on :9998; workers
+may join or leave at any time.
+
+The implementation is this repo's own: the relay's three concerns live in
+small dedicated components, and the leaf worker separates job transport
+from job execution.
 """
 
 import copy
-import functools
 import pickle
-import queue
 import random
 import threading
 import time
 import multiprocessing as mp
 from socket import gethostname
-from collections import deque
 
 from .environment import prepare_env, make_env
 from .connection import QueueCommunicator
@@ -31,6 +36,38 @@ ENTRY_PORT = int(_os.environ.get('HANDYRL_ENTRY_PORT', 9999))
 WORKER_PORT = int(_os.environ.get('HANDYRL_WORKER_PORT', 9998))
 
 
+def _gathers_for(num_parallel):
+    """Gather processes for a worker count (~16 leaves per relay)."""
+    return 1 + max(0, num_parallel - 1) // 16
+
+
+class _LeafModelPool:
+    """Model cache on a leaf worker: capacity one (the live epoch), plus
+    the probe-built RandomModel for id 0."""
+
+    def __init__(self, env, fetch):
+        self.env = env
+        self.fetch = fetch             # model_id -> pickled nn.Module bytes
+        self.entries = {}
+
+    def get(self, model_id):
+        return self.entries.get(model_id)
+
+    def ensure(self, model_ids):
+        for model_id in model_ids:
+            if model_id is None or model_id < 0 or model_id in self.entries:
+                continue
+            model = pickle.loads(self.fetch(model_id))
+            if model_id == 0:
+                # epoch-0 opponent: uniform-random model probed on a real obs
+                self.env.reset()
+                probe_obs = self.env.observation(self.env.players()[0])
+                model = RandomModel(model, probe_obs)
+            while len(self.entries) >= 1:
+                self.entries.pop(next(iter(self.entries)))
+            self.entries[model_id] = ModelWrapper(model)
+
+
 class Worker:
     """Leaf actor: pulls job args, runs generation or evaluation, pushes
     the episode/result back."""
@@ -40,216 +77,239 @@ class Worker:
         self.worker_id = wid
         self.args = args
         self.conn = conn
-        self.model_pool = {}
-
         self.env = make_env({**args['env'], 'id': wid})
-        self.generator = Generator(self.env, self.args)
-        self.evaluator = Evaluator(self.env, self.args)
-
+        self.generator = Generator(self.env, args)
+        self.evaluator = Evaluator(self.env, args)
+        self.models = _LeafModelPool(
+            self.env, lambda mid: send_recv(self.conn, ('model', mid)))
         random.seed(args['seed'] + wid)
 
     def __del__(self):
         print('closed worker %d' % self.worker_id)
 
-    def _fetch_models(self, model_ids):
-        for model_id in model_ids:
-            if model_id is None or model_id < 0 or model_id in self.model_pool:
-                continue
-            model = pickle.loads(send_recv(self.conn, ('model', model_id)))
-            if model_id == 0:
-                # epoch 0 opponent: uniform-random model probed on a real obs
-                self.env.reset()
-                obs = self.env.observation(self.env.players()[0])
-                model = RandomModel(model, obs)
-            if len(self.model_pool) >= 1:      # keep a pool of one
-                self.model_pool.pop(next(iter(self.model_pool)))
-            self.model_pool[model_id] = ModelWrapper(model)
+    def _job_models(self, job):
+        if 'model_id' not in job:
+            return {}
+        self.models.ensure(job['model_id'].values())
+        return {p: self.models.get(mid)
+                for p, mid in job['model_id'].items()}
 
     def run(self):
         while True:
-            args = send_recv(self.conn, ('args', None))
-            if args is None:
+            job = send_recv(self.conn, ('args', None))
+            if job is None:
                 break
-            role = args['role']
-
-            models = {}
-            if 'model_id' in args:
-                self._fetch_models(list(args['model_id'].values()))
-                for p, model_id in args['model_id'].items():
-                    models[p] = self.model_pool.get(model_id, None)
-
-            if role == 'g':
-                episode = self.generator.execute(models, args)
-                send_recv(self.conn, ('episode', episode))
-            elif role == 'e':
-                result = self.evaluator.execute(models, args)
-                send_recv(self.conn, ('result', result))
+            models = self._job_models(job)
+            if job['role'] == 'g':
+                payload = ('episode', self.generator.execute(models, job))
+            elif job['role'] == 'e':
+                payload = ('result', self.evaluator.execute(models, job))
+            else:
+                continue
+            send_recv(self.conn, payload)
 
 
 def make_worker_args(args, n_ga, gaid, base_wid, wid, conn):
+    # worker ids interleave across gathers: gather g's k-th worker gets
+    # base + k*n_gathers + g, so ids are globally distinct and dense
     return args, conn, base_wid + wid * n_ga + gaid
 
 
 def open_worker(args, conn, wid):
-    worker = Worker(args, conn, wid)
-    worker.run()
+    Worker(args, conn, wid).run()
+
+
+# -- gather relay ------------------------------------------------------------
+
+class _ArgsPrefetcher:
+    """Pulls job-args from upstream in bursts and hands them out one by
+    one (cuts per-job round trips by the burst factor)."""
+
+    def __init__(self, upstream, burst):
+        self.upstream = upstream
+        self.burst = burst
+        self.pending = []
+
+    def next(self):
+        if not self.pending:
+            self.upstream.send(('args', [None] * self.burst))
+            self.pending = list(self.upstream.recv())
+        return self.pending.pop(0)
+
+
+class _UploadBuffer:
+    """Accumulates episode/result uploads and flushes them upstream in
+    batches (one round trip per burst, per command kind)."""
+
+    def __init__(self, upstream, burst):
+        self.upstream = upstream
+        self.burst = burst
+        self.held = {}
+        self.count = 0
+
+    def add(self, command, payload):
+        self.held.setdefault(command, []).append(payload)
+        self.count += 1
+        if self.count >= self.burst:
+            self.flush()
+
+    def flush(self):
+        for command, payloads in self.held.items():
+            self.upstream.send((command, payloads))
+            self.upstream.recv()
+        self.held = {}
+        self.count = 0
 
 
 class Gather(QueueCommunicator):
-    """Aggregation node: batches args requests upstream, caches model
-    responses by id, and buffers episode/result uploads."""
+    """Relay node between the learner and ~16 leaf workers: prefetches job
+    args, caches model pulls by id, batches uploads."""
 
     def __init__(self, args, conn, gaid):
         print('started gather %d' % gaid)
         super().__init__()
         self.gather_id = gaid
         self.server_conn = conn
-        self.args_queue = deque()
-        self.data_map = {'model': {}}
-        self.result_send_map = {}
-        self.result_send_cnt = 0
 
-        n_pro, n_ga = args['worker']['num_parallel'], args['worker']['num_gathers']
-        num_workers_here = (n_pro // n_ga) + int(gaid < n_pro % n_ga)
-        base_wid = args['worker'].get('base_worker_id', 0)
+        wcfg = args['worker']
+        n_pro, n_ga = wcfg['num_parallel'], wcfg['num_gathers']
+        local_workers = n_pro // n_ga + (1 if gaid < n_pro % n_ga else 0)
+        base_wid = wcfg.get('base_worker_id', 0)
+        for leaf in open_multiprocessing_connections(
+                local_workers, open_worker,
+                lambda wid, c: make_worker_args(args, n_ga, gaid, base_wid,
+                                                wid, c)):
+            self.add_connection(leaf)
 
-        worker_conns = open_multiprocessing_connections(
-            num_workers_here, open_worker,
-            functools.partial(make_worker_args, args, n_ga, gaid, base_wid))
-        for conn_ in worker_conns:
-            self.add_connection(conn_)
-
-        self.buffer_length = 1 + len(worker_conns) // 4
+        burst = 1 + local_workers // 4
+        self.args_feed = _ArgsPrefetcher(conn, burst)
+        self.uploads = _UploadBuffer(conn, burst)
+        self.model_cache = {}
 
     def __del__(self):
         print('finished gather %d' % self.gather_id)
 
+    def _pull_model(self, model_id):
+        if model_id not in self.model_cache:
+            self.server_conn.send(('model', model_id))
+            self.model_cache[model_id] = self.server_conn.recv()
+        return self.model_cache[model_id]
+
     def run(self):
+        import queue as _queue
         while self.connection_count() > 0:
             try:
-                conn, (command, args) = self.recv(timeout=0.3)
-            except queue.Empty:
+                conn, (command, payload) = self.recv(timeout=0.3)
+            except _queue.Empty:
                 continue
-
             if command == 'args':
-                if len(self.args_queue) == 0:
-                    self.server_conn.send((command, [None] * self.buffer_length))
-                    self.args_queue += self.server_conn.recv()
-                self.send(conn, self.args_queue.popleft())
-
-            elif command in self.data_map:
-                data_id = args
-                if data_id not in self.data_map[command]:
-                    self.server_conn.send((command, args))
-                    self.data_map[command][data_id] = self.server_conn.recv()
-                self.send(conn, self.data_map[command][data_id])
-
-            else:
-                # ack first, upload in buffered bursts
-                self.send(conn, None)
-                self.result_send_map.setdefault(command, []).append(args)
-                self.result_send_cnt += 1
-                if self.result_send_cnt >= self.buffer_length:
-                    for cmd, args_list in self.result_send_map.items():
-                        self.server_conn.send((cmd, args_list))
-                        self.server_conn.recv()
-                    self.result_send_map = {}
-                    self.result_send_cnt = 0
+                self.send(conn, self.args_feed.next())
+            elif command == 'model':
+                self.send(conn, self._pull_model(payload))
+            else:                       # episode / result upload
+                self.send(conn, None)   # ack immediately, flush in bursts
+                self.uploads.add(command, payload)
 
 
 def gather_loop(args, conn, gaid):
-    gather = Gather(args, conn, gaid)
-    gather.run()
+    Gather(args, conn, gaid).run()
 
+
+# -- cluster frontends -------------------------------------------------------
 
 class WorkerCluster(QueueCommunicator):
-    """Local mode: gathers+workers as child processes over pipes."""
+    """Local mode: gather+worker tree as child processes over pipes."""
 
     def __init__(self, args):
         super().__init__()
         self.args = args
 
     def run(self):
-        if 'num_gathers' not in self.args['worker']:
-            self.args['worker']['num_gathers'] = \
-                1 + max(0, self.args['worker']['num_parallel'] - 1) // 16
-        for i in range(self.args['worker']['num_gathers']):
-            conn0, conn1 = mp.Pipe(duplex=True)
-            mp.Process(target=gather_loop, args=(self.args, conn1, i)).start()
-            conn1.close()
-            self.add_connection(conn0)
+        wcfg = self.args['worker']
+        if 'num_gathers' not in wcfg:
+            wcfg['num_gathers'] = _gathers_for(wcfg['num_parallel'])
+        for gaid in range(wcfg['num_gathers']):
+            here, there = mp.Pipe(duplex=True)
+            mp.Process(target=gather_loop,
+                       args=(self.args, there, gaid)).start()
+            there.close()
+            self.add_connection(here)
 
 
 class WorkerServer(QueueCommunicator):
-    """Remote mode server: entry handshake (:9999) assigns worker id ranges
-    and returns the full config; gather data connections accepted on :9998."""
+    """Remote mode server.
+
+    Entry thread (:9999): each connecting machine sends its worker args,
+    receives the full config back with a disjoint ``base_worker_id``
+    range.  Data thread (:9998): accepts gather connections into the hub.
+    """
 
     def __init__(self, args):
         super().__init__()
         self.args = args
         self.total_worker_count = 0
 
+    def _entry_loop(self, port):
+        print('started entry server %d' % port)
+        for conn in accept_socket_connections(port=port):
+            worker_args = conn.recv()
+            print('accepted connection from %s!' % worker_args['address'])
+            worker_args['base_worker_id'] = self.total_worker_count
+            self.total_worker_count += worker_args['num_parallel']
+            reply = copy.deepcopy(self.args)
+            reply['worker'] = worker_args
+            conn.send(reply)
+            conn.close()
+
+    def _data_loop(self, port):
+        print('started worker server %d' % port)
+        for conn in accept_socket_connections(port=port):
+            self.add_connection(conn)
+
     def run(self):
-        def entry_server(port):
-            print('started entry server %d' % port)
-            acceptor = accept_socket_connections(port=port)
-            while True:
-                conn = next(acceptor)
-                worker_args = conn.recv()
-                print('accepted connection from %s!' % worker_args['address'])
-                worker_args['base_worker_id'] = self.total_worker_count
-                self.total_worker_count += worker_args['num_parallel']
-                args = copy.deepcopy(self.args)
-                args['worker'] = worker_args
-                conn.send(args)
-                conn.close()
-
-        def worker_server(port):
-            print('started worker server %d' % port)
-            acceptor = accept_socket_connections(port=port)
-            while True:
-                self.add_connection(next(acceptor))
-
-        threading.Thread(target=entry_server, args=(ENTRY_PORT,), daemon=True).start()
-        threading.Thread(target=worker_server, args=(WORKER_PORT,), daemon=True).start()
+        threading.Thread(target=self._entry_loop, args=(ENTRY_PORT,),
+                         daemon=True).start()
+        threading.Thread(target=self._data_loop, args=(WORKER_PORT,),
+                         daemon=True).start()
 
 
 def entry(worker_args):
-    conn = connect_socket_connection(worker_args['server_address'], ENTRY_PORT)
-    conn.send(worker_args)
-    args = conn.recv()
+    conn = connect_socket_connection(worker_args['server_address'],
+                                     ENTRY_PORT)
+    full_args = send_recv(conn, worker_args)
     conn.close()
-    return args
+    return full_args
 
 
 class RemoteWorkerCluster:
-    """Remote machine side: handshake, then one gather process per data
-    connection to the learner."""
+    """Remote machine side: entry handshake, then one gather process per
+    data connection back to the learner."""
 
     def __init__(self, args):
         args['address'] = gethostname()
         if 'num_gathers' not in args:
-            args['num_gathers'] = 1 + max(0, args['num_parallel'] - 1) // 16
+            args['num_gathers'] = _gathers_for(args['num_parallel'])
         self.args = args
 
     def run(self):
-        args = entry(self.args)
-        print(args)
-        prepare_env(args['env'])
-
-        processes = []
+        full_args = entry(self.args)
+        print(full_args)
+        prepare_env(full_args['env'])
+        children = []
         try:
-            for i in range(self.args['num_gathers']):
-                conn = connect_socket_connection(self.args['server_address'], WORKER_PORT)
-                p = mp.Process(target=gather_loop, args=(args, conn, i))
-                p.start()
+            for gaid in range(self.args['num_gathers']):
+                conn = connect_socket_connection(
+                    self.args['server_address'], WORKER_PORT)
+                proc = mp.Process(target=gather_loop,
+                                  args=(full_args, conn, gaid))
+                proc.start()
                 conn.close()
-                processes.append(p)
+                children.append(proc)
             while True:
                 time.sleep(100)
         finally:
-            for p in processes:
-                p.terminate()
+            for proc in children:
+                proc.terminate()
 
 
 def worker_main(args, argv):
