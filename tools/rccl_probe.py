@@ -34,7 +34,8 @@ def build_step(dist_on, seed=17):
     torch.manual_seed(seed)
     model = GeeseNet().cuda().train()
     params = list(model.parameters())
-    opt = torch.optim.Adam(params, lr=1e-3, weight_decay=1e-5)
+    opt = torch.optim.Adam(params, lr=torch.tensor(1e-3, device='cuda'),
+                           weight_decay=1e-5, capturable=True)
     reducer = hdist.GradReducer(params)
 
     torch.manual_seed(seed + 1)
