@@ -623,18 +623,41 @@ class MultiProcGeesePool:
         self.conns[wid].send(('go', slot))
         self.timing['sample'] += time.time() - t0    # event sync + go
 
+    def _poll_completions(self, force_first=False):
+        """Complete every in-flight round whose GPU work already finished
+        (event fired) — the worker's 'go' goes out the moment its results
+        are ready instead of waiting for the next service.  Rounds finish
+        in issue order (one engine, one stream), so scanning the FIFO head
+        suffices.  ``force_first`` blocks on the oldest round."""
+        while self._fifo:
+            wid, slot = self._fifo[0]
+            M = self.inflight[(wid, slot)]
+            if M and self.graphed is not None and not force_first \
+                    and not self._events[wid][slot].query():
+                break
+            self._complete(*self._fifo.pop(0))
+            force_first = False
+
     def step_once(self):
         """Service one child request (whichever is ready first — a
         jittering child never stalls the sweep): collect its obs, issue
-        its inference, and complete the oldest in-flight round.  With
-        ``slots`` > 1 each child keeps another half-shard stepping while
-        this one's round trip is in flight.  Returns frames reported."""
+        its inference, and hand every finished round back the moment its
+        GPU event fires (polled between waits).  With ``slots`` > 1 each
+        child keeps another half-shard stepping while this one's round
+        trip is in flight.  Returns frames reported."""
         import time
         import multiprocessing.connection as mpc
         t0 = time.time()
-        waitable = [c for i, c in enumerate(self.conns)
-                    if self._inflight_cnt(i) < self.slots]
-        ready = mpc.wait(waitable)
+        while True:
+            self._poll_completions()
+            waitable = [c for i, c in enumerate(self.conns)
+                        if self._inflight_cnt(i) < self.slots]
+            if not waitable:
+                self._poll_completions(force_first=True)
+                continue
+            ready = mpc.wait(waitable, timeout=0.0005 if self._fifo else None)
+            if ready:
+                break
         conn = self.conns[self.rr] if self.conns[self.rr] in ready else ready[0]
         wid = self.conns.index(conn)
         self.rr = (wid + 1) % self.workers
@@ -684,9 +707,11 @@ class MultiProcGeesePool:
         # HANDYRL_ACTOR_SYNC=1: complete every service immediately (no
         # overlap at all) — a race-diagnosis probe for
         # tools/repro_transport_race.py
-        depth = 0 if os.environ.get('HANDYRL_ACTOR_SYNC') == '1' else 1
-        while len(self._fifo) > depth:
-            self._complete(*self._fifo.pop(0))
+        if os.environ.get('HANDYRL_ACTOR_SYNC') == '1':
+            while self._fifo:
+                self._complete(*self._fifo.pop(0))
+        else:
+            self._poll_completions()
         return frames
 
     def refresh_weights(self):
