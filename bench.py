@@ -96,12 +96,20 @@ def main():
     # end-to-end (it steers self-play into the short-episode regime;
     # BASELINE.md post-fix slots comparison)
     os.environ.setdefault('HANDYRL_ACTOR_SLOTS', '1')
+    # device-side trajectory recording (handyrl_amd/traj): the actor graph
+    # scatters obs/alive/(action,prob,value) into HBM rings; finished
+    # episodes commit D2D into the replay ring — env workers ship only
+    # metadata, never observation arrays.  HANDYRL_TRAJ=0 restores the
+    # host-recorded episode path.
+    traj_mode = torch.cuda.is_available() and device_replay and \
+        os.environ.get('HANDYRL_TRAJ', '1') == '1'
     mpool = None
     if actor_procs > 0:
         from handyrl_amd.actor import MultiProcGeesePool
         mpool = MultiProcGeesePool(args, n_games=cli.envs,
                                    seed=1000 + hdist.env_rank() * 31,
-                                   workers=actor_procs)
+                                   workers=actor_procs,
+                                   traj_mode=traj_mode)
 
     use_cuda = torch.cuda.is_available()
     device = torch.device('cuda', local_rank) if use_cuda else torch.device('cpu')
@@ -125,7 +133,7 @@ def main():
 
     if mpool is not None:
         # env work in child processes; parent runs the inference engine
-        mpool.attach(trainer.model, device)
+        mpool.attach(trainer.model, device, replay=replay)
         pool = mpool
     else:
         # in-process fallback: pipelined two-shard pool on GPU, plain on CPU

@@ -29,13 +29,19 @@ class GeeseActorPool:
     """Self-play actor pool for Hungry Geese on one GPU."""
 
     def __init__(self, model, args, n_games=256, device=None, seed=0,
-                 use_graphs=True, engine=None, block_episodes=False):
+                 use_graphs=True, engine=None, block_episodes=False,
+                 record_host=True):
         # block_episodes: package ALL games finishing in a step as ONE
         # concatenated columnar block (env-worker processes; the parent's
         # drain thread splits it back into zero-copy per-episode views via
         # split_episode_block) — replaces per-episode dict building and
         # slice copies on the worker hot loop
+        # record_host=False: device-side trajectory mode (handyrl_amd/traj):
+        # the GPU records obs/alive/action/prob/value in HBM rings inside
+        # the actor graph; this pool only tracks per-game step counters and
+        # reports finished-episode metadata (take_meta / take_finished)
         self.block_episodes = block_episodes
+        self.record_host = record_host
         self.args = args
         self.device = device if device is not None else (
             torch.device('cuda') if torch.cuda.is_available() else torch.device('cpu'))
@@ -47,12 +53,15 @@ class GeeseActorPool:
         G, CAPT = n_games, MAX_STEPS
         # obs are stored CANONICAL (per game, not per seat, 4x smaller);
         # the seat channel-gather happens on the GPU (CHMAP)
-        self.rec_obs = np.zeros((G, CAPT, 17, 7, 11), dtype=np.uint8)
-        self.rec_alive = np.zeros((G, CAPT, N_PLAYERS), dtype=bool)
-        self.rec_act = np.zeros((G, CAPT, N_PLAYERS), dtype=np.int32)
-        self.rec_prob = np.zeros((G, CAPT, N_PLAYERS), dtype=np.float32)
-        self.rec_val = np.zeros((G, CAPT, N_PLAYERS), dtype=np.float32)
+        if record_host:
+            self.rec_obs = np.zeros((G, CAPT, 17, 7, 11), dtype=np.uint8)
+            self.rec_alive = np.zeros((G, CAPT, N_PLAYERS), dtype=bool)
+            self.rec_act = np.zeros((G, CAPT, N_PLAYERS), dtype=np.int32)
+            self.rec_prob = np.zeros((G, CAPT, N_PLAYERS), dtype=np.float32)
+            self.rec_val = np.zeros((G, CAPT, N_PLAYERS), dtype=np.float32)
         self.rec_len = np.zeros(G, dtype=np.int32)
+        self._meta = None              # (rows, t_idx) of the last prepare
+        self._finished = None          # (g_local, lens, outcomes) backlog
         self.completed = []
         self.frames = 0          # env transitions executed (sum over games)
         self.episodes_done = 0
@@ -113,15 +122,31 @@ class GeeseActorPool:
         lg = np.nonzero(live.any(axis=1))[0]
         if len(lg) == 0:
             self._pending = None
+            self._meta = (lg.astype(np.int64), np.empty(0, dtype=np.int64))
             return 0
         t_idx = self.rec_len[lg]
-        self.rec_obs[lg, t_idx] = obs_u8[lg]
-        self.rec_alive[lg, t_idx] = live[lg]
+        if self.record_host:
+            self.rec_obs[lg, t_idx] = obs_u8[lg]
+            self.rec_alive[lg, t_idx] = live[lg]
+        else:
+            self._meta = (lg.astype(np.int64), t_idx.astype(np.int64))
         self._rec_slot = (lg, t_idx)
         M = len(lg)
         out_buf[:M] = obs_u8[lg]
         self._pending = ('ext', lg, live, M)
         return M
+
+    def take_meta(self):
+        """(game rows, step indices) of the last prepare_step (traj mode)."""
+        meta, self._meta = self._meta, None
+        return meta
+
+    def take_finished(self):
+        """Finished-episode metadata accumulated since the last call
+        (traj mode): (g_local int64[K], lens int64[K], outcomes f32[K,4])
+        or None."""
+        fin, self._finished = self._finished, None
+        return fin
 
     def complete_step(self, actions, probs, values):
         """Apply externally computed inference results (4*M game-major
@@ -225,16 +250,17 @@ class GeeseActorPool:
         act_grid = np.zeros((self.n_games, N_PLAYERS), dtype=np.int32)
         act_grid[lg] = np.where(live_lg, actions.reshape(M, N_PLAYERS), 0)
 
-        prob_row = np.zeros((self.n_games, N_PLAYERS), dtype=np.float32)
-        val_row = np.zeros((self.n_games, N_PLAYERS), dtype=np.float32)
-        prob_row[lg] = np.where(live_lg, probs.reshape(M, N_PLAYERS), 0.0)
-        val_row[lg] = np.where(live_lg, values.reshape(M, N_PLAYERS), 0.0)
-
-        # finish the columnar step record started in _phase1
+        # finish the columnar step record started in _phase1 (host mode;
+        # in traj mode the GPU already recorded everything in-graph)
         lg, t_idx = self._rec_slot
-        self.rec_act[lg, t_idx] = act_grid[lg]
-        self.rec_prob[lg, t_idx] = prob_row[lg]
-        self.rec_val[lg, t_idx] = val_row[lg]
+        if self.record_host:
+            prob_row = np.zeros((self.n_games, N_PLAYERS), dtype=np.float32)
+            val_row = np.zeros((self.n_games, N_PLAYERS), dtype=np.float32)
+            prob_row[lg] = np.where(live_lg, probs.reshape(M, N_PLAYERS), 0.0)
+            val_row[lg] = np.where(live_lg, values.reshape(M, N_PLAYERS), 0.0)
+            self.rec_act[lg, t_idx] = act_grid[lg]
+            self.rec_prob[lg, t_idx] = prob_row[lg]
+            self.rec_val[lg, t_idx] = val_row[lg]
         self.rec_len[lg] += 1
 
         tm['record'] += time.time() - t0
@@ -247,7 +273,19 @@ class GeeseActorPool:
         finished = np.nonzero(done)[0]
         if len(finished):
             outcomes = vec.outcomes(finished)
-            if self.block_episodes:
+            if not self.record_host:
+                lens = self.rec_len[finished].astype(np.int64)
+                item = (finished.astype(np.int64), lens,
+                        outcomes.astype(np.float32))
+                if self._finished is None:
+                    self._finished = item
+                else:                  # merge backlog (rare)
+                    pf, pl, po = self._finished
+                    self._finished = (np.concatenate([pf, item[0]]),
+                                      np.concatenate([pl, item[1]]),
+                                      np.concatenate([po, item[2]]))
+                self.rec_len[finished] = 0
+            elif self.block_episodes:
                 self.completed.append(self._package_block(finished, outcomes))
                 self.rec_len[finished] = 0
             else:
@@ -395,15 +433,22 @@ def split_episode_block(item):
 
 
 def _geese_env_worker(conn, ep_conn, obs_name, res_name, n_games, args, seed,
-                      slots=2):
-    """Env-side child process: vectorized stepping, columnar recording and
-    episode packaging on host cores; observations/results move through
-    shared memory, inference runs in the parent (GPU).
+                      slots=2, traj_mode=False):
+    """Env-side child process: vectorized stepping (plus columnar
+    recording and episode packaging in host-record mode) on host cores;
+    observations/results move through shared memory, inference runs in
+    the parent (GPU).
 
     The shard is split into ``slots`` software-pipelined halves: while one
     half's observations are away at the parent (GPU forward + service
     latency), the worker steps/records the other half — the inference
     round trip hides entirely behind env CPU work.
+
+    With ``traj_mode`` the GPU records trajectories in HBM rings inside
+    the actor graph (handyrl_amd/traj): this worker records NOTHING —
+    each obs message carries the (game row, step index) metadata for the
+    in-graph scatter plus finished-episode (rows, lens, outcomes) info,
+    and the episode pipe goes unused.
     """
     from multiprocessing import shared_memory
     obs_shm = shared_memory.SharedMemory(name=obs_name)
@@ -420,13 +465,20 @@ def _geese_env_worker(conn, ep_conn, obs_name, res_name, n_games, args, seed,
         pools.append(GeeseActorPool(
             None, args, n_games=per, device=torch.device('cpu'),
             use_graphs=False, seed=seed + 131 * s,
+            record_host=not traj_mode,
             block_episodes=os.environ.get('HANDYRL_BLOCK_EPISODES',
                                           '1') == '1'))
+
+    def obs_msg(s, m, frames):
+        if traj_mode:
+            return ('obs', s, m, frames, pools[s].take_meta(),
+                    pools[s].take_finished())
+        return ('obs', s, m, frames)
 
     m_inflight = [0] * slots
     for s in range(slots):                    # prime the pipeline
         m_inflight[s] = pools[s].prepare_step(obs_views[s])
-        conn.send(('obs', s, m_inflight[s], 0))
+        conn.send(obs_msg(s, m_inflight[s], 0))
     while True:
         cmd = conn.recv()
         if cmd == 'quit':
@@ -435,15 +487,19 @@ def _geese_env_worker(conn, ep_conn, obs_name, res_name, n_games, args, seed,
         M, frames = m_inflight[s], 0
         if M:
             r = res_views[s][:M * N_PLAYERS]      # 4 seat rows per game
-            frames = pools[s].complete_step(
-                r[:, 0].astype(np.int64), r[:, 1].copy(), r[:, 2].copy())
+            if traj_mode:                  # probs/values recorded on device
+                frames = pools[s].complete_step(
+                    r[:, 0].astype(np.int64), None, None)
+            else:
+                frames = pools[s].complete_step(
+                    r[:, 0].astype(np.int64), r[:, 1].copy(), r[:, 2].copy())
             eps = pools[s].harvest()
             if eps:
                 # episodes travel on their own pipe, drained by a parent
                 # background thread: the service path never deserializes them
                 ep_conn.send(eps)
         m_inflight[s] = pools[s].prepare_step(obs_views[s])
-        conn.send(('obs', s, m_inflight[s], frames))
+        conn.send(obs_msg(s, m_inflight[s], frames))
 
 
 class MultiProcGeesePool:
@@ -457,11 +513,16 @@ class MultiProcGeesePool:
     exists in the parent; attach(model, device) wires the engine after.
     """
 
-    def __init__(self, args, n_games=768, seed=0, workers=3, slots=None):
+    def __init__(self, args, n_games=768, seed=0, workers=3, slots=None,
+                 traj_mode=False):
         import multiprocessing as mp
         from multiprocessing import shared_memory
         self.args = args
         self.workers = workers
+        # traj_mode: device-side trajectory recording (handyrl_amd/traj) —
+        # requires attach(model, device, replay=DeviceReplay) on CUDA;
+        # workers then ship only step/episode METADATA, never obs arrays
+        self.traj_mode = traj_mode
         if slots is None:
             # >1 slots (double-buffered half-shards) measures faster but a
             # GPU-side transport race poisons recorded values (NaN losses
@@ -496,7 +557,7 @@ class MultiProcGeesePool:
             proc = mp.Process(target=_geese_env_worker,
                               args=(child_conn, ep_child, obs_shm.name,
                                     res_shm.name, per, args, seed + 977 * w,
-                                    slots),
+                                    slots, traj_mode),
                               daemon=True)
             proc.start()
             child_conn.close()
@@ -524,15 +585,26 @@ class MultiProcGeesePool:
                        ('obs', 'fwd', 'sample', 'record', 'env', 'package')}
         self.timing['n'] = 1
 
-    def attach(self, model, device):
-        """Wire the inference engine (after CUDA init)."""
+    def attach(self, model, device, replay=None):
+        """Wire the inference engine (after CUDA init).  In traj mode,
+        ``replay`` (a DeviceReplay) receives finished episodes
+        device-to-device via commit_traj."""
         self.model = model
         self.device = device
+        self.replay = replay
+        self.traj = None
         if device.type == 'cuda':
             from .models.geese_net import GeeseFusedEval
             from .hipgraph import GraphedActorForward
             self.fused = GeeseFusedEval(model, device)
-            self.graphed = GraphedActorForward(model, device, fused=self.fused)
+            if self.traj_mode:
+                assert replay is not None, 'traj mode needs a DeviceReplay'
+                from .traj import TrajRecorder
+                self.traj = TrajRecorder(
+                    self.workers * self.slots * self.n_per, device)
+            self.graphed = GraphedActorForward(model, device,
+                                               fused=self.fused,
+                                               traj=self.traj)
             cap = self.n_per * N_PLAYERS
             mk = lambda shape, dt: [[torch.empty(*shape, dtype=dt,
                                                  pin_memory=True)
@@ -544,6 +616,11 @@ class MultiProcGeesePool:
             self._out_pin_np = [[t.numpy() for t in row] for row in self._out_pin]
             self._events = [[torch.cuda.Event() for _ in range(self.slots)]
                             for _ in range(self.workers)]
+            if self.traj is not None:
+                bucket = self.graphed._bucket(self.n_per)
+                self._idx_pin = mk((2, bucket), torch.int64)
+                self._idx_pin_np = [[t.numpy() for t in row]
+                                    for row in self._idx_pin]
             self._register_shm()
 
     def _register_shm(self):
@@ -623,6 +700,28 @@ class MultiProcGeesePool:
         self.conns[wid].send(('go', slot))
         self.timing['sample'] += time.time() - t0    # event sync + go
 
+    def _commit_finished(self, base, fin):
+        """Traj mode: move finished device-recorded episodes into the
+        replay ring (D2D) and queue lightweight stubs for stats.  Runs
+        BEFORE issuing the worker's next forward, whose in-graph scatter
+        would overwrite these trajectory rows; the main stream waits on
+        the commit event to keep that ordering on device."""
+        g_local, lens, outcomes = fin
+        event = self.replay.commit_traj(self.traj, base + g_local, lens,
+                                        outcomes)
+        if event is not None:
+            torch.cuda.current_stream().wait_event(event)
+        job_args = {'player': list(range(N_PLAYERS)),
+                    'model_id': {p: -1 for p in range(N_PLAYERS)}}
+        stubs = [{'args': job_args, 'steps': int(lens[k]),
+                  'outcome': {p: float(outcomes[k, p])
+                              for p in range(N_PLAYERS)},
+                  'committed': True}
+                 for k in range(len(g_local))]
+        with self._completed_lock:
+            self.completed.extend(stubs)
+            self.episodes_done += len(stubs)
+
     def _poll_completions(self, force_first=False):
         """Complete every in-flight round whose GPU work already finished
         (event fired) — the worker's 'go' goes out the moment its results
@@ -661,26 +760,44 @@ class MultiProcGeesePool:
         conn = self.conns[self.rr] if self.conns[self.rr] in ready else ready[0]
         wid = self.conns.index(conn)
         self.rr = (wid + 1) % self.workers
-        tag, slot, M, frames = conn.recv()
+        msg = conn.recv()
+        tag, slot, M, frames = msg[:4]
         assert tag == 'obs'
         self.frames += frames
         self.timing['obs'] += time.time() - t0       # wait + recv
         self.timing['n'] += 1
 
         t0 = time.time()
+        idx_pin = None
+        if self.traj is not None:
+            _tag, _slot, _m, _f, meta, fin = msg
+            base = (wid * self.slots + slot) * self.n_per
+            if fin is not None:
+                self._commit_finished(base, fin)
+            if M:
+                rows, tidx = meta
+                bucket = self.graphed._bucket(M)
+                idx_np = self._idx_pin_np[wid][slot]
+                idx_np[0, :M] = base + rows
+                idx_np[0, M:bucket] = self.traj.scratch_row
+                idx_np[1, :M] = tidx
+                idx_np[1, M:bucket] = 0
+                idx_pin = self._idx_pin[wid][slot]
         if M and self.graphed is not None:
             if self._use_registered:
                 # shm is hipHostRegister'd: DMA straight from/to it, no
                 # staging memcpy on the service path
                 self.graphed.run_async(self._obs_src[wid][slot], M,
                                        self._res_dst[wid][slot],
-                                       self._events[wid][slot])
+                                       self._events[wid][slot],
+                                       idx_pinned=idx_pin)
             else:
                 np.copyto(self._obs_pin_np[wid][slot][:M],
                           self.obs_views[wid][slot][:M])
                 self.graphed.run_async(self._obs_pin[wid][slot], M,
                                        self._out_pin[wid][slot],
-                                       self._events[wid][slot])
+                                       self._events[wid][slot],
+                                       idx_pinned=idx_pin)
             self.inflight[(wid, slot)] = M
         elif M:
             # CPU fallback (tests): synchronous eager inference on the

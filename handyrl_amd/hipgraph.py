@@ -98,7 +98,7 @@ class GraphedActorForward:
     sample -> packed (action, prob, value)} for the GPU actor pool."""
 
     def __init__(self, model, device, warmup_iters=2, fused=None,
-                 canonical=True):
+                 canonical=True, traj=None):
         from . import ops
         from .envs.vec_geese import CHMAP
         self._ops = ops
@@ -106,6 +106,7 @@ class GraphedActorForward:
         self.device = device
         self.fused = fused        # GeeseFusedEval: hand-written MFMA path
         self.canonical = canonical  # obs rows are per GAME; out rows x4
+        self.traj = traj          # TrajRecorder: in-graph HBM recording
         self._chmap = torch.from_numpy(CHMAP).to(device)
         self.graphs = {}
 
@@ -140,18 +141,34 @@ class GraphedActorForward:
                                  device=self.device)
         rows = bucket * 4 if self.canonical else bucket
         zero_mask = torch.zeros(rows, n_actions, device=self.device)
+        static_idx = None
+        if self.traj is not None:
+            # in-graph HBM trajectory recording: game-row / step-index
+            # inputs (padded bucket rows point at the scratch row)
+            static_idx = (torch.full((bucket,), self.traj.scratch_row,
+                                     dtype=torch.int64, device=self.device),
+                          torch.zeros(bucket, dtype=torch.int64,
+                                      device=self.device))
+
+        def run_once():
+            packed = self._fwd_sample(static_obs, zero_mask)
+            if self.traj is not None:
+                self.traj.record_(static_obs, packed,
+                                  static_idx[0], static_idx[1])
+            return packed
+
         stream = torch.cuda.Stream()
         stream.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(stream):
             for _ in range(2):
-                self._fwd_sample(static_obs, zero_mask)
+                run_once()
         torch.cuda.current_stream().wait_stream(stream)
         graph = torch.cuda.CUDAGraph()
         with torch.cuda.graph(graph):
-            packed = self._fwd_sample(static_obs, zero_mask)
+            packed = run_once()
         if was_training:
             self.model.train()
-        self.graphs[bucket] = (graph, static_obs, packed)
+        self.graphs[bucket] = (graph, static_obs, packed, static_idx)
 
     def run(self, obs_u8_cpu, n_actions=4):
         """obs_u8_cpu: torch uint8 tensor (M, 17, 7, 11) on CPU (M <= bucket
@@ -161,7 +178,7 @@ class GraphedActorForward:
         bucket = self._bucket(M)
         if bucket not in self.graphs:
             self._capture(bucket, n_actions)
-        graph, static_obs, packed = self.graphs[bucket]
+        graph, static_obs, packed, _idx = self.graphs[bucket]
         static_obs[:M].copy_(obs_u8_cpu, non_blocking=True)
         # rows [M:bucket) keep stale data; every op is row-independent and
         # the outputs are sliced to the live rows
@@ -175,16 +192,24 @@ class GraphedActorForward:
         q = 32 if self.canonical else 256
         return q * ((M + q - 1) // q)
 
-    def run_async(self, obs_pinned, M, out_pinned, event, n_actions=4):
+    def run_async(self, obs_pinned, M, out_pinned, event, n_actions=4,
+                  idx_pinned=None):
         """Pipelined variant: H2D from a pinned staging tensor, replay, and
         an async D2H of the packed result into ``out_pinned``; ``event``
         records completion.  No host sync — the caller overlaps CPU work
-        and waits on the event."""
+        and waits on the event.
+
+        With in-graph trajectory recording, ``idx_pinned`` is a pinned
+        (2, bucket) int64 tensor of (game row, step index) per obs row,
+        padded rows pointing at the recorder's scratch row."""
         bucket = self._bucket(M)
         if bucket not in self.graphs:
             self._capture(bucket, n_actions)
-        graph, static_obs, packed = self.graphs[bucket]
+        graph, static_obs, packed, static_idx = self.graphs[bucket]
         static_obs[:M].copy_(obs_pinned[:M], non_blocking=True)
+        if static_idx is not None:
+            static_idx[0].copy_(idx_pinned[0, :bucket], non_blocking=True)
+            static_idx[1].copy_(idx_pinned[1, :bucket], non_blocking=True)
         graph.replay()
         R = M * 4 if self.canonical else M
         out_pinned[:R].copy_(packed[:R], non_blocking=True)

@@ -80,11 +80,12 @@ class ColumnRingReplay:
         self.total_added = 0
         self._pin = {}
         self._ingest = None
+        self._ready = deque()
+        self._ready_lock = threading.Lock()
+        if device.type == 'cuda':
+            self._ingest_stream = torch.cuda.Stream()
         if ingest_thread and device.type == 'cuda':
             self._ingest_q = queue_mod.Queue(maxsize=64)
-            self._ready = deque()
-            self._ready_lock = threading.Lock()
-            self._ingest_stream = torch.cuda.Stream()
             self._ingest = threading.Thread(target=self._ingest_loop,
                                             daemon=True)
             self._ingest.start()
@@ -137,8 +138,10 @@ class ColumnRingReplay:
             yield chunk
 
     def publish(self, stream=None):
-        """Make finished background writes sampleable (consumer thread)."""
-        if self._ingest is None:
+        """Make finished background/committed writes sampleable (consumer
+        thread): extends the table after a stream-ordered wait on each
+        write's completion event."""
+        if self.device.type != 'cuda':
             return
         if stream is None:
             stream = torch.cuda.current_stream()
@@ -153,9 +156,10 @@ class ColumnRingReplay:
 
     def flush(self):
         """Block until every queued episode block is published."""
-        if self._ingest is None:
+        if self.device.type != 'cuda':
             return
-        self._ingest_q.join()
+        if self._ingest is not None:
+            self._ingest_q.join()
         torch.cuda.synchronize()
         self.publish()
 
@@ -273,7 +277,11 @@ class ColumnRingReplay:
             self.table.popleft()
 
     def extend(self, episodes):
-        """Append columnar episodes (numpy fields) to the device ring."""
+        """Append columnar episodes (numpy fields) to the device ring.
+        Episode stubs already committed device-side (traj mode,
+        ep['committed']) pass through untouched — their rows are in the
+        ring already."""
+        episodes = [ep for ep in episodes if not ep.get('committed')]
         if not episodes:
             return
         limit = max(512, self.ring_T // 4)
@@ -319,6 +327,80 @@ class DeviceReplay(ColumnRingReplay):
         from .envs.vec_geese import CHMAP
         self._chmap = torch.from_numpy(CHMAP).to(device)     # (4, 17)
         self._arange_cache = {}
+
+    # -- device-side ingest (traj mode) --------------------------------------
+    def commit_traj(self, traj, g_rows, lens, outcomes):
+        """Copy finished device-recorded episodes (handyrl_amd/traj) into
+        the ring DEVICE-TO-DEVICE and publish their table entries.
+
+        g_rows/lens: int64[K] global trajectory rows and episode lengths;
+        outcomes: float32[K, 4].  Returns the completion event (the caller
+        must make the next trajectory write to these rows wait on it) or
+        None when K == 0.  Runs on the ingest stream, ordered after all
+        issued main-stream work (the in-graph scatters of these rows)."""
+        K = len(g_rows)
+        if K == 0:
+            return None
+        lens = np.asarray(lens, dtype=np.int64)
+        n = int(lens.sum())
+        assert n <= self.ring_T // 4, 'episode burst larger than ring/4'
+        import time as _time
+        waited = 0.0
+        while True:                    # reader-floor back-pressure
+            floor = self._reader_floor
+            if floor is None or self.head + n - self.ring_T <= floor:
+                break
+            _time.sleep(0.001)
+            waited += 0.001
+            if waited > 5.0:
+                raise RuntimeError('replay ring writer starved (floor %r)'
+                                   % (floor,))
+        g_flat, t_flat = traj.episode_indices(np.asarray(g_rows), lens)
+        dev = self.device
+        head0 = self.head
+        if dev.type == 'cuda':
+            stream = self._ingest_stream if self._ingest is not None \
+                else torch.cuda.current_stream()
+            stream.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(stream):
+                event = self._commit_copy(traj, g_flat, t_flat, head0, n)
+        else:
+            event = None
+            self._commit_copy(traj, g_flat, t_flat, head0, n)
+        entries, pos = [], head0
+        for k in range(K):
+            entries.append((pos, int(lens[k]),
+                            np.asarray(outcomes[k], dtype=np.float32)))
+            pos += int(lens[k])
+        self.head = pos
+        self.head_planned = max(self.head_planned, pos)
+        with self._table_lock:
+            if event is None:
+                self.table.extend(entries)
+                self.total_added += len(entries)
+                self._evict()
+        if event is not None:
+            with self._ready_lock:
+                self._ready.append((entries, n, event))
+        return event
+
+    def _commit_copy(self, traj, g_flat, t_flat, head0, n):
+        dev = self.device
+        g_t = torch.from_numpy(g_flat).to(dev, non_blocking=True)
+        t_t = torch.from_numpy(t_flat).to(dev, non_blocking=True)
+        dst = (torch.arange(n, device=dev, dtype=torch.int64) + head0) \
+            % self.ring_T
+        self.obs[dst] = traj.obs[g_t, t_t]
+        self.alive[dst] = traj.alive[g_t, t_t]
+        rec = traj.rec[g_t, t_t]                     # (n, 4, 3)
+        self.action[dst] = rec[..., 0].to(torch.int32)
+        self.prob[dst] = rec[..., 1]
+        self.value[dst] = rec[..., 2]
+        if dev.type != 'cuda':
+            return None
+        event = torch.cuda.Event()
+        event.record()
+        return event
 
     # -- sample path ---------------------------------------------------------
     def sample_indices(self, batch_size):

@@ -428,3 +428,125 @@ def test_turn_device_replay_recurrent_training():
         assert float(dcnt) > 0
     assert any(not torch.equal(b, p.detach())
                for b, p in zip(before, trainer.params))
+
+
+@requires_gpu
+def test_traj_mode_pool_end_to_end():
+    """Device-side trajectory recording: the traj-mode multiproc pool
+    fills the replay ring via in-graph scatters + commit_traj (workers
+    ship only metadata), and the captured train step runs on it."""
+    import os
+    from handyrl_amd.actor import MultiProcGeesePool
+    from handyrl_amd.models.geese_net import GeeseNet
+    from handyrl_amd.replay import DeviceReplay
+    from handyrl_amd.train import Trainer
+    from handyrl_amd.hipgraph import GraphedReplayTrainStep
+
+    args = {
+        'turn_based_training': False, 'observation': False, 'gamma': 0.8,
+        'forward_steps': 8, 'burn_in_steps': 0, 'compress_steps': 4,
+        'entropy_regularization': 0.1, 'entropy_regularization_decay': 0.1,
+        'batch_size': 16, 'minimum_episodes': 2, 'maximum_episodes': 500,
+        'num_batchers': 1, 'lambda': 0.7, 'policy_target': 'VTRACE',
+        'value_target': 'VTRACE', 'seed': 0, 'bf16': True,
+        'compress_episodes': False,
+    }
+    pool = MultiProcGeesePool(args, n_games=128, seed=3, workers=2,
+                              traj_mode=True)
+    device = torch.device('cuda', 0)
+    trainer = Trainer(args, GeeseNet(), device=device, batcher=False)
+    replay = DeviceReplay(args, device, bytes_budget=256 << 20,
+                          ingest_thread=True)
+    pool.attach(trainer.model, device, replay=replay)
+    try:
+        trainer.model.eval()
+        while pool.episodes_done < 40:
+            pool.step_once()
+        replay.flush()
+        assert len(replay) >= 40
+        # stubs carry stats but no data
+        stubs = pool.harvest()
+        assert stubs and all(s.get('committed') for s in stubs)
+        assert all(s['steps'] > 0 for s in stubs)
+        oc = [sum(s['outcome'].values()) for s in stubs]
+        assert all(abs(v) < 1e-5 for v in oc)      # zero-sum ranks
+
+        step = GraphedReplayTrainStep(trainer, replay, args['batch_size'])
+        assert step.graph is not None
+        for _ in range(3):
+            losses, dcnt = step.step()
+            torch.cuda.synchronize()
+            assert torch.isfinite(losses['total']), losses
+            assert float(dcnt) > 0
+    finally:
+        pool.shutdown()
+
+
+@requires_gpu
+def test_traj_ring_matches_worker_host_recording():
+    """The in-graph recorder reproduces the host-recorded trajectory
+    bit-for-bit for a single in-process pool driven both ways on the
+    same action stream (the graph's sampled actions feed both)."""
+    from handyrl_amd.actor import GeeseActorPool
+    from handyrl_amd.models.geese_net import GeeseNet
+    from handyrl_amd.traj import TrajRecorder
+    from handyrl_amd.hipgraph import GraphedActorForward
+    from handyrl_amd.models.geese_net import GeeseFusedEval
+
+    device = torch.device('cuda', 0)
+    torch.manual_seed(0)
+    model = GeeseNet().to(device).eval()
+    args = {'compress_episodes': False, 'compress_steps': 4, 'gamma': 0.8,
+            'observation': False, 'turn_based_training': False}
+
+    n = 32
+    traj = TrajRecorder(n, device)
+    fused = GeeseFusedEval(model, device)
+    graphed = GraphedActorForward(model, device, fused=fused, traj=traj)
+
+    pool = GeeseActorPool(None, args, n_games=n, device=torch.device('cpu'),
+                          use_graphs=False, seed=5, record_host=True)
+    obs_buf = np.zeros((n, 17, 7, 11), dtype=np.uint8)
+    obs_pin = torch.empty(n, 17, 7, 11, dtype=torch.uint8, pin_memory=True)
+    out_pin = torch.empty(n * 4, 3, dtype=torch.float32, pin_memory=True)
+    bucket = graphed._bucket(n)
+    idx_pin = torch.empty(2, bucket, dtype=torch.int64, pin_memory=True)
+    ev = torch.cuda.Event()
+
+    for _ in range(10):
+        M = pool.prepare_step(obs_buf)
+        if M == 0:
+            break
+        lg, t_idx = pool._rec_slot
+        idx_pin[0, :M] = torch.from_numpy(lg.astype(np.int64))
+        idx_pin[0, M:] = traj.scratch_row
+        idx_pin[1, :M] = torch.from_numpy(t_idx.astype(np.int64))
+        idx_pin[1, M:] = 0
+        obs_pin[:M].copy_(torch.from_numpy(obs_buf[:M]))
+        graphed.run_async(obs_pin, M, out_pin, ev, idx_pinned=idx_pin)
+        ev.synchronize()
+        r = out_pin.numpy()[:M * 4]
+        pool.complete_step(r[:, 0].astype(np.int64), r[:, 1].copy(),
+                           r[:, 2].copy())
+
+    # compare device rings vs the pool's host recording for rows still
+    # in progress (not yet reset)
+    t_obs = traj.obs.cpu().numpy()
+    t_alive = traj.alive.cpu().numpy()
+    t_rec = traj.rec.cpu().numpy()
+    checked = 0
+    for g in range(n):
+        S = int(pool.rec_len[g])
+        if S == 0:
+            continue
+        np.testing.assert_array_equal(t_obs[g, :S], pool.rec_obs[g, :S])
+        np.testing.assert_array_equal(t_alive[g, :S], pool.rec_alive[g, :S])
+        a = pool.rec_alive[g, :S]
+        np.testing.assert_array_equal(
+            t_rec[g, :S, :, 0].astype(np.int32)[a], pool.rec_act[g, :S][a])
+        np.testing.assert_allclose(t_rec[g, :S, :, 1][a],
+                                   pool.rec_prob[g, :S][a], rtol=0, atol=0)
+        np.testing.assert_allclose(t_rec[g, :S, :, 2][a],
+                                   pool.rec_val[g, :S][a], rtol=0, atol=0)
+        checked += 1
+    assert checked > 0
