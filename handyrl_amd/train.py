@@ -438,9 +438,56 @@ class Trainer:
             self.update_queue.put((model, self.steps))
 
 
+class ScoreBook:
+    """(count, sum, sum-of-squares) accumulators keyed by (epoch, tag).
+
+    Backs the win-rate and generation-stats reports; the stdout line
+    grammar is a compatibility contract with the plot tools
+    (scripts/win_rate_plot.py, stats_plot.py)."""
+
+    def __init__(self):
+        self._acc = {}
+
+    def add(self, key, value, tag=None):
+        n, s, s2 = self._acc.get((key, tag), (0, 0.0, 0.0))
+        self._acc[(key, tag)] = (n + 1, s + value, s2 + value * value)
+
+    def tags(self, key):
+        return sorted(t for (k, t) in self._acc if k == key and t is not None)
+
+    def has(self, key, tag=None):
+        return (key, tag) in self._acc
+
+    def total(self, key, tag=None):
+        return self._acc.get((key, tag), (0, 0.0, 0.0))
+
+    def print_win_rate(self, key, tag=None, label=''):
+        n, r, _ = self.total(key, tag)
+        mean = r / (n + 1e-6)
+        name_tag = ' (%s)' % label if label != '' else ''
+        print('win rate%s = %.3f (%.1f / %d)'
+              % (name_tag, (mean + 1) / 2, (r + n) / 2, n))
+
+    def print_generation(self, key):
+        n, r, r2 = self.total(key)
+        mean = r / (n + 1e-6)
+        std = (r2 / (n + 1e-6) - mean ** 2) ** 0.5
+        print('generation stats = %.3f +- %.3f' % (mean, std))
+
+
 class Learner:
     """Central conductor: owns the model epoch, serves worker requests
-    (job args / episodes / results / model pulls) and rolls training epochs."""
+    (job args / episodes / results / model pulls) and rolls training
+    epochs.
+
+    Protocol and report-line parity with the reference Learner
+    (reference train.py:403-645): job dicts carry role/player/model_id,
+    the eval share follows eval_rate with seat rotation, model pulls
+    return pickled modules, and the epoch report prints the win-rate /
+    generation-stats grammar the plot tools parse.  MI355X extensions:
+    a GPU actor mode (self-play as batched inference on the learner GPU,
+    CPU workers demoted to evaluation), optimizer-state sidecars, and
+    reference-layout checkpoint export."""
 
     def __init__(self, args, net=None, remote=False):
         train_args = args['train_args']
@@ -452,8 +499,9 @@ class Learner:
         random.seed(args['seed'])
 
         self.env = make_env(env_args)
-        eval_modify_rate = (args['update_episodes'] ** 0.85) / args['update_episodes']
-        self.eval_rate = max(args['eval_rate'], eval_modify_rate)
+        # never let the eval share starve at tiny update_episodes
+        floor = (args['update_episodes'] ** 0.85) / args['update_episodes']
+        self.eval_rate = max(args['eval_rate'], floor)
         self.shutdown_flag = False
         self.flags = set()
 
@@ -463,12 +511,10 @@ class Learner:
             self.model.load_state_dict(
                 torch.load(self.model_path(self.model_epoch)), strict=False)
 
-        # generation / evaluation statistics: {model_id: (n, sum_r, sum_r2)}
-        self.generation_results = {}
+        self.generation_stats = ScoreBook()
+        self.eval_stats = ScoreBook()
         self.num_episodes = 0
         self.num_returned_episodes = 0
-        self.results = {}
-        self.results_per_opponent = {}
         self.num_results = 0
 
         self.worker = WorkerServer(args) if remote else WorkerCluster(args)
@@ -490,6 +536,8 @@ class Learner:
         # processes; any CPU workers then serve evaluation jobs only.
         self.gpu_actor = bool(args['worker'].get('type') == 'gpu')
         self.feed_lock = threading.Lock()
+
+    # -- checkpoint files --------------------------------------------------
 
     def model_path(self, model_id):
         return os.path.join('models', str(model_id) + '.pth')
@@ -524,35 +572,47 @@ class Learner:
                         'data_cnt_ema': self.trainer.data_cnt_ema},
                        self.optimizer_path(self.model_epoch))
 
+    # -- data intake -------------------------------------------------------
+
     def feed_episodes(self, episodes):
+        live = [e for e in episodes if e is not None]
         with self.feed_lock:
-            for episode in episodes:
-                if episode is None:
-                    continue
+            for episode in live:
                 for p in episode['args']['player']:
-                    model_id = self.model_epoch
-                    outcome = episode['outcome'][p]
-                    n, r, r2 = self.generation_results.get(model_id, (0, 0, 0))
-                    self.generation_results[model_id] = \
-                        n + 1, r + outcome, r2 + outcome ** 2
+                    self.generation_stats.add(self.model_epoch,
+                                              episode['outcome'][p])
                 self.num_returned_episodes += 1
-                self.num_episodes += 1 if self.gpu_actor else 0
+                if self.gpu_actor:
+                    self.num_episodes += 1
                 if self.num_returned_episodes % 100 == 0:
                     print(self.num_returned_episodes, end=' ', flush=True)
+            self.trainer.episodes.extend(live)
+            self._ram_watchdog()
 
-            self.trainer.episodes.extend([e for e in episodes if e is not None])
-
-            mem_percent = psutil.virtual_memory().percent if psutil is not None else 0
-            mem_ok = mem_percent <= 95
-            maximum_episodes = self.args['maximum_episodes'] if mem_ok else \
-                int(len(self.trainer.episodes) * 95 / mem_percent)
-
-            if not mem_ok and 'memory_over' not in self.flags:
-                warnings.warn('memory usage %.1f%% with buffer size %d' %
-                              (mem_percent, len(self.trainer.episodes)))
+    def _ram_watchdog(self):
+        """Shrink the episode buffer under host-RAM pressure (reference
+        train.py:474-483 semantics); called under feed_lock."""
+        mem_percent = psutil.virtual_memory().percent if psutil else 0
+        limit = self.args['maximum_episodes']
+        if mem_percent > 95:
+            limit = int(len(self.trainer.episodes) * 95 / mem_percent)
+            if 'memory_over' not in self.flags:
+                warnings.warn('memory usage %.1f%% with buffer size %d'
+                              % (mem_percent, len(self.trainer.episodes)))
                 self.flags.add('memory_over')
+        self.trainer.episodes.trim(limit)
 
-            self.trainer.episodes.trim(maximum_episodes)
+    def feed_results(self, results):
+        for result in results:
+            if result is None:
+                continue
+            for p in result['args']['player']:
+                res = result['result'][p]
+                self.eval_stats.add(self.model_epoch, res)
+                self.eval_stats.add(self.model_epoch, res,
+                                    tag=result['opponent'])
+
+    # -- GPU actor mode ----------------------------------------------------
 
     def _gpu_actor_loop(self):
         """Self-play generation on the learner GPU (worker type 'gpu'):
@@ -594,134 +654,119 @@ class Learner:
                 self.feed_episodes(episodes)
         print('finished gpu actor pool')
 
-    def feed_results(self, results):
-        for result in results:
-            if result is None:
-                continue
-            for p in result['args']['player']:
-                model_id = self.model_epoch
-                res = result['result'][p]
-                n, r, r2 = self.results.get(model_id, (0, 0, 0))
-                self.results[model_id] = n + 1, r + res, r2 + res ** 2
-                opp_map = self.results_per_opponent.setdefault(model_id, {})
-                opponent = result['opponent']
-                n, r, r2 = opp_map.get(opponent, (0, 0, 0))
-                opp_map[opponent] = n + 1, r + res, r2 + res ** 2
+    # -- epoch rollover ----------------------------------------------------
 
     def update(self):
         print()
         print('epoch %d' % self.model_epoch)
-
-        if self.model_epoch not in self.results:
-            print('win rate = Nan (0)')
-        else:
-            def output_wp(name, results):
-                n, r, r2 = results
-                mean = r / (n + 1e-6)
-                name_tag = ' (%s)' % name if name != '' else ''
-                print('win rate%s = %.3f (%.1f / %d)' % (name_tag, (mean + 1) / 2, (r + n) / 2, n))
-
-            keys = self.results_per_opponent[self.model_epoch]
-            if len(self.args.get('eval', {}).get('opponent', [])) <= 1 and len(keys) <= 1:
-                output_wp('', self.results[self.model_epoch])
-            else:
-                output_wp('total', self.results[self.model_epoch])
-                for key in sorted(list(self.results_per_opponent[self.model_epoch])):
-                    output_wp(key, self.results_per_opponent[self.model_epoch][key])
-
-        if self.model_epoch not in self.generation_results:
-            print('generation stats = Nan (0)')
-        else:
-            n, r, r2 = self.generation_results[self.model_epoch]
-            mean = r / (n + 1e-6)
-            std = (r2 / (n + 1e-6) - mean ** 2) ** 0.5
-            print('generation stats = %.3f +- %.3f' % (mean, std))
-
+        self._print_report()
         model, steps = self.trainer.update()
         if model is None:
             model = self.model
         self.update_model(model, steps)
         self.flags = set()
 
+    def _print_report(self):
+        ep = self.model_epoch
+        if not self.eval_stats.has(ep):
+            print('win rate = Nan (0)')
+        else:
+            tags = self.eval_stats.tags(ep)
+            single_opponent = \
+                len(self.args.get('eval', {}).get('opponent', [])) <= 1
+            if single_opponent and len(tags) <= 1:
+                self.eval_stats.print_win_rate(ep)
+            else:
+                self.eval_stats.print_win_rate(ep, label='total')
+                for tag in tags:
+                    self.eval_stats.print_win_rate(ep, tag=tag, label=tag)
+        if not self.generation_stats.has(ep):
+            print('generation stats = Nan (0)')
+        else:
+            self.generation_stats.print_generation(ep)
+
+    # -- request serving ---------------------------------------------------
+
+    def _next_job(self):
+        """Assign one worker job (reference train.py:566-587 policy):
+        evaluation until its share catches up with eval_rate — always,
+        when GPU actors generate — otherwise generation; eval jobs rotate
+        the evaluated seat and mark opponent seats with model_id -1."""
+        players = self.env.players()
+        job = {'model_id': {}}
+        evaluate = self.gpu_actor or \
+            self.num_results < self.eval_rate * self.num_episodes
+        if evaluate:
+            job['role'] = 'e'
+            seat = players[self.num_results % len(players)]
+            job['player'] = [seat]
+            for p in players:
+                job['model_id'][p] = self.model_epoch if p == seat else -1
+            self.num_results += 1
+        else:
+            job['role'] = 'g'
+            job['player'] = players
+            for p in players:
+                job['model_id'][p] = self.model_epoch
+            self.num_episodes += 1
+        return job
+
+    def _pickled_model(self, model_id):
+        model = self.model
+        if model_id != self.model_epoch and model_id > 0:
+            try:
+                model = copy.deepcopy(self.model)
+                model.load_state_dict(
+                    torch.load(self.model_path(model_id)), strict=False)
+            except Exception:
+                pass   # fall back to the latest model
+        import pickle
+        return pickle.dumps(model)
+
+    def _serve(self, req, payloads):
+        if req == 'args':
+            if self.shutdown_flag:
+                return [None] * len(payloads)
+            return [self._next_job() for _ in payloads]
+        if req == 'episode':
+            self.feed_episodes(payloads)
+            return [None] * len(payloads)
+        if req == 'result':
+            self.feed_results(payloads)
+            return [None] * len(payloads)
+        if req == 'model':
+            return [self._pickled_model(mid) for mid in payloads]
+        return [None] * len(payloads)
+
+    def _maybe_rollover(self, state):
+        """Epoch rollover bookkeeping; returns True to leave the server
+        loop (GPU-actor shutdown with no workers attached)."""
+        if self.num_returned_episodes < state['next_update']:
+            return False
+        state['next_update'] += self.args['update_episodes']
+        self.update()
+        if 0 <= self.args['epochs'] <= self.model_epoch:
+            self.shutdown_flag = True
+            if self.gpu_actor and self.worker.connection_count() == 0:
+                return True
+        return False
+
     def server(self):
         print('started server')
-        prev_update_episodes = self.args['minimum_episodes']
-        next_update_episodes = prev_update_episodes + self.args['update_episodes']
-
+        state = {'next_update': self.args['minimum_episodes'] +
+                 self.args['update_episodes']}
         while self.worker.connection_count() > 0 or not self.shutdown_flag:
-            # epoch rollover: checked every iteration (with GPU actors the
-            # buffer fills without any worker request traffic)
-            if self.num_returned_episodes >= next_update_episodes:
-                prev_update_episodes = next_update_episodes
-                next_update_episodes = prev_update_episodes + self.args['update_episodes']
-                self.update()
-                if self.args['epochs'] >= 0 and self.model_epoch >= self.args['epochs']:
-                    self.shutdown_flag = True
-                    if self.gpu_actor and self.worker.connection_count() == 0:
-                        break
+            # rollover is checked every iteration: with GPU actors the
+            # buffer fills without any worker request traffic
+            if self._maybe_rollover(state):
+                break
             try:
-                conn, (req, data) = self.worker.recv(timeout=0.3)
+                conn, (req, payload) = self.worker.recv(timeout=0.3)
             except queue.Empty:
                 continue
-
-            multi_req = isinstance(data, list)
-            if not multi_req:
-                data = [data]
-            send_data = []
-
-            if req == 'args':
-                if self.shutdown_flag:
-                    send_data = [None] * len(data)
-                else:
-                    for _ in data:
-                        job = {'model_id': {}}
-                        # evaluation share of jobs per eval_rate; with GPU
-                        # actors generating, CPU workers only evaluate
-                        if self.gpu_actor or \
-                                self.num_results < self.eval_rate * self.num_episodes:
-                            job['role'] = 'e'
-                        else:
-                            job['role'] = 'g'
-
-                        if job['role'] == 'g':
-                            job['player'] = self.env.players()
-                            for p in self.env.players():
-                                job['model_id'][p] = self.model_epoch
-                            self.num_episodes += 1
-                        else:
-                            # rotate the evaluated seat
-                            job['player'] = [self.env.players()[
-                                self.num_results % len(self.env.players())]]
-                            for p in self.env.players():
-                                job['model_id'][p] = self.model_epoch \
-                                    if p in job['player'] else -1
-                            self.num_results += 1
-                        send_data.append(job)
-
-            elif req == 'episode':
-                self.feed_episodes(data)
-                send_data = [None] * len(data)
-
-            elif req == 'result':
-                self.feed_results(data)
-                send_data = [None] * len(data)
-
-            elif req == 'model':
-                import pickle
-                for model_id in data:
-                    model = self.model
-                    if model_id != self.model_epoch and model_id > 0:
-                        try:
-                            model = copy.deepcopy(self.model)
-                            model.load_state_dict(
-                                torch.load(self.model_path(model_id)), strict=False)
-                        except Exception:
-                            pass   # fall back to the latest model
-                    send_data.append(pickle.dumps(model))
-
-            if not multi_req and len(send_data) == 1:
-                send_data = send_data[0]
-            self.worker.send(conn, send_data)
+            batched = isinstance(payload, list)
+            replies = self._serve(req, payload if batched else [payload])
+            self.worker.send(conn, replies if batched else replies[0])
         print('finished server')
 
     def run(self):
