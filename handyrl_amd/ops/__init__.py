@@ -174,6 +174,46 @@ def pack_torus_weights(conv_weight, scale):
     return frag.to(torch.bfloat16)
 
 
+def convlstm_cell(x, h, c, wfrag, bias, nbr, h_out, c_out):
+    """Fused ConvLSTM cell (Geister DRC core, one kernel per cell eval):
+    implicit-GEMM 3x3 zero-pad conv over the K-ordered (x | h) halves with
+    the i/f/o/g gate + state update fused into the epilogue.
+
+    x, h, h_out: (B, 36, 32) bf16 NHWC; c, c_out: (B, 36, 32) fp32;
+    wfrag: pack_convlstm_weights output; bias: (128,) fp32;
+    nbr: (36, 9) int32 from convlstm_neighbor_table (-1 = zero pad).
+    Writes h_out/c_out in place and returns them."""
+    return require().convlstm_cell(x, h, c, wfrag, bias, nbr, h_out, c_out)
+
+
+def convlstm_neighbor_table(device=None):
+    """(36, 9) int32 neighbor table for a 6x6 ZERO-padded 3x3 conv
+    (cross-correlation tap order); -1 marks out-of-board taps."""
+    rows = cols = 6
+    tbl = torch.empty(rows * cols, 9, dtype=torch.int32)
+    for r in range(rows):
+        for c in range(cols):
+            for ky in range(3):
+                for kx in range(3):
+                    rr, cc = r + ky - 1, c + kx - 1
+                    ok = 0 <= rr < rows and 0 <= cc < cols
+                    tbl[r * cols + c, ky * 3 + kx] = rr * cols + cc if ok else -1
+    return tbl.to(device) if device is not None else tbl
+
+
+def pack_convlstm_weights(conv_weight):
+    """Pack a ConvLSTM conv weight (128, 64, 3, 3) fp32 into the fused
+    cell kernel's B-fragment layout (2, 9, 8, 4, 16, 8) bf16:
+    frag[src][tap][cotile][khi][lo][e] = W[n = cotile*16+lo]
+                                          [ci = src*32 + khi*8+e][tap]."""
+    co, ci = conv_weight.shape[0], conv_weight.shape[1]
+    assert co == 128 and ci == 64, (co, ci)
+    w9 = conv_weight.float().reshape(128, 64, 9)
+    wv = w9.reshape(8, 16, 2, 4, 8, 9)        # (cotile, lo, src, khi, e, tap)
+    frag = wv.permute(2, 5, 0, 3, 1, 4).contiguous()
+    return frag.to(torch.bfloat16)
+
+
 def masked_sample(logits, action_mask, uniform):
     """Sample one action per row from softmax(logits - action_mask).
 

@@ -668,6 +668,96 @@ __global__ __launch_bounds__(256) void torus_conv_fused_kernel(
     }
 }
 
+// --- fused ConvLSTM cell (Geister DRC core) -------------------------------
+// One kernel per cell evaluation: implicit-GEMM 3x3 zero-pad conv over the
+// K-ordered (x | h) input halves (concat never materialized) with the
+// 4-gate sigmoid/tanh state update fused into the epilogue.  Replaces the
+// reference cell's concat + conv + split + 8 pointwise ops
+// (reference envs/geister.py:18-58) with ONE launch.
+//
+// Shapes: x, h (B, 36, 32) NHWC bf16; c (B, 36, 32) fp32; gates N = 128
+// (i | f | o | g quarters, reference split order geister.py:46).
+// Geometry: 256 threads = 4 waves; each wave computes a 16-position x
+// 128-channel tile (8 MFMA accumulators); K = 2 halves x 9 taps x 32
+// channels = 18 MFMA K-steps, 144 MFMA per wave.  The gate quartets land
+// in the SAME lane (channel hc sits in acc tiles {hc/16, +2, +4, +6} at
+// lane lo = hc % 16), so the state update needs no cross-lane traffic.
+__global__ __launch_bounds__(256) void convlstm_cell_kernel(
+    const short* __restrict__ x,        // (B*36, 32) bf16
+    const short* __restrict__ h,        // (B*36, 32) bf16
+    const float* __restrict__ c,        // (B*36, 32) fp32
+    const short* __restrict__ wfrag,    // (2, 9, 8, 4, 16, 8) bf16
+    const float* __restrict__ bias,     // (128,)
+    const int* __restrict__ nbr,        // (36, 9) zero-pad: -1 = outside
+    short* __restrict__ h_out,          // (B*36, 32) bf16
+    float* __restrict__ c_out,          // (B*36, 32) fp32
+    long total_pos)                     // B*36
+{
+    const int wid = threadIdx.x >> 6;
+    const int lane = threadIdx.x & 63;
+    const int khi = lane >> 4, lo = lane & 15;
+    const long pos_base = (long)blockIdx.x * 64 + wid * 16;
+
+    const long p = pos_base + lo;                  // this lane's A row
+    const long pp = (p < total_pos) ? p : 0;
+    const long n = pp / 36;
+    const int cell = (int)(pp % 36);
+
+    const bf16x8* wf = (const bf16x8*)wfrag;
+    const bf16x8 azero = {};
+    f32x4 acc[8];
+#pragma unroll
+    for (int t = 0; t < 8; ++t) acc[t] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+#pragma unroll
+    for (int src = 0; src < 2; ++src) {
+        const short* in = src == 0 ? x : h;
+#pragma unroll
+        for (int tap = 0; tap < 9; ++tap) {
+            const int nb = nbr[cell * 9 + tap];
+            const bf16x8 a = (nb >= 0)
+                ? *(const bf16x8*)(in + (n * 36 + nb) * 32 + khi * 8)
+                : azero;
+            const bf16x8* wrow = wf + ((src * 9 + tap) * 8) * 64 + khi * 16 + lo;
+#pragma unroll
+            for (int t = 0; t < 8; ++t)
+                acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    a, wrow[t * 64], acc[t], 0, 0, 0);
+        }
+    }
+
+    float b_i0 = bias[lo],       b_i1 = bias[16 + lo];
+    float b_f0 = bias[32 + lo],  b_f1 = bias[48 + lo];
+    float b_o0 = bias[64 + lo],  b_o1 = bias[80 + lo];
+    float b_g0 = bias[96 + lo],  b_g1 = bias[112 + lo];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+        const long prow = pos_base + khi * 4 + r;
+        if (prow >= total_pos) continue;
+        const long off = prow * 32;
+        // hidden channel lo
+        {
+            const float gi = 1.f / (1.f + __expf(-(acc[0][r] + b_i0)));
+            const float gf = 1.f / (1.f + __expf(-(acc[2][r] + b_f0)));
+            const float go = 1.f / (1.f + __expf(-(acc[4][r] + b_o0)));
+            const float gg = tanhf(acc[6][r] + b_g0);
+            const float cn = gf * c[off + lo] + gi * gg;
+            c_out[off + lo] = cn;
+            h_out[off + lo] = f2bf(go * tanhf(cn));
+        }
+        // hidden channel 16 + lo
+        {
+            const float gi = 1.f / (1.f + __expf(-(acc[1][r] + b_i1)));
+            const float gf = 1.f / (1.f + __expf(-(acc[3][r] + b_f1)));
+            const float go = 1.f / (1.f + __expf(-(acc[5][r] + b_o1)));
+            const float gg = tanhf(acc[7][r] + b_g1);
+            const float cn = gf * c[off + 16 + lo] + gi * gg;
+            c_out[off + 16 + lo] = cn;
+            h_out[off + 16 + lo] = f2bf(go * tanhf(cn));
+        }
+    }
+}
+
 }  // namespace
 
 template <typename T>
@@ -820,6 +910,30 @@ static torch::Tensor torus_conv_fused(
     return y;
 }
 
+static std::vector<torch::Tensor> convlstm_cell(
+    torch::Tensor x, torch::Tensor h, torch::Tensor c,
+    torch::Tensor wfrag, torch::Tensor bias, torch::Tensor nbr,
+    torch::Tensor h_out, torch::Tensor c_out) {
+    TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kBFloat16 &&
+                x.dim() == 3 && x.size(1) == 36 && x.size(2) == 32,
+                "convlstm_cell: x must be (B,36,32) bf16");
+    TORCH_CHECK(h.sizes() == x.sizes() &&
+                h.scalar_type() == torch::kBFloat16);
+    TORCH_CHECK(c.sizes() == x.sizes() &&
+                c.scalar_type() == torch::kFloat32);
+    TORCH_CHECK(wfrag.scalar_type() == torch::kBFloat16 &&
+                wfrag.numel() == 2 * 9 * 8 * 4 * 16 * 8);
+    const long total_pos = x.size(0) * 36;
+    const long grid = (total_pos + 63) / 64;
+    auto stream = at::cuda::getCurrentCUDAStream();
+    hipLaunchKernelGGL(convlstm_cell_kernel, dim3(grid), dim3(256), 0, stream,
+        (const short*)x.data_ptr(), (const short*)h.data_ptr(),
+        c.data_ptr<float>(), (const short*)wfrag.data_ptr(),
+        bias.data_ptr<float>(), nbr.data_ptr<int>(),
+        (short*)h_out.data_ptr(), c_out.data_ptr<float>(), total_pos);
+    return {h_out, c_out};
+}
+
 static torch::Tensor pack_torus_weights_hip(
     torch::Tensor w, c10::optional<torch::Tensor> scale, bool dgrad) {
     TORCH_CHECK(w.is_cuda() && w.scalar_type() == torch::kFloat32 && w.dim() == 4);
@@ -965,6 +1079,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "fused wrap-around conv3x3 + BN-fold + residual + relu (MFMA)");
     m.def("torus_wgrad", &torus_wgrad,
           "torus conv weight gradient (MFMA, LDS-staged per image)");
+    m.def("convlstm_cell", &convlstm_cell,
+          "fused ConvLSTM cell: implicit-GEMM conv(x|h) + 4-gate state "
+          "update, one kernel per cell eval (Geister DRC core)");
     m.def("pack_torus_weights_hip", &pack_torus_weights_hip,
           "pack conv weights into MFMA fragment layout (fwd or dgrad)");
     m.def("bn_nhwc_fwd", &bn_nhwc_fwd,
