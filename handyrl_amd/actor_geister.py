@@ -282,43 +282,127 @@ class BatchedDRCEngine:
     fixed under capture); eager on CPU / when capture fails."""
 
     def __init__(self, model, device, n_games, use_graphs=True):
+        import os
         self.model = model
         self.device = device
         self.n_per = n_games
-        hs, cs = model.init_hidden([n_games * 2])
-        self.hidden = ([h.to(device) for h in hs], [c.to(device) for c in cs])
+        # fused DRC: one hand-written CDNA4 kernel per ConvLSTM cell eval
+        # (ops/src/ext.hip::convlstm_cell) with NHWC bf16 hidden state
+        # (HANDYRL_DRC_FUSED=0 opts back into the eager cell chain)
+        self.fused_drc = (device.type == 'cuda' and ops.available() and
+                          os.environ.get('HANDYRL_DRC_FUSED', '1') == '1')
+        if self.fused_drc:
+            G2 = n_games * 2
+            self.hidden = (
+                [torch.zeros(G2, 36, 32, dtype=torch.bfloat16, device=device)
+                 for _ in range(3)],
+                [torch.zeros(G2, 36, 32, dtype=torch.float32, device=device)
+                 for _ in range(3)])
+            self._nbr = ops.convlstm_neighbor_table(device)
+            self._wfrag = [torch.zeros(2, 9, 8, 4, 16, 8,
+                                       dtype=torch.bfloat16, device=device)
+                           for _ in range(3)]
+            self._wbias = [torch.zeros(128, device=device) for _ in range(3)]
+            self.refresh()
+        else:
+            hs, cs = model.init_hidden([n_games * 2])
+            self.hidden = ([h.to(device) for h in hs],
+                           [c.to(device) for c in cs])
         self._arange2 = torch.arange(n_games, device=device) * 2
         self._graph = None
         if use_graphs and device.type == 'cuda' and \
                 __import__('os').environ.get('HANDYRL_NO_GRAPHS') != '1':
             self._capture()
 
+    @torch.no_grad()
+    def refresh(self):
+        """Repack the cell conv weights into MFMA fragments (call after a
+        model-weight push; the packed buffers are graph-stable)."""
+        if not self.fused_drc:
+            return
+        for l, cell in enumerate(self.model.body.blocks):
+            self._wfrag[l].copy_(
+                ops.pack_convlstm_weights(cell.conv.weight.detach()))
+            self._wbias[l].copy_(cell.conv.bias.detach().float())
+
     def _static_in(self):
         per, dev = self.n_per, self.device
-        return {
+        st = {
             'scalar': torch.zeros(per, SCALAR_DIM, device=dev),
             'board': torch.zeros((per,) + BOARD_SHAPE, device=dev),
             'mask': torch.full((per, N_ACTIONS), 1e32, device=dev),
             'parity': torch.zeros(per, dtype=torch.int64, device=dev),
             'keep': torch.ones(per * 2, 1, 1, 1, device=dev),
         }
+        if self.fused_drc:
+            st['keep'] = torch.ones(per * 2, 1, 1, device=dev)
+            st['keep_h'] = torch.ones(per * 2, 1, 1, dtype=torch.bfloat16,
+                                      device=dev)
+        return st
+
+    def _drc_fused(self, x_nchw, rows):
+        """The DRC core on the fused cell kernel: 3 layers x 3 repeats,
+        ping-pong buffers, per-(game, player) hidden gathered by parity."""
+        hs, cs = self.hidden
+        B = x_nchw.shape[0]
+        x = x_nchw.permute(0, 2, 3, 1).reshape(B, 36, 32) \
+            .to(torch.bfloat16).contiguous()
+        h_cur = [h.index_select(0, rows).contiguous() for h in hs]
+        c_cur = [c.index_select(0, rows).contiguous() for c in cs]
+        h_alt = [torch.empty_like(t) for t in h_cur]
+        c_alt = [torch.empty_like(t) for t in c_cur]
+        for _rep in range(3):
+            for l in range(3):
+                src = x if l == 0 else h_alt[l - 1]
+                ops.convlstm_cell(src, h_cur[l], c_cur[l], self._wfrag[l],
+                                  self._wbias[l], self._nbr,
+                                  h_alt[l], c_alt[l])
+            h_cur, h_alt = h_alt, h_cur
+            c_cur, c_alt = c_alt, c_cur
+        for i in range(3):
+            hs[i].index_copy_(0, rows, h_cur[i])
+            cs[i].index_copy_(0, rows, c_cur[i])
+        return h_cur[2].float().reshape(B, 6, 6, 32) \
+            .permute(0, 3, 1, 2).contiguous()
 
     def _infer_body(self, st):
         """One batched recurrent step on static tensors; hidden updates are
         IN-PLACE so the storages stay fixed under graph capture."""
+        import torch.nn.functional as F
+        from .models.common import apply_bn
         hs, cs = self.hidden
         rows = self._arange2 + st['parity']
-        for i in range(len(hs)):
-            hs[i].mul_(st['keep'])
-            cs[i].mul_(st['keep'])
-        h_in = ([h.index_select(0, rows) for h in hs],
-                [c.index_select(0, rows) for c in cs])
-        out = self.model({'scalar': st['scalar'], 'board': st['board']}, h_in)
-        h_out, c_out = out['hidden']
-        for i in range(len(hs)):
-            hs[i].index_copy_(0, rows, h_out[i])
-            cs[i].index_copy_(0, rows, c_out[i])
-        policy = out['policy'].float()
+        if self.fused_drc:
+            model = self.model
+            for i in range(len(hs)):
+                hs[i].mul_(st['keep_h'])
+                cs[i].mul_(st['keep'])
+            B = st['scalar'].shape[0]
+            planes = st['scalar'].view(B, SCALAR_DIM, 1, 1) \
+                .expand(B, SCALAR_DIM, 6, 6)
+            stem = F.relu(apply_bn(model.bn1, model.conv1(
+                torch.cat([planes, st['board']], dim=1))))
+            h_last = self._drc_fused(stem, rows)
+            p_move = model.head_p_move(h_last)
+            p_set = model.head_p_set(st['scalar'][:, :1])
+            policy = torch.cat([p_move, p_set], dim=-1).float()
+            value = torch.tanh(model.head_v(h_last)).float()
+            ret = model.head_r(h_last).float()
+        else:
+            for i in range(len(hs)):
+                hs[i].mul_(st['keep'])
+                cs[i].mul_(st['keep'])
+            h_in = ([h.index_select(0, rows) for h in hs],
+                    [c.index_select(0, rows) for c in cs])
+            out = self.model({'scalar': st['scalar'],
+                              'board': st['board']}, h_in)
+            h_out, c_out = out['hidden']
+            for i in range(len(hs)):
+                hs[i].index_copy_(0, rows, h_out[i])
+                cs[i].index_copy_(0, rows, c_out[i])
+            policy = out['policy'].float()
+            value = out['value'].float()
+            ret = out['return'].float()
         uniform = torch.rand(policy.shape[0], device=self.device)
         if self.device.type == 'cuda':
             actions, probs = ops.masked_sample(policy, st['mask'], uniform)
@@ -327,8 +411,8 @@ class BatchedDRCEngine:
             actions = torch.multinomial(pr, 1).squeeze(-1)
             probs = pr.gather(-1, actions.unsqueeze(-1)).squeeze(-1)
         return torch.stack([actions.float(), probs,
-                            out['value'].float().squeeze(-1),
-                            out['return'].float().squeeze(-1)], dim=1)
+                            value.squeeze(-1),
+                            ret.squeeze(-1)], dim=1)
 
     @torch.no_grad()
     def _capture(self):
@@ -359,8 +443,8 @@ class BatchedDRCEngine:
         """numpy in (shard-sized arrays), numpy out: fills out[:, 0:4] with
         (action, prob, value, return) rows."""
         per, dev = self.n_per, self.device
-        keep_np = (1.0 - reset.astype(np.float32)).repeat(2) \
-            .reshape(per * 2, 1, 1, 1)
+        shape = (per * 2, 1, 1) if self.fused_drc else (per * 2, 1, 1, 1)
+        keep_np = (1.0 - reset.astype(np.float32)).repeat(2).reshape(shape)
         if self._graph is not None:
             graph, st, packed = self._graph
             st['scalar'].copy_(torch.from_numpy(scalar), non_blocking=True)
@@ -369,6 +453,8 @@ class BatchedDRCEngine:
             st['parity'].copy_(torch.from_numpy(parity.astype(np.int64)),
                                non_blocking=True)
             st['keep'].copy_(torch.from_numpy(keep_np), non_blocking=True)
+            if self.fused_drc:
+                st['keep_h'].copy_(st['keep'])
             graph.replay()
             np.copyto(out, packed.cpu().numpy())
             return
@@ -379,6 +465,8 @@ class BatchedDRCEngine:
             'parity': torch.from_numpy(parity.astype(np.int64)).to(dev),
             'keep': torch.from_numpy(keep_np).to(dev),
         }
+        if self.fused_drc:
+            st['keep_h'] = st['keep'].to(torch.bfloat16)
         np.copyto(out, self._infer_body(st).cpu().numpy())
 
 
@@ -449,7 +537,7 @@ class GeisterActorPool:
         return out
 
     def refresh_weights(self):
-        pass                       # shares the live module directly
+        self.engine.refresh()      # repack fused-DRC weight fragments
 
     def shutdown(self):
         pass
@@ -576,7 +664,9 @@ class GeisterMultiProcPool:
         return out
 
     def refresh_weights(self):
-        pass                       # shares the live module directly
+        if self.engines:
+            for e in self.engines:
+                e.refresh()        # repack fused-DRC weight fragments
 
     def shutdown(self):
         for conn in self.conns:

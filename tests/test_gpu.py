@@ -550,3 +550,93 @@ def test_traj_ring_matches_worker_host_recording():
                                    pool.rec_val[g, :S][a], rtol=0, atol=0)
         checked += 1
     assert checked > 0
+
+
+@requires_gpu
+def test_convlstm_cell_kernel_matches_eager():
+    """Fused ConvLSTM cell kernel vs the eager fp32 cell on identical
+    bf16-quantized inputs: single cell eval, tight tolerance."""
+    from handyrl_amd import ops
+    from handyrl_amd.models.geister_net import ConvLSTMCell
+
+    torch.manual_seed(3)
+    dev = torch.device('cuda', 0)
+    cell = ConvLSTMCell(32, 32).to(dev)
+    for p in cell.parameters():
+        p.data.uniform_(-0.3, 0.3)
+    B = 96
+    x_bf = (torch.randn(B, 36, 32, device=dev) * 0.7).to(torch.bfloat16)
+    h_bf = (torch.randn(B, 36, 32, device=dev) * 0.5).to(torch.bfloat16)
+    c_f = torch.randn(B, 36, 32, device=dev) * 0.8
+
+    # eager fp32 reference on the SAME quantized values (NCHW layout)
+    to_nchw = lambda t: t.float().reshape(B, 6, 6, 32).permute(0, 3, 1, 2)
+    with torch.no_grad():
+        h_ref, c_ref = cell(to_nchw(x_bf), (to_nchw(h_bf), to_nchw(c_f)))
+
+    wfrag = ops.pack_convlstm_weights(cell.conv.weight.detach())
+    bias = cell.conv.bias.detach().float()
+    nbr = ops.convlstm_neighbor_table(dev)
+    h_out = torch.empty_like(h_bf)
+    c_out = torch.empty_like(c_f)
+    ops.convlstm_cell(x_bf, h_bf, c_f, wfrag, bias, nbr, h_out, c_out)
+    torch.cuda.synchronize()
+
+    back = lambda t: t.float().reshape(B, 6, 6, 32).permute(0, 3, 1, 2)
+    torch.testing.assert_close(back(c_out), c_ref, rtol=2e-2, atol=2e-2)
+    torch.testing.assert_close(back(h_out), h_ref, rtol=2e-2, atol=2e-2)
+
+
+@requires_gpu
+def test_drc_fused_sequence_matches_eager():
+    """Full DRC (3 layers x 3 repeats = 9 fused cell evals) through the
+    engine's _drc_fused vs the eager DRC module, including the hidden
+    scatter back into the resident rows."""
+    from handyrl_amd.actor_geister import BatchedDRCEngine
+    from handyrl_amd.envs.geister import Environment
+    import os
+
+    torch.manual_seed(5)
+    dev = torch.device('cuda', 0)
+    model = Environment().net().to(dev).eval()
+    for p in model.parameters():
+        p.data.uniform_(-0.2, 0.2)
+
+    n = 48
+    os.environ['HANDYRL_DRC_FUSED'] = '1'
+    eng = BatchedDRCEngine(model, dev, n, use_graphs=False)
+    assert eng.fused_drc
+
+    B = n
+    x = torch.randn(B, 32, 6, 6, device=dev) * 0.6
+    rows = torch.arange(B, device=dev) * 2          # parity 0
+
+    # seed the resident hidden with nonzero state
+    hs, cs = eng.hidden
+    for t in hs:
+        t.normal_(0, 0.4)
+    for t in cs:
+        t.normal_(0, 0.6)
+    h0 = [t.clone() for t in hs]
+    c0 = [t.clone() for t in cs]
+
+    with torch.no_grad():
+        out_fused = eng._drc_fused(x, rows)
+    torch.cuda.synchronize()
+
+    # eager reference on the same (bf16-quantized) starting state
+    nhwc2nchw = lambda t: t.float().reshape(-1, 6, 6, 32).permute(0, 3, 1, 2)
+    h_in = [nhwc2nchw(t.index_select(0, rows)) for t in h0]
+    c_in = [nhwc2nchw(t.index_select(0, rows)) for t in c0]
+    x_q = x.permute(0, 2, 3, 1).to(torch.bfloat16).float() \
+        .permute(0, 3, 1, 2).contiguous()
+    with torch.no_grad():
+        h_last, (hs_ref, cs_ref) = model.body(x_q, (h_in, c_in),
+                                              num_repeats=3)
+    torch.testing.assert_close(out_fused, h_last, rtol=5e-2, atol=5e-2)
+    # resident hidden rows were updated to the eager-equivalent state
+    for i in range(3):
+        got_h = nhwc2nchw(hs[i].index_select(0, rows))
+        got_c = nhwc2nchw(cs[i].index_select(0, rows))
+        torch.testing.assert_close(got_h, hs_ref[i], rtol=5e-2, atol=5e-2)
+        torch.testing.assert_close(got_c, cs_ref[i], rtol=8e-2, atol=8e-2)

@@ -102,6 +102,7 @@ def main():
             losses, dcnt = trainer.train_step(batch)
         # actors follow the trained weights (model push each step)
         actor_model.load_state_dict(trainer.model.state_dict())
+        pool.refresh_weights()     # repack fused-DRC fragments
         return frames
 
     for _ in range(cli.warmup):
