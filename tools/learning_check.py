@@ -12,19 +12,21 @@ from bench import bench_args
 
 torch.set_num_threads(1)
 args = bench_args(128, 16)
-mpool = MultiProcGeesePool(args, n_games=1024, seed=7, workers=4)
+traj = os.environ.get('HANDYRL_TRAJ', '1') == '1'
+mpool = MultiProcGeesePool(args, n_games=1024, seed=7, workers=4,
+                           traj_mode=traj)
 device = torch.device('cuda', 0)
 torch.cuda.set_device(device)
 torch.manual_seed(0)
 trainer = Trainer(args, GeeseNet(), device=device, batcher=False)
-mpool.attach(trainer.model, device)
 # HANDYRL_INGEST_THREAD=0: synchronous main-stream ingest (race probe)
 ingest = os.environ.get('HANDYRL_INGEST_THREAD', '1') == '1'
 replay = DeviceReplay(args, device, bytes_budget=2 << 30,
                       ingest_thread=ingest)
-print('# slots=%d sync=%s ingest_thread=%s register=%s' % (
+mpool.attach(trainer.model, device, replay=replay)
+print('# slots=%d sync=%s ingest_thread=%s register=%s traj=%s' % (
     mpool.slots, os.environ.get('HANDYRL_ACTOR_SYNC', '0'), ingest,
-    getattr(mpool, '_use_registered', False)), flush=True)
+    getattr(mpool, '_use_registered', False), traj), flush=True)
 
 VALIDATE = os.environ.get('HANDYRL_VALIDATE') == '1'
 min_prob_seen = [1.0]
@@ -113,12 +115,15 @@ for i in range(N):
     if (i + 1) % 50 == 0:
         torch.cuda.synchronize()
         ml = sum(window_lens) / max(1, len(window_lens))
-        print('step %4d | mean_ep_len %.2f | eps %d | p %.4f v %.4f ent %.3f | %.1fs'
+        print('step %4d | mean_ep_len %.2f | eps %d | p %.4f v %.4f '
+              'ent %.3f | guard_fires %d | %.1fs'
               % (i + 1, ml, len(window_lens),
                  float(losses['p']) / max(1.0, float(dcnt)),
                  float(losses['v']) / max(1.0, float(dcnt)),
                  float(losses['ent']) / max(1.0, float(dcnt)),
+                 int(trainer.guard_fires.item()),
                  time.time() - t0), flush=True)
         window_lens = []
 mpool.shutdown()
+print('GUARD_FIRES_TOTAL %d' % int(trainer.guard_fires.item()))
 print('LEARNING_CHECK_DONE')
