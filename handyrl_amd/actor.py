@@ -475,9 +475,18 @@ def _geese_env_worker(conn, ep_conn, obs_name, res_name, n_games, args, seed,
                     pools[s].take_finished())
         return ('obs', s, m, frames)
 
+    def prep(s):
+        m = pools[s].prepare_step(obs_views[s])
+        if traj_mode and m != per:
+            # the parent's whole-service graphs assume full shards (every
+            # game alive until reset, which _phase2 does immediately)
+            raise RuntimeError('traj mode shard not full: %d != %d'
+                               % (m, per))
+        return m
+
     m_inflight = [0] * slots
     for s in range(slots):                    # prime the pipeline
-        m_inflight[s] = pools[s].prepare_step(obs_views[s])
+        m_inflight[s] = prep(s)
         conn.send(obs_msg(s, m_inflight[s], 0))
     while True:
         cmd = conn.recv()
@@ -498,7 +507,7 @@ def _geese_env_worker(conn, ep_conn, obs_name, res_name, n_games, args, seed,
                 # episodes travel on their own pipe, drained by a parent
                 # background thread: the service path never deserializes them
                 ep_conn.send(eps)
-        m_inflight[s] = pools[s].prepare_step(obs_views[s])
+        m_inflight[s] = prep(s)
         conn.send(obs_msg(s, m_inflight[s], frames))
 
 
@@ -622,6 +631,53 @@ class MultiProcGeesePool:
                 self._idx_pin_np = [[t.numpy() for t in row]
                                     for row in self._idx_pin]
             self._register_shm()
+            self._svc_graphs = None
+            if self.traj is not None:
+                self._capture_services()
+
+    def _capture_services(self):
+        """Whole-service graphs, one per (worker, slot): H2D (from
+        registered shm when available, else the pinned staging buffer) +
+        fused forward + sample + in-graph trajectory scatter with
+        device-resident step counters + D2H.  A service becomes one
+        graph.replay()."""
+        import sys
+        bucket = self.graphed._bucket(self.n_per)
+        dev = self.device
+        graphs, tidxs = [], []
+        try:
+            for w in range(self.workers):
+                grow, trow = [], []
+                for s in range(self.slots):
+                    base = (w * self.slots + s) * self.n_per
+                    gidx = torch.full((bucket,), self.traj.scratch_row,
+                                      dtype=torch.int64, device=dev)
+                    gidx[:self.n_per] = base + torch.arange(
+                        self.n_per, device=dev)
+                    tidx = torch.zeros(bucket, dtype=torch.int64, device=dev)
+                    if self._use_registered:
+                        obs_src = self._obs_src[w][s]
+                        out_dst = self._res_dst[w][s]
+                    else:
+                        obs_src = self._obs_pin[w][s]
+                        out_dst = self._out_pin[w][s]
+                    grow.append(self.graphed.capture_service(
+                        obs_src, out_dst, gidx, tidx, self.n_per))
+                    trow.append(tidx)
+                graphs.append(grow)
+                tidxs.append(trow)
+            self._svc_graphs = graphs
+            self._tidx_dev = tidxs
+            print('# actor service pipeline captured (%dx%d graphs, '
+                  'obs source: %s)' % (self.workers, self.slots,
+                                       'registered shm'
+                                       if self._use_registered
+                                       else 'pinned staging'),
+                  file=sys.stderr, flush=True)
+        except Exception as e:        # noqa: BLE001 - run the op-by-op path
+            print('service-pipeline capture failed (%r); per-op path'
+                  % (e,), file=sys.stderr, flush=True)
+            self._svc_graphs = None
 
     def _register_shm(self):
         """hipHostRegister the shared-memory obs/result buffers so the DMA
@@ -629,9 +685,14 @@ class MultiProcGeesePool:
         service path).  Falls back to the staging copy if registration is
         refused (e.g. exotic shm mounts)."""
         self._use_registered = False
-        # direct-DMA from hipHostRegister'd shm also showed a slow-burning
-        # value-corruption race (NaN by ~400 steps) — opt-in only.
-        if os.environ.get('HANDYRL_SHM_REGISTER', '0') != '1':
+        # Round-1 history: direct-DMA from hipHostRegister'd shm was
+        # exonerated of the NaN (BASELINE.md root-cause chain) but measured
+        # slower on the per-op service path.  With the whole-service graph
+        # (traj mode) it removes the staging memcpy from a much smaller
+        # service cost, so traj mode defaults it ON; HANDYRL_SHM_REGISTER=0
+        # forces staging copies.
+        default = '1' if self.traj_mode else '0'
+        if os.environ.get('HANDYRL_SHM_REGISTER', default) != '1':
             return
         flat = [v for row in self.obs_views for v in row] + \
                [v for row in self.res_views for v in row]
@@ -700,17 +761,23 @@ class MultiProcGeesePool:
         self.conns[wid].send(('go', slot))
         self.timing['sample'] += time.time() - t0    # event sync + go
 
-    def _commit_finished(self, base, fin):
+    def _commit_finished(self, base, fin, tidx_dev=None):
         """Traj mode: move finished device-recorded episodes into the
         replay ring (D2D) and queue lightweight stubs for stats.  Runs
         BEFORE issuing the worker's next forward, whose in-graph scatter
         would overwrite these trajectory rows; the main stream waits on
-        the commit event to keep that ordering on device."""
+        the commit event to keep that ordering on device.  With
+        whole-service graphs, the finished games' device step counters
+        are zeroed here (the worker reset them host-side already)."""
         g_local, lens, outcomes = fin
         event = self.replay.commit_traj(self.traj, base + g_local, lens,
                                         outcomes)
         if event is not None:
             torch.cuda.current_stream().wait_event(event)
+        if tidx_dev is not None and len(g_local):
+            rows = torch.from_numpy(np.ascontiguousarray(g_local)).to(
+                self.device, non_blocking=True)
+            tidx_dev.index_fill_(0, rows, 0)
         job_args = {'player': list(range(N_PLAYERS)),
                     'model_id': {p: -1 for p in range(N_PLAYERS)}}
         stubs = [{'args': job_args, 'steps': int(lens[k]),
@@ -773,7 +840,27 @@ class MultiProcGeesePool:
             _tag, _slot, _m, _f, meta, fin = msg
             base = (wid * self.slots + slot) * self.n_per
             if fin is not None:
-                self._commit_finished(base, fin)
+                tidx_dev = self._tidx_dev[wid][slot] \
+                    if self._svc_graphs is not None else None
+                self._commit_finished(base, fin, tidx_dev)
+            if M and self._svc_graphs is not None:
+                # whole-service graph: H2D + forward + sample + trajectory
+                # scatter + D2H as ONE replay (step counters live on device)
+                assert M == self.n_per, (M, self.n_per)
+                if not self._use_registered:
+                    np.copyto(self._obs_pin_np[wid][slot][:M],
+                              self.obs_views[wid][slot][:M])
+                self._svc_graphs[wid][slot].replay()
+                self._events[wid][slot].record()
+                self.inflight[(wid, slot)] = M
+                self._fifo.append((wid, slot))
+                self.timing['fwd'] += time.time() - t0
+                if os.environ.get('HANDYRL_ACTOR_SYNC') == '1':
+                    while self._fifo:
+                        self._complete(*self._fifo.pop(0))
+                else:
+                    self._poll_completions()
+                return frames
             if M:
                 rows, tidx = meta
                 bucket = self.graphed._bucket(M)

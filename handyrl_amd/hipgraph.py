@@ -192,6 +192,52 @@ class GraphedActorForward:
         q = 32 if self.canonical else 256
         return q * ((M + q - 1) // q)
 
+    def capture_service(self, obs_src, out_dst, gidx_const, tidx_dev, M,
+                        n_actions=4):
+        """Whole-service capture for one (worker, slot) of the multiproc
+        actor pool: ONE replay executes H2D of the observations (from a
+        pinned staging tensor or a hipHostRegister'd shm view), the fused
+        forward + masked sample, the in-graph HBM trajectory scatter with
+        a device-resident step counter, and the D2H of the packed results.
+
+        gidx_const: (bucket,) int64 device — global trajectory rows
+        (bucket padding points at the scratch row); tidx_dev: (bucket,)
+        int64 device step counters, incremented in-graph and clamped (the
+        caller zeroes finished games' entries before the next replay).
+        Returns the graph; replaying it IS the service."""
+        bucket = self._bucket(M)
+        static_obs = torch.zeros(bucket, 17, 7, 11, dtype=torch.uint8,
+                                 device=self.device)
+        rows = bucket * 4 if self.canonical else bucket
+        zero_mask = torch.zeros(rows, n_actions, device=self.device)
+        R = M * 4 if self.canonical else M
+        max_t = self.traj.max_steps if self.traj is not None else 1
+
+        def run_once():
+            static_obs[:M].copy_(obs_src[:M], non_blocking=True)
+            packed = self._fwd_sample(static_obs, zero_mask)
+            if self.traj is not None:
+                self.traj.record_(static_obs, packed, gidx_const, tidx_dev)
+                tidx_dev.add_(1).clamp_(0, max_t - 1)
+            out_dst[:R].copy_(packed[:R], non_blocking=True)
+
+        was_training = self.model.training
+        self.model.eval()
+        stream = torch.cuda.Stream()
+        stream.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(stream):
+            for _ in range(2):
+                run_once()
+        torch.cuda.current_stream().wait_stream(stream)
+        graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(graph):
+            run_once()
+        if was_training:
+            self.model.train()
+        tidx_dev.zero_()          # warmup replays polluted the counters
+        torch.cuda.synchronize()
+        return graph
+
     def run_async(self, obs_pinned, M, out_pinned, event, n_actions=4,
                   idx_pinned=None):
         """Pipelined variant: H2D from a pinned staging tensor, replay, and

@@ -35,7 +35,7 @@ def build_step(dist_on, seed=17):
     torch.manual_seed(seed)
     model = GeeseNet().cuda().train()
     params = list(model.parameters())
-    opt = torch.optim.Adam(params, lr=torch.tensor(1e-3, device='cuda'),
+    opt = torch.optim.Adam(params, lr=torch.tensor(1e-5, device='cuda'),
                            weight_decay=1e-5, capturable=True)
     reducer = hdist.GradReducer(params)
 
@@ -96,19 +96,20 @@ def main():
     # -- 2/3. capture the step WITH the all-reduce node ------------------
     g, loss_t, model = build_step(dist_on=True)
     losses = []
-    for _ in range(5):
+    for _ in range(3):
         g.replay()
         torch.cuda.synchronize()
         losses.append(float(loss_t.item()))
     verdict['captured_losses_with_allreduce'] = losses
     p_dist = [p.detach().clone() for p in model.parameters()]
-    assert all(torch.isfinite(p).all() for p in p_dist)
+    verdict['weights_finite'] = bool(
+        all(torch.isfinite(p).all() for p in p_dist))
     dist.destroy_process_group()
 
     # control: identical run with NO dist group / no all-reduce node
     g2, loss_t2, model2 = build_step(dist_on=False)
     losses2 = []
-    for _ in range(5):
+    for _ in range(3):
         g2.replay()
         torch.cuda.synchronize()
         losses2.append(float(loss_t2.item()))
@@ -125,6 +126,7 @@ def main():
     verdict['first_step_rel_err'] = rel[0]
     import math
     verdict['ok'] = bool(rel[0] < 0.05 and max_diff < 0.05 and
+                         verdict['weights_finite'] and
                          all(math.isfinite(v) for v in losses))
 
     with open('gpurun_out/rccl_probe.json', 'w') as f:
