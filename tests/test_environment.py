@@ -89,3 +89,10 @@ def test_observation_shapes_stable(env):
             map_r(obs, lambda o: np.asarray(o))
         actions = {p: random.choice(e.legal_actions(p)) for p in e.turns()}
         e.step(actions)
+
+
+def test_check_env_validator():
+    """The runtime contract validator passes every bundled game."""
+    from handyrl_amd.environment import make_env, check_env
+    for name in ['TicTacToe', 'ParallelTicTacToe', 'Geister']:
+        assert check_env(make_env({'env': name}))
