@@ -644,6 +644,8 @@ class MultiProcGeesePool:
         import sys
         bucket = self.graphed._bucket(self.n_per)
         dev = self.device
+        if os.environ.get('HANDYRL_SVC_GRAPH', '1') != '1':
+            return
         graphs, tidxs = [], []
         try:
             for w in range(self.workers):
@@ -655,20 +657,14 @@ class MultiProcGeesePool:
                     gidx[:self.n_per] = base + torch.arange(
                         self.n_per, device=dev)
                     tidx = torch.zeros(bucket, dtype=torch.int64, device=dev)
-                    if self._use_registered:
-                        obs_src = self._obs_src[w][s]
-                        out_dst = self._res_dst[w][s]
-                    else:
-                        obs_src = self._obs_pin[w][s]
-                        out_dst = self._out_pin[w][s]
-                    grow.append(self.graphed.capture_service(
-                        obs_src, out_dst, gidx, tidx, self.n_per))
+                    grow.append(self.graphed.capture_service_core(
+                        gidx, tidx, self.n_per))
                     trow.append(tidx)
                 graphs.append(grow)
                 tidxs.append(trow)
             self._svc_graphs = graphs
             self._tidx_dev = tidxs
-            print('# actor service pipeline captured (%dx%d graphs, '
+            print('# actor service core captured (%dx%d graphs, '
                   'obs source: %s)' % (self.workers, self.slots,
                                        'registered shm'
                                        if self._use_registered
@@ -844,13 +840,25 @@ class MultiProcGeesePool:
                     if self._svc_graphs is not None else None
                 self._commit_finished(base, fin, tidx_dev)
             if M and self._svc_graphs is not None:
-                # whole-service graph: H2D + forward + sample + trajectory
-                # scatter + D2H as ONE replay (step counters live on device)
+                # service-core graph: forward + sample + trajectory
+                # scatter + device step counters as ONE replay; H2D in
+                # (from registered shm when available) and D2H out stay
+                # eager — ~4 host calls per service
                 assert M == self.n_per, (M, self.n_per)
-                if not self._use_registered:
+                graph, static_obs, packed = self._svc_graphs[wid][slot]
+                if self._use_registered:
+                    static_obs[:M].copy_(self._obs_src[wid][slot][:M],
+                                         non_blocking=True)
+                else:
                     np.copyto(self._obs_pin_np[wid][slot][:M],
                               self.obs_views[wid][slot][:M])
-                self._svc_graphs[wid][slot].replay()
+                    static_obs[:M].copy_(self._obs_pin[wid][slot][:M],
+                                         non_blocking=True)
+                graph.replay()
+                R = M * N_PLAYERS
+                dst = self._res_dst[wid][slot] if self._use_registered \
+                    else self._out_pin[wid][slot]
+                dst[:R].copy_(packed[:R], non_blocking=True)
                 self._events[wid][slot].record()
                 self.inflight[(wid, slot)] = M
                 self._fifo.append((wid, slot))

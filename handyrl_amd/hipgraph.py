@@ -192,6 +192,45 @@ class GraphedActorForward:
         q = 32 if self.canonical else 256
         return q * ((M + q - 1) // q)
 
+    def capture_service_core(self, gidx_const, tidx_dev, M, n_actions=4):
+        """Service CORE capture: fused forward + masked sample + in-graph
+        trajectory scatter + device step-counter update, with the pinned
+        H2D/D2H left to the caller (captured host-memcpy nodes faulted on
+        hardware — HSA exception at replay — so the copies stay eager:
+        still only ~4 host calls per service).
+
+        Returns (graph, static_obs, packed)."""
+        bucket = self._bucket(M)
+        static_obs = torch.zeros(bucket, 17, 7, 11, dtype=torch.uint8,
+                                 device=self.device)
+        rows = bucket * 4 if self.canonical else bucket
+        zero_mask = torch.zeros(rows, n_actions, device=self.device)
+        max_t = self.traj.max_steps if self.traj is not None else 1
+
+        def run_once():
+            packed = self._fwd_sample(static_obs, zero_mask)
+            if self.traj is not None:
+                self.traj.record_(static_obs, packed, gidx_const, tidx_dev)
+                tidx_dev.add_(1).clamp_(0, max_t - 1)
+            return packed
+
+        was_training = self.model.training
+        self.model.eval()
+        stream = torch.cuda.Stream()
+        stream.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(stream):
+            for _ in range(2):
+                run_once()
+        torch.cuda.current_stream().wait_stream(stream)
+        graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(graph):
+            packed = run_once()
+        if was_training:
+            self.model.train()
+        tidx_dev.zero_()          # warmup replays polluted the counters
+        torch.cuda.synchronize()
+        return graph, static_obs, packed
+
     def capture_service(self, obs_src, out_dst, gidx_const, tidx_dev, M,
                         n_actions=4):
         """Whole-service capture for one (worker, slot) of the multiproc
