@@ -1,8 +1,10 @@
-"""Build the in-tree CPU env-core extension (g++ + pybind11).
+"""Build the in-tree CPU env-core extensions (g++ + pybind11).
 
-handyrl_amd/envs/src/vec_geese_core.cpp -> handyrl_amd/envs/_vec_geese_core.so
-— a host-side native module (no GPU, no torch linkage): the env workers'
-hot loop.  The .so travels with repo snapshots like the HIP extension.
+handyrl_amd/envs/src/vec_geese_core.cpp  -> envs/_vec_geese_core.so
+handyrl_amd/envs/src/vec_geister_core.cpp -> envs/_vec_geister_core.so
+— host-side native modules (no GPU, no torch linkage): the env workers'
+hot loops.  The .so files travel with repo snapshots like the HIP
+extension.
 """
 
 import os
@@ -10,12 +12,14 @@ import subprocess
 import sys
 import sysconfig
 
+CORES = ['vec_geese_core', 'vec_geister_core']
 
-def build(verbose=False):
+
+def _build_one(name, verbose=False):
     import pybind11
     here = os.path.dirname(os.path.abspath(__file__))
-    src = os.path.join(here, 'src', 'vec_geese_core.cpp')
-    out = os.path.join(here, '_vec_geese_core.so')
+    src = os.path.join(here, 'src', name + '.cpp')
+    out = os.path.join(here, '_%s.so' % name)
     if os.path.exists(out) and os.path.getmtime(out) > os.path.getmtime(src):
         return out
     cmd = [
@@ -30,6 +34,10 @@ def build(verbose=False):
     subprocess.run(cmd, check=True)
     print('built %s' % out)
     return out
+
+
+def build(verbose=False):
+    return [_build_one(name, verbose) for name in CORES]
 
 
 if __name__ == '__main__':

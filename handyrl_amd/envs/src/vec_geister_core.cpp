@@ -1,0 +1,230 @@
+// CPU core for the vectorized Geister engine (host env workers).
+//
+// Implements GeisterVecEnv's three hot phases — legal_masks(),
+// observations() and step() — as tight per-game scalar loops over the
+// SAME struct-of-arrays numpy buffers the python engine uses (in place).
+// Every phase is RNG-free, so the python and native engines are
+// bit-equal state machines (asserted in tests/test_vec_geister_native.py).
+//
+// Rule semantics follow handyrl_amd/envs/geister.py (the single-game
+// oracle; reference semantics from handyrl/envs/geister.py:361-522).
+// Move tables (A_FROM / A_TO / A_GOAL / LAYOUT_BLUE / START_CELLS) are
+// precomputed in python and passed in once via set_tables().
+
+#include <pybind11/pybind11.h>
+#include <pybind11/numpy.h>
+
+#include <cstdint>
+#include <cstring>
+#include <vector>
+
+namespace py = pybind11;
+
+static constexpr int N_CELLS = 36;
+static constexpr int N_MOVE = 144;
+static constexpr int N_LAYOUTS = 70;
+static constexpr int N_ACTIONS = N_MOVE + N_LAYOUTS;   // 214
+static constexpr int MAX_TURNS = 200;
+static constexpr float ILLEGAL = 1e32f;
+
+// tables, filled by set_tables()
+static int64_t T_FROM[2][N_MOVE];
+static int64_t T_TO[2][N_MOVE];
+static bool T_GOAL[2][N_MOVE];
+static bool T_LAYOUT[N_LAYOUTS][8];
+static int64_t T_START[2][8];
+static bool tables_ready = false;
+
+void set_tables(py::array_t<int64_t> a_from, py::array_t<int64_t> a_to,
+                py::array_t<bool> a_goal, py::array_t<bool> layout_blue,
+                py::array_t<int64_t> start_cells) {
+    auto f = a_from.unchecked<2>();
+    auto t = a_to.unchecked<2>();
+    auto g = a_goal.unchecked<2>();
+    auto l = layout_blue.unchecked<2>();
+    auto s = start_cells.unchecked<2>();
+    for (int c = 0; c < 2; ++c)
+        for (int a = 0; a < N_MOVE; ++a) {
+            T_FROM[c][a] = f(c, a);
+            T_TO[c][a] = t(c, a);
+            T_GOAL[c][a] = g(c, a);
+        }
+    for (int i = 0; i < N_LAYOUTS; ++i)
+        for (int j = 0; j < 8; ++j) T_LAYOUT[i][j] = l(i, j);
+    for (int c = 0; c < 2; ++c)
+        for (int j = 0; j < 8; ++j) T_START[c][j] = s(c, j);
+    tables_ready = true;
+}
+
+void legal_masks_core(
+    py::array_t<int8_t> board_a,      // (G, 36)
+    py::array_t<int64_t> color_a,     // (G,)
+    py::array_t<int16_t> turn_count_a,// (G,)
+    py::array_t<int8_t> win_a,        // (G,)
+    py::array_t<float> mask_a)        // (G, 214) out
+{
+    auto board = board_a.unchecked<2>();
+    auto color = color_a.unchecked<1>();
+    auto tc = turn_count_a.unchecked<1>();
+    auto win = win_a.unchecked<1>();
+    auto mask = mask_a.mutable_unchecked<2>();
+    const py::ssize_t G = board_a.shape(0);
+    for (py::ssize_t g = 0; g < G; ++g) {
+        float* row = &mask(g, 0);
+        for (int a = 0; a < N_ACTIONS; ++a) row[a] = ILLEGAL;
+        if (win(g) >= 0) continue;
+        if (tc(g) < 0) {                          // layout turn
+            for (int a = N_MOVE; a < N_ACTIONS; ++a) row[a] = 0.0f;
+            continue;
+        }
+        const int c = (int)color(g);
+        for (int a = 0; a < N_MOVE; ++a) {
+            const int fcode = board(g, T_FROM[c][a]);
+            if (fcode < 0 || (fcode >> 1) != c) continue;
+            const int64_t to = T_TO[c][a];
+            bool ok;
+            if (to >= 0) {
+                const int tcode = board(g, to);
+                ok = !(tcode >= 0 && (tcode >> 1) == c);
+            } else {
+                ok = T_GOAL[c][a] && (fcode & 1) == 0;   // blue through goal
+            }
+            if (ok) row[a] = 0.0f;
+        }
+    }
+}
+
+void observations_core(
+    py::array_t<int8_t> board_a,      // (G, 36)
+    py::array_t<int8_t> piece_cnt_a,  // (G, 4)
+    py::array_t<int64_t> color_a,     // (G,)
+    py::array_t<float> scalar_a,      // (G, 18) out
+    py::array_t<float> planes_a)      // (G, 7, 6, 6) out (contiguous)
+{
+    auto board = board_a.unchecked<2>();
+    auto cnt = piece_cnt_a.unchecked<2>();
+    auto color = color_a.unchecked<1>();
+    auto scalar = scalar_a.mutable_unchecked<2>();
+    float* planes = planes_a.mutable_data();
+    const py::ssize_t G = board_a.shape(0);
+    for (py::ssize_t g = 0; g < G; ++g) {
+        const int me = (int)color(g);
+        float* p = planes + g * 7 * N_CELLS;
+        std::memset(p, 0, sizeof(float) * 7 * N_CELLS);
+        for (int cell = 0; cell < N_CELLS; ++cell) {
+            // WHITE sees the 180-degree rotated board
+            const int view = me == 0 ? cell : (N_CELLS - 1 - cell);
+            p[view] = 1.0f;                        // plane 0: ones
+            const int b = board(g, cell);
+            if (b < 0) continue;
+            const int col = b >> 1;
+            if (col == me) {
+                p[1 * N_CELLS + view] = 1.0f;
+                if (b == me * 2) p[3 * N_CELLS + view] = 1.0f;
+                else p[4 * N_CELLS + view] = 1.0f;
+            } else {
+                p[2 * N_CELLS + view] = 1.0f;
+            }
+            // planes 5/6 (true opponent types) stay zero: partial view
+        }
+        float* s = &scalar(g, 0);
+        std::memset(s, 0, sizeof(float) * 18);
+        s[0] = me == 0 ? 1.0f : 0.0f;
+        s[1] = 1.0f;
+        const int codes[4] = {me * 2, me * 2 + 1, (me ^ 1) * 2,
+                              (me ^ 1) * 2 + 1};
+        for (int grp = 0; grp < 4; ++grp) {
+            const int n = cnt(g, codes[grp]);
+            if (n >= 1 && n <= 4) s[2 + 4 * grp + (n - 1)] = 1.0f;
+        }
+    }
+}
+
+void step_core(
+    py::array_t<int8_t> board_a,      // (G, 36)
+    py::array_t<int8_t> slot_of_a,    // (G, 36)
+    py::array_t<int8_t> piece_pos_a,  // (G, 16)
+    py::array_t<int8_t> piece_cnt_a,  // (G, 4)
+    py::array_t<int64_t> color_a,     // (G,)
+    py::array_t<int16_t> turn_count_a,// (G,)
+    py::array_t<int8_t> win_a,        // (G,)
+    py::array_t<bool> over_a,         // (G,)
+    py::array_t<int64_t> act_a,       // (G,)
+    py::array_t<bool> done_a)         // (G,) out
+{
+    auto board = board_a.mutable_unchecked<2>();
+    auto slot_of = slot_of_a.mutable_unchecked<2>();
+    auto pos = piece_pos_a.mutable_unchecked<2>();
+    auto cnt = piece_cnt_a.mutable_unchecked<2>();
+    auto color = color_a.mutable_unchecked<1>();
+    auto tc = turn_count_a.mutable_unchecked<1>();
+    auto win = win_a.mutable_unchecked<1>();
+    auto over = over_a.mutable_unchecked<1>();
+    auto act = act_a.unchecked<1>();
+    auto done = done_a.mutable_unchecked<1>();
+    const py::ssize_t G = board_a.shape(0);
+    for (py::ssize_t g = 0; g < G; ++g) {
+        done(g) = false;
+        if (win(g) >= 0) { over(g) = true; continue; }
+        const int c = (int)color(g);
+        if (tc(g) < 0) {                          // layout turn
+            const bool* blues = T_LAYOUT[act(g) - N_MOVE];
+            for (int j = 0; j < 8; ++j) {
+                const int code = c * 2 + (blues[j] ? 0 : 1);
+                const int cell = (int)T_START[c][j];
+                const int slot = c * 8 + j;
+                board(g, cell) = (int8_t)code;
+                slot_of(g, cell) = (int8_t)slot;
+                pos(g, slot) = (int8_t)cell;
+            }
+            cnt(g, c * 2) = 4;
+            cnt(g, c * 2 + 1) = 4;
+        } else {
+            const int a = (int)act(g);
+            const int fcell = (int)T_FROM[c][a];
+            const int fcode = board(g, fcell);
+            const int fslot = slot_of(g, fcell);
+            const int64_t tcell = T_TO[c][a];
+            if (tcell < 0) {                      // blue exits via goal
+                board(g, fcell) = -1;
+                slot_of(g, fcell) = -1;
+                pos(g, fslot) = -1;
+                cnt(g, fcode) -= 1;
+                win(g) = (int8_t)c;
+            } else {
+                const int tcode = board(g, tcell);
+                if (tcode >= 0) {                 // capture
+                    const int tslot = slot_of(g, tcell);
+                    pos(g, tslot) = -1;
+                    cnt(g, tcode) -= 1;
+                    if (cnt(g, tcode) == 0) {
+                        // all enemy blues captured: mover wins; capturing
+                        // all enemy reds makes the CAPTURER lose
+                        win(g) = (int8_t)((tcode & 1) == 0 ? c : (c ^ 1));
+                    }
+                }
+                board(g, fcell) = -1;
+                slot_of(g, fcell) = -1;
+                board(g, tcell) = (int8_t)fcode;
+                slot_of(g, tcell) = (int8_t)fslot;
+                pos(g, fslot) = (int8_t)tcell;
+            }
+        }
+        color(g) ^= 1;
+        tc(g) += 1;
+        if (tc(g) >= MAX_TURNS && win(g) < 0) win(g) = 2;
+        done(g) = win(g) >= 0;
+        over(g) = win(g) >= 0;
+    }
+}
+
+bool ready() { return tables_ready; }
+
+PYBIND11_MODULE(_vec_geister_core, m) {
+    m.doc() = "native CPU core for the vectorized Geister engine";
+    m.def("set_tables", &set_tables);
+    m.def("legal_masks_core", &legal_masks_core);
+    m.def("observations_core", &observations_core);
+    m.def("step_core", &step_core);
+    m.def("ready", &ready);
+}

@@ -20,8 +20,33 @@ python.
 """
 
 import itertools
+import os
 
 import numpy as np
+
+
+def _load_native_core():
+    """In-tree C++ env core (envs/src/vec_geister_core.cpp): legality,
+    observation build and step as native per-game loops, bit-equal to the
+    numpy engine (tests/test_vec_geister_native.py).
+    HANDYRL_NO_NATIVE_ENV=1 or a missing .so falls back to numpy."""
+    if os.environ.get('HANDYRL_NO_NATIVE_ENV') == '1':
+        return None
+    so = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                      '_vec_geister_core.so')
+    if not os.path.exists(so):
+        return None
+    try:
+        import importlib.machinery
+        import importlib.util
+        loader = importlib.machinery.ExtensionFileLoader(
+            '_vec_geister_core', so)
+        spec = importlib.util.spec_from_loader('_vec_geister_core', loader)
+        mod = importlib.util.module_from_spec(spec)
+        loader.exec_module(mod)
+        return mod
+    except Exception:            # noqa: BLE001 - numpy fallback
+        return None
 
 BLACK, WHITE = 0, 1
 BLUE, RED = 0, 1
@@ -69,6 +94,14 @@ START_CELLS = np.array([
         ['E5', 'D5', 'C5', 'B5', 'E6', 'D6', 'C6', 'B6'])],
     dtype=np.int64)                             # (2, 8)
 
+_CORE = _load_native_core()
+if _CORE is not None:
+    _CORE.set_tables(np.ascontiguousarray(A_FROM),
+                     np.ascontiguousarray(A_TO),
+                     np.ascontiguousarray(A_GOAL),
+                     np.ascontiguousarray(LAYOUT_BLUE),
+                     np.ascontiguousarray(START_CELLS))
+
 
 class GeisterVecEnv:
     """G simultaneous Geister games (layout turns included: actions
@@ -111,6 +144,10 @@ class GeisterVecEnv:
         G = self.G
         mask = out if out is not None else \
             np.empty((G, N_ACTIONS), dtype=np.float32)
+        if _CORE is not None and mask.flags.c_contiguous:
+            _CORE.legal_masks_core(self.board, self.color, self.turn_count,
+                                   self.win, mask)
+            return mask
         mask[:] = ILLEGAL
         live = self.win < 0
         lay = live & (self.turn_count < 0)
@@ -134,6 +171,14 @@ class GeisterVecEnv:
         scalar (G, 18) float32 + board planes (G, 7, 6, 6) float32 —
         geister.py Environment.observation(player=turn) semantics."""
         G = self.G
+        if _CORE is not None:
+            if not hasattr(self, '_obs_scalar'):
+                self._obs_scalar = np.empty((G, 18), dtype=np.float32)
+                self._obs_planes = np.empty((G, 7, BOARD_N, BOARD_N),
+                                            dtype=np.float32)
+            _CORE.observations_core(self.board, self.piece_cnt, self.color,
+                                    self._obs_scalar, self._obs_planes)
+            return self._obs_scalar, self._obs_planes
         me = self.color
         b = self.board
         col = b >> 1
@@ -164,6 +209,14 @@ class GeisterVecEnv:
         actions are assumed legal).  Returns the (G,) bool mask of games
         that finished this step."""
         act = np.asarray(actions, dtype=np.int64)
+        if _CORE is not None:
+            if not hasattr(self, '_done_buf'):
+                self._done_buf = np.empty(self.G, dtype=bool)
+            _CORE.step_core(self.board, self.slot_of, self.piece_pos,
+                            self.piece_cnt, self.color, self.turn_count,
+                            self.win, self.over, np.ascontiguousarray(act),
+                            self._done_buf)
+            return self._done_buf
         active = self.win < 0
         lay = active & (self.turn_count < 0)
 
