@@ -199,7 +199,7 @@ class GeeseActorPool:
                 obs_in = np.concatenate([obs_sel, pad], axis=0)
             else:
                 obs_in = obs_sel
-            obs_t = torch.from_numpy(obs_in).to(self.device, non_blocking=True)
+            obs_t = torch.from_numpy(obs_in).to(self.device)
             policy, value = self._policy_forward(obs_t.float())
             policy, value = policy[:R], value[:R]
             A = policy.shape[1]
@@ -771,8 +771,10 @@ class MultiProcGeesePool:
         if event is not None:
             torch.cuda.current_stream().wait_event(event)
         if tidx_dev is not None and len(g_local):
+            # blocking H2D: g_local is a temporary pageable array (see
+            # replay.commit_traj) — async copy risks garbage indices
             rows = torch.from_numpy(np.ascontiguousarray(g_local)).to(
-                self.device, non_blocking=True)
+                self.device)
             tidx_dev.index_fill_(0, rows, 0)
         job_args = {'player': list(range(N_PLAYERS)),
                     'model_id': {p: -1 for p in range(N_PLAYERS)}}

@@ -447,21 +447,24 @@ class BatchedDRCEngine:
         keep_np = (1.0 - reset.astype(np.float32)).repeat(2).reshape(shape)
         if self._graph is not None:
             graph, st, packed = self._graph
-            st['scalar'].copy_(torch.from_numpy(scalar), non_blocking=True)
-            st['board'].copy_(torch.from_numpy(board), non_blocking=True)
-            st['mask'].copy_(torch.from_numpy(mask), non_blocking=True)
-            st['parity'].copy_(torch.from_numpy(parity.astype(np.int64)),
-                               non_blocking=True)
-            st['keep'].copy_(torch.from_numpy(keep_np), non_blocking=True)
+            # scalar/board/mask are shm views (stable until 'go'), but
+            # parity.astype and keep_np are TEMPORARIES: all copies stay
+            # blocking — hipMemcpyAsync from pageable memory can outlive
+            # a freed source and read garbage
+            st['scalar'].copy_(torch.from_numpy(scalar))
+            st['board'].copy_(torch.from_numpy(board))
+            st['mask'].copy_(torch.from_numpy(mask))
+            st['parity'].copy_(torch.from_numpy(parity.astype(np.int64)))
+            st['keep'].copy_(torch.from_numpy(keep_np))
             if self.fused_drc:
                 st['keep_h'].copy_(st['keep'])
             graph.replay()
             np.copyto(out, packed.cpu().numpy())
             return
         st = {
-            'scalar': torch.from_numpy(scalar).to(dev, non_blocking=True),
-            'board': torch.from_numpy(board).to(dev, non_blocking=True),
-            'mask': torch.from_numpy(mask).to(dev, non_blocking=True),
+            'scalar': torch.from_numpy(scalar).to(dev),
+            'board': torch.from_numpy(board).to(dev),
+            'mask': torch.from_numpy(mask).to(dev),
             'parity': torch.from_numpy(parity.astype(np.int64)).to(dev),
             'keep': torch.from_numpy(keep_np).to(dev),
         }

@@ -362,7 +362,10 @@ class GraphedReplayTrainStep:
         for key, arr in (('pos0', pos0), ('start', start), ('length', length),
                          ('seat', seat), ('outcome', outcome),
                          ('inv_total', inv_total)):
-            self.idx[key].copy_(torch.from_numpy(arr), non_blocking=True)
+            # blocking H2D: `arr` is a temporary pageable numpy array; an
+            # async copy that outlives it feeds garbage SAMPLE INDICES to
+            # the captured ring gather (intermittent batch corruption)
+            self.idx[key].copy_(torch.from_numpy(arr))
 
     def _run(self):
         tr = self.trainer
@@ -461,7 +464,10 @@ class GraphedRecurrentTrainStep:
             (('lead',) if self.burn_in else ())
         self._last_host_idx = dict(zip(keys, out))
         for key, arr in self._last_host_idx.items():
-            self.idx[key].copy_(torch.from_numpy(arr), non_blocking=True)
+            # blocking H2D: `arr` is a temporary pageable numpy array; an
+            # async copy that outlives it feeds garbage SAMPLE INDICES to
+            # the captured ring gather (intermittent batch corruption)
+            self.idx[key].copy_(torch.from_numpy(arr))
 
     def _run(self):
         tr = self.trainer

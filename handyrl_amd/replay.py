@@ -386,8 +386,13 @@ class DeviceReplay(ColumnRingReplay):
 
     def _commit_copy(self, traj, g_flat, t_flat, head0, n):
         dev = self.device
-        g_t = torch.from_numpy(g_flat).to(dev, non_blocking=True)
-        t_t = torch.from_numpy(t_flat).to(dev, non_blocking=True)
+        # BLOCKING H2D: the sources are temporary pageable numpy arrays;
+        # hipMemcpyAsync from pageable memory can return before the source
+        # is fully read, and a freed temporary then feeds garbage INDICES
+        # to the gather below (observed as HSAIL memory faults; same
+        # hazard class as the round-1 staging race)
+        g_t = torch.from_numpy(g_flat).to(dev)
+        t_t = torch.from_numpy(t_flat).to(dev)
         dst = (torch.arange(n, device=dev, dtype=torch.int64) + head0) \
             % self.ring_T
         self.obs[dst] = traj.obs[g_t, t_t]
