@@ -668,6 +668,139 @@ __global__ __launch_bounds__(256) void torus_conv_fused_kernel(
     }
 }
 
+// --- fused loss head (flagship FF solo family) ----------------------------
+// Replaces the ~40 eager elementwise/reduce ops of compute_loss around the
+// target scans (reference train.py:229-267, 189-215) with three kernels:
+//   pre : log-softmax gather, importance ratios, value/outcome splice
+//   fwd : reduced loss sums (p, v, H, H*decay) + data count
+//   bwd : d(policy logits), d(value) from the four loss-component grads
+// One thread per (b, t) row; the action dimension is a register loop.
+
+__global__ void loss_head_pre_kernel(
+    const float* __restrict__ policy,   // (n, A) masked logits
+    const long* __restrict__ action,    // (n,)
+    const float* __restrict__ mu,       // (n,) behavior prob
+    const float* __restrict__ emask,    // (n,)
+    const float* __restrict__ value,    // (n,)
+    const float* __restrict__ outcome,  // (n,)
+    float* __restrict__ log_sel,        // (n,) out
+    float* __restrict__ rho_clip,       // (n,) out (clip 1.0; == c)
+    float* __restrict__ v_spliced,      // (n,) out
+    long n, int A)
+{
+    const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= n) return;
+    const float* row = policy + i * A;
+    float mx = row[0];
+    for (int j = 1; j < A; ++j) mx = fmaxf(mx, row[j]);
+    float se = 0.f;
+    for (int j = 0; j < A; ++j) se += __expf(row[j] - mx);
+    const float lse = mx + __logf(se);
+    const float em = emask[i];
+    const float ls = (row[action[i]] - lse) * em;
+    log_sel[i] = ls;
+    const float lb = __logf(fminf(fmaxf(mu[i], 1e-16f), 1.f)) * em;
+    rho_clip[i] = fminf(__expf(ls - lb), 1.f);
+    v_spliced[i] = value[i] * em + outcome[i] * (1.f - em);
+}
+
+__global__ void loss_head_fwd_kernel(
+    const float* __restrict__ policy,   // (n, A)
+    const float* __restrict__ log_sel,  // (n,)
+    const float* __restrict__ ta,       // (n,) total advantage (rho*adv)
+    const float* __restrict__ tmask,    // (n,)
+    const float* __restrict__ omask,    // (n,)
+    const float* __restrict__ value,    // (n,)
+    const float* __restrict__ vtarget,  // (n,)
+    const float* __restrict__ progress, // (n,)
+    float* __restrict__ accum,          // (5,): p, v, ent, ent_decay, dcnt
+    float ent_decay, long n, int A)
+{
+    const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    float p_l = 0.f, v_l = 0.f, h_l = 0.f, hd_l = 0.f, dc = 0.f;
+    if (i < n) {
+        const float tm = tmask[i];
+        p_l = -log_sel[i] * ta[i] * tm;
+        const float dv = value[i] - vtarget[i];
+        v_l = 0.5f * dv * dv * omask[i];
+        // entropy of the masked logits
+        const float* row = policy + i * A;
+        float mx = row[0];
+        for (int j = 1; j < A; ++j) mx = fmaxf(mx, row[j]);
+        float se = 0.f;
+        for (int j = 0; j < A; ++j) se += __expf(row[j] - mx);
+        const float lse = mx + __logf(se);
+        float H = 0.f;
+        for (int j = 0; j < A; ++j) {
+            const float lp = row[j] - lse;
+            H -= __expf(lp) * lp;
+        }
+        h_l = H * tm;
+        hd_l = h_l * (1.f - progress[i] * (1.f - ent_decay));
+        dc = tm;
+    }
+    // block reduce + atomics
+    __shared__ float s[256][5];
+    s[threadIdx.x][0] = p_l; s[threadIdx.x][1] = v_l; s[threadIdx.x][2] = h_l;
+    s[threadIdx.x][3] = hd_l; s[threadIdx.x][4] = dc;
+    __syncthreads();
+    for (int step = 128; step > 0; step >>= 1) {
+        if ((int)threadIdx.x < step)
+            for (int k = 0; k < 5; ++k)
+                s[threadIdx.x][k] += s[threadIdx.x + step][k];
+        __syncthreads();
+    }
+    if (threadIdx.x == 0)
+        for (int k = 0; k < 5; ++k) atomicAdd(&accum[k], s[0][k]);
+}
+
+__global__ void loss_head_bwd_kernel(
+    const float* __restrict__ policy,   // (n, A)
+    const long* __restrict__ action,    // (n,)
+    const float* __restrict__ ta,       // (n,)
+    const float* __restrict__ tmask,    // (n,)
+    const float* __restrict__ omask,    // (n,)
+    const float* __restrict__ emask,    // (n,)
+    const float* __restrict__ value,    // (n,)
+    const float* __restrict__ vtarget,  // (n,)
+    const float* __restrict__ progress, // (n,)
+    const float* __restrict__ g,        // (5,): d/d accum[k] (see fwd)
+    float* __restrict__ dpolicy,        // (n, A) out
+    float* __restrict__ dvalue,         // (n,) out
+    float ent_decay, long n, int A)
+{
+    const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= n) return;
+    const float gp = g[0];
+    const float gv = g[1];
+    const float tm = tmask[i];
+    dvalue[i] = gv * (value[i] - vtarget[i]) * omask[i];
+
+    const float* row = policy + i * A;
+    float mx = row[0];
+    for (int j = 1; j < A; ++j) mx = fmaxf(mx, row[j]);
+    float se = 0.f;
+    for (int j = 0; j < A; ++j) se += __expf(row[j] - mx);
+    const float lse = mx + __logf(se);
+    float H = 0.f;
+    for (int j = 0; j < A; ++j) {
+        const float lp = row[j] - lse;
+        H -= __expf(lp) * lp;
+    }
+    const float w_pg = gp * ta[i] * tm * emask[i];
+    const float decay = 1.f - progress[i] * (1.f - ent_decay);
+    const float e_w = (g[2] + g[3] * decay) * tm;
+    const int a = (int)action[i];
+    float* drow = dpolicy + i * A;
+    for (int j = 0; j < A; ++j) {
+        const float lp = row[j] - lse;
+        const float pj = __expf(lp);
+        float d = w_pg * (pj - (j == a ? 1.f : 0.f));
+        d += e_w * (-pj * (lp + H));
+        drow[j] = d;
+    }
+}
+
 // --- fused ConvLSTM cell (Geister DRC core) -------------------------------
 // One kernel per cell evaluation: implicit-GEMM 3x3 zero-pad conv over the
 // K-ordered (x | h) input halves (concat never materialized) with the
@@ -910,6 +1043,63 @@ static torch::Tensor torus_conv_fused(
     return y;
 }
 
+static std::vector<torch::Tensor> loss_head_pre(
+    torch::Tensor policy, torch::Tensor action, torch::Tensor mu,
+    torch::Tensor emask, torch::Tensor value, torch::Tensor outcome) {
+    const long n = policy.size(0);
+    const int A = policy.size(1);
+    auto opts = policy.options();
+    auto log_sel = torch::empty({n}, opts);
+    auto rho = torch::empty({n}, opts);
+    auto v_spl = torch::empty({n}, opts);
+    auto stream = at::cuda::getCurrentCUDAStream();
+    hipLaunchKernelGGL(loss_head_pre_kernel, dim3((n + 255) / 256), dim3(256),
+        0, stream, policy.data_ptr<float>(), action.data_ptr<long>(),
+        mu.data_ptr<float>(), emask.data_ptr<float>(),
+        value.data_ptr<float>(), outcome.data_ptr<float>(),
+        log_sel.data_ptr<float>(), rho.data_ptr<float>(),
+        v_spl.data_ptr<float>(), n, A);
+    return {log_sel, rho, v_spl};
+}
+
+static torch::Tensor loss_head_fwd(
+    torch::Tensor policy, torch::Tensor log_sel, torch::Tensor ta,
+    torch::Tensor tmask, torch::Tensor omask, torch::Tensor value,
+    torch::Tensor vtarget, torch::Tensor progress, double ent_decay) {
+    const long n = policy.size(0);
+    const int A = policy.size(1);
+    auto accum = torch::zeros({5}, policy.options());
+    auto stream = at::cuda::getCurrentCUDAStream();
+    hipLaunchKernelGGL(loss_head_fwd_kernel, dim3((n + 255) / 256), dim3(256),
+        0, stream, policy.data_ptr<float>(), log_sel.data_ptr<float>(),
+        ta.data_ptr<float>(), tmask.data_ptr<float>(),
+        omask.data_ptr<float>(), value.data_ptr<float>(),
+        vtarget.data_ptr<float>(), progress.data_ptr<float>(),
+        accum.data_ptr<float>(), (float)ent_decay, n, A);
+    return accum;
+}
+
+static std::vector<torch::Tensor> loss_head_bwd(
+    torch::Tensor policy, torch::Tensor action, torch::Tensor ta,
+    torch::Tensor tmask, torch::Tensor omask, torch::Tensor emask,
+    torch::Tensor value, torch::Tensor vtarget, torch::Tensor progress,
+    torch::Tensor g, double ent_decay) {
+    const long n = policy.size(0);
+    const int A = policy.size(1);
+    auto dpolicy = torch::empty_like(policy);
+    auto dvalue = torch::empty({n}, policy.options());
+    auto stream = at::cuda::getCurrentCUDAStream();
+    hipLaunchKernelGGL(loss_head_bwd_kernel, dim3((n + 255) / 256), dim3(256),
+        0, stream, policy.data_ptr<float>(), action.data_ptr<long>(),
+        ta.data_ptr<float>(), tmask.data_ptr<float>(),
+        omask.data_ptr<float>(), emask.data_ptr<float>(),
+        value.data_ptr<float>(), vtarget.data_ptr<float>(),
+        progress.data_ptr<float>(), g.data_ptr<float>(),
+        dpolicy.data_ptr<float>(), dvalue.data_ptr<float>(),
+        (float)ent_decay, n, A);
+    return {dpolicy, dvalue};
+}
+
 static std::vector<torch::Tensor> convlstm_cell(
     torch::Tensor x, torch::Tensor h, torch::Tensor c,
     torch::Tensor wfrag, torch::Tensor bias, torch::Tensor nbr,
@@ -1079,6 +1269,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "fused wrap-around conv3x3 + BN-fold + residual + relu (MFMA)");
     m.def("torus_wgrad", &torus_wgrad,
           "torus conv weight gradient (MFMA, LDS-staged per image)");
+    m.def("loss_head_pre", &loss_head_pre,
+          "fused loss pipeline: log-softmax gather + importance ratios + "
+          "value/outcome splice");
+    m.def("loss_head_fwd", &loss_head_fwd,
+          "fused loss pipeline: reduced loss sums (p, v, H, H*decay, dcnt)");
+    m.def("loss_head_bwd", &loss_head_bwd,
+          "fused loss pipeline backward: d(policy), d(value)");
     m.def("convlstm_cell", &convlstm_cell,
           "fused ConvLSTM cell: implicit-GEMM conv(x|h) + 4-gate state "
           "update, one kernel per cell eval (Geister DRC core)");

@@ -128,11 +128,97 @@ def compose_losses(outputs, log_selected_policies, total_advantages, targets, ba
     return losses, dcnt
 
 
+class _FusedLossHead(torch.autograd.Function):
+    """Fused loss pipeline for the flagship family (feed-forward solo,
+    value head only): the ~40 eager ops around the target scans
+    (importance ratios, value/outcome splice, pg/value/entropy sums and
+    their backward) run as three HIP kernels + the fused scan
+    (ops/src/ext.hip loss_head_*; reference semantics train.py:229-267,
+    189-215).  forward returns accum (5,) = [p, v, ent, ent*decay, dcnt];
+    the caller assembles total = p + v - ent_reg * ent_decay_sum."""
+
+    @staticmethod
+    def forward(ctx, policy, value, action, mu, emask, tmask, omask,
+                outcome, ret, progress, shapes, args):
+        from . import ops as _ops
+        ext = _ops.require()
+        B, T = shapes
+        log_sel, rho, v_spl = ext.loss_head_pre(
+            policy, action, mu, emask, value.detach(), outcome)
+        view = lambda t: t.view(B, T, 1, 1)
+        oc4 = outcome.view(B, T, 1, 1)
+        targets_v, adv_v = compute_target(
+            args['value_target'], view(v_spl), oc4, None,
+            args['lambda'], 1, view(rho), view(rho), view(omask))
+        if args['policy_target'] != args['value_target']:
+            _, adv_v = compute_target(
+                args['policy_target'], view(v_spl), oc4, None,
+                args['lambda'], 1, view(rho), view(rho), view(omask))
+        targets_v = targets_v.reshape(-1).contiguous()
+        # 'return'-head-less family: target == advantage == batch return
+        ta = (rho * (adv_v.reshape(-1) + ret)).contiguous()
+        accum = ext.loss_head_fwd(policy, log_sel, ta, tmask, omask,
+                                  value.detach(), targets_v, progress,
+                                  args['entropy_regularization_decay'])
+        ctx.save_for_backward(policy, action, ta, tmask, omask, emask,
+                              value.detach(), targets_v, progress)
+        ctx.ent_decay = args['entropy_regularization_decay']
+        return accum
+
+    @staticmethod
+    def backward(ctx, gaccum):
+        from . import ops as _ops
+        ext = _ops.require()
+        (policy, action, ta, tmask, omask, emask, value, targets_v,
+         progress) = ctx.saved_tensors
+        dpolicy, dvalue = ext.loss_head_bwd(
+            policy, action, ta, tmask, omask, emask, value, targets_v,
+            progress, gaccum.contiguous(), ctx.ent_decay)
+        return (dpolicy, dvalue) + (None,) * 10
+
+
+def _fused_loss_ok(outputs, batch, args):
+    return (outputs['policy'].is_cuda
+            and os.environ.get('HANDYRL_FUSED_LOSS', '1') == '1'
+            and 'value' in outputs and 'return' not in outputs
+            and outputs['policy'].dim() == 4
+            and outputs['policy'].size(2) == 1       # solo seat axis
+            and not args['turn_based_training'])
+
+
+def _compute_loss_fused(outputs, batch, args):
+    """compute_loss tail on the fused loss-head kernels (flagship
+    family); numerics parity vs the eager tail is pinned by
+    tests/test_gpu.py::test_fused_loss_head_matches_eager."""
+    policy = outputs['policy'].float()
+    value = outputs['value'].float()
+    B, T = policy.size(0), policy.size(1)
+    A = policy.size(-1)
+    flat = lambda t: t.reshape(-1).contiguous().float()
+    oc = (batch['outcome'] * torch.ones_like(batch['value'])).reshape(-1)
+    accum = _FusedLossHead.apply(
+        policy.reshape(-1, A).contiguous(), value.reshape(-1),
+        batch['action'].reshape(-1).contiguous().long(),
+        flat(batch['selected_prob']), flat(batch['episode_mask']),
+        flat(batch['turn_mask']), flat(batch['observation_mask']),
+        oc.contiguous().float(), flat(batch['return']),
+        flat(batch['progress']), (B, T), args)
+    losses = {'p': accum[0], 'v': accum[1], 'ent': accum[2]}
+    losses['total'] = accum[0] + accum[1] \
+        - accum[3] * args['entropy_regularization']
+    return losses, accum[4]
+
+
 def compute_loss(batch, model, hidden, args):
     outputs = forward_prediction(model, hidden, batch, args)
     if args['burn_in_steps'] > 0:
         batch = map_r(batch, lambda v: v[:, args['burn_in_steps']:] if v.size(1) > 1 else v)
         outputs = map_r(outputs, lambda v: v[:, args['burn_in_steps']:])
+
+    if _fused_loss_ok(outputs, batch, args):
+        from . import ops as _ops
+        if _ops.available():
+            return _compute_loss_fused(outputs, batch, args)
 
     # loss math in fp32 regardless of autocast compute dtype
     outputs = {k: o.float() for k, o in outputs.items()}

@@ -722,3 +722,94 @@ def test_custom_training_path_tight_parity():
         if b_ref.dtype.is_floating_point:
             torch.testing.assert_close(b_mine, b_ref, rtol=1e-3, atol=1e-4,
                                        msg=lambda m, n=name: '%s: %s' % (n, m))
+
+
+@requires_gpu
+@pytest.mark.parametrize('ptarget,vtarget', [('VTRACE', 'VTRACE'),
+                                             ('UPGO', 'TD'),
+                                             ('TD', 'TD'),
+                                             ('MC', 'MC')])
+def test_fused_loss_head_matches_eager(ptarget, vtarget):
+    """Fused loss pipeline (loss_head_* kernels + fused scans) vs the
+    eager compute_loss tail: loss components, dcnt, and the policy/value
+    input gradients, all fp32 -> tight tolerance."""
+    import os
+    from handyrl_amd import train as htrain
+
+    torch.manual_seed(2)
+    B, T, A = 24, 16, 4
+    dev = torch.device('cuda', 0)
+    args = {'turn_based_training': False, 'observation': False,
+            'gamma': 0.8, 'forward_steps': T, 'burn_in_steps': 0,
+            'lambda': 0.7, 'policy_target': ptarget, 'value_target': vtarget,
+            'entropy_regularization': 0.1,
+            'entropy_regularization_decay': 0.1}
+
+    policy = (torch.randn(B, T, 1, A, device=dev) * 2).requires_grad_()
+    value = torch.randn(B, T, 1, 1, device=dev).requires_grad_()
+    em = (torch.rand(B, T, 1, 1, device=dev) < 0.8).float()
+    batch = {
+        'action': torch.randint(0, A, (B, T, 1, 1), device=dev),
+        'selected_prob': torch.rand(B, T, 1, 1, device=dev) * 0.9 + 0.05,
+        'episode_mask': em,
+        'turn_mask': em.clone(),
+        'observation_mask': em.clone(),
+        'outcome': torch.randn(B, 1, 1, 1, device=dev).clamp(-1, 1),
+        'return': torch.zeros(B, T, 1, 1, device=dev),
+        'reward': torch.zeros(B, T, 1, 1, device=dev),
+        'progress': torch.rand(B, T, 1, device=dev),
+        'value': torch.zeros(B, T, 1, 1, device=dev),
+    }
+
+    class _Identity:
+        training = False
+        def __call__(self, obs, hidden):
+            return {'policy': policy, 'value': value}
+
+    outputs = {'policy': policy, 'value': value}
+    assert htrain._fused_loss_ok(outputs, batch, args)
+    losses_f, dcnt_f = htrain._compute_loss_fused(outputs, batch, args)
+    losses_f['total'].backward()
+    gp_f, gv_f = policy.grad.clone(), value.grad.clone()
+    torch.cuda.synchronize()
+
+    policy.grad = None
+    value.grad = None
+    os.environ['HANDYRL_FUSED_LOSS'] = '0'
+    try:
+        # eager tail on identical inputs (skip forward_prediction)
+        outputs2 = {'policy': policy, 'value': value}
+        import torch.nn.functional as F
+        from handyrl_amd.losses import compute_target
+        actions = batch['action']
+        emasks = batch['episode_mask']
+        omasks = batch['observation_mask']
+        log_b = torch.log(torch.clamp(batch['selected_prob'], 1e-16, 1)) * emasks
+        log_t = F.log_softmax(policy, dim=-1).gather(-1, actions) * emasks
+        rhos = torch.exp(log_t.detach() - log_b)
+        crhos = torch.clamp(rhos, 0, 1)
+        cs = torch.clamp(rhos, 0, 1)
+        v_ng = value.detach() * emasks + batch['outcome'] * (1 - emasks)
+        tv, av = compute_target(vtarget, v_ng, batch['outcome'], None,
+                                args['lambda'], 1, crhos, cs, omasks)
+        if ptarget != vtarget:
+            _, av = compute_target(ptarget, v_ng, batch['outcome'], None,
+                                   args['lambda'], 1, crhos, cs, omasks)
+        tr, ar = batch['return'], batch['return']
+        ta = crhos * (av + ar)
+        losses_e, dcnt_e = htrain.compose_losses(
+            outputs2, log_t, ta, {'value': tv, 'return': tr}, batch, args)
+        losses_e['total'].backward()
+    finally:
+        os.environ.pop('HANDYRL_FUSED_LOSS')
+    torch.cuda.synchronize()
+
+    for k in ('p', 'v', 'ent', 'total'):
+        torch.testing.assert_close(losses_f[k], losses_e[k],
+                                   rtol=1e-4, atol=1e-3,
+                                   msg=lambda m, k=k: '%s: %s' % (k, m))
+    torch.testing.assert_close(dcnt_f, dcnt_e, rtol=0, atol=0)
+    torch.testing.assert_close(gp_f.reshape_as(policy.grad), policy.grad,
+                               rtol=1e-4, atol=1e-5)
+    torch.testing.assert_close(gv_f.reshape_as(value.grad), value.grad,
+                               rtol=1e-4, atol=1e-5)
