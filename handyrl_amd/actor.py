@@ -657,8 +657,13 @@ class MultiProcGeesePool:
                     gidx[:self.n_per] = base + torch.arange(
                         self.n_per, device=dev)
                     tidx = torch.zeros(bucket, dtype=torch.int64, device=dev)
-                    grow.append(self.graphed.capture_service_core(
-                        gidx, tidx, self.n_per))
+                    # the bundle keeps EVERY tensor the graph reads alive
+                    # (graph nodes hold raw pointers; a freed gidx block
+                    # reused by a later allocation turns replays into OOB
+                    # scatters — the round-2 HSAIL-fault root cause)
+                    bundle = self.graphed.capture_service_core(
+                        gidx, tidx, self.n_per) + (gidx,)
+                    grow.append(bundle)
                     trow.append(tidx)
                 graphs.append(grow)
                 tidxs.append(trow)
@@ -847,7 +852,8 @@ class MultiProcGeesePool:
                 # (from registered shm when available) and D2H out stay
                 # eager — ~4 host calls per service
                 assert M == self.n_per, (M, self.n_per)
-                graph, static_obs, packed = self._svc_graphs[wid][slot]
+                (graph, static_obs, packed,
+                 _zm, _gidx) = self._svc_graphs[wid][slot]
                 if self._use_registered:
                     static_obs[:M].copy_(self._obs_src[wid][slot][:M],
                                          non_blocking=True)

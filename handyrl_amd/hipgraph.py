@@ -168,7 +168,13 @@ class GraphedActorForward:
             packed = run_once()
         if was_training:
             self.model.train()
-        self.graphs[bucket] = (graph, static_obs, packed, static_idx)
+        # CRITICAL: every tensor a captured graph reads must stay
+        # python-referenced — the graph holds raw device pointers, and a
+        # freed block gets reused by later allocations (the train-graph
+        # capture reliably recycled a freed index/mask tensor here, turning
+        # replays into OOB scatters / garbage sampling masks)
+        self.graphs[bucket] = (graph, static_obs, packed, static_idx,
+                               zero_mask)
 
     def run(self, obs_u8_cpu, n_actions=4):
         """obs_u8_cpu: torch uint8 tensor (M, 17, 7, 11) on CPU (M <= bucket
@@ -178,7 +184,7 @@ class GraphedActorForward:
         bucket = self._bucket(M)
         if bucket not in self.graphs:
             self._capture(bucket, n_actions)
-        graph, static_obs, packed, _idx = self.graphs[bucket]
+        graph, static_obs, packed, _idx, _zm = self.graphs[bucket]
         static_obs[:M].copy_(obs_u8_cpu, non_blocking=True)
         # rows [M:bucket) keep stale data; every op is row-independent and
         # the outputs are sliced to the live rows
@@ -229,7 +235,8 @@ class GraphedActorForward:
             self.model.train()
         tidx_dev.zero_()          # warmup replays polluted the counters
         torch.cuda.synchronize()
-        return graph, static_obs, packed
+        # zero_mask rides along: the graph reads it every replay
+        return graph, static_obs, packed, zero_mask
 
     def capture_service(self, obs_src, out_dst, gidx_const, tidx_dev, M,
                         n_actions=4):
@@ -275,6 +282,7 @@ class GraphedActorForward:
             self.model.train()
         tidx_dev.zero_()          # warmup replays polluted the counters
         torch.cuda.synchronize()
+        graph._keep_alive = (static_obs, zero_mask, obs_src, out_dst)
         return graph
 
     def run_async(self, obs_pinned, M, out_pinned, event, n_actions=4,
@@ -290,7 +298,7 @@ class GraphedActorForward:
         bucket = self._bucket(M)
         if bucket not in self.graphs:
             self._capture(bucket, n_actions)
-        graph, static_obs, packed, static_idx = self.graphs[bucket]
+        graph, static_obs, packed, static_idx, _zm = self.graphs[bucket]
         static_obs[:M].copy_(obs_pinned[:M], non_blocking=True)
         if static_idx is not None:
             static_idx[0].copy_(idx_pinned[0, :bucket], non_blocking=True)
