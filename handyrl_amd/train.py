@@ -603,6 +603,24 @@ class Learner:
         self.num_returned_episodes = 0
         self.num_results = 0
 
+        # GPU actor pool (worker: {type: 'gpu'}): self-play generation runs
+        # as batched inference on the learner's GPU instead of CPU worker
+        # processes; any CPU workers then serve evaluation jobs only.
+        # HungryGeese uses the multiprocess env-worker pool (the flagship
+        # bench architecture: vectorized native env cores in child
+        # processes + one shared inference engine) — it must FORK before
+        # the Trainer creates a HIP context.
+        self.gpu_actor = bool(args['worker'].get('type') == 'gpu')
+        self._mpool = None
+        if self.gpu_actor and str(env_args.get('env')) == 'HungryGeese' \
+                and torch.cuda.is_available():
+            from .actor import MultiProcGeesePool
+            n_envs = args['worker'].get('num_envs', 2048)
+            procs = int(args['worker'].get('num_actor_procs', 8))
+            self._mpool = MultiProcGeesePool(
+                args, n_games=n_envs, seed=args['seed'] + 1, workers=procs,
+                traj_mode=args.get('replay') == 'device')
+
         self.worker = WorkerServer(args) if remote else WorkerCluster(args)
         self.trainer = Trainer(args, copy.deepcopy(self.model))
         # [amd] save_optimizer: restore the Adam state / step counters
@@ -617,10 +635,6 @@ class Learner:
                 self.trainer.data_cnt_ema = st['data_cnt_ema']
                 print('restored optimizer state at epoch %d' % self.model_epoch)
 
-        # GPU actor pool (worker: {type: 'gpu'}): self-play generation runs
-        # as batched inference on the learner's GPU instead of CPU worker
-        # processes; any CPU workers then serve evaluation jobs only.
-        self.gpu_actor = bool(args['worker'].get('type') == 'gpu')
         self.feed_lock = threading.Lock()
 
     # -- checkpoint files --------------------------------------------------
@@ -707,7 +721,12 @@ class Learner:
         replay buffer — no per-env worker processes, no pickled models."""
         n_envs = self.args['worker'].get('num_envs', 256)
         env_name = str(self.args.get('env', {}).get('env', ''))
-        if env_name == 'Geister':
+        if self._mpool is not None:
+            pool = self._mpool
+            pool.attach(self.trainer.model, self.trainer.device,
+                        replay=self.trainer.device_replay)
+            n_envs = self.args['worker'].get('num_envs', 2048)
+        elif env_name == 'Geister':
             from .actor_geister import GeisterActorPool
             pool = GeisterActorPool(self.trainer.model, self.args,
                                     n_games=n_envs,
