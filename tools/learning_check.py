@@ -12,6 +12,8 @@ from bench import bench_args
 
 torch.set_num_threads(1)
 args = bench_args(128, 16)
+args['entropy_regularization'] = float(
+    os.environ.get('HANDYRL_ENT_REG', args['entropy_regularization']))
 traj = os.environ.get('HANDYRL_TRAJ', '1') == '1'
 mpool = MultiProcGeesePool(args, n_games=1024, seed=7, workers=4,
                            traj_mode=traj)
