@@ -179,15 +179,21 @@ class MultiProcessJobExecutor:
         threading.Thread(target=self._dispatch, daemon=True).start()
 
     def _dispatch(self):
-        for conn in self.conns:
-            conn.send(next(self._jobs))
-        while not self.shutdown_flag:
-            for conn in mp_connection.wait(self.conns, timeout=0.3):
-                result = conn.recv()
-                conn.send(next(self._jobs))      # refill before postprocess
-                if self._post is not None:
-                    result = self._post(result)
-                self.output_queue.put(result)
+        try:
+            for conn in self.conns:
+                conn.send(next(self._jobs))
+            while not self.shutdown_flag:
+                for conn in mp_connection.wait(self.conns, timeout=0.3):
+                    result = conn.recv()
+                    conn.send(next(self._jobs))  # refill before postprocess
+                    if self._post is not None:
+                        result = self._post(result)
+                    self.output_queue.put(result)
+        except (OSError, EOFError, BrokenPipeError, StopIteration):
+            return      # workers gone / generator closed: quiet shutdown
+        except Exception:   # noqa: BLE001 - interpreter teardown
+            if not self.shutdown_flag:
+                raise
 
 
 # -- async connection hub ----------------------------------------------------
