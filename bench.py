@@ -31,7 +31,7 @@ from handyrl_amd.batch import Batcher
 from handyrl_amd.models.geese_net import GeeseNet
 from handyrl_amd.train import Trainer
 
-N_ENVS = 2048           # self-play games per GPU (8 env-worker processes)
+N_ENVS = 4096           # self-play games per GPU (12 env-worker processes)
 ACTOR_VEC_STEPS = 16    # env transitions (per game) per learner step
 
 
@@ -91,9 +91,13 @@ def main():
         os.environ.get('HANDYRL_DEVICE_REPLAY', '1') == '1'
     batcher = False if device_replay else Batcher(args, buffer)
 
-    # 8 workers measured best with event-polled completion (round-2 sweep:
-    # w8 512-549k vs w6 400-414k frames/s interleaved on one box)
-    actor_procs = int(os.environ.get('HANDYRL_ACTOR_PROCS', '8'))
+    # measured round-2 defaults (interleaved sweeps, profiles/b9_*):
+    # envs=4096 + 12 workers gives 752-812k frames/s at ~12 learner
+    # steps/s on one MI355X; multi-rank runs keep 8 workers per GPU so an
+    # 8-GPU node stays within its core budget
+    actor_procs = int(os.environ.get(
+        'HANDYRL_ACTOR_PROCS',
+        '12' if hdist.env_world_size() == 1 else '8'))
     # single slot per worker: 2-slot double-buffering measured SLOWER
     # end-to-end (it steers self-play into the short-episode regime;
     # BASELINE.md post-fix slots comparison)
