@@ -1,0 +1,52 @@
+"""End-to-end GPU run of the CONFIG-DRIVEN stack: main.py-style training
+(Learner + GPU actor pool + device replay + CPU eval workers) for a few
+epochs, then the offline evaluation harness on the saved checkpoint vs
+random opponents.
+
+Exercises in one go: Learner orchestration with the multiprocess
+HungryGeese pool (traj mode), epoch rollover + reference-layout
+checkpoint save, the in-training eval role split, and `--eval`-style
+evaluate_mp on models/latest.pth.  Run from a scratch directory.
+"""
+
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+
+def main():
+    epochs = int(sys.argv[1]) if len(sys.argv) > 1 else 3
+    args = {
+        'env_args': {'env': 'HungryGeese'},
+        'train_args': {
+            'turn_based_training': False, 'observation': False,
+            'gamma': 0.8, 'forward_steps': 16, 'burn_in_steps': 0,
+            'compress_steps': 4, 'entropy_regularization': 0.1,
+            'entropy_regularization_decay': 0.1,
+            'update_episodes': 20000, 'batch_size': 128,
+            'minimum_episodes': 2000, 'maximum_episodes': 40000,
+            'epochs': epochs, 'num_batchers': 2, 'eval_rate': 0.1,
+            'worker': {'type': 'gpu', 'num_envs': 2048,
+                       'num_actor_procs': 8, 'num_parallel': 2},
+            'replay': 'device',
+            'lambda': 0.7, 'policy_target': 'VTRACE',
+            'value_target': 'VTRACE',
+            'eval': {'opponent': ['random']},
+            'seed': 0, 'restart_epoch': 0, 'bf16': True,
+            'compress_episodes': False,
+        },
+    }
+    from handyrl_amd.train import train_main
+    train_main(args)
+    print('TRAIN_DONE', flush=True)
+
+    # offline eval: trained checkpoint vs random opponents (4-player)
+    from handyrl_amd.evaluation import eval_main
+    eval_args = {'env_args': {'env': 'HungryGeese'}}
+    eval_main(eval_args, ['models/latest.pth', '64', '4'])
+    print('EVAL_DONE', flush=True)
+
+
+if __name__ == '__main__':
+    main()
