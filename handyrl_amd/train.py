@@ -846,6 +846,11 @@ class Learner:
     def _maybe_rollover(self, state):
         """Epoch rollover bookkeeping; returns True to leave the server
         loop (GPU-actor shutdown with no workers attached)."""
+        if self.shutdown_flag:
+            # GPU actors can buffer tens of thousands of episodes; without
+            # this guard the backlog keeps re-firing rollovers past the
+            # configured epoch count during the drain
+            return self.gpu_actor and self.worker.connection_count() == 0
         if self.num_returned_episodes < state['next_update']:
             return False
         state['next_update'] += self.args['update_episodes']

@@ -41,10 +41,17 @@ def main():
     train_main(args)
     print('TRAIN_DONE', flush=True)
 
-    # offline eval: trained checkpoint vs random opponents (4-player)
-    from handyrl_amd.evaluation import eval_main
-    eval_args = {'env_args': {'env': 'HungryGeese'}}
-    eval_main(eval_args, ['models/latest.pth', '64', '4'])
+    # offline eval in a FRESH interpreter: forking eval workers from a
+    # CUDA-initialized process deadlocks in HIP
+    import subprocess
+    code = (
+        "import sys; sys.path.insert(0, %r);"
+        "from handyrl_amd.evaluation import eval_main;"
+        "eval_main({'env_args': {'env': 'HungryGeese'}},"
+        "          ['models/latest.pth', '64', '4'])"
+        % os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    proc = subprocess.run([sys.executable, '-c', code], timeout=600)
+    assert proc.returncode == 0, proc.returncode
     print('EVAL_DONE', flush=True)
 
 
