@@ -439,6 +439,34 @@ class BatchedDRCEngine:
             self._graph = None
 
     @torch.no_grad()
+    def infer_async(self, scalar, board, mask, parity, reset, out_pin,
+                    event):
+        """Issue one batched recurrent step WITHOUT a host sync: the
+        packed (action, prob, value, return) rows land in pinned
+        ``out_pin`` and ``event`` records completion.  Returns False when
+        no captured graph is available (caller falls back to infer()).
+        scalar/board/mask must outlive the copy (they are shm views the
+        worker will not touch until 'go'); the parity/keep temporaries
+        are copied blocking."""
+        if self._graph is None:
+            return False
+        per = self.n_per
+        shape = (per * 2, 1, 1) if self.fused_drc else (per * 2, 1, 1, 1)
+        keep_np = (1.0 - reset.astype(np.float32)).repeat(2).reshape(shape)
+        graph, st, packed = self._graph
+        st['scalar'].copy_(torch.from_numpy(scalar), non_blocking=True)
+        st['board'].copy_(torch.from_numpy(board), non_blocking=True)
+        st['mask'].copy_(torch.from_numpy(mask), non_blocking=True)
+        st['parity'].copy_(torch.from_numpy(parity.astype(np.int64)))
+        st['keep'].copy_(torch.from_numpy(keep_np))
+        if self.fused_drc:
+            st['keep_h'].copy_(st['keep'])
+        graph.replay()
+        out_pin.copy_(packed, non_blocking=True)
+        event.record()
+        return True
+
+    @torch.no_grad()
     def infer(self, scalar, board, mask, parity, reset, out):
         """numpy in (shard-sized arrays), numpy out: fills out[:, 0:4] with
         (action, prob, value, return) rows."""
@@ -619,11 +647,24 @@ class GeisterMultiProcPool:
         drain.start()
 
     def attach(self, model, device):
+        import os
         self.model = model
         self.device = device
         # one engine (resident hidden + captured graph) per worker shard
         self.engines = [BatchedDRCEngine(model, device, self.n_per)
                         for _ in range(self.workers)]
+        # event-polled async service (the geese-pool design): the DRC
+        # round trip of one worker overlaps every other worker's env work;
+        # HANDYRL_GEISTER_ASYNC=0 restores the synchronous service
+        self._async = (device.type == 'cuda'
+                       and os.environ.get('HANDYRL_GEISTER_ASYNC', '1') == '1')
+        if self._async:
+            self._out_pin = [torch.empty(self.n_per, 4, dtype=torch.float32,
+                                         pin_memory=True)
+                             for _ in range(self.workers)]
+            self._out_pin_np = [t.numpy() for t in self._out_pin]
+            self._events = [torch.cuda.Event() for _ in range(self.workers)]
+        self._fifo = []
 
     @torch.no_grad()
     def _infer(self, wid):
@@ -631,17 +672,64 @@ class GeisterMultiProcPool:
         self.engines[wid].infer(v['scalar'], v['board'], v['mask'],
                                 v['parity'], v['reset'], v['res'])
 
+    def _complete(self, wid):
+        self._events[wid].synchronize()
+        np.copyto(self.views[wid]['res'], self._out_pin_np[wid])
+        self.conns[wid].send('go')
+
+    def _poll_completions(self, force_first=False):
+        """'go' goes out the moment a worker's DRC results are ready
+        (events fire in issue order: one model, one stream)."""
+        while self._fifo:
+            wid = self._fifo[0]
+            if not force_first and not self._events[wid].query():
+                break
+            self._complete(self._fifo.pop(0))
+            force_first = False
+
     def step_once(self):
         import multiprocessing.connection as mpc
-        ready = mpc.wait(self.conns)
+        if not self._async:
+            ready = mpc.wait(self.conns)
+            conn = self.conns[self.rr] \
+                if self.conns[self.rr] in ready else ready[0]
+            wid = self.conns.index(conn)
+            self.rr = (wid + 1) % self.workers
+            tag, G, frames = conn.recv()
+            assert tag == 'obs'
+            self.frames += frames
+            self._infer(wid)
+            conn.send('go')
+            return frames
+
+        inflight = set(self._fifo)
+        while True:
+            self._poll_completions()
+            inflight = set(self._fifo)
+            waitable = [c for i, c in enumerate(self.conns)
+                        if i not in inflight]
+            if not waitable:
+                self._poll_completions(force_first=True)
+                continue
+            ready = mpc.wait(waitable, timeout=0.0005 if self._fifo else None)
+            if ready:
+                break
         conn = self.conns[self.rr] if self.conns[self.rr] in ready else ready[0]
         wid = self.conns.index(conn)
         self.rr = (wid + 1) % self.workers
         tag, G, frames = conn.recv()
         assert tag == 'obs'
         self.frames += frames
-        self._infer(wid)
-        conn.send('go')
+        v = self.views[wid]
+        if self.engines[wid].infer_async(v['scalar'], v['board'], v['mask'],
+                                         v['parity'], v['reset'],
+                                         self._out_pin[wid],
+                                         self._events[wid]):
+            self._fifo.append(wid)
+            self._poll_completions()
+        else:                      # capture unavailable: synchronous path
+            self._infer(wid)
+            conn.send('go')
         return frames
 
     def _drain_episodes(self):
