@@ -21,6 +21,7 @@ import torch
 
 from . import ops
 from .environment import make_env
+from .hipgraph import CAPTURE_LOCK
 
 N_ACTIONS = 214
 SCALAR_DIM = 18
@@ -425,7 +426,7 @@ class BatchedDRCEngine:
                     self._infer_body(st)
             torch.cuda.current_stream().wait_stream(stream)
             graph = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(graph):
+            with CAPTURE_LOCK, torch.cuda.graph(graph):
                 packed = self._infer_body(st)
             self._graph = (graph, st, packed)
             # captured warmups corrupted the hidden state: reset it

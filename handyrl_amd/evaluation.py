@@ -415,7 +415,14 @@ def load_model(model_path, model=None):
     assert model is not None
     import torch
     from .model import ModelWrapper
-    model.load_state_dict(torch.load(model_path))
+    sd = torch.load(model_path)
+    if hasattr(model, 'load_reference_state_dict'):
+        # nets whose saved layout is the REFERENCE layout (e.g. GeeseNet
+        # exports explicit under-BN conv biases) load through their
+        # fold-aware loader; it is exact for both layouts
+        model.load_reference_state_dict(sd)
+    else:
+        model.load_state_dict(sd)
     model.eval()
     return ModelWrapper(model)
 

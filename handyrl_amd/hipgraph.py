@@ -15,10 +15,20 @@ Constraints honored here:
   if capture fails anywhere the caller falls back to the eager path.
 """
 
+import threading
+
 import torch
 import torch.nn as nn
 
 from .util import map_r, bimap_r
+
+# Graph CAPTURE must never overlap a replay from another thread: torch's
+# graph-safe RNG is process-global, and a replay issued mid-capture fails
+# with "Offset increment outside graph capture" (seen when the Learner's
+# GPU-actor thread replayed service graphs while the trainer thread
+# captured its train step).  Capture sites hold this lock; so does the
+# Learner's actor loop around its replay bursts.
+CAPTURE_LOCK = threading.RLock()
 
 
 def apply_grad_guard(params, counter=None):
@@ -57,7 +67,7 @@ class GraphedTrainStep:
         torch.cuda.current_stream().wait_stream(stream)
 
         self.graph = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(self.graph):
+        with CAPTURE_LOCK, torch.cuda.graph(self.graph):
             self.losses, self.dcnt = self._run()
 
     def _run(self):
@@ -164,7 +174,7 @@ class GraphedActorForward:
                 run_once()
         torch.cuda.current_stream().wait_stream(stream)
         graph = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(graph):
+        with CAPTURE_LOCK, torch.cuda.graph(graph):
             packed = run_once()
         if was_training:
             self.model.train()
@@ -229,7 +239,7 @@ class GraphedActorForward:
                 run_once()
         torch.cuda.current_stream().wait_stream(stream)
         graph = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(graph):
+        with CAPTURE_LOCK, torch.cuda.graph(graph):
             packed = run_once()
         if was_training:
             self.model.train()
@@ -276,7 +286,7 @@ class GraphedActorForward:
                 run_once()
         torch.cuda.current_stream().wait_stream(stream)
         graph = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(graph):
+        with CAPTURE_LOCK, torch.cuda.graph(graph):
             run_once()
         if was_training:
             self.model.train()
@@ -349,7 +359,7 @@ class GraphedReplayTrainStep:
         try:
             self.graph = torch.cuda.CUDAGraph()
             self._fill()
-            with torch.cuda.graph(self.graph):
+            with CAPTURE_LOCK, torch.cuda.graph(self.graph):
                 self.losses, self.dcnt = self._run()
         except Exception as e:      # noqa: BLE001 - run eager if capture fails
             import sys
@@ -457,7 +467,7 @@ class GraphedRecurrentTrainStep:
                 torch.cuda.current_stream().wait_stream(stream)
                 self.graph = torch.cuda.CUDAGraph()
                 self._fill()
-                with torch.cuda.graph(self.graph):
+                with CAPTURE_LOCK, torch.cuda.graph(self.graph):
                     self.losses, self.dcnt = self._run()
             except Exception as e:  # noqa: BLE001 - run eager if capture fails
                 import sys

@@ -100,6 +100,12 @@ class GeeseNet(nn.Module):
         """Reference-layout export hook (used by Learner.update_model)."""
         return export_reference_state_dict(self)
 
+    def load_reference_state_dict(self, sd):
+        """Load EITHER layout exactly: reference checkpoints (with
+        under-BN conv biases) fold the bias into running_mean; this
+        repo's own exports (zero biases) reduce to a strict load."""
+        load_reference_state_dict(self, sd)
+
     def forward(self, x, hidden=None):
         if x.is_cuda and self.training and self._custom_train_ok():
             return self._forward_nhwc_train(x)

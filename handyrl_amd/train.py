@@ -594,8 +594,11 @@ class Learner:
         self.model_epoch = self.args['restart_epoch']
         self.model = net if net is not None else self.env.net()
         if self.model_epoch > 0:
-            self.model.load_state_dict(
-                torch.load(self.model_path(self.model_epoch)), strict=False)
+            sd = torch.load(self.model_path(self.model_epoch))
+            if hasattr(self.model, 'load_reference_state_dict'):
+                self.model.load_reference_state_dict(sd)
+            else:
+                self.model.load_state_dict(sd, strict=False)
 
         self.generation_stats = ScoreBook()
         self.eval_stats = ScoreBook()
@@ -750,8 +753,11 @@ class Learner:
                 last_epoch = self.model_epoch
             was_training = self.trainer.model.training
             self.trainer.model.eval()
-            for _ in range(8):
-                pool.step_once()
+            from .hipgraph import CAPTURE_LOCK
+            with CAPTURE_LOCK:
+                # never replay while the trainer thread captures its graph
+                for _ in range(8):
+                    pool.step_once()
             if was_training:
                 self.trainer.model.train()
             episodes = pool.harvest()
