@@ -115,7 +115,8 @@ def main():
         mpool = MultiProcGeesePool(args, n_games=cli.envs,
                                    seed=1000 + hdist.env_rank() * 31,
                                    workers=actor_procs,
-                                   traj_mode=traj_mode)
+                                   traj_mode=traj_mode,
+                                   make_stubs=not traj_mode)
 
     use_cuda = torch.cuda.is_available()
     device = torch.device('cuda', local_rank) if use_cuda else torch.device('cpu')

@@ -523,7 +523,7 @@ class MultiProcGeesePool:
     """
 
     def __init__(self, args, n_games=768, seed=0, workers=3, slots=None,
-                 traj_mode=False):
+                 traj_mode=False, make_stubs=True):
         import multiprocessing as mp
         from multiprocessing import shared_memory
         self.args = args
@@ -532,6 +532,9 @@ class MultiProcGeesePool:
         # requires attach(model, device, replay=DeviceReplay) on CUDA;
         # workers then ship only step/episode METADATA, never obs arrays
         self.traj_mode = traj_mode
+        # make_stubs=False skips per-episode stat stubs on the service
+        # path (consumers that only need episodes_done, e.g. bench.py)
+        self.make_stubs = make_stubs
         if slots is None:
             # >1 slots (double-buffered half-shards) measures faster but a
             # GPU-side transport race poisons recorded values (NaN losses
@@ -781,16 +784,20 @@ class MultiProcGeesePool:
             rows = torch.from_numpy(np.ascontiguousarray(g_local)).to(
                 self.device)
             tidx_dev.index_fill_(0, rows, 0)
-        job_args = {'player': list(range(N_PLAYERS)),
-                    'model_id': {p: -1 for p in range(N_PLAYERS)}}
-        stubs = [{'args': job_args, 'steps': int(lens[k]),
-                  'outcome': {p: float(outcomes[k, p])
-                              for p in range(N_PLAYERS)},
-                  'committed': True}
-                 for k in range(len(g_local))]
-        with self._completed_lock:
-            self.completed.extend(stubs)
-            self.episodes_done += len(stubs)
+        if self.make_stubs:
+            job_args = {'player': list(range(N_PLAYERS)),
+                        'model_id': {p: -1 for p in range(N_PLAYERS)}}
+            stubs = [{'args': job_args, 'steps': int(lens[k]),
+                      'outcome': {p: float(outcomes[k, p])
+                                  for p in range(N_PLAYERS)},
+                      'committed': True}
+                     for k in range(len(g_local))]
+            with self._completed_lock:
+                self.completed.extend(stubs)
+                self.episodes_done += len(stubs)
+        else:
+            with self._completed_lock:
+                self.episodes_done += len(g_local)
 
     def _poll_completions(self, force_first=False):
         """Complete every in-flight round whose GPU work already finished

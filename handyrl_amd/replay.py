@@ -384,15 +384,34 @@ class DeviceReplay(ColumnRingReplay):
                 self._ready.append((entries, n, event))
         return event
 
+    def _stage_idx(self, key, arr):
+        """Pinned staging for commit index arrays: async H2D made SAFE
+        (the pinned buffer persists; a refill waits on the previous
+        copy's event).  Runs inside the ingest-stream context.  Temporary
+        pageable sources must never feed a non-blocking copy — a freed
+        temporary under a late-reading hipMemcpyAsync supplies garbage
+        INDICES to the gather below (HSAIL-fault class)."""
+        if self.device.type != 'cuda':
+            return torch.from_numpy(arr).to(self.device)
+        n = arr.shape[0]
+        pin, ev = self._pin.get(key, (None, None))
+        if pin is None or pin.shape[0] < n:
+            pin = torch.empty(max(int(n * 1.5), 4096), dtype=torch.int64,
+                              pin_memory=True)
+            ev = None
+        if ev is not None:
+            ev.synchronize()
+        pin.numpy()[:n] = arr
+        dev_t = pin[:n].to(self.device, non_blocking=True)
+        ev = torch.cuda.Event()
+        ev.record()
+        self._pin[key] = (pin, ev)
+        return dev_t
+
     def _commit_copy(self, traj, g_flat, t_flat, head0, n):
         dev = self.device
-        # BLOCKING H2D: the sources are temporary pageable numpy arrays;
-        # hipMemcpyAsync from pageable memory can return before the source
-        # is fully read, and a freed temporary then feeds garbage INDICES
-        # to the gather below (observed as HSAIL memory faults; same
-        # hazard class as the round-1 staging race)
-        g_t = torch.from_numpy(g_flat).to(dev)
-        t_t = torch.from_numpy(t_flat).to(dev)
+        g_t = self._stage_idx('commit_g', g_flat)
+        t_t = self._stage_idx('commit_t', t_flat)
         dst = (torch.arange(n, device=dev, dtype=torch.int64) + head0) \
             % self.ring_T
         self.obs[dst] = traj.obs[g_t, t_t]
