@@ -471,8 +471,10 @@ def _geese_env_worker(conn, ep_conn, obs_name, res_name, n_games, args, seed,
 
     def obs_msg(s, m, frames):
         if traj_mode:
-            return ('obs', s, m, frames, pools[s].take_meta(),
-                    pools[s].take_finished())
+            # no per-step meta: shards are always full (asserted below)
+            # and the parent mirrors the step counters itself
+            pools[s].take_meta()
+            return ('obs', s, m, frames, pools[s].take_finished())
         return ('obs', s, m, frames)
 
     def prep(s):
@@ -633,6 +635,9 @@ class MultiProcGeesePool:
                 self._idx_pin = mk((2, bucket), torch.int64)
                 self._idx_pin_np = [[t.numpy() for t in row]
                                     for row in self._idx_pin]
+                self._tidx_np = [[np.zeros(self.n_per, dtype=np.int64)
+                                  for _ in range(self.slots)]
+                                 for _ in range(self.workers)]
             self._register_shm()
             self._svc_graphs = None
             if self.traj is not None:
@@ -847,12 +852,14 @@ class MultiProcGeesePool:
         t0 = time.time()
         idx_pin = None
         if self.traj is not None:
-            _tag, _slot, _m, _f, meta, fin = msg
+            _tag, _slot, _m, _f, fin = msg
             base = (wid * self.slots + slot) * self.n_per
             if fin is not None:
                 tidx_dev = self._tidx_dev[wid][slot] \
                     if self._svc_graphs is not None else None
                 self._commit_finished(base, fin, tidx_dev)
+                if self._svc_graphs is None:
+                    self._tidx_np[wid][slot][fin[0]] = 0
             if M and self._svc_graphs is not None:
                 # service-core graph: forward + sample + trajectory
                 # scatter + device step counters as ONE replay; H2D in
@@ -885,13 +892,16 @@ class MultiProcGeesePool:
                     self._poll_completions()
                 return frames
             if M:
-                rows, tidx = meta
+                # per-op fallback: the parent's mirrored step counters
+                # replace the meta the worker used to ship
+                assert M == self.n_per, (M, self.n_per)
                 bucket = self.graphed._bucket(M)
                 idx_np = self._idx_pin_np[wid][slot]
-                idx_np[0, :M] = base + rows
+                idx_np[0, :M] = base + np.arange(M)
                 idx_np[0, M:bucket] = self.traj.scratch_row
-                idx_np[1, :M] = tidx
+                idx_np[1, :M] = self._tidx_np[wid][slot]
                 idx_np[1, M:bucket] = 0
+                self._tidx_np[wid][slot] += 1
                 idx_pin = self._idx_pin[wid][slot]
         if M and self.graphed is not None:
             if self._use_registered:
