@@ -253,6 +253,17 @@ def _geister_vec_worker(conn, ep_conn, shm_names, n_games, args, seed):
             break
 
         actions = res_v[:, 0].astype(np.int64)
+        # visibility: sampled actions that violate the mask we shipped
+        # (a NaN policy row or an all-illegal row samples arbitrarily;
+        # the env treats them as a pass, but they signal upstream NaN)
+        bad = (actions < 0) | (actions >= N_ACTIONS)
+        ok = ~bad
+        bad[ok] |= mask_v[gar[ok], actions[ok]] != 0.0
+        nbad = int(bad.sum())
+        if nbad:
+            import sys as _sys
+            print('# geister worker: %d illegal sampled actions'
+                  % nbad, file=_sys.stderr, flush=True)
         if columnar:
             _record_round(rec, rec_len, gar, scalar_v, board_v, mask_v,
                           parity_v, actions, res_v[:, 1], res_v[:, 2])

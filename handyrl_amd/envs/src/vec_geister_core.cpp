@@ -167,8 +167,19 @@ void step_core(
         done(g) = false;
         if (win(g) >= 0) { over(g) = true; continue; }
         const int c = (int)color(g);
+        // Defensive bounds checks: the numpy engine wraps negative
+        // indices harmlessly, but a raw C index corrupts adjacent
+        // memory.  Illegal actions CAN arrive (e.g. a NaN policy row
+        // sampled over an all-illegal mask) — treat them as a pass:
+        // no board change, turn still advances.
         if (tc(g) < 0) {                          // layout turn
-            const bool* blues = T_LAYOUT[act(g) - N_MOVE];
+            const int64_t li = act(g) - N_MOVE;
+            if (li < 0 || li >= N_LAYOUTS) {
+                color(g) ^= 1;
+                tc(g) += 1;
+                continue;
+            }
+            const bool* blues = T_LAYOUT[li];
             for (int j = 0; j < 8; ++j) {
                 const int code = c * 2 + (blues[j] ? 0 : 1);
                 const int cell = (int)T_START[c][j];
@@ -180,10 +191,27 @@ void step_core(
             cnt(g, c * 2) = 4;
             cnt(g, c * 2 + 1) = 4;
         } else {
-            const int a = (int)act(g);
+            const int64_t a64 = act(g);
+            if (a64 < 0 || a64 >= N_MOVE) {
+                color(g) ^= 1;
+                tc(g) += 1;
+                if (tc(g) >= MAX_TURNS && win(g) < 0) win(g) = 2;
+                done(g) = win(g) >= 0;
+                over(g) = win(g) >= 0;
+                continue;
+            }
+            const int a = (int)a64;
             const int fcell = (int)T_FROM[c][a];
             const int fcode = board(g, fcell);
             const int fslot = slot_of(g, fcell);
+            if (fcode < 0 || (fcode >> 1) != c || fslot < 0) {
+                color(g) ^= 1;                    // illegal move: pass
+                tc(g) += 1;
+                if (tc(g) >= MAX_TURNS && win(g) < 0) win(g) = 2;
+                done(g) = win(g) >= 0;
+                over(g) = win(g) >= 0;
+                continue;
+            }
             const int64_t tcell = T_TO[c][a];
             if (tcell < 0) {                      // blue exits via goal
                 board(g, fcell) = -1;
