@@ -615,14 +615,21 @@ class Learner:
         # the Trainer creates a HIP context.
         self.gpu_actor = bool(args['worker'].get('type') == 'gpu')
         self._mpool = None
-        if self.gpu_actor and str(env_args.get('env')) == 'HungryGeese' \
-                and torch.cuda.is_available():
-            from .actor import MultiProcGeesePool
+        env_name = str(env_args.get('env'))
+        if self.gpu_actor and torch.cuda.is_available():
             n_envs = args['worker'].get('num_envs', 2048)
             procs = int(args['worker'].get('num_actor_procs', 8))
-            self._mpool = MultiProcGeesePool(
-                args, n_games=n_envs, seed=args['seed'] + 1, workers=procs,
-                traj_mode=args.get('replay') == 'device')
+            if env_name == 'HungryGeese':
+                from .actor import MultiProcGeesePool
+                self._mpool = MultiProcGeesePool(
+                    args, n_games=n_envs, seed=args['seed'] + 1,
+                    workers=procs,
+                    traj_mode=args.get('replay') == 'device')
+            elif env_name == 'Geister':
+                from .actor_geister import GeisterMultiProcPool
+                self._mpool = GeisterMultiProcPool(
+                    args, n_games=n_envs, seed=args['seed'] + 1,
+                    workers=procs)
 
         self.worker = WorkerServer(args) if remote else WorkerCluster(args)
         self.trainer = Trainer(args, copy.deepcopy(self.model))
@@ -726,8 +733,11 @@ class Learner:
         env_name = str(self.args.get('env', {}).get('env', ''))
         if self._mpool is not None:
             pool = self._mpool
-            pool.attach(self.trainer.model, self.trainer.device,
-                        replay=self.trainer.device_replay)
+            if env_name == 'HungryGeese':
+                pool.attach(self.trainer.model, self.trainer.device,
+                            replay=self.trainer.device_replay)
+            else:
+                pool.attach(self.trainer.model, self.trainer.device)
             n_envs = self.args['worker'].get('num_envs', 2048)
         elif env_name == 'Geister':
             from .actor_geister import GeisterActorPool
