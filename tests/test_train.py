@@ -350,3 +350,42 @@ def test_device_replay_config_validation():
     bad['burn_in_steps'] = 2
     with _pytest.raises(ValueError, match='burn_in'):
         Trainer(bad, SimpleConv2dModel(), device=torch.device('cpu'))
+
+
+def test_gpu_actor_mode_respects_epoch_limit():
+    """GPU actors buffer many episodes; the backlog must not re-fire
+    epoch rollovers past the configured count during the shutdown drain
+    (regression: epochs=1 once rolled to 7 on hardware)."""
+    script = textwrap.dedent('''
+        import sys
+        sys.path.insert(0, %r)
+        from handyrl_amd.train import train_main
+        args = {
+            'env_args': {'env': 'HungryGeese'},
+            'train_args': {
+                'turn_based_training': False, 'observation': False,
+                'gamma': 0.8, 'forward_steps': 8, 'burn_in_steps': 0,
+                'compress_steps': 4, 'entropy_regularization': 0.1,
+                'entropy_regularization_decay': 0.1, 'update_episodes': 10,
+                'batch_size': 4, 'minimum_episodes': 10,
+                'maximum_episodes': 300,
+                'epochs': 1, 'num_batchers': 1, 'eval_rate': 0.1,
+                'worker': {'type': 'gpu', 'num_parallel': 0, 'num_envs': 64},
+                'lambda': 0.7, 'policy_target': 'VTRACE',
+                'value_target': 'VTRACE',
+                'eval': {'opponent': ['random']}, 'seed': 0,
+                'restart_epoch': 0, 'bf16': False,
+            },
+        }
+        train_main(args)
+        print('TRAIN_DONE')
+    ''') % REPO
+    with tempfile.TemporaryDirectory() as tmp:
+        res = subprocess.run([sys.executable, '-c', script], cwd=tmp,
+                             capture_output=True, text=True, timeout=300)
+        out = res.stdout
+        assert 'TRAIN_DONE' in out, (out[-3000:], res.stderr[-3000:])
+        # 64 envs with update_episodes=10 buffer a large backlog; exactly
+        # ONE epoch must be written
+        assert os.path.exists(os.path.join(tmp, 'models', '1.pth'))
+        assert not os.path.exists(os.path.join(tmp, 'models', '2.pth')), out
