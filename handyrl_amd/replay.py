@@ -359,11 +359,12 @@ class DeviceReplay(ColumnRingReplay):
         dev = self.device
         head0 = self.head
         if dev.type == 'cuda':
+            main = torch.cuda.current_stream()
             stream = self._ingest_stream if self._ingest is not None \
-                else torch.cuda.current_stream()
-            stream.wait_stream(torch.cuda.current_stream())
+                else main
             with torch.cuda.stream(stream):
-                event = self._commit_copy(traj, g_flat, t_flat, head0, n)
+                event = self._commit_copy(traj, g_flat, t_flat, head0, n,
+                                          gate=main)
         else:
             event = None
             self._commit_copy(traj, g_flat, t_flat, head0, n)
@@ -408,12 +409,19 @@ class DeviceReplay(ColumnRingReplay):
         self._pin[key] = (pin, ev)
         return dev_t
 
-    def _commit_copy(self, traj, g_flat, t_flat, head0, n):
+    def _commit_copy(self, traj, g_flat, t_flat, head0, n, gate=None):
         dev = self.device
+        # stage the index H2Ds BEFORE gating on the main stream: gating
+        # them too makes the next commit's pinned-buffer reuse sync wait
+        # behind every in-flight forward (measured as the short-episode
+        # regime's 2x service cost); only the ring GATHERS below must
+        # order after the main stream's trajectory scatters
         g_t = self._stage_idx('commit_g', g_flat)
         t_t = self._stage_idx('commit_t', t_flat)
         dst = (torch.arange(n, device=dev, dtype=torch.int64) + head0) \
             % self.ring_T
+        if gate is not None:
+            torch.cuda.current_stream().wait_stream(gate)
         self.obs[dst] = traj.obs[g_t, t_t]
         self.alive[dst] = traj.alive[g_t, t_t]
         rec = traj.rec[g_t, t_t]                     # (n, 4, 3)
