@@ -167,6 +167,86 @@ class ColumnRingReplay:
         with self._table_lock:
             return len(self.table)
 
+    def commit_traj(self, traj, g_rows, lens, outcomes):
+        """Copy finished device-recorded episodes (handyrl_amd/traj) into
+        the ring DEVICE-TO-DEVICE and publish their table entries.
+
+        g_rows/lens: int64[K] global trajectory rows and episode lengths;
+        outcomes: float32[K, 4].  Returns the completion event (the caller
+        must make the next trajectory write to these rows wait on it) or
+        None when K == 0.  Runs on the ingest stream, ordered after all
+        issued main-stream work (the in-graph scatters of these rows)."""
+        K = len(g_rows)
+        if K == 0:
+            return None
+        lens = np.asarray(lens, dtype=np.int64)
+        n = int(lens.sum())
+        assert n <= self.ring_T // 4, 'episode burst larger than ring/4'
+        import time as _time
+        waited = 0.0
+        while True:                    # reader-floor back-pressure
+            floor = self._reader_floor
+            if floor is None or self.head + n - self.ring_T <= floor:
+                break
+            _time.sleep(0.001)
+            waited += 0.001
+            if waited > 5.0:
+                raise RuntimeError('replay ring writer starved (floor %r)'
+                                   % (floor,))
+        g_flat, t_flat = traj.episode_indices(np.asarray(g_rows), lens)
+        dev = self.device
+        head0 = self.head
+        if dev.type == 'cuda':
+            main = torch.cuda.current_stream()
+            stream = self._ingest_stream if self._ingest is not None \
+                else main
+            with torch.cuda.stream(stream):
+                event = self._commit_copy(traj, g_flat, t_flat, head0, n,
+                                          gate=main)
+        else:
+            event = None
+            self._commit_copy(traj, g_flat, t_flat, head0, n)
+        entries, pos = [], head0
+        for k in range(K):
+            entries.append((pos, int(lens[k]),
+                            np.asarray(outcomes[k], dtype=np.float32)))
+            pos += int(lens[k])
+        self.head = pos
+        self.head_planned = max(self.head_planned, pos)
+        with self._table_lock:
+            if event is None:
+                self.table.extend(entries)
+                self.total_added += len(entries)
+                self._evict()
+        if event is not None:
+            with self._ready_lock:
+                self._ready.append((entries, n, event))
+        return event
+
+    def _stage_idx(self, key, arr):
+        """Pinned staging for commit index arrays: async H2D made SAFE
+        (the pinned buffer persists; a refill waits on the previous
+        copy's event).  Runs inside the ingest-stream context.  Temporary
+        pageable sources must never feed a non-blocking copy — a freed
+        temporary under a late-reading hipMemcpyAsync supplies garbage
+        INDICES to the gather below (HSAIL-fault class)."""
+        if self.device.type != 'cuda':
+            return torch.from_numpy(arr).to(self.device)
+        n = arr.shape[0]
+        pin, ev = self._pin.get(key, (None, None))
+        if pin is None or pin.shape[0] < n:
+            pin = torch.empty(max(int(n * 1.5), 4096), dtype=torch.int64,
+                              pin_memory=True)
+            ev = None
+        if ev is not None:
+            ev.synchronize()
+        pin.numpy()[:n] = arr
+        dev_t = pin[:n].to(self.device, non_blocking=True)
+        ev = torch.cuda.Event()
+        ev.record()
+        self._pin[key] = (pin, ev)
+        return dev_t
+
     # -- write path --------------------------------------------------------
     def _stage(self, key, parts, n, dtype, tail_shape):
         """Fill a pinned staging buffer from per-episode arrays (one memcpy
@@ -329,86 +409,6 @@ class DeviceReplay(ColumnRingReplay):
         self._arange_cache = {}
 
     # -- device-side ingest (traj mode) --------------------------------------
-    def commit_traj(self, traj, g_rows, lens, outcomes):
-        """Copy finished device-recorded episodes (handyrl_amd/traj) into
-        the ring DEVICE-TO-DEVICE and publish their table entries.
-
-        g_rows/lens: int64[K] global trajectory rows and episode lengths;
-        outcomes: float32[K, 4].  Returns the completion event (the caller
-        must make the next trajectory write to these rows wait on it) or
-        None when K == 0.  Runs on the ingest stream, ordered after all
-        issued main-stream work (the in-graph scatters of these rows)."""
-        K = len(g_rows)
-        if K == 0:
-            return None
-        lens = np.asarray(lens, dtype=np.int64)
-        n = int(lens.sum())
-        assert n <= self.ring_T // 4, 'episode burst larger than ring/4'
-        import time as _time
-        waited = 0.0
-        while True:                    # reader-floor back-pressure
-            floor = self._reader_floor
-            if floor is None or self.head + n - self.ring_T <= floor:
-                break
-            _time.sleep(0.001)
-            waited += 0.001
-            if waited > 5.0:
-                raise RuntimeError('replay ring writer starved (floor %r)'
-                                   % (floor,))
-        g_flat, t_flat = traj.episode_indices(np.asarray(g_rows), lens)
-        dev = self.device
-        head0 = self.head
-        if dev.type == 'cuda':
-            main = torch.cuda.current_stream()
-            stream = self._ingest_stream if self._ingest is not None \
-                else main
-            with torch.cuda.stream(stream):
-                event = self._commit_copy(traj, g_flat, t_flat, head0, n,
-                                          gate=main)
-        else:
-            event = None
-            self._commit_copy(traj, g_flat, t_flat, head0, n)
-        entries, pos = [], head0
-        for k in range(K):
-            entries.append((pos, int(lens[k]),
-                            np.asarray(outcomes[k], dtype=np.float32)))
-            pos += int(lens[k])
-        self.head = pos
-        self.head_planned = max(self.head_planned, pos)
-        with self._table_lock:
-            if event is None:
-                self.table.extend(entries)
-                self.total_added += len(entries)
-                self._evict()
-        if event is not None:
-            with self._ready_lock:
-                self._ready.append((entries, n, event))
-        return event
-
-    def _stage_idx(self, key, arr):
-        """Pinned staging for commit index arrays: async H2D made SAFE
-        (the pinned buffer persists; a refill waits on the previous
-        copy's event).  Runs inside the ingest-stream context.  Temporary
-        pageable sources must never feed a non-blocking copy — a freed
-        temporary under a late-reading hipMemcpyAsync supplies garbage
-        INDICES to the gather below (HSAIL-fault class)."""
-        if self.device.type != 'cuda':
-            return torch.from_numpy(arr).to(self.device)
-        n = arr.shape[0]
-        pin, ev = self._pin.get(key, (None, None))
-        if pin is None or pin.shape[0] < n:
-            pin = torch.empty(max(int(n * 1.5), 4096), dtype=torch.int64,
-                              pin_memory=True)
-            ev = None
-        if ev is not None:
-            ev.synchronize()
-        pin.numpy()[:n] = arr
-        dev_t = pin[:n].to(self.device, non_blocking=True)
-        ev = torch.cuda.Event()
-        ev.record()
-        self._pin[key] = (pin, ev)
-        return dev_t
-
     def _commit_copy(self, traj, g_flat, t_flat, head0, n, gate=None):
         dev = self.device
         # stage the index H2Ds BEFORE gating on the main stream: gating
@@ -550,6 +550,43 @@ class TurnDeviceReplay(ColumnRingReplay):
         'ret': (torch.float32, (2,), 'return'),
     }
     OUTCOME_P = 2
+
+    def _commit_copy(self, traj, g_flat, t_flat, head0, n, gate=None):
+        """Turn-based commit (GeisterTrajRecorder rings -> ring columns).
+        reward is the constant -0.01 per step (geister.py reward());
+        the discounted-return column is computed in closed form:
+        ret[t] = -0.01 * (1 - gamma^(S-t)) / (1 - gamma) — identical math
+        to the worker's backward scan."""
+        dev = self.device
+        g_t = self._stage_idx('commit_g', g_flat)
+        t_t = self._stage_idx('commit_t', t_flat)
+        # steps-to-go per flat row: S - t (episode boundaries at t == 0)
+        starts = np.flatnonzero(t_flat == 0)
+        lens_ep = np.diff(np.append(starts, len(t_flat)))
+        k_np = (np.repeat(lens_ep, lens_ep) - t_flat).astype(np.int64)
+        k_t = self._stage_idx('commit_k', k_np)
+        dst = (torch.arange(n, device=dev, dtype=torch.int64) + head0) \
+            % self.ring_T
+        if gate is not None and dev.type == 'cuda':
+            torch.cuda.current_stream().wait_stream(gate)
+        self.scalar[dst] = traj.scalar[g_t, t_t]
+        self.board[dst] = traj.board[g_t, t_t]
+        self.mask[dst] = traj.mask[g_t, t_t]
+        self.turn[dst] = traj.turn[g_t, t_t]
+        apv = traj.apv[g_t, t_t]                      # (n, 3)
+        self.action[dst] = apv[:, 0].to(torch.int16)
+        self.prob[dst] = apv[:, 1]
+        self.value[dst] = apv[:, 2]
+        gamma = float(self.args['gamma'])
+        ret = (-0.01) * (1.0 - torch.pow(
+            torch.full((n,), gamma, device=dev), k_t.float())) / (1.0 - gamma)
+        self.reward[dst] = torch.full((n, 2), -0.01, device=dev)
+        self.ret[dst] = ret.unsqueeze(1).expand(n, 2)
+        if dev.type != 'cuda':
+            return None
+        event = torch.cuda.Event()
+        event.record()
+        return event
 
     def sample_indices(self, batch_size):
         """Recency-biased picks + window cuts (no seat choice: the mover

@@ -69,12 +69,10 @@ class TrajRecorder:
         self.alive[gidx, tidx] = alive
         self.rec[gidx, tidx] = packed.reshape(B, 4, 3)
 
-    def episode_indices(self, g_rows, lens):
+    @staticmethod
+    def flat_indices(g_rows, lens):
         """Flat (sum(lens),) gather indices for finished episodes:
-        (g repeated len(g) times, t = 0..len(g)-1).  Fully vectorized —
-        the short-episode regime commits hundreds of episodes per
-        service, so a per-episode python loop here costs ~20% of the
-        whole actor phase."""
+        (g repeated len(g) times, t = 0..len(g)-1).  Fully vectorized."""
         lens = np.asarray(lens, dtype=np.int64)
         if lens.size == 0:
             empty = np.empty(0, dtype=np.int64)
@@ -83,3 +81,49 @@ class TrajRecorder:
         starts = np.repeat(np.cumsum(lens) - lens, lens)
         t_flat = np.arange(int(lens.sum()), dtype=np.int64) - starts
         return g_flat, t_flat
+
+    def episode_indices(self, g_rows, lens):
+        """See flat_indices (kept as a method for the replay commit)."""
+        return self.flat_indices(g_rows, lens)
+
+
+class GeisterTrajRecorder:
+    """Turn-based device trajectory rings for the Geister DRC actors.
+
+    Row g holds the in-progress episode of global game g (the MOVER's
+    view per step, matching the columnar turn-based episode format);
+    row ``n_games`` is the scratch target for padded rows.  record_ runs
+    INSIDE the engine's captured graph."""
+
+    MAX_STEPS = 202                       # 2 layout turns + 200 moves
+
+    def __init__(self, n_games, device, max_steps=MAX_STEPS):
+        self.n_games = n_games
+        self.max_steps = max_steps
+        self.device = device
+        R = n_games + 1
+        self.scalar = torch.zeros(R, max_steps, 18, dtype=torch.uint8,
+                                  device=device)
+        self.board = torch.zeros(R, max_steps, 7, 6, 6, dtype=torch.uint8,
+                                 device=device)
+        self.mask = torch.zeros(R, max_steps, 214, dtype=torch.bool,
+                                device=device)
+        self.turn = torch.zeros(R, max_steps, dtype=torch.int8,
+                                device=device)
+        # (action, prob, value) as produced by the engine's packed output
+        self.apv = torch.zeros(R, max_steps, 3, dtype=torch.float32,
+                               device=device)
+        self.scratch_row = n_games
+
+    def record_(self, scalar_f, board_f, mask_f, parity, packed, gidx, tidx):
+        """Scatter one engine service's rows (B = shard games) into the
+        rings; every op is graph-capturable.  mask_f is the additive
+        float mask (0 legal / 1e32 illegal)."""
+        self.scalar[gidx, tidx] = scalar_f.to(torch.uint8)
+        self.board[gidx, tidx] = board_f.to(torch.uint8)
+        self.mask[gidx, tidx] = mask_f == 0.0
+        self.turn[gidx, tidx] = parity.to(torch.int8)
+        self.apv[gidx, tidx] = packed[:, :3]
+
+    episode_indices = TrajRecorder.episode_indices
+    flat_indices = staticmethod(TrajRecorder.flat_indices)
