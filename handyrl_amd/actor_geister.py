@@ -168,7 +168,8 @@ def _geister_env_worker(conn, ep_conn, shm_names, n_games, args, seed):
             frames_prev += 1
 
 
-def _geister_vec_worker(conn, ep_conn, shm_names, n_games, args, seed):
+def _geister_vec_worker(conn, ep_conn, shm_names, n_games, args, seed,
+                        traj_mode=False):
     """Vectorized variant of _geister_env_worker: one GeisterVecEnv steps
     the whole shard (legality, observations, moves, captures and win
     detection as batched numpy ops — envs/vec_geister.py, parity-tested
@@ -202,7 +203,9 @@ def _geister_vec_worker(conn, ep_conn, shm_names, n_games, args, seed):
     columnar = os.environ.get('HANDYRL_GEISTER_COLUMNAR', '1') == '1'
     moments = [[] for _ in range(G)]
     gar = np.arange(G)
-    if columnar:
+    if traj_mode:
+        rec, rec_len = None, np.zeros(G, np.int32)   # counters only
+    elif columnar:
         rec, rec_len = _make_rec_buffers(G)
 
     def package(g, outcome):
@@ -221,19 +224,29 @@ def _geister_vec_worker(conn, ep_conn, shm_names, n_games, args, seed):
 
     frames_prev = 0
     eps_out = []
+    fin = None
     while True:
         reset_v[:] = 0
         done_idx = np.nonzero(vec.over)[0]
         if len(done_idx):
             ocs = vec.outcomes(done_idx)
-            for k, g in enumerate(done_idx):
-                ep = package(g, {0: float(ocs[k, 0]), 1: float(ocs[k, 1])})
-                if ep is not None:
-                    eps_out.append(ep)
-                if columnar:
-                    rec_len[g] = 0
-                else:
-                    moments[g] = []
+            if traj_mode:
+                # fin metadata only: the parent commits the device-recorded
+                # trajectories; lens exclude rows past the recorder cap
+                lens = np.minimum(rec_len[done_idx], REC_CAP).astype(np.int64)
+                fin = (done_idx.astype(np.int64), lens,
+                       ocs.astype(np.float32))
+                rec_len[done_idx] = 0
+            else:
+                for k, g in enumerate(done_idx):
+                    ep = package(g, {0: float(ocs[k, 0]),
+                                     1: float(ocs[k, 1])})
+                    if ep is not None:
+                        eps_out.append(ep)
+                    if columnar:
+                        rec_len[g] = 0
+                    else:
+                        moments[g] = []
             vec.reset_games(done_idx)
             reset_v[done_idx] = 1
         scalar, board = vec.observations()
@@ -246,7 +259,11 @@ def _geister_vec_worker(conn, ep_conn, shm_names, n_games, args, seed):
             # episodes travel on their own pipe, drained by a parent
             # background thread: the service path never deserializes them
             ep_conn.send(eps_out)
-        conn.send(('obs', G, frames_prev))
+        if traj_mode:
+            conn.send(('obs', G, frames_prev, fin))
+            fin = None
+        else:
+            conn.send(('obs', G, frames_prev))
         frames_prev, eps_out = 0, []
         cmd = conn.recv()
         if cmd == 'quit':
@@ -264,7 +281,9 @@ def _geister_vec_worker(conn, ep_conn, shm_names, n_games, args, seed):
             import sys as _sys
             print('# geister worker: %d illegal sampled actions'
                   % nbad, file=_sys.stderr, flush=True)
-        if columnar:
+        if traj_mode:
+            rec_len += 1               # step counter only (device records)
+        elif columnar:
             _record_round(rec, rec_len, gar, scalar_v, board_v, mask_v,
                           parity_v, actions, res_v[:, 1], res_v[:, 2])
         else:
@@ -293,11 +312,20 @@ class BatchedDRCEngine:
     hipGraph-captured on GPU (in-place hidden updates keep the storages
     fixed under capture); eager on CPU / when capture fails."""
 
-    def __init__(self, model, device, n_games, use_graphs=True):
+    def __init__(self, model, device, n_games, use_graphs=True,
+                 traj=None, traj_base=0):
         import os
         self.model = model
         self.device = device
         self.n_per = n_games
+        # device-side trajectory recording (handyrl_amd/traj
+        # GeisterTrajRecorder): the captured service scatters the mover's
+        # scalar/board/mask/turn and the packed outputs into HBM rings
+        self.traj = traj
+        if traj is not None:
+            self._gidx = traj_base + torch.arange(n_games, dtype=torch.int64,
+                                                  device=device)
+            self.tidx = torch.zeros(n_games, dtype=torch.int64, device=device)
         # fused DRC: one hand-written CDNA4 kernel per ConvLSTM cell eval
         # (ops/src/ext.hip::convlstm_cell) with NHWC bf16 hidden state
         # (HANDYRL_DRC_FUSED=0 opts back into the eager cell chain)
@@ -422,9 +450,14 @@ class BatchedDRCEngine:
             pr = torch.softmax(policy - st['mask'], dim=-1)
             actions = torch.multinomial(pr, 1).squeeze(-1)
             probs = pr.gather(-1, actions.unsqueeze(-1)).squeeze(-1)
-        return torch.stack([actions.float(), probs,
-                            value.squeeze(-1),
-                            ret.squeeze(-1)], dim=1)
+        packed = torch.stack([actions.float(), probs,
+                              value.squeeze(-1),
+                              ret.squeeze(-1)], dim=1)
+        if self.traj is not None:
+            self.traj.record_(st['scalar'], st['board'], st['mask'],
+                              st['parity'], packed, self._gidx, self.tidx)
+            self.tidx.add_(1).clamp_(0, self.traj.max_steps - 1)
+        return packed
 
     @torch.no_grad()
     def _capture(self):
@@ -444,6 +477,8 @@ class BatchedDRCEngine:
             hs, cs = self.hidden
             for t in hs + cs:
                 t.zero_()
+            if self.traj is not None:
+                self.tidx.zero_()     # warmups advanced the counters
         except Exception as e:     # noqa: BLE001 - eager fallback
             import sys
             print('geister actor graph capture failed, running eager: %r'
@@ -590,14 +625,22 @@ class GeisterMultiProcPool:
     """256-actor-style Geister self-play on one GPU: W env-worker processes
     + batched recurrent inference with GPU-resident DRC hidden state."""
 
-    def __init__(self, args, n_games=256, seed=0, workers=8, vec=None):
+    def __init__(self, args, n_games=256, seed=0, workers=8, vec=None,
+                 traj_mode=False, make_stubs=True):
         import multiprocessing as mp
         import os
         from multiprocessing import shared_memory
         self.args = args
         self.workers = workers
+        # traj_mode: GeisterTrajRecorder rings + TurnDeviceReplay.commit_traj
+        # (requires attach(..., replay=TurnDeviceReplay) on CUDA): workers
+        # ship only fin metadata, never episode arrays
+        self.traj_mode = traj_mode
+        self.make_stubs = make_stubs
         if vec is None:
             vec = os.environ.get('HANDYRL_GEISTER_VEC', '1') == '1'
+        if traj_mode:
+            vec = True                  # traj workers are the vec variant
         worker_fn = _geister_vec_worker if vec else _geister_env_worker
         per = max(1, n_games // workers)
         self.n_per = per
@@ -632,12 +675,12 @@ class GeisterMultiProcPool:
             self.views.append(views)
             parent_conn, child_conn = mp.Pipe(duplex=True)
             ep_parent, ep_child = mp.Pipe(duplex=False)
-            proc = mp.Process(
-                target=worker_fn,
-                args=(child_conn, ep_child,
-                      {k: s.name for k, s in shm.items()},
-                      per, args, seed + 131 * w),
-                daemon=True)
+            wargs = (child_conn, ep_child,
+                     {k: s.name for k, s in shm.items()},
+                     per, args, seed + 131 * w)
+            if worker_fn is _geister_vec_worker:
+                wargs = wargs + (traj_mode,)
+            proc = mp.Process(target=worker_fn, args=wargs, daemon=True)
             proc.start()
             child_conn.close()
             ep_child.close()
@@ -658,13 +701,22 @@ class GeisterMultiProcPool:
                                                daemon=True)
         drain.start()
 
-    def attach(self, model, device):
+    def attach(self, model, device, replay=None):
         import os
         self.model = model
         self.device = device
+        self.replay = replay
+        self.traj = None
+        if self.traj_mode:
+            assert replay is not None, 'traj mode needs a TurnDeviceReplay'
+            from .traj import GeisterTrajRecorder
+            self.traj = GeisterTrajRecorder(self.workers * self.n_per,
+                                            device)
         # one engine (resident hidden + captured graph) per worker shard
-        self.engines = [BatchedDRCEngine(model, device, self.n_per)
-                        for _ in range(self.workers)]
+        self.engines = [BatchedDRCEngine(model, device, self.n_per,
+                                         traj=self.traj,
+                                         traj_base=w * self.n_per)
+                        for w in range(self.workers)]
         # event-polled async service (the geese-pool design): the DRC
         # round trip of one worker overlaps every other worker's env work;
         # HANDYRL_GEISTER_ASYNC=0 restores the synchronous service
@@ -683,6 +735,33 @@ class GeisterMultiProcPool:
         v = self.views[wid]
         self.engines[wid].infer(v['scalar'], v['board'], v['mask'],
                                 v['parity'], v['reset'], v['res'])
+
+    def _commit_finished(self, wid, fin):
+        """Traj mode: commit finished device-recorded episodes into the
+        replay ring and zero the engine's step counters for those games
+        (ordering: the engine's next replay waits on the commit event)."""
+        g_local, lens, outcomes = fin
+        base = wid * self.n_per
+        event = self.replay.commit_traj(self.traj, base + g_local, lens,
+                                        outcomes)
+        if event is not None:
+            torch.cuda.current_stream().wait_event(event)
+        if len(g_local):
+            rows = torch.from_numpy(np.ascontiguousarray(g_local)).to(
+                self.device)          # blocking: temporary pageable source
+            self.engines[wid].tidx.index_fill_(0, rows, 0)
+        if self.make_stubs:
+            stubs = [{'args': _JOB_ARGS, 'steps': int(lens[k]),
+                      'outcome': {0: float(outcomes[k, 0]),
+                                  1: float(outcomes[k, 1])},
+                      'committed': True}
+                     for k in range(len(g_local))]
+            with self._completed_lock:
+                self.completed.extend(stubs)
+                self.episodes_done += len(stubs)
+        else:
+            with self._completed_lock:
+                self.episodes_done += len(g_local)
 
     def _complete(self, wid):
         self._events[wid].synchronize()
@@ -707,9 +786,12 @@ class GeisterMultiProcPool:
                 if self.conns[self.rr] in ready else ready[0]
             wid = self.conns.index(conn)
             self.rr = (wid + 1) % self.workers
-            tag, G, frames = conn.recv()
+            msg = conn.recv()
+            tag, G, frames = msg[:3]
             assert tag == 'obs'
             self.frames += frames
+            if self.traj is not None and msg[3] is not None:
+                self._commit_finished(wid, msg[3])
             self._infer(wid)
             conn.send('go')
             return frames
@@ -729,9 +811,12 @@ class GeisterMultiProcPool:
         conn = self.conns[self.rr] if self.conns[self.rr] in ready else ready[0]
         wid = self.conns.index(conn)
         self.rr = (wid + 1) % self.workers
-        tag, G, frames = conn.recv()
+        msg = conn.recv()
+        tag, G, frames = msg[:3]
         assert tag == 'obs'
         self.frames += frames
+        if self.traj is not None and msg[3] is not None:
+            self._commit_finished(wid, msg[3])
         v = self.views[wid]
         if self.engines[wid].infer_async(v['scalar'], v['board'], v['mask'],
                                          v['parity'], v['reset'],

@@ -627,9 +627,17 @@ class Learner:
                     traj_mode=args.get('replay') == 'device')
             elif env_name == 'Geister':
                 from .actor_geister import GeisterMultiProcPool
+                # mirrors the Trainer's device-replay auto-default: traj
+                # mode records trajectories on device when the replay ring
+                # will be device-resident
+                will_device = args.get('replay') == 'device' or (
+                    args.get('replay') is None
+                    and args['turn_based_training']
+                    and not args['observation']
+                    and not args.get('burn_in_steps', 0))
                 self._mpool = GeisterMultiProcPool(
                     args, n_games=n_envs, seed=args['seed'] + 1,
-                    workers=procs)
+                    workers=procs, traj_mode=will_device)
 
         self.worker = WorkerServer(args) if remote else WorkerCluster(args)
         self.trainer = Trainer(args, copy.deepcopy(self.model))
@@ -733,11 +741,8 @@ class Learner:
         env_name = str(self.args.get('env', {}).get('env', ''))
         if self._mpool is not None:
             pool = self._mpool
-            if env_name == 'HungryGeese':
-                pool.attach(self.trainer.model, self.trainer.device,
-                            replay=self.trainer.device_replay)
-            else:
-                pool.attach(self.trainer.model, self.trainer.device)
+            pool.attach(self.trainer.model, self.trainer.device,
+                        replay=self.trainer.device_replay)
             n_envs = self.args['worker'].get('num_envs', 2048)
         elif env_name == 'Geister':
             from .actor_geister import GeisterActorPool

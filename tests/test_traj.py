@@ -130,3 +130,94 @@ def test_extend_skips_committed_stubs():
     r.extend([{'committed': True, 'steps': 5,
                'outcome': {p: 0.0 for p in range(4)}}])
     assert len(r) == 0
+
+
+def _record_turn_episode(traj, g, steps, rng, gamma=0.8):
+    """Record one synthetic turn-based episode into trajectory row g and
+    return the equivalent columnar turn-based episode dict."""
+    scalar = rng.integers(0, 2, (steps, 18)).astype(np.uint8)
+    board = rng.integers(0, 2, (steps, 7, 6, 6)).astype(np.uint8)
+    mask = rng.random((steps, 214)) < 0.2
+    mask[:, 0] = True                               # >=1 legal
+    turn = (np.arange(steps) % 2).astype(np.int8)
+    act = rng.integers(0, 214, steps).astype(np.int16)
+    prob = rng.random(steps).astype(np.float32)
+    val = rng.standard_normal(steps).astype(np.float32)
+
+    for t in range(steps):
+        packed = np.stack([act[t].astype(np.float32), prob[t], val[t],
+                           np.float32(0)]).reshape(1, 4)
+        traj.record_(
+            torch.from_numpy(scalar[t:t + 1].astype(np.float32)),
+            torch.from_numpy(board[t:t + 1].astype(np.float32)),
+            torch.from_numpy(np.where(mask[t:t + 1], 0.0, 1e32)
+                             .astype(np.float32)),
+            torch.tensor([int(turn[t])], dtype=torch.int64),
+            torch.from_numpy(packed),
+            torch.tensor([g], dtype=torch.int64),
+            torch.tensor([t], dtype=torch.int64))
+
+    acc, rets = 0.0, np.empty(steps, np.float32)
+    for t in range(steps - 1, -1, -1):
+        acc = -0.01 + gamma * acc
+        rets[t] = acc
+    return {'args': {'player': [0, 1], 'model_id': {0: -1, 1: -1}},
+            'steps': steps, 'columnar': True, 'turn_based': True,
+            'n_actions': 214, 'n_players': 2,
+            'outcome': {0: 1.0, 1: -1.0},
+            'scalar': scalar, 'board': board, 'mask': mask, 'turn': turn,
+            'action': act, 'prob': prob, 'value': val,
+            'reward': np.full((steps, 2), -0.01, np.float32),
+            'return': np.stack([rets, rets], axis=1)}
+
+
+def test_turn_commit_traj_matches_host_extend():
+    from handyrl_amd.replay import TurnDeviceReplay
+    from handyrl_amd.traj import GeisterTrajRecorder
+
+    rng = np.random.default_rng(7)
+    dev = torch.device('cpu')
+    traj = GeisterTrajRecorder(8, dev, max_steps=32)
+    args = {'turn_based_training': True, 'observation': False,
+            'gamma': 0.8, 'forward_steps': 8, 'burn_in_steps': 0,
+            'compress_steps': 4, 'batch_size': 4, 'minimum_episodes': 2,
+            'maximum_episodes': 64, 'lambda': 0.7,
+            'policy_target': 'UPGO', 'value_target': 'TD'}
+
+    eps = [_record_turn_episode(traj, g, steps, rng)
+           for g, steps in [(0, 9), (2, 5), (6, 12)]]
+    outcomes = np.array([[ep['outcome'][p] for p in range(2)] for ep in eps],
+                        dtype=np.float32)
+
+    r_traj = TurnDeviceReplay(args, dev, bytes_budget=64 << 20)
+    r_traj.commit_traj(traj, np.array([0, 2, 6], dtype=np.int64),
+                       np.array([9, 5, 12], dtype=np.int64), outcomes)
+    r_host = TurnDeviceReplay(args, dev, bytes_budget=64 << 20)
+    r_host.extend(eps)
+
+    assert len(r_traj) == len(r_host) == 3
+    for (p0a, sa, oca), (p0b, sb, ocb) in zip(r_traj.table, r_host.table):
+        assert (p0a, sa) == (p0b, sb)
+        np.testing.assert_array_equal(oca, ocb)
+    n = sum(ep['steps'] for ep in eps)
+    for col in ('scalar', 'board', 'mask', 'turn', 'action'):
+        assert torch.equal(getattr(r_traj, col)[:n],
+                           getattr(r_host, col)[:n]), col
+    torch.testing.assert_close(r_traj.prob[:n], r_host.prob[:n])
+    torch.testing.assert_close(r_traj.value[:n], r_host.value[:n])
+    torch.testing.assert_close(r_traj.reward[:n], r_host.reward[:n])
+    # closed-form discounted return vs the serial float backfill
+    torch.testing.assert_close(r_traj.ret[:n], r_host.ret[:n],
+                               rtol=1e-5, atol=1e-5)
+
+    import random as pyrandom
+    to_t = lambda arrs: [torch.from_numpy(np.asarray(a)) for a in arrs]
+    pyrandom.seed(3)
+    idx_a = r_traj.sample_indices(4)
+    pyrandom.seed(3)
+    idx_b = r_host.sample_indices(4)
+    ba = r_traj.gather_batch(*to_t(idx_a))
+    bb = r_host.gather_batch(*to_t(idx_b))
+    for k in bb:
+        torch.testing.assert_close(ba[k], bb[k], rtol=1e-5, atol=1e-5,
+                                   msg=lambda m, k=k: '%s: %s' % (k, m))

@@ -44,9 +44,13 @@ def main():
     from handyrl_amd.batch import EpisodeBuffer, Batcher
     buffer = EpisodeBuffer(args)
     batcher = None if cli.device_replay else Batcher(args, buffer)
+    import torch as _torch
+    traj_mode = cli.device_replay and _torch.cuda.is_available() and \
+        os.environ.get('HANDYRL_GEISTER_TRAJ', '1') == '1'
     from handyrl_amd.actor_geister import GeisterMultiProcPool
     pool = GeisterMultiProcPool(args, n_games=cli.actors, seed=17,
-                                workers=cli.workers)
+                                workers=cli.workers, traj_mode=traj_mode,
+                                make_stubs=False)
 
     use_cuda = torch.cuda.is_available()
     device = torch.device('cuda', 0) if use_cuda else torch.device('cpu')
@@ -59,7 +63,6 @@ def main():
     actor_model = env.net().to(device)
     actor_model.load_state_dict(trainer.model.state_dict())
     actor_model.eval()
-    pool.attach(actor_model, device)
 
     dreplay = dstep = None
     if cli.device_replay:
@@ -67,6 +70,8 @@ def main():
         from handyrl_amd.hipgraph import GraphedRecurrentTrainStep
         dreplay = TurnDeviceReplay(args, device, bytes_budget=2 << 30,
                                    ingest_thread=use_cuda)
+    pool.attach(actor_model, device,
+                replay=dreplay if traj_mode else None)
 
     def pump(n):
         frames = 0
