@@ -585,6 +585,7 @@ class MultiProcGeesePool:
         self.graphed = None
         self.fused = None
         self.inflight = {}
+        self._gate_event = None
         self._use_registered = False
         self._fifo = []
         self.rr = 0
@@ -780,7 +781,7 @@ class MultiProcGeesePool:
         are zeroed here (the worker reset them host-side already)."""
         g_local, lens, outcomes = fin
         event = self.replay.commit_traj(self.traj, base + g_local, lens,
-                                        outcomes)
+                                        outcomes, gate=self._gate_event)
         if event is not None:
             torch.cuda.current_stream().wait_event(event)
         if tidx_dev is not None and len(g_local):
@@ -857,6 +858,10 @@ class MultiProcGeesePool:
             if fin is not None:
                 tidx_dev = self._tidx_dev[wid][slot] \
                     if self._svc_graphs is not None else None
+                # the worker's previous service event HAS fired (it sent
+                # this message only after 'go'): a free, sufficient gate
+                self._gate_event = self._events[wid][slot] \
+                    if self.graphed is not None else None
                 self._commit_finished(base, fin, tidx_dev)
                 if self._svc_graphs is None:
                     self._tidx_np[wid][slot][fin[0]] = 0

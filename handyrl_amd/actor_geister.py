@@ -742,8 +742,9 @@ class GeisterMultiProcPool:
         (ordering: the engine's next replay waits on the commit event)."""
         g_local, lens, outcomes = fin
         base = wid * self.n_per
+        gate = self._events[wid] if self._async else None
         event = self.replay.commit_traj(self.traj, base + g_local, lens,
-                                        outcomes)
+                                        outcomes, gate=gate)
         if event is not None:
             torch.cuda.current_stream().wait_event(event)
         if len(g_local):
