@@ -627,17 +627,14 @@ class Learner:
                     traj_mode=args.get('replay') == 'device')
             elif env_name == 'Geister':
                 from .actor_geister import GeisterMultiProcPool
-                # mirrors the Trainer's device-replay auto-default: traj
-                # mode records trajectories on device when the replay ring
-                # will be device-resident
-                will_device = args.get('replay') == 'device' or (
-                    args.get('replay') is None
-                    and args['turn_based_training']
-                    and not args['observation']
-                    and not args.get('burn_in_steps', 0))
+                # Geister ships episodes over the pipe into the device
+                # replay (measured faster than device trajectory
+                # recording for this env: 151k vs 146k at 2048 actors;
+                # HANDYRL_GEISTER_TRAJ=1 opts into the traj rings)
+                traj = os.environ.get('HANDYRL_GEISTER_TRAJ', '0') == '1'
                 self._mpool = GeisterMultiProcPool(
                     args, n_games=n_envs, seed=args['seed'] + 1,
-                    workers=procs, traj_mode=will_device)
+                    workers=procs, traj_mode=traj)
 
         self.worker = WorkerServer(args) if remote else WorkerCluster(args)
         self.trainer = Trainer(args, copy.deepcopy(self.model))

@@ -46,7 +46,7 @@ def main():
     batcher = None if cli.device_replay else Batcher(args, buffer)
     import torch as _torch
     traj_mode = cli.device_replay and _torch.cuda.is_available() and \
-        os.environ.get('HANDYRL_GEISTER_TRAJ', '1') == '1'
+        os.environ.get('HANDYRL_GEISTER_TRAJ', '0') == '1'
     from handyrl_amd.actor_geister import GeisterMultiProcPool
     pool = GeisterMultiProcPool(args, n_games=cli.actors, seed=17,
                                 workers=cli.workers, traj_mode=traj_mode,
