@@ -470,7 +470,8 @@ class BatchedDRCEngine:
                     self._infer_body(st)
             torch.cuda.current_stream().wait_stream(stream)
             graph = torch.cuda.CUDAGraph()
-            with CAPTURE_LOCK, torch.cuda.graph(graph):
+            with CAPTURE_LOCK, torch.cuda.graph(
+                graph, capture_error_mode='thread_local'):
                 packed = self._infer_body(st)
             self._graph = (graph, st, packed)
             # captured warmups corrupted the hidden state: reset it

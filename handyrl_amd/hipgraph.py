@@ -67,7 +67,8 @@ class GraphedTrainStep:
         torch.cuda.current_stream().wait_stream(stream)
 
         self.graph = torch.cuda.CUDAGraph()
-        with CAPTURE_LOCK, torch.cuda.graph(self.graph):
+        with CAPTURE_LOCK, torch.cuda.graph(
+                    self.graph, capture_error_mode='thread_local'):
             self.losses, self.dcnt = self._run()
 
     def _run(self):
@@ -174,7 +175,8 @@ class GraphedActorForward:
                 run_once()
         torch.cuda.current_stream().wait_stream(stream)
         graph = torch.cuda.CUDAGraph()
-        with CAPTURE_LOCK, torch.cuda.graph(graph):
+        with CAPTURE_LOCK, torch.cuda.graph(
+                graph, capture_error_mode='thread_local'):
             packed = run_once()
         if was_training:
             self.model.train()
@@ -239,7 +241,8 @@ class GraphedActorForward:
                 run_once()
         torch.cuda.current_stream().wait_stream(stream)
         graph = torch.cuda.CUDAGraph()
-        with CAPTURE_LOCK, torch.cuda.graph(graph):
+        with CAPTURE_LOCK, torch.cuda.graph(
+                graph, capture_error_mode='thread_local'):
             packed = run_once()
         if was_training:
             self.model.train()
@@ -286,7 +289,8 @@ class GraphedActorForward:
                 run_once()
         torch.cuda.current_stream().wait_stream(stream)
         graph = torch.cuda.CUDAGraph()
-        with CAPTURE_LOCK, torch.cuda.graph(graph):
+        with CAPTURE_LOCK, torch.cuda.graph(
+                graph, capture_error_mode='thread_local'):
             run_once()
         if was_training:
             self.model.train()
@@ -359,7 +363,8 @@ class GraphedReplayTrainStep:
         try:
             self.graph = torch.cuda.CUDAGraph()
             self._fill()
-            with CAPTURE_LOCK, torch.cuda.graph(self.graph):
+            with CAPTURE_LOCK, torch.cuda.graph(
+                    self.graph, capture_error_mode='thread_local'):
                 self.losses, self.dcnt = self._run()
         except Exception as e:      # noqa: BLE001 - run eager if capture fails
             import sys
@@ -467,7 +472,8 @@ class GraphedRecurrentTrainStep:
                 torch.cuda.current_stream().wait_stream(stream)
                 self.graph = torch.cuda.CUDAGraph()
                 self._fill()
-                with CAPTURE_LOCK, torch.cuda.graph(self.graph):
+                with CAPTURE_LOCK, torch.cuda.graph(
+                    self.graph, capture_error_mode='thread_local'):
                     self.losses, self.dcnt = self._run()
             except Exception as e:  # noqa: BLE001 - run eager if capture fails
                 import sys
