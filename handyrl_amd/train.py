@@ -906,7 +906,11 @@ class Learner:
         if isinstance(self.worker, WorkerServer) or \
                 self.args['worker'].get('num_parallel', 0) > 0:
             self.worker.run()
-        self.server()
+        try:
+            self.server()
+        finally:
+            if self._mpool is not None:
+                self._mpool.shutdown()    # quit the env workers cleanly
 
 
 def train_main(args):
