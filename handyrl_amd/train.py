@@ -899,8 +899,11 @@ class Learner:
 
     def run(self):
         threading.Thread(target=self.trainer.run, daemon=True).start()
+        self._actor_thread = None
         if self.gpu_actor:
-            threading.Thread(target=self._gpu_actor_loop, daemon=True).start()
+            self._actor_thread = threading.Thread(
+                target=self._gpu_actor_loop, daemon=True)
+            self._actor_thread.start()
         # a LOCAL cluster with zero workers has nothing to spawn; the remote
         # WorkerServer must always run (it accepts workers that join later)
         if isinstance(self.worker, WorkerServer) or \
@@ -909,6 +912,9 @@ class Learner:
         try:
             self.server()
         finally:
+            self.shutdown_flag = True
+            if self._actor_thread is not None:
+                self._actor_thread.join(timeout=10)
             if self._mpool is not None:
                 self._mpool.shutdown()    # quit the env workers cleanly
 
