@@ -110,3 +110,46 @@ def test_geister_pool_python_env_fallback(monkeypatch):
         _train_on(eps, env, args)
     finally:
         pool.shutdown()
+
+
+def test_geister_multiproc_pool_traj_mode_cpu():
+    """Traj-mode pool end to end on CPU: workers ship fin metadata only,
+    the engine records into GeisterTrajRecorder rings eagerly, and
+    commit_traj fills the TurnDeviceReplay ring."""
+    import torch
+    from handyrl_amd.actor_geister import GeisterMultiProcPool
+    from handyrl_amd.replay import TurnDeviceReplay
+    from handyrl_amd.envs.geister import Environment
+
+    args = {'turn_based_training': True, 'observation': False,
+            'gamma': 0.8, 'forward_steps': 8, 'burn_in_steps': 0,
+            'compress_steps': 4, 'batch_size': 4, 'minimum_episodes': 2,
+            'maximum_episodes': 200, 'lambda': 0.7,
+            'policy_target': 'UPGO', 'value_target': 'TD',
+            'compress_episodes': False}
+    pool = GeisterMultiProcPool(args, n_games=8, seed=11, workers=2,
+                                traj_mode=True)
+    dev = torch.device('cpu')
+    replay = TurnDeviceReplay(args, dev, bytes_budget=32 << 20)
+    try:
+        torch.manual_seed(0)
+        model = Environment().net()
+        model.eval()
+        pool.attach(model, dev, replay=replay)
+        for _ in range(3000):
+            pool.step_once()
+            if pool.episodes_done >= 4:
+                break
+        assert pool.episodes_done >= 4
+        assert len(replay) >= 4
+        stubs = pool.harvest()
+        assert stubs and all(s.get('committed') for s in stubs)
+        # the committed rows sample and gather
+        import numpy as np
+        idx = replay.sample_indices(4)
+        batch = replay.gather_batch(*[torch.from_numpy(np.asarray(a))
+                                      for a in idx])
+        assert torch.isfinite(batch['selected_prob']).all()
+        assert batch['observation']['board'].shape[0] == 4
+    finally:
+        pool.shutdown()
