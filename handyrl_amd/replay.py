@@ -171,11 +171,13 @@ class ColumnRingReplay:
         """Copy finished device-recorded episodes (handyrl_amd/traj) into
         the ring DEVICE-TO-DEVICE and publish their table entries.
 
-        g_rows/lens: int64[K] global trajectory rows and episode lengths;
-        outcomes: float32[K, 4].  Returns the completion event (the caller
-        must make the next trajectory write to these rows wait on it) or
-        None when K == 0.  Runs on the ingest stream, ordered after all
-        issued main-stream work (the in-graph scatters of these rows)."""
+        g_rows/lens: int64[K] global trajectory rows and episode
+        lengths; outcomes: float32[K, OUTCOME_P].  ``gate`` is an event
+        recorded after the finishing worker's last service (its scatters
+        to these rows) — by the pool protocol it has already fired, so
+        the wait is free.  Returns the completion event (the caller must
+        make the next trajectory write to these rows wait on it) or None
+        when K == 0.  Runs on the ingest stream."""
         K = len(g_rows)
         if K == 0:
             return None
