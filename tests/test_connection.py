@@ -58,3 +58,13 @@ def test_queue_communicator_pipe():
     assert conn1.recv() == 'world'
     hub.disconnect(conn)
     assert hub.connection_count() == 0
+
+
+def test_queue_communicator_close():
+    """close() retires every connection and stops the pumps."""
+    import multiprocessing as mp
+    conn0, conn1 = mp.Pipe(duplex=True)
+    hub = QueueCommunicator([conn0])
+    assert hub.connection_count() == 1
+    hub.close()
+    assert hub.connection_count() == 0

@@ -92,7 +92,9 @@ def test_observation_shapes_stable(env):
 
 
 def test_check_env_validator():
-    """The runtime contract validator passes every bundled game."""
+    """The runtime contract validator passes every bundled game
+    (turn-based, simultaneous and 4-player survival alike)."""
     from handyrl_amd.environment import make_env, check_env
-    for name in ['TicTacToe', 'ParallelTicTacToe', 'Geister']:
+    for name in ['TicTacToe', 'ParallelTicTacToe', 'Geister',
+                 'HungryGeese']:
         assert check_env(make_env({'env': name}))
