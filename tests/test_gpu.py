@@ -241,8 +241,11 @@ def test_gpu_train_step_bf16():
     trainer = Trainer(args, GeeseNet(), device=device)
     pool = GeeseActorPool(trainer.model, args, n_games=32, device=device, seed=0)
     trainer.model.eval()
-    while pool.episodes_done < 8:
+    for _ in range(20000):
         pool.step_once()
+        if pool.episodes_done >= 8:
+            break
+    assert pool.episodes_done >= 8, 'no episodes finished'
     trainer.episodes.extend(pool.harvest())
     for _ in range(3):
         sel = [trainer.episodes.select_episode() for _ in range(args['batch_size'])]
@@ -276,8 +279,11 @@ def test_device_replay_graphed_training():
     trainer = Trainer(args, GeeseNet(), device=device, batcher=False)
     pool = GeeseActorPool(trainer.model, args, n_games=64, device=device, seed=0)
     trainer.model.eval()
-    while pool.episodes_done < 40:
+    for _ in range(40000):
         pool.step_once()
+        if pool.episodes_done >= 40:
+            break
+    assert pool.episodes_done >= 40, 'no episodes finished'
     replay = DeviceReplay(args, device, bytes_budget=256 << 20)
     replay.extend(pool.harvest())
 
@@ -413,8 +419,11 @@ def test_turn_device_replay_recurrent_training():
     trainer.model.eval()
     pool = GeisterActorPool(trainer.model, args, n_games=64, device=device,
                             seed=0)
-    while pool.episodes_done < 30:
+    for _ in range(40000):
         pool.step_once()
+        if pool.episodes_done >= 30:
+            break
+    assert pool.episodes_done >= 30, 'no episodes finished'
     replay = TurnDeviceReplay(args, device, bytes_budget=256 << 20)
     replay.extend(pool.harvest())
 
@@ -460,8 +469,11 @@ def test_traj_mode_pool_end_to_end():
     pool.attach(trainer.model, device, replay=replay)
     try:
         trainer.model.eval()
-        while pool.episodes_done < 40:
+        for _ in range(40000):
             pool.step_once()
+            if pool.episodes_done >= 40:
+                break
+        assert pool.episodes_done >= 40, 'no episodes finished'
         replay.flush()
         assert len(replay) >= 40
         # stubs carry stats but no data
